@@ -1,0 +1,130 @@
+"""Window-batch dataset: scenario traces -> training batches.
+
+Each training example is one 30 s sliding window rendered as
+(graph tensors + per-file sequences + labels).  Batches are prebuilt on the
+CPU (numpy) and staged to the device as a handful of contiguous copies.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, Iterator, List, Optional
+
+import numpy as np
+import torch
+
+from ..graph.constructor import build_graph, sliding_windows
+from ..graph.sampling import sample_fanout, to_csr
+from .labels import event_labels
+from .sequences import build_sequences
+from .synth import AttackWindow, SynthConfig, generate
+from .trace import EventArray
+
+
+@dataclass
+class WindowBatch:
+    """One window's tensors (numpy, CPU)."""
+
+    x: np.ndarray
+    nbr_idx: np.ndarray
+    nbr_w: np.ndarray
+    edge_index: np.ndarray
+    edge_weight: np.ndarray
+    edge_ts: np.ndarray
+    y_node: np.ndarray
+    y_edge: np.ndarray
+    seq_feats: np.ndarray
+    seq_lengths: np.ndarray
+    y_seq: np.ndarray
+    n_events: int
+
+    def to_torch(self, device="cpu", dtype=torch.float32) -> Dict[str, torch.Tensor]:
+        def t(a, dt=None):
+            x = torch.from_numpy(a)
+            if dt is not None:
+                x = x.to(dt)
+            return x.to(device, non_blocking=True)
+
+        return {
+            "x": t(self.x, dtype),
+            "nbr_idx": t(self.nbr_idx),
+            "nbr_w": t(self.nbr_w, torch.float32),
+            "edge_index": t(self.edge_index),
+            "edge_weight": t(self.edge_weight, torch.float32),
+            "edge_ts": t(self.edge_ts, torch.float32),
+            "y_node": t(self.y_node, torch.float32),
+            "y_edge": t(self.y_edge, torch.float32),
+            "seq_feats": t(self.seq_feats, dtype),
+            "seq_lengths": t(self.seq_lengths),
+            "y_seq": t(self.y_seq, torch.float32),
+            "n_events": torch.tensor(self.n_events),
+        }
+
+
+def window_to_batch(
+    events: EventArray,
+    window: Optional[AttackWindow],
+    fanout: int = 16,
+    seq_len: int = 100,
+    seed: int = 0,
+) -> WindowBatch:
+    y_ev = event_labels(events, window)
+    g = build_graph(events, window, y_ev)
+    csr = to_csr(g.edge_index, g.num_nodes, g.edge_weight)
+    nbr_idx, nbr_w = sample_fanout(csr, fanout, seed=seed)
+    seqs = build_sequences(events, y_ev, seq_len=seq_len)
+    return WindowBatch(
+        x=g.x,
+        nbr_idx=nbr_idx,
+        nbr_w=nbr_w,
+        edge_index=g.edge_index,
+        edge_weight=g.edge_weight,
+        edge_ts=g.edge_ts,
+        y_node=g.y_node if g.y_node is not None else np.zeros(g.num_nodes, dtype=np.float32),
+        y_edge=g.y_edge if g.y_edge is not None else np.zeros(g.num_edges, dtype=np.float32),
+        seq_feats=seqs.feats,
+        seq_lengths=seqs.lengths,
+        y_seq=seqs.labels if seqs.labels is not None else np.zeros(len(seqs.lengths), dtype=np.float32),
+        n_events=len(events),
+    )
+
+
+def synth_window_batches(
+    n_scenarios: int = 8,
+    window_s: float = 30.0,
+    stride_s: float = 15.0,
+    duration_s: float = 120.0,
+    benign_rate_hz: float = 800.0,
+    attack_fraction: float = 0.6,
+    fanout: int = 16,
+    seq_len: int = 100,
+    base_seed: int = 0,
+) -> List[WindowBatch]:
+    """Prebuild window batches from synthetic scenarios."""
+    batches: List[WindowBatch] = []
+    for i in range(n_scenarios):
+        cfg = SynthConfig(
+            duration_s=duration_s,
+            benign_rate_hz=benign_rate_hz,
+            attack=(i % 100) < int(attack_fraction * 100),
+            seed=base_seed + 7919 * i,
+            attack_start_frac=0.2 + 0.5 * ((i * 13) % 10) / 10.0,
+        )
+        arr, win = generate(cfg)
+        for j, (t0, evw) in enumerate(sliding_windows(arr, window_s, stride_s)):
+            batches.append(window_to_batch(evw, win, fanout=fanout, seq_len=seq_len, seed=base_seed + i * 131 + j))
+    return batches
+
+
+def iterate_epochs(
+    batches: List[WindowBatch],
+    epochs: int,
+    device="cpu",
+    dtype=torch.float32,
+    shuffle: bool = True,
+    seed: int = 0,
+) -> Iterator[Dict[str, torch.Tensor]]:
+    rng = np.random.default_rng(seed)
+    for _ in range(epochs):
+        order = rng.permutation(len(batches)) if shuffle else np.arange(len(batches))
+        for i in order:
+            yield batches[int(i)].to_torch(device=device, dtype=dtype)

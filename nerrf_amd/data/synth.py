@@ -1,0 +1,262 @@
+"""Deterministic synthetic trace generation (benign load + LockBit-style attack).
+
+The upstream project validates against a deterministic, reversible ransomware
+simulator run inside minikube (reference behavior:
+/root/reference/benchmarks/m1/scripts/sim_lockbit_m1.py — 5-phase attack:
+recon burst, file seeding, chunked encrypt/rename/unlink at a rate limit,
+ransom note) with ground truth expressed as an attack time-window CSV
+(/root/reference/benchmarks/m1/scripts/m1_minikube_bootstrap.sh:219-224).
+
+Here the same scenario is a pure in-process generator producing columnar
+:class:`~nerrf_amd.data.trace.EventArray` batches — no k8s needed — so it can
+drive CPU tests, GPU benches, and multi-hour training-set synthesis at
+millions of events/second (vectorised numpy, no per-event Python loop).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional, Tuple
+
+import numpy as np
+
+from .trace import SYSCALL_IDS, EventArray, StringTable
+
+
+@dataclass
+class AttackWindow:
+    """Ground truth: [t_start, t_end) and the directory under attack."""
+
+    t_start: float
+    t_end: float
+    target_dir: str
+    encrypted_ext: str = ".lockbit3"
+
+
+@dataclass
+class SynthConfig:
+    duration_s: float = 120.0
+    # benign background
+    n_benign_procs: int = 24
+    n_benign_files: int = 400
+    benign_rate_hz: float = 800.0  # aggregate benign events/second
+    # attack
+    attack: bool = True
+    attack_start_frac: float = 0.35
+    n_victim_files: int = 48
+    victim_file_mb: float = 2.4
+    encrypt_rate_mbps: float = 2.0  # reference sim rate limit (sim_lockbit_m1.py:17)
+    chunk_kb: int = 256
+    recon_burst: int = 40
+    target_dir: str = "/app/uploads"
+    encrypted_ext: str = ".lockbit3"
+    seed: int = 0
+
+
+_BENIGN_DIRS = ["/var/www/html", "/home/svc/data", "/app/cache", "/var/log/app"]
+_BENIGN_EXTS = [".html", ".log", ".json", ".tmp", ".dat", ".css"]
+_RECON_BINS = ["/bin/ps", "/bin/netstat", "/usr/bin/whoami", "/bin/df", "/bin/mount", "/bin/ss"]
+
+
+def _interleave(cols: List[Tuple[np.ndarray, ...]]) -> Tuple[np.ndarray, ...]:
+    """Concatenate column tuples and sort all by the first column (time)."""
+    merged = [np.concatenate([c[i] for c in cols]) for i in range(len(cols[0]))]
+    order = np.argsort(merged[0], kind="stable")
+    return tuple(m[order] for m in merged)
+
+
+def generate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
+    rng = np.random.default_rng(cfg.seed)
+    paths = StringTable()
+    comms = StringTable()
+
+    benign_comm_ids = np.array(
+        [comms.intern(c) for c in ("nginx", "python3", "node", "postgres", "cron", "rsyslogd")],
+        dtype=np.int64,
+    )
+    benign_path_ids = np.array(
+        [
+            paths.intern(f"{rng.choice(_BENIGN_DIRS)}/f{j:05d}{_BENIGN_EXTS[j % len(_BENIGN_EXTS)]}")
+            for j in range(cfg.n_benign_files)
+        ],
+        dtype=np.int64,
+    )
+
+    blocks: List[Tuple[np.ndarray, ...]] = []
+
+    # ---- benign background: open/read/write/close micro-sessions ----------
+    n_benign = max(1, int(cfg.benign_rate_hz * cfg.duration_s))
+    ts = np.sort(rng.uniform(0.0, cfg.duration_s, size=n_benign))
+    pid = rng.integers(100, 100 + cfg.n_benign_procs, size=n_benign).astype(np.int64)
+    # syscall mix: mostly read/write/openat/close, few renames (benign
+    # write-to-rename ratio stays far below the 0.8 indicator threshold).
+    sys_choices = np.array(
+        [
+            SYSCALL_IDS["openat"],
+            SYSCALL_IDS["read"],
+            SYSCALL_IDS["write"],
+            SYSCALL_IDS["close"],
+            SYSCALL_IDS["rename"],
+        ],
+        dtype=np.int8,
+    )
+    syscall = sys_choices[
+        rng.choice(len(sys_choices), p=[0.22, 0.36, 0.28, 0.13, 0.01], size=n_benign)
+    ]
+    path_id = benign_path_ids[rng.integers(0, len(benign_path_ids), size=n_benign)]
+    new_path_id = np.full(n_benign, -1, dtype=np.int64)
+    ren = syscall == SYSCALL_IDS["rename"]
+    new_path_id[ren] = benign_path_ids[rng.integers(0, len(benign_path_ids), size=int(ren.sum()))]
+    nbytes = np.zeros(n_benign, dtype=np.int64)
+    rw = (syscall == SYSCALL_IDS["read"]) | (syscall == SYSCALL_IDS["write"])
+    nbytes[rw] = np.exp(rng.normal(8.5, 1.5, size=int(rw.sum()))).astype(np.int64)  # ~5 KB median
+    comm_id = benign_comm_ids[pid % len(benign_comm_ids)]
+    blocks.append((ts, pid, syscall, path_id, new_path_id, nbytes, comm_id))
+
+    window: Optional[AttackWindow] = None
+
+    if cfg.attack:
+        t0 = cfg.attack_start_frac * cfg.duration_s
+        atk_pid = np.int64(6666)
+        atk_comm = comms.intern("lockbit")
+
+        # phase 1 — recon burst (exec of enumeration binaries)
+        n_recon = cfg.recon_burst
+        rts = t0 + np.sort(rng.uniform(0.0, 1.5, size=n_recon))
+        rpaths = np.array(
+            [paths.intern(_RECON_BINS[i % len(_RECON_BINS)]) for i in range(n_recon)],
+            dtype=np.int64,
+        )
+        blocks.append(
+            (
+                rts,
+                np.full(n_recon, atk_pid),
+                np.full(n_recon, SYSCALL_IDS["exec"], dtype=np.int8),
+                rpaths,
+                np.full(n_recon, -1, dtype=np.int64),
+                np.zeros(n_recon, dtype=np.int64),
+                np.full(n_recon, atk_comm, dtype=np.int64),
+            )
+        )
+
+        # phase 2+3 — per-file chunked encrypt: openat, read xk, write xk,
+        # rename(.dat -> .lockbit3), unlink(original)
+        chunk_bytes = cfg.chunk_kb * 1024
+        file_bytes = int(cfg.victim_file_mb * 1e6)
+        k = max(1, file_bytes // chunk_bytes)
+        per_file_s = (file_bytes / 1e6) / cfg.encrypt_rate_mbps  # rate-limited
+        victim_ids = np.array(
+            [paths.intern(f"{cfg.target_dir}/doc_{j:04d}.dat") for j in range(cfg.n_victim_files)],
+            dtype=np.int64,
+        )
+        enc_ids = np.array(
+            [
+                paths.intern(f"{cfg.target_dir}/doc_{j:04d}.dat{cfg.encrypted_ext}")
+                for j in range(cfg.n_victim_files)
+            ],
+            dtype=np.int64,
+        )
+        ev_per_file = 1 + k + k + 1 + 1  # open, k reads, k writes, rename, unlink
+        n_atk = cfg.n_victim_files * ev_per_file
+        f_start = t0 + 2.0 + np.arange(cfg.n_victim_files) * per_file_s
+        # per-file intra-times: open at 0, read i at (i+0.25)/k, write i at
+        # (i+0.75)/k of the file's slice, rename/unlink at the end
+        intra = np.concatenate(
+            [
+                np.zeros(1),
+                (np.arange(k) + 0.25) / k * per_file_s,
+                (np.arange(k) + 0.75) / k * per_file_s,
+                np.array([per_file_s * 1.001, per_file_s * 1.002]),
+            ]
+        )
+        a_ts = (f_start[:, None] + intra[None, :]).reshape(-1)
+        sys_pat = np.concatenate(
+            [
+                np.array([SYSCALL_IDS["openat"]], dtype=np.int8),
+                np.full(k, SYSCALL_IDS["read"], dtype=np.int8),
+                np.full(k, SYSCALL_IDS["write"], dtype=np.int8),
+                np.array([SYSCALL_IDS["rename"], SYSCALL_IDS["unlink"]], dtype=np.int8),
+            ]
+        )
+        a_sys = np.tile(sys_pat, cfg.n_victim_files)
+        # reads hit the victim file, writes hit the encrypted copy
+        pat_path = np.empty(ev_per_file, dtype=np.int64)
+        a_path = np.empty(n_atk, dtype=np.int64)
+        a_newp = np.full(n_atk, -1, dtype=np.int64)
+        for j in range(cfg.n_victim_files):
+            pat_path[0] = victim_ids[j]
+            pat_path[1 : 1 + k] = victim_ids[j]
+            pat_path[1 + k : 1 + 2 * k] = enc_ids[j]
+            pat_path[1 + 2 * k] = victim_ids[j]  # rename src
+            pat_path[2 + 2 * k] = victim_ids[j]  # unlink original
+            a_path[j * ev_per_file : (j + 1) * ev_per_file] = pat_path
+            a_newp[j * ev_per_file + 1 + 2 * k] = enc_ids[j]
+        a_bytes = np.zeros(n_atk, dtype=np.int64)
+        rw_pat = (a_sys == SYSCALL_IDS["read"]) | (a_sys == SYSCALL_IDS["write"])
+        a_bytes[rw_pat] = chunk_bytes
+        blocks.append(
+            (
+                a_ts,
+                np.full(n_atk, atk_pid),
+                a_sys,
+                a_path,
+                a_newp,
+                a_bytes,
+                np.full(n_atk, atk_comm, dtype=np.int64),
+            )
+        )
+
+        # phase 4 — ransom note
+        note_id = paths.intern(f"{cfg.target_dir}/README_LOCKBIT.txt")
+        t_note = float(a_ts[-1]) + 0.05
+        blocks.append(
+            (
+                np.array([t_note, t_note + 0.01]),
+                np.full(2, atk_pid),
+                np.array([SYSCALL_IDS["openat"], SYSCALL_IDS["write"]], dtype=np.int8),
+                np.array([note_id, note_id], dtype=np.int64),
+                np.full(2, -1, dtype=np.int64),
+                np.array([0, 1337], dtype=np.int64),
+                np.full(2, atk_comm, dtype=np.int64),
+            )
+        )
+        window = AttackWindow(
+            t_start=float(t0),
+            t_end=t_note + 0.02,
+            target_dir=cfg.target_dir,
+            encrypted_ext=cfg.encrypted_ext,
+        )
+
+    ts_, pid_, sys_, path_, newp_, bytes_, comm_ = _interleave(blocks)
+    arr = EventArray(
+        ts=ts_,
+        pid=pid_,
+        syscall=sys_,
+        path_id=path_,
+        new_path_id=newp_,
+        nbytes=bytes_,
+        ret_val=np.zeros(len(ts_), dtype=np.int64),
+        comm_id=comm_,
+        paths=paths,
+        comms=comms,
+    )
+    return arr, window
+
+
+def generate_mixed_dataset(
+    n_scenarios: int,
+    base_seed: int = 0,
+    attack_fraction: float = 0.6,
+    duration_s: float = 120.0,
+) -> List[Tuple[EventArray, Optional[AttackWindow]]]:
+    """A set of scenario traces, some with attacks, some clean."""
+    out = []
+    for i in range(n_scenarios):
+        cfg = SynthConfig(
+            duration_s=duration_s,
+            attack=(i % 100) < int(attack_fraction * 100),
+            seed=base_seed + 1000 * i,
+            n_victim_files=24 + (i * 7) % 40,
+            attack_start_frac=0.2 + 0.5 * ((i * 13) % 10) / 10.0,
+        )
+        out.append(generate(cfg))
+    return out

@@ -1,0 +1,348 @@
+"""Temporal dependency-graph construction from syscall event windows.
+
+Behavioral spec (reference docs architecture.mdx:36-43, threat-model.mdx:179-188):
+  * sliding window of 30-60 s over the event stream,
+  * nodes = processes and files, files deduped across renames (the upstream
+    spec dedups by inode; traces without inodes get rename-union instead,
+    which is the same equivalence for the attack pattern),
+  * edge weight = causality confidence (recency-decayed interaction count),
+  * node features: in/out degree, temporal deltas, byte-count ratios,
+    extension-pattern confidence, read/write/rename counters, plus the
+    documented detection indicators (write-to-rename ratio, ransom-note name,
+    recon burst) as explicit feature channels.
+
+Everything is vectorised numpy; `TemporalGraph.to_torch()` produces the
+contiguous tensors the GPU path stages into HBM.
+"""
+from __future__ import annotations
+
+import re
+from dataclasses import dataclass
+from typing import Dict, Optional, Tuple
+
+import numpy as np
+
+from ..data.synth import AttackWindow
+from ..data.trace import SYSCALL_IDS, EventArray
+
+NUM_NODE_FEATURES = 32
+# syscalls that imply process->file data flow vs file->process
+_WRITE_LIKE = (SYSCALL_IDS["write"], SYSCALL_IDS["openat"], SYSCALL_IDS["chmod"])
+_READ_LIKE = (SYSCALL_IDS["read"], SYSCALL_IDS["exec"])
+
+_SUSPICIOUS_EXT = re.compile(r"\.(lockbit\w*|encrypted|locked|crypt\w*)$", re.IGNORECASE)
+_RANSOM_NOTE = re.compile(r"(^|/)(README|HOW_TO|RESTORE)[-_]", re.IGNORECASE)
+_RECON_BIN = re.compile(r"^/(usr/)?s?bin/")
+_DOUBLE_EXT = re.compile(r"\.\w{1,5}\.\w{1,9}$")
+
+
+class _UnionFind:
+    def __init__(self, n: int) -> None:
+        self.parent = np.arange(n, dtype=np.int64)
+
+    def find(self, x: int) -> int:
+        root = x
+        while self.parent[root] != root:
+            root = self.parent[root]
+        while self.parent[x] != root:
+            self.parent[x], x = root, self.parent[x]
+        return root
+
+    def union(self, a: int, b: int) -> None:
+        ra, rb = self.find(a), self.find(b)
+        if ra != rb:
+            self.parent[max(ra, rb)] = min(ra, rb)
+
+
+@dataclass
+class TemporalGraph:
+    """One 30-60 s window's dependency graph, columnar."""
+
+    x: np.ndarray  # [N, F] float32 node features
+    edge_index: np.ndarray  # [2, E] int64 (src, dst) in node ids
+    edge_weight: np.ndarray  # [E] float32 causality confidence
+    edge_ts: np.ndarray  # [E] float32 last-interaction time (window-relative)
+    node_kind: np.ndarray  # [N] int8: 0 = process, 1 = file
+    node_key: np.ndarray  # [N] int64: pid for processes, path root id for files
+    y_node: Optional[np.ndarray] = None  # [N] float32 anomaly ground truth
+    y_edge: Optional[np.ndarray] = None  # [E] float32
+    t0: float = 0.0
+    t1: float = 0.0
+
+    @property
+    def num_nodes(self) -> int:
+        return int(self.x.shape[0])
+
+    @property
+    def num_edges(self) -> int:
+        return int(self.edge_index.shape[1])
+
+    def to_torch(self, device=None, dtype=None):
+        import torch
+
+        dev = device if device is not None else "cpu"
+        x = torch.from_numpy(self.x)
+        if dtype is not None:
+            x = x.to(dtype)
+        return {
+            "x": x.to(dev),
+            "edge_index": torch.from_numpy(self.edge_index).to(dev),
+            "edge_weight": torch.from_numpy(self.edge_weight).to(dev),
+            "edge_ts": torch.from_numpy(self.edge_ts).to(dev),
+            "node_kind": torch.from_numpy(self.node_kind).to(dev),
+            "y_node": None if self.y_node is None else torch.from_numpy(self.y_node).to(dev),
+            "y_edge": None if self.y_edge is None else torch.from_numpy(self.y_edge).to(dev),
+        }
+
+
+def _path_flags(
+    paths, path_root: np.ndarray, root_to_file: np.ndarray, n_files: int
+) -> Tuple[np.ndarray, ...]:
+    """Per file-node string-pattern indicator features.
+
+    Every alias of a file (pre- and post-rename paths) contributes its flags
+    to the rename-union node, so a `.lockbit3` rename marks the merged node.
+    """
+    suspicious = np.zeros(n_files, dtype=np.float32)
+    note = np.zeros(n_files, dtype=np.float32)
+    recon = np.zeros(n_files, dtype=np.float32)
+    double_ext = np.zeros(n_files, dtype=np.float32)
+    for path_idx, s in enumerate(paths.strings):
+        node = root_to_file[path_root[path_idx]]
+        if node < 0:
+            continue
+        if _SUSPICIOUS_EXT.search(s):
+            suspicious[node] = 1.0
+        if _RANSOM_NOTE.search(s):
+            note[node] = 1.0
+        if _RECON_BIN.search(s):
+            recon[node] = 1.0
+        if _DOUBLE_EXT.search(s):
+            double_ext[node] = 1.0
+    return suspicious, note, recon, double_ext
+
+
+def build_graph(
+    events: EventArray,
+    window: Optional[AttackWindow] = None,
+    y_event: Optional[np.ndarray] = None,
+    causality_tau_s: float = 10.0,
+) -> TemporalGraph:
+    """Build the dependency graph for one event window."""
+    n_ev = len(events)
+    t0 = float(events.ts[0]) if n_ev else 0.0
+    t1 = float(events.ts[-1]) if n_ev else 0.0
+    span = max(t1 - t0, 1e-6)
+
+    n_paths = len(events.paths)
+
+    # ---- file identity: union path <-> new_path over renames --------------
+    uf = _UnionFind(n_paths)
+    ren_mask = (events.syscall == SYSCALL_IDS["rename"]) & (events.new_path_id >= 0)
+    for a, b in zip(events.path_id[ren_mask], events.new_path_id[ren_mask]):
+        if a >= 0 and b >= 0:
+            uf.union(int(a), int(b))
+    path_root = np.array([uf.find(i) for i in range(n_paths)], dtype=np.int64)
+
+    # file nodes = distinct roots actually touched
+    touched = np.concatenate(
+        [events.path_id[events.path_id >= 0], events.new_path_id[events.new_path_id >= 0]]
+    )
+    touched_roots = np.unique(path_root[touched]) if touched.size else np.empty(0, dtype=np.int64)
+    file_node_of_root: Dict[int, int] = {int(r): i for i, r in enumerate(touched_roots)}
+    n_files = len(touched_roots)
+
+    # process nodes
+    upids = np.unique(events.pid) if n_ev else np.empty(0, dtype=np.int64)
+    proc_node_of_pid: Dict[int, int] = {int(p): n_files + i for i, p in enumerate(upids)}
+    n_procs = len(upids)
+    n_nodes = n_files + n_procs
+
+    # per-event node ids (vectorised via lookup tables)
+    root_to_file = np.full(n_paths, -1, dtype=np.int64)
+    for r, i in file_node_of_root.items():
+        root_to_file[r] = i
+    ev_file = np.where(events.path_id >= 0, root_to_file[path_root[np.clip(events.path_id, 0, None)]], -1)
+    pid_sorted = np.argsort(upids)
+    ev_proc = n_files + pid_sorted[np.searchsorted(upids[pid_sorted], events.pid)] if n_ev else np.empty(0, dtype=np.int64)
+
+    # ---- per-node counters (vectorised bincount) --------------------------
+    def _count(mask: np.ndarray, ids: np.ndarray) -> np.ndarray:
+        out = np.zeros(n_nodes, dtype=np.float64)
+        sel = mask & (ids >= 0)
+        if sel.any():
+            np.add.at(out, ids[sel], 1.0)
+        return out
+
+    def _sum(mask: np.ndarray, ids: np.ndarray, vals: np.ndarray) -> np.ndarray:
+        out = np.zeros(n_nodes, dtype=np.float64)
+        sel = mask & (ids >= 0)
+        if sel.any():
+            np.add.at(out, ids[sel], vals[sel].astype(np.float64))
+        return out
+
+    sc = events.syscall
+    is_read = sc == SYSCALL_IDS["read"]
+    is_write = sc == SYSCALL_IDS["write"]
+    is_rename = sc == SYSCALL_IDS["rename"]
+    is_unlink = sc == SYSCALL_IDS["unlink"]
+    is_open = sc == SYSCALL_IDS["openat"]
+    is_exec = sc == SYSCALL_IDS["exec"]
+
+    # counts touch both the file node and the process node of each event
+    cnt_read = _count(is_read, ev_file) + _count(is_read, ev_proc)
+    cnt_write = _count(is_write, ev_file) + _count(is_write, ev_proc)
+    cnt_rename = _count(is_rename, ev_file) + _count(is_rename, ev_proc)
+    cnt_unlink = _count(is_unlink, ev_file) + _count(is_unlink, ev_proc)
+    cnt_open = _count(is_open, ev_file) + _count(is_open, ev_proc)
+    cnt_exec = _count(is_exec, ev_file) + _count(is_exec, ev_proc)
+    cnt_total = _count(np.ones(n_ev, dtype=bool), ev_file) + _count(np.ones(n_ev, dtype=bool), ev_proc)
+
+    bytes_read = _sum(is_read, ev_file, events.nbytes) + _sum(is_read, ev_proc, events.nbytes)
+    bytes_write = _sum(is_write, ev_file, events.nbytes) + _sum(is_write, ev_proc, events.nbytes)
+
+    # first/last event time per node
+    t_first = np.full(n_nodes, np.inf)
+    t_last = np.full(n_nodes, -np.inf)
+    for ids in (ev_file, ev_proc):
+        sel = ids >= 0
+        if sel.any():
+            np.minimum.at(t_first, ids[sel], events.ts[sel])
+            np.maximum.at(t_last, ids[sel], events.ts[sel])
+    t_first[~np.isfinite(t_first)] = t0
+    t_last[~np.isfinite(t_last)] = t0
+
+    # ---- edges: (proc, file, direction) aggregation -----------------------
+    valid = (ev_file >= 0) & (ev_proc >= 0)
+    direction = np.where(np.isin(sc, _READ_LIKE), 1, 0)  # 0: proc->file, 1: file->proc
+    key = (ev_proc.astype(np.int64) * n_nodes + ev_file.astype(np.int64)) * 2 + direction
+    key = key[valid]
+    if key.size:
+        uk, inv = np.unique(key, return_inverse=True)
+        e_cnt = np.bincount(inv).astype(np.float64)
+        e_bytes = np.bincount(inv, weights=events.nbytes[valid].astype(np.float64))
+        # causality confidence: recency-decayed count, saturating
+        rec = np.exp(-(t1 - events.ts[valid]) / causality_tau_s)
+        e_conf = np.bincount(inv, weights=rec)
+        e_last = np.zeros(len(uk))
+        np.maximum.at(e_last, inv, events.ts[valid])
+        dirs = uk % 2
+        pf = uk // 2
+        e_proc = (pf // n_nodes).astype(np.int64)
+        e_file = (pf % n_nodes).astype(np.int64)
+        src = np.where(dirs == 0, e_proc, e_file)
+        dst = np.where(dirs == 0, e_file, e_proc)
+        edge_index = np.stack([src, dst]).astype(np.int64)
+        edge_weight = (1.0 - np.exp(-e_conf)).astype(np.float32)  # saturate to (0,1)
+        edge_ts = ((e_last - t0) / span).astype(np.float32)
+    else:
+        edge_index = np.zeros((2, 0), dtype=np.int64)
+        edge_weight = np.zeros(0, dtype=np.float32)
+        edge_ts = np.zeros(0, dtype=np.float32)
+        e_cnt = np.zeros(0)
+        e_bytes = np.zeros(0)
+
+    # distinct peers per node
+    peer = np.zeros(n_nodes, dtype=np.float64)
+    if edge_index.shape[1]:
+        np.add.at(peer, edge_index[0], 1.0)
+        np.add.at(peer, edge_index[1], 1.0)
+
+    in_deg = np.zeros(n_nodes, dtype=np.float64)
+    out_deg = np.zeros(n_nodes, dtype=np.float64)
+    if edge_index.shape[1]:
+        np.add.at(out_deg, edge_index[0], 1.0)
+        np.add.at(in_deg, edge_index[1], 1.0)
+
+    # ---- string-pattern indicator features --------------------------------
+    suspicious, note, recon, double_ext = _path_flags(events.paths, path_root, root_to_file, n_files)
+    pad = np.zeros(n_procs, dtype=np.float32)
+    suspicious = np.concatenate([suspicious, pad])
+    note = np.concatenate([note, pad])
+    recon = np.concatenate([recon, pad])
+    double_ext = np.concatenate([double_ext, pad])
+
+    # ---- assemble feature matrix ------------------------------------------
+    x = np.zeros((n_nodes, NUM_NODE_FEATURES), dtype=np.float32)
+    node_kind = np.concatenate(
+        [np.ones(n_files, dtype=np.int8), np.zeros(n_procs, dtype=np.int8)]
+    )
+    dur = np.maximum(t_last - t_first, 0.0)
+    x[:, 0] = node_kind == 0  # is_process
+    x[:, 1] = node_kind == 1  # is_file
+    x[:, 2] = np.log1p(in_deg)
+    x[:, 3] = np.log1p(out_deg)
+    x[:, 4] = np.log1p(cnt_read)
+    x[:, 5] = np.log1p(cnt_write)
+    x[:, 6] = np.log1p(cnt_rename)
+    x[:, 7] = np.log1p(cnt_unlink)
+    x[:, 8] = np.log1p(cnt_open)
+    x[:, 9] = np.log1p(bytes_read) / 16.0
+    x[:, 10] = np.log1p(bytes_write) / 16.0
+    x[:, 11] = bytes_write / np.maximum(bytes_read + bytes_write, 1.0)
+    x[:, 12] = cnt_rename / np.maximum(cnt_write + cnt_rename, 1.0)  # write-to-rename mix
+    x[:, 13] = suspicious
+    x[:, 14] = double_ext
+    x[:, 15] = dur / span
+    x[:, 16] = (t_first - t0) / span
+    x[:, 17] = (t_last - t0) / span
+    x[:, 18] = np.log1p(cnt_total / np.maximum(dur, 1.0))  # burstiness
+    x[:, 19] = np.log1p(dur / np.maximum(cnt_total, 1.0))  # mean interarrival
+    x[:, 20] = note
+    x[:, 21] = recon
+    x[:, 22] = ((cnt_rename > 0) & (cnt_unlink > 0)).astype(np.float32)
+    x[:, 23] = np.log1p(peer)
+    x[:, 24] = np.log1p(cnt_total)
+    x[:, 25] = np.log1p(bytes_write / np.maximum(cnt_write, 1.0)) / 16.0
+    x[:, 26] = np.log1p(cnt_exec)
+    # 27..31 reserved
+
+    # ---- labels -----------------------------------------------------------
+    y_node = None
+    y_edge = None
+    if y_event is not None and n_ev:
+        y_node_f = np.zeros(n_nodes, dtype=np.float32)
+        sel = (ev_file >= 0) & (y_event > 0.5)
+        if sel.any():
+            y_node_f[np.unique(ev_file[sel])] = 1.0
+        selp = (ev_proc >= 0) & (y_event > 0.5)
+        if selp.any():
+            y_node_f[np.unique(ev_proc[selp])] = 1.0
+        y_node = y_node_f
+        if edge_index.shape[1]:
+            y_edge = (y_node[edge_index[0]] * y_node[edge_index[1]]).astype(np.float32)
+
+    node_key = np.concatenate(
+        [touched_roots, upids.astype(np.int64)]
+    ) if n_nodes else np.empty(0, dtype=np.int64)
+
+    return TemporalGraph(
+        x=x,
+        edge_index=edge_index,
+        edge_weight=edge_weight,
+        edge_ts=edge_ts,
+        node_kind=node_kind,
+        node_key=node_key,
+        y_node=y_node,
+        y_edge=y_edge,
+        t0=t0,
+        t1=t1,
+    )
+
+
+def sliding_windows(
+    events: EventArray,
+    window_s: float = 30.0,
+    stride_s: float = 15.0,
+):
+    """Yield (t_start, EventArray) windows over a time-sorted trace."""
+    if len(events) == 0:
+        return
+    t0 = float(events.ts[0])
+    t_end = float(events.ts[-1])
+    t = t0
+    while t <= t_end:
+        win = events.time_window(t, t + window_s)
+        if len(win):
+            yield t, win
+        t += stride_s

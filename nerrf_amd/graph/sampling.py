@@ -1,0 +1,86 @@
+"""CSR adjacency + fixed-fanout neighbor sampling.
+
+GraphSAGE-style fixed fanout turns irregular neighborhood aggregation into a
+dense gather over an [N, K] index matrix — the shape the CDNA4 gather-GEMM
+kernel (nerrf_amd/ops/hip/gather_sage.hip) is built around.  Sampling is
+deterministic given a seed so CPU reference and GPU runs see identical
+neighborhoods.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional, Tuple
+
+import numpy as np
+
+
+@dataclass
+class CSRGraph:
+    indptr: np.ndarray  # [N+1] int64
+    indices: np.ndarray  # [E] int64 neighbor node ids
+    weights: np.ndarray  # [E] float32
+
+    @property
+    def num_nodes(self) -> int:
+        return int(self.indptr.shape[0] - 1)
+
+
+def to_csr(
+    edge_index: np.ndarray,
+    num_nodes: int,
+    edge_weight: Optional[np.ndarray] = None,
+    symmetric: bool = True,
+) -> CSRGraph:
+    """Destination-indexed CSR (row = dst, cols = incoming srcs).
+
+    GraphSAGE aggregates *incoming* messages; `symmetric=True` adds the
+    reverse direction so information flows both ways along each edge.
+    """
+    src, dst = edge_index[0], edge_index[1]
+    w = edge_weight if edge_weight is not None else np.ones(src.shape[0], dtype=np.float32)
+    if symmetric:
+        src = np.concatenate([src, edge_index[1]])
+        dst = np.concatenate([dst, edge_index[0]])
+        w = np.concatenate([w, w])
+    order = np.argsort(dst, kind="stable")
+    dst_s, src_s, w_s = dst[order], src[order], w[order]
+    counts = np.bincount(dst_s, minlength=num_nodes)
+    indptr = np.zeros(num_nodes + 1, dtype=np.int64)
+    np.cumsum(counts, out=indptr[1:])
+    return CSRGraph(indptr=indptr, indices=src_s.astype(np.int64), weights=w_s.astype(np.float32))
+
+
+def sample_fanout(
+    csr: CSRGraph,
+    fanout: int,
+    seed: int = 0,
+    self_fill: bool = True,
+) -> Tuple[np.ndarray, np.ndarray]:
+    """Sample `fanout` incoming neighbors per node.
+
+    Returns (idx [N, K] int64, w [N, K] float32) — neighbor ids and their
+    causality weights.  Nodes with fewer than K neighbors repeat what they
+    have (sampling with replacement); isolated nodes point at themselves with
+    weight 1 when `self_fill` (aggregation degenerates to the node's own
+    features, the GraphSAGE convention for isolated vertices).
+    """
+    n = csr.num_nodes
+    k = fanout
+    rng = np.random.default_rng(seed)
+    deg = np.diff(csr.indptr)
+    if len(csr.indices) == 0:
+        idx = np.tile(np.arange(n, dtype=np.int64)[:, None], (1, k))
+        return idx, np.ones((n, k), dtype=np.float32)
+    # vectorised: random offsets modulo degree
+    rand = rng.integers(0, 1 << 62, size=(n, k))
+    safe_deg = np.maximum(deg, 1)
+    offs = (rand % safe_deg[:, None]).astype(np.int64)
+    flat = np.clip(csr.indptr[:-1][:, None] + offs, 0, len(csr.indices) - 1)
+    idx = csr.indices[flat]
+    w = csr.weights[flat].astype(np.float32)
+    if self_fill:
+        isolated = deg == 0
+        if isolated.any():
+            idx[isolated] = np.arange(n, dtype=np.int64)[isolated, None]
+            w[isolated] = 1.0
+    return idx, w

@@ -1,0 +1,138 @@
+// Torch extension bindings for the nerrf-amd CDNA4 kernels.
+#include <torch/extension.h>
+
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+namespace nerrf {
+void launch_gather_mean_fwd(const void*, const long*, const float*, void*, int,
+                            int, int, bool, hipStream_t);
+void launch_gather_mean_bwd(const void*, const long*, const float*, float*,
+                            int, int, int, bool, hipStream_t);
+void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
+                               const float*, void*, void*, void*, long, int,
+                               bool, hipStream_t);
+void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
+                               const void*, const float*, void*, void*, void*,
+                               long, int, bool, hipStream_t);
+}  // namespace nerrf
+
+namespace {
+
+bool is_bf16(const torch::Tensor& t) {
+  TORCH_CHECK(
+      t.scalar_type() == torch::kBFloat16 || t.scalar_type() == torch::kFloat32,
+      "nerrf kernels support bf16/fp32, got ", t.scalar_type());
+  return t.scalar_type() == torch::kBFloat16;
+}
+
+void check_gpu_contig(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+torch::Tensor gather_mean_fwd(torch::Tensor h, torch::Tensor idx,
+                              torch::Tensor w) {
+  check_gpu_contig(h, "h");
+  check_gpu_contig(idx, "idx");
+  check_gpu_contig(w, "w");
+  TORCH_CHECK(idx.scalar_type() == torch::kInt64, "idx must be int64");
+  const int n = idx.size(0);
+  const int k = idx.size(1);
+  const int dim = h.size(1);
+  TORCH_CHECK(k <= 64, "fanout K must be <= 64 (wave-resident), got ", k);
+  TORCH_CHECK(dim <= 512, "feature dim must be <= 512, got ", dim);
+  auto wf = w.scalar_type() == torch::kFloat32 ? w : w.to(torch::kFloat32);
+  auto out = torch::empty({n, dim}, h.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_gather_mean_fwd(h.data_ptr(), idx.data_ptr<long>(),
+                                wf.data_ptr<float>(), out.data_ptr(), n, dim,
+                                k, is_bf16(h), stream.stream());
+  return out;
+}
+
+torch::Tensor gather_mean_bwd(torch::Tensor grad_out, torch::Tensor idx,
+                              torch::Tensor w, long num_nodes) {
+  check_gpu_contig(grad_out, "grad_out");
+  check_gpu_contig(idx, "idx");
+  const int n = idx.size(0);
+  const int k = idx.size(1);
+  const int dim = grad_out.size(1);
+  auto wf = w.scalar_type() == torch::kFloat32 ? w.contiguous()
+                                               : w.to(torch::kFloat32).contiguous();
+  auto ws = torch::zeros({num_nodes, dim},
+                         grad_out.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_gather_mean_bwd(grad_out.data_ptr(), idx.data_ptr<long>(),
+                                wf.data_ptr<float>(), ws.data_ptr<float>(), n,
+                                dim, k, is_bf16(grad_out), stream.stream());
+  return ws.to(grad_out.scalar_type());
+}
+
+std::vector<torch::Tensor> lstm_pointwise_fwd(torch::Tensor gates_pre,
+                                              torch::Tensor c_prev,
+                                              torch::Tensor h_prev,
+                                              torch::Tensor mask) {
+  check_gpu_contig(gates_pre, "gates_pre");
+  check_gpu_contig(c_prev, "c_prev");
+  check_gpu_contig(h_prev, "h_prev");
+  const long batch = c_prev.size(0);
+  const int hdim = c_prev.size(1);
+  TORCH_CHECK(gates_pre.size(1) == 4 * hdim, "gates_pre must be [B, 4H]");
+  const float* mask_ptr = nullptr;
+  torch::Tensor mf;
+  if (mask.numel() > 0) {
+    mf = mask.scalar_type() == torch::kFloat32 ? mask.contiguous()
+                                               : mask.to(torch::kFloat32).contiguous();
+    TORCH_CHECK(mf.numel() == batch, "mask must be [B]");
+    mask_ptr = mf.data_ptr<float>();
+  }
+  auto h_out = torch::empty_like(c_prev);
+  auto c_out = torch::empty_like(c_prev);
+  auto gates_act = torch::empty_like(gates_pre);
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_lstm_pointwise_fwd(
+      gates_pre.data_ptr(), c_prev.data_ptr(), h_prev.data_ptr(), mask_ptr,
+      h_out.data_ptr(), c_out.data_ptr(), gates_act.data_ptr(), batch, hdim,
+      is_bf16(gates_pre), stream.stream());
+  return {h_out, c_out, gates_act};
+}
+
+std::vector<torch::Tensor> lstm_pointwise_bwd(torch::Tensor grad_h,
+                                              torch::Tensor grad_c,
+                                              torch::Tensor gates_act,
+                                              torch::Tensor c_prev,
+                                              torch::Tensor mask) {
+  check_gpu_contig(grad_h, "grad_h");
+  check_gpu_contig(grad_c, "grad_c");
+  check_gpu_contig(gates_act, "gates_act");
+  check_gpu_contig(c_prev, "c_prev");
+  const long batch = c_prev.size(0);
+  const int hdim = c_prev.size(1);
+  const float* mask_ptr = nullptr;
+  torch::Tensor mf;
+  if (mask.numel() > 0) {
+    mf = mask.scalar_type() == torch::kFloat32 ? mask.contiguous()
+                                               : mask.to(torch::kFloat32).contiguous();
+    mask_ptr = mf.data_ptr<float>();
+  }
+  auto grad_gates = torch::empty_like(gates_act);
+  auto grad_c_prev = torch::empty_like(c_prev);
+  auto grad_h_pass = torch::empty_like(c_prev);
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_lstm_pointwise_bwd(
+      grad_h.data_ptr(), grad_c.data_ptr(), gates_act.data_ptr(),
+      c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
+      grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
+      is_bf16(grad_h), stream.stream());
+  return {grad_gates, grad_c_prev, grad_h_pass};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gather_mean_fwd", &gather_mean_fwd, "weighted neighbor gather-mean");
+  m.def("gather_mean_bwd", &gather_mean_bwd, "gather-mean backward");
+  m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "fused LSTM gate pointwise fwd");
+  m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "fused LSTM gate pointwise bwd");
+}

@@ -1,0 +1,164 @@
+// Fused LSTM gate pointwise + state update — CDNA4 (gfx950).
+//
+// Forward:  given pre-activation gates [B, 4H] (= x W_ih + h W_hh + b,
+// the GEMMs stay in hipBLASLt), previous (h, c) and a per-row validity mask,
+// computes in ONE launch what eager PyTorch does in ~12 elementwise kernels:
+//   i,f,o = sigmoid; g = tanh; c' = f*c + i*g; h' = o*tanh(c');
+//   masked rows pass (h, c) through unchanged.
+// Also writes the post-activation gates (saved for backward).
+//
+// Backward: the matching fused gate-gradient kernel; the two outer GEMMs
+// (grad_gates @ W_hh, grad_gates^T @ h) stay in hipBLASLt.
+//
+// Counterpart of the "fused LSTM cell" obligation in SURVEY.md §2a;
+// validated against nerrf_amd/ops/reference.py::lstm_pointwise_{fwd,bwd}_ref.
+#include "common.h"
+
+namespace nerrf {
+
+template <typename T>
+__global__ void lstm_pointwise_fwd_kernel(
+    const T* __restrict__ gates_pre,  // [B, 4H]
+    const T* __restrict__ c_prev,     // [B, H]
+    const T* __restrict__ h_prev,     // [B, H]
+    const float* __restrict__ mask,   // [B] or nullptr
+    T* __restrict__ h_out,            // [B, H]
+    T* __restrict__ c_out,            // [B, H]
+    T* __restrict__ gates_act,        // [B, 4H]
+    long batch, int hdim) {
+  const long total = batch * hdim;
+  for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long b = t / hdim;
+    const int d = (int)(t % hdim);
+    const long g0 = b * 4 * hdim + d;
+    const float ip = to_f32(gates_pre[g0]);
+    const float fp = to_f32(gates_pre[g0 + hdim]);
+    const float gp = to_f32(gates_pre[g0 + 2 * hdim]);
+    const float op = to_f32(gates_pre[g0 + 3 * hdim]);
+    const float i = sigmoidf_(ip);
+    const float f = sigmoidf_(fp);
+    const float g = tanhf(gp);
+    const float o = sigmoidf_(op);
+    const float cp = to_f32(c_prev[t]);
+    float cn = f * cp + i * g;
+    float hn = o * tanhf(cn);
+    if (mask != nullptr) {
+      const float m = mask[b];
+      cn = m * cn + (1.0f - m) * cp;
+      hn = m * hn + (1.0f - m) * to_f32(h_prev[t]);
+    }
+    c_out[t] = from_f32<T>(cn);
+    h_out[t] = from_f32<T>(hn);
+    gates_act[g0] = from_f32<T>(i);
+    gates_act[g0 + hdim] = from_f32<T>(f);
+    gates_act[g0 + 2 * hdim] = from_f32<T>(g);
+    gates_act[g0 + 3 * hdim] = from_f32<T>(o);
+  }
+}
+
+template <typename T>
+__global__ void lstm_pointwise_bwd_kernel(
+    const T* __restrict__ grad_h,     // [B, H]
+    const T* __restrict__ grad_c,     // [B, H]
+    const T* __restrict__ gates_act,  // [B, 4H]
+    const T* __restrict__ c_prev,     // [B, H]
+    const float* __restrict__ mask,   // [B] or nullptr
+    T* __restrict__ grad_gates,       // [B, 4H]
+    T* __restrict__ grad_c_prev,      // [B, H]
+    T* __restrict__ grad_h_pass,      // [B, H]
+    long batch, int hdim) {
+  const long total = batch * hdim;
+  for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long b = t / hdim;
+    const int d = (int)(t % hdim);
+    const long g0 = b * 4 * hdim + d;
+    const float i = to_f32(gates_act[g0]);
+    const float f = to_f32(gates_act[g0 + hdim]);
+    const float g = to_f32(gates_act[g0 + 2 * hdim]);
+    const float o = to_f32(gates_act[g0 + 3 * hdim]);
+    const float cp = to_f32(c_prev[t]);
+    const float m = (mask != nullptr) ? mask[b] : 1.0f;
+    const float tcn = tanhf(f * cp + i * g);  // tanh of UNMASKED c_new
+    const float gh_in = to_f32(grad_h[t]);
+    const float gc_in = to_f32(grad_c[t]);
+    const float gh = gh_in * m;
+    const float gc = gc_in * m;
+    const float d_o = gh * tcn;
+    const float d_c = gc + gh * o * (1.0f - tcn * tcn);
+    const float d_i = d_c * g;
+    const float d_f = d_c * cp;
+    const float d_g = d_c * i;
+    grad_c_prev[t] = from_f32<T>(d_c * f + gc_in * (1.0f - m));
+    grad_h_pass[t] = from_f32<T>(gh_in * (1.0f - m));
+    grad_gates[g0] = from_f32<T>(d_i * i * (1.0f - i));
+    grad_gates[g0 + hdim] = from_f32<T>(d_f * f * (1.0f - f));
+    grad_gates[g0 + 2 * hdim] = from_f32<T>(d_g * (1.0f - g * g));
+    grad_gates[g0 + 3 * hdim] = from_f32<T>(d_o * o * (1.0f - o));
+  }
+}
+
+template __global__ void lstm_pointwise_fwd_kernel<float>(
+    const float*, const float*, const float*, const float*, float*, float*, float*, long, int);
+template __global__ void lstm_pointwise_fwd_kernel<__hip_bfloat16>(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int);
+template __global__ void lstm_pointwise_bwd_kernel<float>(
+    const float*, const float*, const float*, const float*, const float*,
+    float*, float*, float*, long, int);
+template __global__ void lstm_pointwise_bwd_kernel<__hip_bfloat16>(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int);
+
+// ---------------------------------------------------------------------------
+// host launchers
+// ---------------------------------------------------------------------------
+
+static inline int grid_elems(long total, int block) {
+  long blocks = (total + block - 1) / block;
+  if (blocks > 8192) blocks = 8192;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+void launch_lstm_pointwise_fwd(const void* gates_pre, const void* c_prev,
+                               const void* h_prev, const float* mask,
+                               void* h_out, void* c_out, void* gates_act,
+                               long batch, int hdim, bool bf16, hipStream_t s) {
+  const int block = 256;
+  const int grid = grid_elems(batch * hdim, block);
+  if (bf16) {
+    lstm_pointwise_fwd_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
+        (const __hip_bfloat16*)gates_pre, (const __hip_bfloat16*)c_prev,
+        (const __hip_bfloat16*)h_prev, mask, (__hip_bfloat16*)h_out,
+        (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim);
+  } else {
+    lstm_pointwise_fwd_kernel<float><<<grid, block, 0, s>>>(
+        (const float*)gates_pre, (const float*)c_prev, (const float*)h_prev,
+        mask, (float*)h_out, (float*)c_out, (float*)gates_act, batch, hdim);
+  }
+}
+
+void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_c,
+                               const void* gates_act, const void* c_prev,
+                               const float* mask, void* grad_gates,
+                               void* grad_c_prev, void* grad_h_pass,
+                               long batch, int hdim, bool bf16, hipStream_t s) {
+  const int block = 256;
+  const int grid = grid_elems(batch * hdim, block);
+  if (bf16) {
+    lstm_pointwise_bwd_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
+        (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_c,
+        (const __hip_bfloat16*)gates_act, (const __hip_bfloat16*)c_prev, mask,
+        (__hip_bfloat16*)grad_gates, (__hip_bfloat16*)grad_c_prev,
+        (__hip_bfloat16*)grad_h_pass, batch, hdim);
+  } else {
+    lstm_pointwise_bwd_kernel<float><<<grid, block, 0, s>>>(
+        (const float*)grad_h, (const float*)grad_c, (const float*)gates_act,
+        (const float*)c_prev, mask, (float*)grad_gates, (float*)grad_c_prev,
+        (float*)grad_h_pass, batch, hdim);
+  }
+}
+
+}  // namespace nerrf

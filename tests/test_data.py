@@ -1,0 +1,96 @@
+"""Data layer tests: synth generator, loaders, labels, sequences."""
+import numpy as np
+
+from nerrf_amd.data.labels import event_labels, malicious_pids
+from nerrf_amd.data.sequences import build_sequences
+from nerrf_amd.data.synth import SynthConfig, generate
+from nerrf_amd.data.trace import SYSCALL_IDS, load_csv, load_trace, write_csv
+
+
+def test_synth_deterministic():
+    a1, w1 = generate(SynthConfig(seed=7, duration_s=30))
+    a2, w2 = generate(SynthConfig(seed=7, duration_s=30))
+    assert len(a1) == len(a2)
+    assert np.array_equal(a1.ts, a2.ts)
+    assert np.array_equal(a1.syscall, a2.syscall)
+    assert w1.t_start == w2.t_start
+
+
+def test_synth_time_sorted():
+    arr, _ = generate(SynthConfig(seed=3, duration_s=40))
+    assert np.all(np.diff(arr.ts) >= 0)
+
+
+def test_synth_attack_pattern():
+    cfg = SynthConfig(seed=1, duration_s=60, n_victim_files=10)
+    arr, win = generate(cfg)
+    assert win is not None
+    y = event_labels(arr, win)
+    assert y.sum() > 0
+    # every victim file sees rename + unlink inside the window
+    n_ren = int(((arr.syscall == SYSCALL_IDS["rename"]) & (y > 0.5)).sum())
+    n_unl = int(((arr.syscall == SYSCALL_IDS["unlink"]) & (y > 0.5)).sum())
+    assert n_ren == cfg.n_victim_files
+    assert n_unl == cfg.n_victim_files
+    assert set(malicious_pids(arr, y).tolist()) == {6666}
+
+
+def test_no_attack_no_labels():
+    arr, win = generate(SynthConfig(seed=2, attack=False, duration_s=30))
+    assert win is None
+    assert event_labels(arr, win).sum() == 0
+
+
+def test_csv_roundtrip(tmp_path):
+    arr, _ = generate(SynthConfig(seed=5, duration_s=20, benign_rate_hz=50))
+    p = tmp_path / "t.csv"
+    write_csv(p, arr)
+    back = load_csv(p)
+    assert len(back) == len(arr)
+    assert np.allclose(back.ts, arr.ts, atol=1e-5)
+    assert np.array_equal(back.syscall, arr.syscall)
+    assert np.array_equal(back.nbytes, arr.nbytes)
+
+
+def test_load_trace_dispatch(tmp_path):
+    arr, _ = generate(SynthConfig(seed=5, duration_s=10, benign_rate_hz=20))
+    p = tmp_path / "t.csv"
+    write_csv(p, arr)
+    assert len(load_trace(p)) == len(arr)
+
+
+def test_jsonl_loader(tmp_path):
+    p = tmp_path / "t.jsonl"
+    p.write_text(
+        '{"timestamp": 1.5, "event": "write", "path": "/a/b.dat", "size": 100, "pid": 3}\n'
+        '{"timestamp": 2.5, "event": "rename", "path": "/a/b.dat", "new_path": "/a/b.lockbit3", "pid": 3}\n'
+        '{"timestamp": 0.5, "event": "open", "path": "/a/b.dat", "pid": 3}\n'
+    )
+    arr = load_trace(p)
+    assert len(arr) == 3
+    # sorted by time; "open" normalised to openat
+    assert arr.syscall[0] == SYSCALL_IDS["openat"]
+    assert arr.syscall[1] == SYSCALL_IDS["write"]
+    assert arr.syscall[2] == SYSCALL_IDS["rename"]
+    assert arr.new_path_id[2] >= 0
+
+
+def test_sequences_shapes_and_labels():
+    arr, win = generate(SynthConfig(seed=9, duration_s=60, n_victim_files=8))
+    y = event_labels(arr, win)
+    seqs = build_sequences(arr, y, seq_len=100)
+    assert seqs.feats.shape[1] == 100
+    assert seqs.feats.shape[2] == 16
+    assert (seqs.lengths >= 2).all()
+    assert (seqs.lengths <= 100).all()
+    assert seqs.labels is not None and seqs.labels.sum() >= 8  # victim files flagged
+    # one-hot rows sum to 1 for valid steps
+    for bi in range(min(4, len(seqs.lengths))):
+        t = seqs.lengths[bi]
+        assert np.allclose(seqs.feats[bi, :t, :10].sum(axis=1), 1.0)
+        assert np.allclose(seqs.feats[bi, t:], 0.0)
+
+
+def test_toy_trace_exists_and_loads():
+    arr = load_trace("datasets/traces/toy_trace.csv")
+    assert len(arr) > 100

@@ -1,0 +1,97 @@
+"""Multi-process data-parallel tests (gloo backend, world_size=2, CPU)."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from nerrf_amd.eval import roc_auc
+
+
+def _find_free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _worker_allreduce(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from nerrf_amd.parallel.ddp import GradAllReducer
+
+        torch.manual_seed(100 + rank)  # different data per rank
+        model = torch.nn.Sequential(
+            torch.nn.Linear(8, 16), torch.nn.GELU(), torch.nn.Linear(16, 1)
+        )
+        # identical init across ranks
+        torch.manual_seed(7)
+        for p in model.parameters():
+            torch.nn.init.normal_(p)
+        reducer = GradAllReducer(model, bucket_bytes=128)  # force several buckets
+        x = torch.randn(4, 8) * (rank + 1)
+        y = model(x).sum()
+        y.backward()
+        reducer.finalize()
+        grads = torch.cat([p.grad.flatten() for p in model.parameters()])
+        results[rank] = grads.numpy()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_grad_allreduce_averages_across_ranks():
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_allreduce, args=(2, port, results), nprocs=2, join=True)
+    g0, g1 = results[0], results[1]
+    # after all-reduce both ranks hold identical averaged grads
+    assert np.allclose(g0, g1, atol=1e-6)
+
+
+def _worker_expected_avg(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from nerrf_amd.parallel.ddp import GradAllReducer
+
+        model = torch.nn.Linear(4, 1, bias=False)
+        with torch.no_grad():
+            model.weight.fill_(1.0)
+        reducer = GradAllReducer(model)
+        x = torch.full((1, 4), float(rank + 1))  # rank0 grads=1, rank1 grads=2
+        model(x).sum().backward()
+        reducer.finalize()
+        results[rank] = model.weight.grad.numpy().copy()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_grad_allreduce_value():
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_expected_avg, args=(2, port, results), nprocs=2, join=True)
+    # average of grad 1 and grad 2 = 1.5
+    assert np.allclose(results[0], 1.5, atol=1e-6)
+    assert np.allclose(results[1], 1.5, atol=1e-6)
+
+
+def test_roc_auc_against_sklearn():
+    try:
+        from sklearn.metrics import roc_auc_score
+    except ImportError:
+        pytest.skip("sklearn unavailable")
+    rng = np.random.default_rng(0)
+    y = rng.integers(0, 2, 500)
+    s = rng.random(500) * 0.5 + y * rng.random(500) * 0.5
+    assert abs(roc_auc(y, s) - roc_auc_score(y, s)) < 1e-9
+    # with heavy ties
+    s_t = np.round(s, 1)
+    assert abs(roc_auc(y, s_t) - roc_auc_score(y, s_t)) < 1e-9

@@ -1,0 +1,98 @@
+"""Graph constructor + sampling tests."""
+import numpy as np
+
+from nerrf_amd.data.labels import event_labels
+from nerrf_amd.data.synth import SynthConfig, generate
+from nerrf_amd.data.trace import EventArrayBuilder
+from nerrf_amd.graph.constructor import NUM_NODE_FEATURES, build_graph, sliding_windows
+from nerrf_amd.graph.sampling import sample_fanout, to_csr
+
+
+def _mini_trace():
+    b = EventArrayBuilder()
+    b.add(ts=1.0, pid=10, syscall="openat", path="/d/a.dat")
+    b.add(ts=2.0, pid=10, syscall="write", path="/d/a.dat", nbytes=100)
+    b.add(ts=3.0, pid=10, syscall="rename", path="/d/a.dat", new_path="/d/a.dat.lockbit3")
+    b.add(ts=4.0, pid=11, syscall="read", path="/d/b.dat", nbytes=50)
+    return b.build()
+
+
+def test_rename_union_merges_file_nodes():
+    arr = _mini_trace()
+    g = build_graph(arr)
+    # /d/a.dat and /d/a.dat.lockbit3 are ONE node; /d/b.dat another; 2 procs
+    n_files = int((g.node_kind == 1).sum())
+    n_procs = int((g.node_kind == 0).sum())
+    assert n_files == 2
+    assert n_procs == 2
+
+
+def test_feature_matrix_shape_and_indicators():
+    arr = _mini_trace()
+    g = build_graph(arr)
+    assert g.x.shape == (g.num_nodes, NUM_NODE_FEATURES)
+    assert np.isfinite(g.x).all()
+    # the merged a.dat node carries the suspicious-extension flag (col 13)
+    file_nodes = np.nonzero(g.node_kind == 1)[0]
+    assert g.x[file_nodes, 13].max() == 1.0
+
+
+def test_edge_directions():
+    arr = _mini_trace()
+    g = build_graph(arr)
+    # writes: proc->file; reads: file->proc => both directions present
+    src_kind = g.node_kind[g.edge_index[0]]
+    dst_kind = g.node_kind[g.edge_index[1]]
+    assert ((src_kind == 0) & (dst_kind == 1)).any()  # proc -> file
+    assert ((src_kind == 1) & (dst_kind == 0)).any()  # file -> proc
+    assert (g.edge_weight > 0).all() and (g.edge_weight <= 1).all()
+
+
+def test_labels_propagate_to_nodes_and_edges():
+    arr, win = generate(SynthConfig(seed=11, duration_s=60, n_victim_files=6))
+    y = event_labels(arr, win)
+    g = build_graph(arr, win, y)
+    assert g.y_node is not None and g.y_node.sum() >= 6
+    assert g.y_edge is not None and g.y_edge.sum() > 0
+    # malicious edges connect two malicious endpoints
+    bad = g.y_edge > 0.5
+    assert (g.y_node[g.edge_index[0][bad]] > 0.5).all()
+
+
+def test_sliding_windows_cover_trace():
+    arr, _ = generate(SynthConfig(seed=4, duration_s=90, benign_rate_hz=100))
+    wins = list(sliding_windows(arr, window_s=30, stride_s=15))
+    assert len(wins) >= 5
+    total = sum(len(w) for _, w in wins)
+    assert total >= len(arr)  # overlap counts events twice
+
+
+def test_csr_and_fanout():
+    arr, _ = generate(SynthConfig(seed=6, duration_s=40))
+    g = build_graph(arr)
+    csr = to_csr(g.edge_index, g.num_nodes, g.edge_weight)
+    assert csr.indptr[-1] == len(csr.indices)
+    idx, w = sample_fanout(csr, 16, seed=1)
+    assert idx.shape == (g.num_nodes, 16)
+    assert w.shape == (g.num_nodes, 16)
+    assert (idx >= 0).all() and (idx < g.num_nodes).all()
+    assert (w >= 0).all()
+    # determinism
+    idx2, w2 = sample_fanout(csr, 16, seed=1)
+    assert np.array_equal(idx, idx2) and np.array_equal(w, w2)
+    # sampled ids are actual in-neighbors (or self for isolated)
+    deg = np.diff(csr.indptr)
+    for n in range(0, g.num_nodes, max(1, g.num_nodes // 17)):
+        if deg[n] == 0:
+            assert (idx[n] == n).all()
+        else:
+            nbrs = set(csr.indices[csr.indptr[n] : csr.indptr[n + 1]].tolist())
+            assert set(idx[n].tolist()) <= nbrs
+
+
+def test_empty_trace():
+    b = EventArrayBuilder()
+    arr = b.build()
+    g = build_graph(arr)
+    assert g.num_nodes == 0
+    assert g.num_edges == 0
