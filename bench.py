@@ -1,0 +1,167 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: trace-events/sec through GraphSAGE-T + BiLSTM training.
+
+Measures the BASELINE.json metric ("trace-events/sec through GraphSAGE-T+LSTM
+at 1/2/4/8 MI355X") on synthetic trace windows (the reference publishes no GPU
+numbers; its headline ingest figure is 1,250 evt/s on a 4-core VM —
+BASELINE.md).  Each timed step is one full bf16 training step (forward +
+backward + gradient all-reduce + optimizer) of the joint model on one 30 s
+window of a synthetic LockBit scenario; events/sec counts the raw trace
+events the processed windows represent.
+
+Weak scaling: every rank owns identically-shaped windows (per-GPU work fixed);
+the whole-job value sums events over ranks and takes the max step time.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+from nerrf_amd.data.dataset import synth_window_batches
+from nerrf_amd.models.graphsage import SageConfig
+from nerrf_amd.models.joint import JointConfig, NerrfJointModel
+from nerrf_amd.models.lstm import LSTMConfig
+from nerrf_amd.parallel.ddp import GradAllReducer, init_distributed
+
+BASELINE_EVT_S = 1250.0  # reference tracker peak throughput
+
+
+def build_bench_batches(rank: int, n_windows: int, scale: str):
+    """Prebuild identically-shaped-per-rank window batches."""
+    shapes = {
+        # benign_rate, duration, n_files, n_victims
+        "small": (2_000.0, 60.0, 2_000, 48),
+        "full": (20_000.0, 60.0, 16_000, 64),
+    }[scale]
+    rate, dur, n_files, n_victims = shapes
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.data.dataset import window_to_batch
+    from nerrf_amd.graph.constructor import sliding_windows
+
+    batches = []
+    i = 0
+    while len(batches) < n_windows:
+        cfg = SynthConfig(
+            duration_s=dur,
+            benign_rate_hz=rate,
+            n_benign_files=n_files,
+            n_victim_files=n_victims,
+            attack=True,
+            seed=1234 + 7919 * (rank * 97 + i),
+        )
+        arr, win = generate(cfg)
+        for t0, evw in sliding_windows(arr, window_s=30.0, stride_s=30.0):
+            batches.append(window_to_batch(evw, win, fanout=16, seq_len=100, seed=i))
+            if len(batches) >= n_windows:
+                break
+        i += 1
+    return batches
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--scale", choices=["small", "full"], default="full")
+    ap.add_argument("--windows", type=int, default=2, help="prebuilt windows per rank")
+    ap.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
+    args = ap.parse_args()
+
+    rank, world, local_rank = init_distributed()
+    has_gpu = torch.cuda.is_available()
+    if has_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device(f"cuda:{local_rank}")
+    else:
+        device = torch.device("cpu")
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and has_gpu) else torch.float32
+    scale = args.scale if has_gpu else "small"
+
+    batches_np = build_bench_batches(rank, args.windows, scale)
+    batches = [b.to_torch(device=device, dtype=dtype) for b in batches_np]
+    events_per_step = [int(b.n_events) for b in batches_np]
+
+    model = NerrfJointModel(JointConfig(sage=SageConfig(), lstm=LSTMConfig())).to(
+        device=device, dtype=dtype
+    )
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3, weight_decay=1e-4, foreach=True)
+    reducer = GradAllReducer(model)
+    reducer.broadcast_params(model)
+
+    def step(i: int) -> int:
+        b = batches[i % len(batches)]
+        node_logit, edge_logit, seq_logit = model(b)
+        losses = model.loss(node_logit, edge_logit, seq_logit, b)
+        opt.zero_grad(set_to_none=False)
+        losses["total"].backward()
+        reducer.finalize()
+        opt.step()
+        return events_per_step[i % len(batches)]
+
+    import torch.distributed as dist
+
+    def barrier_sync():
+        if dist.is_initialized():
+            dist.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    t0 = time.perf_counter()
+    events_done = 0
+    for i in range(args.steps):
+        events_done += step(i)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # whole-job: max time over ranks, sum of events over ranks
+    t_tensor = torch.tensor([elapsed], dtype=torch.float64)
+    e_tensor = torch.tensor([float(events_done)], dtype=torch.float64)
+    if dist.is_initialized():
+        dist.all_reduce(t_tensor, op=dist.ReduceOp.MAX)
+        dist.all_reduce(e_tensor, op=dist.ReduceOp.SUM)
+    t_max = float(t_tensor[0])
+    e_sum = float(e_tensor[0])
+    value = e_sum / t_max
+
+    if rank == 0:
+        result = {
+            "metric": "trace_events_per_sec_graphsage_lstm_train",
+            "value": value,
+            "unit": "events/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": t_max / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / BASELINE_EVT_S,
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "GraphSAGE-T(28x128)+BiLSTM(2x256) joint",
+                "global_batch": world * 1,
+                "seq_len": 100,
+                "window_s": 30,
+                "parallelism": f"dp{world}",
+                "scale": scale,
+                "events_per_window": int(np.mean(events_per_step)),
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()
