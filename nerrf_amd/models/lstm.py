@@ -17,7 +17,7 @@ from dataclasses import dataclass
 import torch
 import torch.nn as nn
 
-from ..ops import lstm_cell
+from ..ops import lstm_sequence
 
 
 @dataclass
@@ -47,20 +47,14 @@ class FusedLSTMDirection(nn.Module):
             self.b[self.hidden : 2 * self.hidden] = 1.0
 
     def forward(self, x: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
-        """x: [B, T, E]; mask: [B, T] (1 while t < length). Returns [B, T, H]."""
-        b, t, _ = x.shape
-        # one big GEMM for all timesteps' input projection
-        xg = torch.matmul(x, self.w_ih.t())  # [B, T, 4H]
-        h = x.new_zeros(b, self.hidden)
-        c = x.new_zeros(b, self.hidden)
-        outs = []
-        steps = range(t - 1, -1, -1) if self.reverse else range(t)
-        for ti in steps:
-            h, c = lstm_cell(xg[:, ti], h, c, self.w_hh, self.b, mask[:, ti])
-            outs.append(h)
-        if self.reverse:
-            outs.reverse()
-        return torch.stack(outs, dim=1)
+        """x: [T, B, E] time-major; mask: [T, B]. Returns [T, B, H] time-major."""
+        t, b, _ = x.shape
+        # one big GEMM for all timesteps' input projection (time-major slices
+        # stay contiguous for the per-step fused kernel)
+        xg = torch.matmul(x.reshape(t * b, -1), self.w_ih.t()).reshape(t, b, 4 * self.hidden)
+        h0 = x.new_zeros(b, self.hidden)
+        c0 = x.new_zeros(b, self.hidden)
+        return lstm_sequence(xg, h0, c0, self.w_hh, self.b, mask, reverse=self.reverse)
 
 
 class BiLSTMDetector(nn.Module):
@@ -87,16 +81,16 @@ class BiLSTMDetector(nn.Module):
         """feats: [B, T, E]; lengths: [B] -> per-sequence logit [B]."""
         b, t, _ = feats.shape
         ar = torch.arange(t, device=feats.device)
-        mask = (ar.unsqueeze(0) < lengths.unsqueeze(1)).to(feats.dtype)  # [B, T]
-        h = feats
+        mask = (ar.unsqueeze(1) < lengths.unsqueeze(0)).to(feats.dtype)  # [T, B]
+        h = feats.transpose(0, 1).contiguous()  # time-major [T, B, E]
         for layer in self.dirs:
             fwd = layer[0](h, mask)
             bwd = layer[1](h, mask)
-            h = torch.cat([fwd, bwd], dim=-1)
+            h = torch.cat([fwd, bwd], dim=-1)  # [T, B, 2H]
         # forward state at t=len-1, backward state at t=0
-        idx = (lengths.clamp(min=1) - 1).view(b, 1, 1).expand(b, 1, self.cfg.hidden)
-        h_fwd = h[:, :, : self.cfg.hidden].gather(1, idx).squeeze(1)
-        h_bwd = h[:, 0, self.cfg.hidden :]
+        idx = (lengths.clamp(min=1) - 1).view(1, b, 1).expand(1, b, self.cfg.hidden)
+        h_fwd = h[:, :, : self.cfg.hidden].gather(0, idx).squeeze(0)  # [B, H]
+        h_bwd = h[0, :, self.cfg.hidden :]
         return self.head(torch.cat([h_fwd, h_bwd], dim=-1)).squeeze(-1)
 
     def num_parameters(self) -> int:

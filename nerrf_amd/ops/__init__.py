@@ -18,7 +18,7 @@ import torch
 from . import reference as _ref
 from .native import get_native, native_available
 
-__all__ = ["gather_mean", "lstm_cell", "native_available"]
+__all__ = ["gather_mean", "lstm_cell", "lstm_sequence", "native_available"]
 
 
 class _GatherMeanFn(torch.autograd.Function):
@@ -61,13 +61,19 @@ class _LSTMCellFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, xg, h, c, w_hh, b, mask):
-        gates_pre = torch.addmm(b, h, w_hh.t()) + xg  # [B, 4H]
         ext = get_native(h)
         if ext is not None:
-            h_new, c_new, gates_act = ext.lstm_pointwise_fwd(
-                gates_pre, c, h, mask if mask is not None else torch.empty(0, device=h.device)
+            hg = torch.mm(h, w_hh.t())
+            h_new = torch.empty_like(c)
+            c_new = torch.empty_like(c)
+            gates_act = torch.empty_like(xg)
+            ext.lstm_pointwise_fwd(
+                hg, xg, b, c, h,
+                mask if mask is not None else torch.empty(0, device=h.device),
+                h_new, c_new, gates_act,
             )
         else:
+            gates_pre = torch.addmm(b, h, w_hh.t()) + xg  # [B, 4H]
             h_new, c_new, gates_act = _ref.lstm_pointwise_fwd_ref(gates_pre, c, h, mask)
         ctx.save_for_backward(gates_act, c, h, w_hh, mask if mask is not None else torch.empty(0))
         return h_new, c_new
@@ -80,9 +86,13 @@ class _LSTMCellFn(torch.autograd.Function):
         grad_h = grad_h.contiguous()
         grad_c = grad_c.contiguous()
         if ext is not None:
-            grad_gates, grad_c_prev, grad_h_pass = ext.lstm_pointwise_bwd(
+            grad_gates = torch.empty_like(gates_act)
+            grad_c_prev = torch.empty_like(c)
+            grad_h_pass = torch.empty_like(c)
+            ext.lstm_pointwise_bwd(
                 grad_h, grad_c, gates_act, c,
                 mask if mask is not None else torch.empty(0, device=grad_h.device),
+                grad_gates, grad_c_prev, grad_h_pass,
             )
         else:
             grad_gates, grad_c_prev, grad_h_pass = _ref.lstm_pointwise_bwd_ref(
@@ -110,3 +120,6 @@ def lstm_cell(
     return _LSTMCellFn.apply(
         xg.contiguous(), h.contiguous(), c.contiguous(), w_hh, b, m
     )
+
+
+from .lstm_seq import lstm_sequence  # noqa: E402  (re-export)

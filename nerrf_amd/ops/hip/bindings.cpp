@@ -10,8 +10,8 @@ void launch_gather_mean_fwd(const void*, const long*, const float*, void*, int,
 void launch_gather_mean_bwd(const void*, const long*, const float*, float*,
                             int, int, int, bool, hipStream_t);
 void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
-                               const float*, void*, void*, void*, long, int,
-                               bool, hipStream_t);
+                               const void*, const void*, const float*, void*,
+                               void*, void*, long, int, bool, hipStream_t);
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
                                const void*, const float*, void*, void*, void*,
                                long, int, bool, hipStream_t);
@@ -69,16 +69,24 @@ torch::Tensor gather_mean_bwd(torch::Tensor grad_out, torch::Tensor idx,
   return ws.to(grad_out.scalar_type());
 }
 
-std::vector<torch::Tensor> lstm_pointwise_fwd(torch::Tensor gates_pre,
-                                              torch::Tensor c_prev,
-                                              torch::Tensor h_prev,
-                                              torch::Tensor mask) {
-  check_gpu_contig(gates_pre, "gates_pre");
+// Writes into caller-provided (h_out, c_out, gates_act) so the sequence loop
+// can target time-major buffers directly (no per-step allocation or copy).
+void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
+                        torch::Tensor c_prev, torch::Tensor h_prev,
+                        torch::Tensor mask, torch::Tensor h_out,
+                        torch::Tensor c_out, torch::Tensor gates_act) {
+  check_gpu_contig(hg, "hg");
+  check_gpu_contig(xg, "xg");
   check_gpu_contig(c_prev, "c_prev");
   check_gpu_contig(h_prev, "h_prev");
+  check_gpu_contig(h_out, "h_out");
+  check_gpu_contig(c_out, "c_out");
+  check_gpu_contig(gates_act, "gates_act");
   const long batch = c_prev.size(0);
   const int hdim = c_prev.size(1);
-  TORCH_CHECK(gates_pre.size(1) == 4 * hdim, "gates_pre must be [B, 4H]");
+  TORCH_CHECK(hg.size(1) == 4 * hdim && xg.size(1) == 4 * hdim,
+              "hg/xg must be [B, 4H]");
+  TORCH_CHECK(bias.numel() == 4 * hdim, "bias must be [4H]");
   const float* mask_ptr = nullptr;
   torch::Tensor mf;
   if (mask.numel() > 0) {
@@ -87,26 +95,25 @@ std::vector<torch::Tensor> lstm_pointwise_fwd(torch::Tensor gates_pre,
     TORCH_CHECK(mf.numel() == batch, "mask must be [B]");
     mask_ptr = mf.data_ptr<float>();
   }
-  auto h_out = torch::empty_like(c_prev);
-  auto c_out = torch::empty_like(c_prev);
-  auto gates_act = torch::empty_like(gates_pre);
+  auto bc = bias.contiguous();
   auto stream = at::hip::getCurrentHIPStream();
   nerrf::launch_lstm_pointwise_fwd(
-      gates_pre.data_ptr(), c_prev.data_ptr(), h_prev.data_ptr(), mask_ptr,
-      h_out.data_ptr(), c_out.data_ptr(), gates_act.data_ptr(), batch, hdim,
-      is_bf16(gates_pre), stream.stream());
-  return {h_out, c_out, gates_act};
+      hg.data_ptr(), xg.data_ptr(), bc.data_ptr(), c_prev.data_ptr(),
+      h_prev.data_ptr(), mask_ptr, h_out.data_ptr(), c_out.data_ptr(),
+      gates_act.data_ptr(), batch, hdim, is_bf16(hg), stream.stream());
 }
 
-std::vector<torch::Tensor> lstm_pointwise_bwd(torch::Tensor grad_h,
-                                              torch::Tensor grad_c,
-                                              torch::Tensor gates_act,
-                                              torch::Tensor c_prev,
-                                              torch::Tensor mask) {
+void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_c,
+                        torch::Tensor gates_act, torch::Tensor c_prev,
+                        torch::Tensor mask, torch::Tensor grad_gates,
+                        torch::Tensor grad_c_prev, torch::Tensor grad_h_pass) {
   check_gpu_contig(grad_h, "grad_h");
   check_gpu_contig(grad_c, "grad_c");
   check_gpu_contig(gates_act, "gates_act");
   check_gpu_contig(c_prev, "c_prev");
+  check_gpu_contig(grad_gates, "grad_gates");
+  check_gpu_contig(grad_c_prev, "grad_c_prev");
+  check_gpu_contig(grad_h_pass, "grad_h_pass");
   const long batch = c_prev.size(0);
   const int hdim = c_prev.size(1);
   const float* mask_ptr = nullptr;
@@ -116,16 +123,12 @@ std::vector<torch::Tensor> lstm_pointwise_bwd(torch::Tensor grad_h,
                                                : mask.to(torch::kFloat32).contiguous();
     mask_ptr = mf.data_ptr<float>();
   }
-  auto grad_gates = torch::empty_like(gates_act);
-  auto grad_c_prev = torch::empty_like(c_prev);
-  auto grad_h_pass = torch::empty_like(c_prev);
   auto stream = at::hip::getCurrentHIPStream();
   nerrf::launch_lstm_pointwise_bwd(
       grad_h.data_ptr(), grad_c.data_ptr(), gates_act.data_ptr(),
       c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
       grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
       is_bf16(grad_h), stream.stream());
-  return {grad_gates, grad_c_prev, grad_h_pass};
 }
 
 }  // namespace

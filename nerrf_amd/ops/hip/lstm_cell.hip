@@ -18,7 +18,9 @@ namespace nerrf {
 
 template <typename T>
 __global__ void lstm_pointwise_fwd_kernel(
-    const T* __restrict__ gates_pre,  // [B, 4H]
+    const T* __restrict__ hg,         // [B, 4H] = h_prev @ W_hh^T (beta=0 GEMM)
+    const T* __restrict__ xg,         // [B, 4H] = x_t @ W_ih^T (time-major slice)
+    const T* __restrict__ bias,       // [4H]
     const T* __restrict__ c_prev,     // [B, H]
     const T* __restrict__ h_prev,     // [B, H]
     const float* __restrict__ mask,   // [B] or nullptr
@@ -32,10 +34,10 @@ __global__ void lstm_pointwise_fwd_kernel(
     const long b = t / hdim;
     const int d = (int)(t % hdim);
     const long g0 = b * 4 * hdim + d;
-    const float ip = to_f32(gates_pre[g0]);
-    const float fp = to_f32(gates_pre[g0 + hdim]);
-    const float gp = to_f32(gates_pre[g0 + 2 * hdim]);
-    const float op = to_f32(gates_pre[g0 + 3 * hdim]);
+    const float ip = to_f32(hg[g0]) + to_f32(xg[g0]) + to_f32(bias[d]);
+    const float fp = to_f32(hg[g0 + hdim]) + to_f32(xg[g0 + hdim]) + to_f32(bias[d + hdim]);
+    const float gp = to_f32(hg[g0 + 2 * hdim]) + to_f32(xg[g0 + 2 * hdim]) + to_f32(bias[d + 2 * hdim]);
+    const float op = to_f32(hg[g0 + 3 * hdim]) + to_f32(xg[g0 + 3 * hdim]) + to_f32(bias[d + 3 * hdim]);
     const float i = sigmoidf_(ip);
     const float f = sigmoidf_(fp);
     const float g = tanhf(gp);
@@ -100,10 +102,12 @@ __global__ void lstm_pointwise_bwd_kernel(
 }
 
 template __global__ void lstm_pointwise_fwd_kernel<float>(
-    const float*, const float*, const float*, const float*, float*, float*, float*, long, int);
+    const float*, const float*, const float*, const float*, const float*, const float*,
+    float*, float*, float*, long, int);
 template __global__ void lstm_pointwise_fwd_kernel<__hip_bfloat16>(
-    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const float*,
-    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int);
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
+    long, int);
 template __global__ void lstm_pointwise_bwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*,
     float*, float*, float*, long, int);
@@ -122,21 +126,24 @@ static inline int grid_elems(long total, int block) {
   return (int)blocks;
 }
 
-void launch_lstm_pointwise_fwd(const void* gates_pre, const void* c_prev,
-                               const void* h_prev, const float* mask,
-                               void* h_out, void* c_out, void* gates_act,
-                               long batch, int hdim, bool bf16, hipStream_t s) {
+void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
+                               const void* c_prev, const void* h_prev,
+                               const float* mask, void* h_out, void* c_out,
+                               void* gates_act, long batch, int hdim, bool bf16,
+                               hipStream_t s) {
   const int block = 256;
   const int grid = grid_elems(batch * hdim, block);
   if (bf16) {
     lstm_pointwise_fwd_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
-        (const __hip_bfloat16*)gates_pre, (const __hip_bfloat16*)c_prev,
+        (const __hip_bfloat16*)hg, (const __hip_bfloat16*)xg,
+        (const __hip_bfloat16*)bias, (const __hip_bfloat16*)c_prev,
         (const __hip_bfloat16*)h_prev, mask, (__hip_bfloat16*)h_out,
         (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim);
   } else {
     lstm_pointwise_fwd_kernel<float><<<grid, block, 0, s>>>(
-        (const float*)gates_pre, (const float*)c_prev, (const float*)h_prev,
-        mask, (float*)h_out, (float*)c_out, (float*)gates_act, batch, hdim);
+        (const float*)hg, (const float*)xg, (const float*)bias,
+        (const float*)c_prev, (const float*)h_prev, mask, (float*)h_out,
+        (float*)c_out, (float*)gates_act, batch, hdim);
   }
 }
 

@@ -1,0 +1,147 @@
+"""Time-major fused LSTM sequence op.
+
+One autograd Function per (layer, direction) covering all T timesteps:
+  forward : per step exactly 1 hipBLASLt GEMM (h @ W_hh^T, beta=0) and 1 fused
+            HIP kernel (add xg + bias, 4-gate pointwise, masked state update)
+            writing straight into time-major output buffers;
+  backward: per step 1 fused gate-gradient kernel + 1 GEMM, then the weight
+            gradients as TWO large batched GEMMs over all timesteps
+            (grad_gates [T*B,4H]^T @ h_in [T*B,H]) and one reduction for bias.
+
+This removes the per-step elementwise adds / stacks / reduces that dominated
+the naive schedule (profiles/: 84% of busy time before the restructure).
+Everything is time-major [T, B, ...]; the model transposes once per sequence.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference as _ref
+from .native import get_native
+
+
+class _LSTMSeqFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xg, h0, c0, w_hh, bias, mask, reverse: bool):
+        """xg: [T, B, 4H] (time-major input projection); mask: [T, B] or None.
+
+        Returns h_all: [T, B, H] (time-major hidden states).
+        """
+        t_len, batch, gdim = xg.shape
+        hdim = gdim // 4
+        dev, dt = xg.device, xg.dtype
+        ext = get_native(xg)
+        h_all = torch.empty(t_len, batch, hdim, device=dev, dtype=dt)
+        c_all = torch.empty(t_len, batch, hdim, device=dev, dtype=dt)
+        gates_all = torch.empty(t_len, batch, gdim, device=dev, dtype=dt)
+        h = h0.contiguous()
+        c = c0.contiguous()
+        steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
+        if ext is not None:
+            w_hh_t = w_hh.t().contiguous()
+            hg = torch.empty(batch, gdim, device=dev, dtype=dt)
+            empty_mask = torch.empty(0, device=dev)
+            for ti in steps:
+                torch.mm(h, w_hh_t, out=hg)
+                ext.lstm_pointwise_fwd(
+                    hg, xg[ti], bias, c, h,
+                    mask[ti] if mask is not None else empty_mask,
+                    h_all[ti], c_all[ti], gates_all[ti],
+                )
+                h = h_all[ti]
+                c = c_all[ti]
+        else:
+            for ti in steps:
+                gates_pre = torch.addmm(bias, h, w_hh.t()) + xg[ti]
+                h_new, c_new, g_act = _ref.lstm_pointwise_fwd_ref(
+                    gates_pre, c, h, mask[ti] if mask is not None else None
+                )
+                h_all[ti] = h_new
+                c_all[ti] = c_new
+                gates_all[ti] = g_act
+                h, c = h_new, c_new
+        ctx.save_for_backward(
+            gates_all, h_all, c_all, h0, c0, w_hh,
+            mask if mask is not None else torch.empty(0, device=dev),
+        )
+        ctx.reverse = reverse
+        return h_all
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        gates_all, h_all, c_all, h0, c0, w_hh, mask_t = ctx.saved_tensors
+        reverse = ctx.reverse
+        mask = mask_t if mask_t.numel() else None
+        t_len, batch, hdim = h_all.shape
+        gdim = 4 * hdim
+        dev, dt = h_all.device, h_all.dtype
+        ext = get_native(h_all)
+        grad_out = grad_out.contiguous()
+
+        grad_gates_all = torch.empty(t_len, batch, gdim, device=dev, dtype=dt)
+        grad_h = torch.zeros(batch, hdim, device=dev, dtype=dt)
+        grad_c = torch.zeros(batch, hdim, device=dev, dtype=dt)
+        grad_h_pass = torch.empty(batch, hdim, device=dev, dtype=dt)
+        grad_c_prev = torch.empty(batch, hdim, device=dev, dtype=dt)
+
+        # iterate in the opposite order of forward
+        steps = range(t_len) if reverse else range(t_len - 1, -1, -1)
+        empty_mask = torch.empty(0, device=dev)
+        for ti in steps:
+            grad_h = grad_h + grad_out[ti]
+            # c/h input of step ti = previous step's output (or h0/c0 at start)
+            first = (ti == t_len - 1) if reverse else (ti == 0)
+            if first:
+                c_in = c0
+            else:
+                c_in = c_all[ti + 1] if reverse else c_all[ti - 1]
+            if ext is not None:
+                ext.lstm_pointwise_bwd(
+                    grad_h.contiguous(), grad_c.contiguous(), gates_all[ti], c_in.contiguous(),
+                    mask[ti] if mask is not None else empty_mask,
+                    grad_gates_all[ti], grad_c_prev, grad_h_pass,
+                )
+            else:
+                gg, gcp, ghp = _ref.lstm_pointwise_bwd_ref(
+                    grad_h, grad_c, gates_all[ti], c_in,
+                    mask[ti] if mask is not None else None,
+                )
+                grad_gates_all[ti] = gg
+                grad_c_prev = gcp
+                grad_h_pass = ghp
+            grad_h = torch.mm(grad_gates_all[ti], w_hh) + grad_h_pass
+            grad_c, grad_c_prev = grad_c_prev, grad_c  # ping-pong buffers
+
+        # weight grads as single large GEMMs over all timesteps
+        # h input of step ti (time-major): shift h_all by one step
+        h_in_all = torch.empty_like(h_all)
+        if reverse:
+            h_in_all[t_len - 1] = h0
+            if t_len > 1:
+                h_in_all[:-1] = h_all[1:]
+        else:
+            h_in_all[0] = h0
+            if t_len > 1:
+                h_in_all[1:] = h_all[:-1]
+        gg2 = grad_gates_all.reshape(t_len * batch, gdim)
+        grad_whh = torch.mm(gg2.t(), h_in_all.reshape(t_len * batch, hdim))
+        grad_bias = gg2.sum(dim=0)
+        grad_xg = grad_gates_all
+        grad_h0 = grad_h
+        grad_c0 = grad_c
+        return grad_xg, grad_h0, grad_c0, grad_whh, grad_bias, None, None
+
+
+def lstm_sequence(
+    xg: torch.Tensor,  # [T, B, 4H]
+    h0: torch.Tensor,
+    c0: torch.Tensor,
+    w_hh: torch.Tensor,
+    bias: torch.Tensor,
+    mask: Optional[torch.Tensor] = None,  # [T, B]
+    reverse: bool = False,
+) -> torch.Tensor:
+    m = mask.detach().contiguous().to(torch.float32) if mask is not None else None
+    return _LSTMSeqFn.apply(xg.contiguous(), h0, c0, w_hh, bias, m, reverse)

@@ -1,3 +1,5 @@
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import time, torch
 from bench import build_bench_batches
 from nerrf_amd.models.joint import NerrfJointModel, JointConfig
