@@ -4,7 +4,15 @@
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_runtime.h>
 
+#include "mcts_params.h"
+
 namespace nerrf {
+void launch_mcts(const float*, const float*, const float*, float, float,
+                 const PlannerParamsDev&, int, int, int*, float*, int*, float*,
+                 hipStream_t);
+void launch_eval_plans(const float*, const float*, const float*, float, float,
+                       const PlannerParamsDev&, const int*, int, int, float*,
+                       hipStream_t);
 void launch_gather_mean_fwd(const void*, const long*, const float*, void*, int,
                             int, int, bool, hipStream_t);
 void launch_gather_mean_bwd(const void*, const long*, const float*, float*,
@@ -131,9 +139,74 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_c,
       is_bf16(grad_h), stream.stream());
 }
 
+nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
+  nerrf::PlannerParamsDev p;
+  p.n_groups = d["n_groups"].cast<int>();
+  p.n_actions = p.n_groups + 2;
+  p.max_depth = d["max_depth"].cast<int>();
+  p.sims_per_tree = d["sims_per_tree"].cast<int>();
+  p.downtime_weight = d["downtime_weight"].cast<float>();
+  p.revert_time_s = d["revert_time_s"].cast<float>();
+  p.kill_time_s = d["kill_time_s"].cast<float>();
+  p.fp_weight = d["fp_weight"].cast<float>();
+  p.attack_rate_mbps = d["attack_rate_mbps"].cast<float>();
+  p.horizon_s = d["horizon_s"].cast<float>();
+  p.ucb_c = d["ucb_c"].cast<float>();
+  p.seed = d["seed"].cast<unsigned>();
+  TORCH_CHECK(p.n_groups <= 16, "n_groups must be <= 16");
+  TORCH_CHECK(p.max_depth <= 16, "max_depth must be <= 16");
+  return p;
+}
+
+std::vector<torch::Tensor> mcts_search(torch::Tensor gscore, torch::Tensor gmb,
+                                       torch::Tensor gfiles, double proc_score,
+                                       double remaining_clean_mb,
+                                       pybind11::dict param_dict, long n_trees) {
+  auto p = params_from_dict(param_dict);
+  check_gpu_contig(gscore, "gscore");
+  check_gpu_contig(gmb, "gmb");
+  check_gpu_contig(gfiles, "gfiles");
+  const int cap = p.sims_per_tree * p.max_depth + 2;
+  auto opts_i = gscore.options().dtype(torch::kInt32);
+  auto opts_f = gscore.options().dtype(torch::kFloat32);
+  auto arena_i = torch::empty({n_trees, (long)(3 * cap + cap * p.n_actions)}, opts_i);
+  auto arena_f = torch::empty({n_trees, (long)cap}, opts_f);
+  auto root_visits = torch::zeros({n_trees, (long)p.n_actions}, opts_i);
+  auto root_value = torch::zeros({n_trees, (long)p.n_actions}, opts_f);
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_mcts(gscore.data_ptr<float>(), gmb.data_ptr<float>(),
+                     gfiles.data_ptr<float>(), (float)proc_score,
+                     (float)remaining_clean_mb, p, (int)n_trees, cap,
+                     arena_i.data_ptr<int>(), arena_f.data_ptr<float>(),
+                     root_visits.data_ptr<int>(), root_value.data_ptr<float>(),
+                     stream.stream());
+  return {root_visits, root_value};
+}
+
+torch::Tensor mcts_eval_plans(torch::Tensor gscore, torch::Tensor gmb,
+                              torch::Tensor gfiles, double proc_score,
+                              double remaining_clean_mb,
+                              pybind11::dict param_dict, torch::Tensor plans) {
+  auto p = params_from_dict(param_dict);
+  check_gpu_contig(plans, "plans");
+  TORCH_CHECK(plans.scalar_type() == torch::kInt32, "plans must be int32");
+  const int n_plans = plans.size(0);
+  const int plan_len = plans.size(1);
+  auto out = torch::empty({(long)n_plans}, gscore.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_eval_plans(gscore.data_ptr<float>(), gmb.data_ptr<float>(),
+                           gfiles.data_ptr<float>(), (float)proc_score,
+                           (float)remaining_clean_mb, p, plans.data_ptr<int>(),
+                           n_plans, plan_len, out.data_ptr<float>(),
+                           stream.stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mcts_search", &mcts_search, "batched root-parallel MCTS");
+  m.def("mcts_eval_plans", &mcts_eval_plans, "batch plan reward evaluation");
   m.def("gather_mean_fwd", &gather_mean_fwd, "weighted neighbor gather-mean");
   m.def("gather_mean_bwd", &gather_mean_bwd, "gather-mean backward");
   m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "fused LSTM gate pointwise fwd");

@@ -204,3 +204,48 @@ def test_lstm_sequence_gpu_backward(gpu_device):
     grads_cpu = [p.grad for p in (xg_c, w_c, b_c)]
     for gg, gc, name in zip(grads_gpu, grads_cpu, ["xg", "w_hh", "bias"]):
         assert torch.allclose(gg, gc, atol=1e-3, rtol=1e-3), f"grad {name}"
+
+
+def test_mcts_eval_plans_matches_cpu(gpu_device):
+    """Device eval_plan vs CPU simulate_plan on enumerated plans."""
+    from nerrf_amd.planner.mcts import UCB_C
+    from nerrf_amd.planner.rewards import PlannerParams, build_state, simulate_plan
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    rng = np.random.default_rng(0)
+    scores = rng.random(40)
+    mb = rng.random(40) * 5
+    st = build_state(scores, mb, proc_score=0.8, remaining_clean_mb=30.0)
+    params = PlannerParams()
+    pd = {
+        "n_groups": st.n_groups, "max_depth": params.max_depth, "sims_per_tree": 1,
+        "downtime_weight": params.downtime_weight, "revert_time_s": params.revert_time_s,
+        "kill_time_s": params.kill_time_s, "fp_weight": params.fp_weight,
+        "attack_rate_mbps": params.attack_rate_mbps, "horizon_s": params.horizon_s,
+        "ucb_c": UCB_C, "seed": 0,
+    }
+    n_actions = st.n_groups + 2
+    plans = rng.integers(0, n_actions, size=(200, 6)).astype(np.int32)
+    gs = torch.from_numpy(st.group_score.astype(np.float32)).to(gpu_device)
+    gm = torch.from_numpy(st.group_mb.astype(np.float32)).to(gpu_device)
+    gf = torch.from_numpy(st.group_files.astype(np.float32)).to(gpu_device)
+    out = ext.mcts_eval_plans(gs, gm, gf, st.proc_score, st.remaining_clean_mb, pd,
+                              torch.from_numpy(plans).to(gpu_device))
+    cpu = np.array([simulate_plan(st, list(p), params) for p in plans])
+    assert np.allclose(out.cpu().numpy(), cpu, rtol=1e-4, atol=1e-3)
+
+
+def test_mcts_search_gpu_agrees_with_cpu(gpu_device):
+    from nerrf_amd.planner.mcts import run_mcts, run_mcts_gpu
+    from nerrf_amd.planner.rewards import A_KILL, build_state
+
+    scores = np.concatenate([np.full(20, 0.95), np.full(20, 0.03)])
+    mb = np.full(40, 2.0)
+    st = build_state(scores, mb, proc_score=0.97, remaining_clean_mb=50.0)
+    cpu = run_mcts(st, n_sims=1024, seed=1)
+    gpu = run_mcts_gpu(st, n_sims=1024, seed=1, device=str(gpu_device))
+    assert gpu.simulations == 1024
+    # same dominant action and comparable plan quality
+    assert gpu.ranked_actions[0][0] == cpu.ranked_actions[0][0] == A_KILL
+    assert abs(gpu.root_value - cpu.root_value) < 1.0
