@@ -1,0 +1,194 @@
+"""nerrf command-line interface.
+
+Commands (spec: reference ROADMAP.md:86, README.md:82 — `nerrf undo`,
+`nerrf status` — plus the operational verbs this engine adds):
+
+  nerrf status                       engine/store/checkpoint status
+  nerrf undo --dir D [--id ID]       detect + plan + sandbox-validate + restore
+  nerrf simulate --dir D             run the reversible LockBit attack sim
+  nerrf scenario --dir D             full e2e: seed -> attack -> detect ->
+                                     plan -> rollback -> sha256 verify
+  nerrf train ...                    training entrypoint (= ai/train.py)
+  nerrf serve --trace T              tracker-sim + streaming engine demo
+
+Entry point: `python -m nerrf_amd.cli <cmd>` (also installed as ./nerrf).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+
+def cmd_status(args) -> int:
+    import torch
+
+    from .ops.native import native_available
+
+    info = {
+        "version": __import__("nerrf_amd").__version__,
+        "torch": torch.__version__,
+        "gpu_available": torch.cuda.is_available(),
+        "native_kernels_built": native_available(),
+    }
+    if args.checkpoint and Path(args.checkpoint, "checkpoint.json").exists():
+        info["checkpoint"] = json.loads(Path(args.checkpoint, "checkpoint.json").read_text())
+    print(json.dumps(info, indent=2))
+    return 0
+
+
+def cmd_simulate(args) -> int:
+    from .harness.attack_sim import run_attack, seed_files
+
+    if args.seed_files:
+        manifest = seed_files(args.dir, n_files=args.n_files, file_kb=args.file_kb)
+        Path(args.dir, ".nerrf_manifest.json").write_text(json.dumps(manifest))
+        print(f"seeded {len(manifest)} files in {args.dir}")
+    report = run_attack(args.dir, trace_path=args.trace_out)
+    print(
+        json.dumps(
+            {
+                "files_attacked": len(report.files_attacked),
+                "bytes_attacked": report.bytes_attacked,
+                "duration_s": report.t_end - report.t_start,
+                "trace_events": len(report.trace_events),
+            }
+        )
+    )
+    return 0
+
+
+def cmd_undo(args) -> int:
+    from .data.trace import load_trace
+    from .serve.engine import StreamingEngine
+
+    t0 = time.time()
+    engine = StreamingEngine(device=args.device)
+    if args.trace:
+        engine.ingest_events(load_trace(args.trace))
+    det = engine.score_window()
+    if not det.alarm and not args.force:
+        print(json.dumps({"alarm": False, "msg": "no attack detected; use --force to undo anyway"}))
+        return 1
+    plan = engine.plan(det, n_sims=args.sims)
+    manifest = None
+    mpath = Path(args.dir, ".nerrf_manifest.json")
+    if mpath.exists():
+        manifest = json.loads(mpath.read_text())
+    result = engine.respond(det, plan, args.dir, manifest=manifest)
+    out = {
+        "alarm": det.alarm,
+        "indicators": det.indicators,
+        "plan": plan.describe(engine.planner_params.n_groups),
+        "plan_value": plan.root_value,
+        "mttr_s": time.time() - t0,
+        **result.as_dict(),
+    }
+    print(json.dumps(out, indent=2))
+    return 0 if result.files_failed == 0 else 2
+
+
+def cmd_scenario(args) -> int:
+    from .harness.scenario import run_scenario
+
+    report = run_scenario(
+        work_dir=args.dir,
+        n_files=args.n_files,
+        file_kb=args.file_kb,
+        device=args.device,
+        n_sims=args.sims,
+    )
+    print(json.dumps(report, indent=2))
+    return 0 if report["recovered_ok"] else 2
+
+
+def cmd_serve(args) -> int:
+    from .data.trace import load_trace
+    from .serve.engine import StreamingEngine
+    from .serve.tracker_sim import TrackerSimServer
+
+    trace = load_trace(args.trace)
+    server = TrackerSimServer(trace, rate_multiplier=args.rate)
+    server.start()
+    print(f"tracker-sim listening on {server.address}")
+    engine = StreamingEngine(device=args.device)
+    n = engine.ingest_from_tracker(server.address, max_events=args.max_events, timeout_s=args.timeout)
+    det = engine.score_window()
+    server.stop()
+    print(
+        json.dumps(
+            {
+                "events_ingested": n,
+                "alarm": det.alarm,
+                "indicators": det.indicators,
+                "suspicious_files": det.encrypted_paths[:10],
+            },
+            indent=2,
+        )
+    )
+    return 0
+
+
+def cmd_train(args, extra) -> int:
+    from .train import main as train_main
+
+    train_main(extra)
+    return 0
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser(prog="nerrf", description="nerrf-amd: undo computing engine")
+    sub = ap.add_subparsers(dest="cmd", required=True)
+
+    p = sub.add_parser("status", help="engine status")
+    p.add_argument("--checkpoint", default=None)
+
+    p = sub.add_parser("simulate", help="run the reversible attack simulator")
+    p.add_argument("--dir", required=True)
+    p.add_argument("--seed-files", action="store_true")
+    p.add_argument("--n-files", type=int, default=24)
+    p.add_argument("--file-kb", type=int, default=64)
+    p.add_argument("--trace-out", default=None)
+
+    p = sub.add_parser("undo", help="detect + plan + rollback a directory")
+    p.add_argument("--dir", required=True)
+    p.add_argument("--id", default=None, help="attack id (informational)")
+    p.add_argument("--trace", default=None, help="trace file to ingest first")
+    p.add_argument("--device", default="cpu")
+    p.add_argument("--sims", type=int, default=1024)
+    p.add_argument("--force", action="store_true")
+
+    p = sub.add_parser("scenario", help="full e2e attack->detect->recover")
+    p.add_argument("--dir", required=True)
+    p.add_argument("--n-files", type=int, default=16)
+    p.add_argument("--file-kb", type=int, default=32)
+    p.add_argument("--device", default="cpu")
+    p.add_argument("--sims", type=int, default=512)
+
+    p = sub.add_parser("serve", help="tracker-sim + streaming engine demo")
+    p.add_argument("--trace", required=True)
+    p.add_argument("--rate", type=float, default=0.0)
+    p.add_argument("--max-events", type=int, default=None)
+    p.add_argument("--timeout", type=float, default=15.0)
+    p.add_argument("--device", default="cpu")
+
+    sub.add_parser("train", help="training entrypoint (args passed through)")
+
+    if argv is None:
+        argv = sys.argv[1:]
+    if argv and argv[0] == "train":
+        return cmd_train(None, argv[1:])
+    args = ap.parse_args(argv)
+    return {
+        "status": cmd_status,
+        "simulate": cmd_simulate,
+        "undo": cmd_undo,
+        "scenario": cmd_scenario,
+        "serve": cmd_serve,
+    }[args.cmd](args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -36,6 +36,7 @@ class WindowBatch:
     seq_lengths: np.ndarray
     y_seq: np.ndarray
     n_events: int
+    seq_path_id: Optional[np.ndarray] = None  # [B] path id per sequence
 
     def to_torch(self, device="cpu", dtype=torch.float32) -> Dict[str, torch.Tensor]:
         def t(a, dt=None):
@@ -85,6 +86,7 @@ def window_to_batch(
         seq_lengths=seqs.lengths,
         y_seq=seqs.labels if seqs.labels is not None else np.zeros(len(seqs.lengths), dtype=np.float32),
         n_events=len(events),
+        seq_path_id=seqs.file_path_id,
     )
 
 

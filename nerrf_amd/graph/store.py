@@ -1,0 +1,95 @@
+"""Streaming delta edge store: 30 s sliding window with delta compaction.
+
+Behavioral spec (reference README.md:114 "RocksDB + 30 s delta compaction",
+architecture.mdx:36-43): the event stream lands in small append-only delta
+chunks; a sliding window (default 30 s) of deltas is compacted on demand into
+one columnar batch for graph construction.  Chunks older than the window are
+evicted.
+
+The store is columnar end-to-end so the GPU path stages each delta as a
+handful of contiguous copies into HBM (288 GB/GPU leaves the whole window
+resident).
+"""
+from __future__ import annotations
+
+import threading
+from collections import deque
+from typing import Deque, List, Optional, Tuple
+
+import numpy as np
+
+from ..data.trace import EventArray, EventArrayBuilder, StringTable
+
+
+class DeltaGraphStore:
+    def __init__(self, window_s: float = 30.0, delta_s: float = 5.0) -> None:
+        self.window_s = window_s
+        self.delta_s = delta_s
+        self.paths = StringTable()
+        self.comms = StringTable()
+        self._deltas: Deque[EventArray] = deque()
+        self._current: Optional[EventArrayBuilder] = None
+        self._current_t0: float = -np.inf
+        self._lock = threading.Lock()
+        self.total_events = 0
+        self.evicted_events = 0
+
+    def append(self, ts, pid, syscall, path="", new_path="", nbytes=0, ret_val=0, comm="") -> None:
+        """Append one event (shares the store-wide string tables)."""
+        with self._lock:
+            if self._current is None or ts - self._current_t0 >= self.delta_s:
+                self._seal_locked()
+                self._current = EventArrayBuilder(self.paths, self.comms)
+                self._current_t0 = ts
+            self._current.add(ts=ts, pid=pid, syscall=syscall, path=path,
+                              new_path=new_path, nbytes=nbytes, ret_val=ret_val, comm=comm)
+            self.total_events += 1
+            self._evict_locked(ts)
+
+    def append_wire_batch(self, events) -> None:
+        for ev in events:
+            self.append(
+                ts=ev.timestamp, pid=ev.pid, syscall=ev.syscall, path=ev.path,
+                new_path=ev.new_path, nbytes=ev.bytes, ret_val=ev.ret_val, comm=ev.comm,
+            )
+
+    def _seal_locked(self) -> None:
+        if self._current is not None and len(self._current):
+            self._deltas.append(self._current.build(sort=False))
+        self._current = None
+
+    def _evict_locked(self, now: float) -> None:
+        while self._deltas and len(self._deltas[0]) and float(self._deltas[0].ts[-1]) < now - self.window_s:
+            gone = self._deltas.popleft()
+            self.evicted_events += len(gone)
+
+    def compact(self, now: Optional[float] = None) -> EventArray:
+        """Merge the window's deltas into one time-sorted columnar batch."""
+        with self._lock:
+            self._seal_locked()
+            deltas = list(self._deltas)
+        if not deltas:
+            return EventArrayBuilder(self.paths, self.comms).build()
+        cols = {
+            "ts": np.concatenate([d.ts for d in deltas]),
+            "pid": np.concatenate([d.pid for d in deltas]),
+            "syscall": np.concatenate([d.syscall for d in deltas]),
+            "path_id": np.concatenate([d.path_id for d in deltas]),
+            "new_path_id": np.concatenate([d.new_path_id for d in deltas]),
+            "nbytes": np.concatenate([d.nbytes for d in deltas]),
+            "ret_val": np.concatenate([d.ret_val for d in deltas]),
+            "comm_id": np.concatenate([d.comm_id for d in deltas]),
+        }
+        if now is not None:
+            keep = cols["ts"] >= now - self.window_s
+            cols = {k: v[keep] for k, v in cols.items()}
+        arr = EventArray(paths=self.paths, comms=self.comms, **cols)
+        return arr.sort_by_time()
+
+    @property
+    def window_event_count(self) -> int:
+        with self._lock:
+            n = sum(len(d) for d in self._deltas)
+            if self._current is not None:
+                n += len(self._current)
+            return n

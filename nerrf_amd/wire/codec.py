@@ -254,12 +254,20 @@ def decode_event_batch(buf: bytes) -> List[Event]:
     return events
 
 
-def encode_empty() -> bytes:
+class Empty:
+    """nerrf.trace.Empty (no fields).  A real instance (not None) because
+    grpcio interprets a None deserializer result as a deserialization error."""
+
+    def __eq__(self, other) -> bool:
+        return isinstance(other, Empty)
+
+
+def encode_empty(_msg: "Empty | None" = None) -> bytes:
     return b""
 
 
-def decode_empty(buf: bytes) -> None:  # noqa: ARG001 - contract signature
-    return None
+def decode_empty(buf: bytes) -> Empty:  # noqa: ARG001 - contract signature
+    return Empty()
 
 
 STREAM_EVENTS_METHOD = "/nerrf.trace.Tracker/StreamEvents"
