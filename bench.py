@@ -82,6 +82,9 @@ def main() -> None:
     if has_gpu:
         torch.cuda.set_device(local_rank)
         device = torch.device(f"cuda:{local_rank}")
+        from nerrf_amd.perf import enable_tuned_gemms
+
+        enable_tuned_gemms()
     else:
         device = torch.device("cpu")
     dtype = torch.bfloat16 if (args.dtype == "bf16" and has_gpu) else torch.float32
