@@ -23,6 +23,9 @@ void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
                                const void*, const float*, void*, void*, void*,
                                long, int, bool, hipStream_t);
+void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
+                            const void*, const float*, void*, void*, void*,
+                            int, bool, hipStream_t);
 }  // namespace nerrf
 
 namespace {
@@ -139,6 +142,35 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_c,
       is_bf16(grad_h), stream.stream());
 }
 
+// Fully-fused MFMA step (bf16, H == 256). Writes into caller buffers.
+void lstm_step_fused(torch::Tensor h_prev, torch::Tensor w_hh, torch::Tensor xg,
+                     torch::Tensor bias, torch::Tensor c_prev,
+                     torch::Tensor mask, torch::Tensor h_out,
+                     torch::Tensor c_out, torch::Tensor gates_act, bool raw) {
+  for (auto* t : {&h_prev, &w_hh, &xg, &c_prev, &h_out, &c_out, &gates_act}) {
+    check_gpu_contig(*t, "lstm_step_fused arg");
+    TORCH_CHECK(t->scalar_type() == torch::kBFloat16,
+                "lstm_step_fused is bf16-only");
+  }
+  const int batch = h_prev.size(0);
+  TORCH_CHECK(h_prev.size(1) == 256 && w_hh.size(0) == 1024 &&
+                  w_hh.size(1) == 256 && xg.size(1) == 1024,
+              "lstm_step_fused requires H=256 (got h ", h_prev.size(1), ")");
+  const float* mask_ptr = nullptr;
+  torch::Tensor mf;
+  if (mask.numel() > 0) {
+    mf = mask.scalar_type() == torch::kFloat32 ? mask.contiguous()
+                                               : mask.to(torch::kFloat32).contiguous();
+    mask_ptr = mf.data_ptr<float>();
+  }
+  auto bc = bias.contiguous();
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_lstm_step_fused(
+      h_prev.data_ptr(), w_hh.data_ptr(), xg.data_ptr(), bc.data_ptr(),
+      c_prev.data_ptr(), mask_ptr, h_out.data_ptr(), c_out.data_ptr(),
+      gates_act.data_ptr(), batch, raw, stream.stream());
+}
+
 nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
   nerrf::PlannerParamsDev p;
   p.n_groups = d["n_groups"].cast<int>();
@@ -210,5 +242,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_mean_fwd", &gather_mean_fwd, "weighted neighbor gather-mean");
   m.def("gather_mean_bwd", &gather_mean_bwd, "gather-mean backward");
   m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "fused LSTM gate pointwise fwd");
+  m.def("lstm_step_fused", &lstm_step_fused, "fully-fused MFMA LSTM step (bf16, H=256)");
   m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "fused LSTM gate pointwise bwd");
 }

@@ -40,18 +40,31 @@ class _LSTMSeqFn(torch.autograd.Function):
         c = c0.contiguous()
         steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
         if ext is not None:
-            w_hh_t = w_hh.t().contiguous()
-            hg = torch.empty(batch, gdim, device=dev, dtype=dt)
             empty_mask = torch.empty(0, device=dev)
-            for ti in steps:
-                torch.mm(h, w_hh_t, out=hg)
-                ext.lstm_pointwise_fwd(
-                    hg, xg[ti], bias, c, h,
-                    mask[ti] if mask is not None else empty_mask,
-                    h_all[ti], c_all[ti], gates_all[ti],
-                )
-                h = h_all[ti]
-                c = c_all[ti]
+            fused = dt == torch.bfloat16 and hdim == 256 and w_hh.is_contiguous()
+            if fused:
+                # fully-fused MFMA step: no separate GEMM, no gates HBM round trip
+                bias_c = bias.contiguous()
+                for ti in steps:
+                    ext.lstm_step_fused(
+                        h, w_hh, xg[ti], bias_c, c,
+                        mask[ti] if mask is not None else empty_mask,
+                        h_all[ti], c_all[ti], gates_all[ti], False,
+                    )
+                    h = h_all[ti]
+                    c = c_all[ti]
+            else:
+                w_hh_t = w_hh.t().contiguous()
+                hg = torch.empty(batch, gdim, device=dev, dtype=dt)
+                for ti in steps:
+                    torch.mm(h, w_hh_t, out=hg)
+                    ext.lstm_pointwise_fwd(
+                        hg, xg[ti], bias, c, h,
+                        mask[ti] if mask is not None else empty_mask,
+                        h_all[ti], c_all[ti], gates_all[ti],
+                    )
+                    h = h_all[ti]
+                    c = c_all[ti]
         else:
             for ti in steps:
                 gates_pre = torch.addmm(bias, h, w_hh.t()) + xg[ti]

@@ -249,3 +249,77 @@ def test_mcts_search_gpu_agrees_with_cpu(gpu_device):
     # same dominant action and comparable plan quality
     assert gpu.ranked_actions[0][0] == cpu.ranked_actions[0][0] == A_KILL
     assert abs(gpu.root_value - cpu.root_value) < 1.0
+
+
+def test_lstm_step_fused_raw_gemm_matches_mm(gpu_device):
+    """RAW mode: fused-kernel gates_pre == h @ W_hh^T + xg + bias (bf16 MFMA
+    vs hipBLASLt).  Asymmetric operands so operand/output transposes are
+    caught (guide rule G9)."""
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(11)
+    b, hd = 333, 256  # non-multiple of 32 exercises row guards
+    h = (torch.randn(b, hd, device=gpu_device) * 0.3).to(torch.bfloat16)
+    w = (torch.randn(4 * hd, hd, device=gpu_device) * 0.1).to(torch.bfloat16)
+    w += torch.arange(4 * hd, device=gpu_device).unsqueeze(1).to(torch.bfloat16) * 1e-4  # asymmetry
+    xg = (torch.randn(b, 4 * hd, device=gpu_device) * 0.2).to(torch.bfloat16)
+    bias = torch.randn(4 * hd, device=gpu_device).to(torch.bfloat16)
+    c = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    h_out = torch.empty_like(c)
+    c_out = torch.empty_like(c)
+    gates = torch.empty_like(xg)
+    ext.lstm_step_fused(h, w, xg, bias, c, torch.empty(0, device=gpu_device),
+                        h_out, c_out, gates, True)
+    ref = (h.float() @ w.float().t() + xg.float() + bias.float())
+    assert torch.allclose(gates.float(), ref, atol=8e-2, rtol=4e-2)
+
+
+def test_lstm_step_fused_full_matches_reference(gpu_device):
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(12)
+    b, hd = 256, 256
+    h = (torch.randn(b, hd, device=gpu_device) * 0.3).to(torch.bfloat16)
+    w = (torch.randn(4 * hd, hd, device=gpu_device) * 0.1).to(torch.bfloat16)
+    xg = (torch.randn(b, 4 * hd, device=gpu_device) * 0.2).to(torch.bfloat16)
+    bias = torch.randn(4 * hd, device=gpu_device).to(torch.bfloat16)
+    c = (torch.randn(b, hd, device=gpu_device) * 0.5).to(torch.bfloat16)
+    mask = (torch.rand(b, device=gpu_device) > 0.3).float()
+    h_out = torch.empty_like(c)
+    c_out = torch.empty_like(c)
+    gates = torch.empty_like(xg)
+    ext.lstm_step_fused(h, w, xg, bias, c, mask, h_out, c_out, gates, False)
+
+    gp = h.float().cpu() @ w.float().cpu().t() + xg.float().cpu() + bias.float().cpu()
+    h_ref, c_ref, g_ref = ref.lstm_pointwise_fwd_ref(gp, c.float().cpu(), h.float().cpu(), mask.cpu())
+    assert torch.allclose(h_out.float().cpu(), h_ref, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(c_out.float().cpu(), c_ref, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(gates.float().cpu(), g_ref, atol=5e-2, rtol=5e-2)
+
+
+def test_lstm_sequence_bf16_fused_path(gpu_device):
+    """bf16 lstm_sequence (fused MFMA step path) vs fp32 CPU loop."""
+    from nerrf_amd.ops import lstm_sequence
+    from nerrf_amd.ops.reference import lstm_pointwise_fwd_ref
+
+    torch.manual_seed(13)
+    t, b, hd = 6, 64, 256
+    xg32 = torch.randn(t, b, 4 * hd, device=gpu_device) * 0.3
+    w32 = torch.randn(4 * hd, hd, device=gpu_device) * 0.1
+    bias32 = torch.randn(4 * hd, device=gpu_device) * 0.3
+    mask = (torch.rand(t, b, device=gpu_device) > 0.2).float()
+    h0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    c0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    out = lstm_sequence(xg32.to(torch.bfloat16), h0, c0, w32.to(torch.bfloat16),
+                        bias32.to(torch.bfloat16), mask, reverse=False)
+    h = torch.zeros(b, hd)
+    c = torch.zeros(b, hd)
+    outs = []
+    for ti in range(t):
+        gp = h @ w32.cpu().t() + xg32[ti].cpu() + bias32.cpu()
+        h, c, _ = lstm_pointwise_fwd_ref(gp, c, h, mask[ti].cpu())
+        outs.append(h)
+    ref_out = torch.stack(outs)
+    assert torch.allclose(out.float().cpu(), ref_out, atol=1e-1, rtol=1e-1)
