@@ -153,9 +153,12 @@ void lstm_step_fused(torch::Tensor h_prev, torch::Tensor w_hh, torch::Tensor xg,
                 "lstm_step_fused is bf16-only");
   }
   const int batch = h_prev.size(0);
-  TORCH_CHECK(h_prev.size(1) == 256 && w_hh.size(0) == 1024 &&
-                  w_hh.size(1) == 256 && xg.size(1) == 1024,
+  TORCH_CHECK(h_prev.size(1) == 256 && xg.size(1) == 1024,
               "lstm_step_fused requires H=256 (got h ", h_prev.size(1), ")");
+  TORCH_CHECK(w_hh.dim() == 3 && w_hh.size(0) == 8 && w_hh.size(1) == 1024 &&
+                  w_hh.size(2) == 32,
+              "w must be k-tiled [8,1024,32] (w_hh.reshape(1024,8,32)"
+              ".permute(1,0,2).contiguous())");
   const float* mask_ptr = nullptr;
   torch::Tensor mf;
   if (mask.numel() > 0) {

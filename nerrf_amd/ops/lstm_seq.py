@@ -45,9 +45,11 @@ class _LSTMSeqFn(torch.autograd.Function):
             if fused:
                 # fully-fused MFMA step: no separate GEMM, no gates HBM round trip
                 bias_c = bias.contiguous()
+                # k-slice-contiguous weight tiling (see lstm_step_fused.hip)
+                w_tiled = w_hh.reshape(gdim, hdim // 32, 32).permute(1, 0, 2).contiguous()
                 for ti in steps:
                     ext.lstm_step_fused(
-                        h, w_hh, xg[ti], bias_c, c,
+                        h, w_tiled, xg[ti], bias_c, c,
                         mask[ti] if mask is not None else empty_mask,
                         h_all[ti], c_all[ti], gates_all[ti], False,
                     )

@@ -269,7 +269,8 @@ def test_lstm_step_fused_raw_gemm_matches_mm(gpu_device):
     h_out = torch.empty_like(c)
     c_out = torch.empty_like(c)
     gates = torch.empty_like(xg)
-    ext.lstm_step_fused(h, w, xg, bias, c, torch.empty(0, device=gpu_device),
+    w_tiled = w.reshape(4 * hd, hd // 32, 32).permute(1, 0, 2).contiguous()
+    ext.lstm_step_fused(h, w_tiled, xg, bias, c, torch.empty(0, device=gpu_device),
                         h_out, c_out, gates, True)
     ref = (h.float() @ w.float().t() + xg.float() + bias.float())
     assert torch.allclose(gates.float(), ref, atol=8e-2, rtol=4e-2)
@@ -290,7 +291,8 @@ def test_lstm_step_fused_full_matches_reference(gpu_device):
     h_out = torch.empty_like(c)
     c_out = torch.empty_like(c)
     gates = torch.empty_like(xg)
-    ext.lstm_step_fused(h, w, xg, bias, c, mask, h_out, c_out, gates, False)
+    w_tiled = w.reshape(4 * hd, hd // 32, 32).permute(1, 0, 2).contiguous()
+    ext.lstm_step_fused(h, w_tiled, xg, bias, c, mask, h_out, c_out, gates, False)
 
     gp = h.float().cpu() @ w.float().cpu().t() + xg.float().cpu() + bias.float().cpu()
     h_ref, c_ref, g_ref = ref.lstm_pointwise_fwd_ref(gp, c.float().cpu(), h.float().cpu(), mask.cpu())
