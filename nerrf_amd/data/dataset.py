@@ -13,7 +13,7 @@ import numpy as np
 import torch
 
 from ..graph.constructor import build_graph, sliding_windows
-from ..graph.sampling import sample_fanout, to_csr
+from ..graph.sampling import reverse_index, sample_fanout, to_csr
 from .labels import event_labels
 from .sequences import build_sequences
 from .synth import AttackWindow, SynthConfig, generate
@@ -37,6 +37,9 @@ class WindowBatch:
     y_seq: np.ndarray
     n_events: int
     seq_path_id: Optional[np.ndarray] = None  # [B] path id per sequence
+    rev_indptr: Optional[np.ndarray] = None  # reverse CSR for gather backward
+    rev_src: Optional[np.ndarray] = None
+    rev_w: Optional[np.ndarray] = None
 
     def to_torch(self, device="cpu", dtype=torch.float32) -> Dict[str, torch.Tensor]:
         def t(a, dt=None):
@@ -58,6 +61,9 @@ class WindowBatch:
             "seq_lengths": t(self.seq_lengths),
             "y_seq": t(self.y_seq, torch.float32),
             "n_events": torch.tensor(self.n_events),
+            "nbr_rev": None
+            if self.rev_indptr is None
+            else (t(self.rev_indptr), t(self.rev_src), t(self.rev_w, torch.float32)),
         }
 
 
@@ -72,6 +78,7 @@ def window_to_batch(
     g = build_graph(events, window, y_ev)
     csr = to_csr(g.edge_index, g.num_nodes, g.edge_weight)
     nbr_idx, nbr_w = sample_fanout(csr, fanout, seed=seed)
+    rev_indptr, rev_src, rev_w = reverse_index(nbr_idx, nbr_w)
     seqs = build_sequences(events, y_ev, seq_len=seq_len)
     return WindowBatch(
         x=g.x,
@@ -87,6 +94,9 @@ def window_to_batch(
         y_seq=seqs.labels if seqs.labels is not None else np.zeros(len(seqs.lengths), dtype=np.float32),
         n_events=len(events),
         seq_path_id=seqs.file_path_id,
+        rev_indptr=rev_indptr,
+        rev_src=rev_src,
+        rev_w=rev_w,
     )
 
 

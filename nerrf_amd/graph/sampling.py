@@ -84,3 +84,23 @@ def sample_fanout(
             idx[isolated] = np.arange(n, dtype=np.int64)[isolated, None]
             w[isolated] = 1.0
     return idx, w
+
+
+def reverse_index(idx: np.ndarray, w: np.ndarray) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Reverse CSR of a sampled-fanout matrix for the deterministic backward.
+
+    Returns (rev_indptr [N+1], rev_src [N*K], rev_w [N*K]) where for target
+    node m, entries rev_indptr[m]:rev_indptr[m+1] list the (source row n,
+    normalised weight w/denom) pairs that gathered from m.  Built once per
+    window batch; reused by every layer and training step on that batch.
+    """
+    n, k = idx.shape
+    denom = np.maximum(w.sum(axis=1, keepdims=True), 1e-6)
+    wn = (w / denom).astype(np.float32).reshape(-1)
+    flat_dst = idx.reshape(-1)
+    src_n = np.repeat(np.arange(n, dtype=np.int64), k)
+    order = np.argsort(flat_dst, kind="stable")
+    counts = np.bincount(flat_dst, minlength=n)
+    rev_indptr = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(counts, out=rev_indptr[1:])
+    return rev_indptr, src_n[order], wn[order]

@@ -18,7 +18,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import gather_mean
+from ..ops import gather_mean, gather_rows
 
 
 @dataclass
@@ -41,8 +41,8 @@ class SageLayer(nn.Module):
         self.norm = nn.LayerNorm(dim)
         self.dropout = dropout
 
-    def forward(self, h: torch.Tensor, nbr_idx: torch.Tensor, nbr_w: torch.Tensor) -> torch.Tensor:
-        agg = gather_mean(h, nbr_idx, nbr_w)
+    def forward(self, h: torch.Tensor, nbr_idx: torch.Tensor, nbr_w: torch.Tensor, rev=None) -> torch.Tensor:
+        agg = gather_mean(h, nbr_idx, nbr_w, rev)
         z = self.w_self(h) + self.w_nbr(agg)
         z = F.gelu(z)
         if self.dropout > 0 and self.training:
@@ -65,10 +65,10 @@ class GraphSAGET(nn.Module):
             nn.Linear(c.edge_head_hidden, 1),
         )
 
-    def encode(self, x: torch.Tensor, nbr_idx: torch.Tensor, nbr_w: torch.Tensor) -> torch.Tensor:
+    def encode(self, x: torch.Tensor, nbr_idx: torch.Tensor, nbr_w: torch.Tensor, rev=None) -> torch.Tensor:
         h = self.input_proj(x)
         for layer in self.layers:
-            h = layer(h, nbr_idx, nbr_w)
+            h = layer(h, nbr_idx, nbr_w, rev)
         return h
 
     def forward(
@@ -79,12 +79,13 @@ class GraphSAGET(nn.Module):
         edge_index: torch.Tensor | None = None,  # [2, E]
         edge_weight: torch.Tensor | None = None,  # [E]
         edge_ts: torch.Tensor | None = None,  # [E]
+        nbr_rev=None,  # optional reverse CSR (sampling.reverse_index tensors)
     ):
-        h = self.encode(x, nbr_idx, nbr_w)
+        h = self.encode(x, nbr_idx, nbr_w, nbr_rev)
         node_logit = self.node_head(h).squeeze(-1)
         edge_logit = None
         if edge_index is not None and edge_index.numel():
-            hs, hd = h[edge_index[0]], h[edge_index[1]]
+            hs, hd = gather_rows(h, edge_index[0]), gather_rows(h, edge_index[1])
             ew = edge_weight if edge_weight is not None else torch.ones(
                 edge_index.shape[1], device=h.device, dtype=h.dtype
             )

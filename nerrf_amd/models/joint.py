@@ -44,6 +44,7 @@ class NerrfJointModel(nn.Module):
             batch.get("edge_index"),
             batch.get("edge_weight"),
             batch.get("edge_ts"),
+            batch.get("nbr_rev"),
         )
         seq_logit = None
         if batch.get("seq_feats") is not None and batch["seq_feats"].shape[0] > 0:

@@ -17,6 +17,9 @@ void launch_gather_mean_fwd(const void*, const long*, const float*, void*, int,
                             int, int, bool, hipStream_t);
 void launch_gather_mean_bwd(const void*, const long*, const float*, float*,
                             int, int, int, bool, hipStream_t);
+void launch_gather_mean_bwd_csr(const void*, const long*, const long*,
+                                const float*, void*, int, int, bool,
+                                hipStream_t);
 void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
                                void*, void*, long, int, bool, hipStream_t);
@@ -78,6 +81,25 @@ torch::Tensor gather_mean_bwd(torch::Tensor grad_out, torch::Tensor idx,
                                 wf.data_ptr<float>(), ws.data_ptr<float>(), n,
                                 dim, k, is_bf16(grad_out), stream.stream());
   return ws.to(grad_out.scalar_type());
+}
+
+torch::Tensor gather_mean_bwd_csr(torch::Tensor grad_out,
+                                  torch::Tensor rev_indptr,
+                                  torch::Tensor rev_src, torch::Tensor rev_w,
+                                  long num_nodes) {
+  check_gpu_contig(grad_out, "grad_out");
+  check_gpu_contig(rev_indptr, "rev_indptr");
+  check_gpu_contig(rev_src, "rev_src");
+  check_gpu_contig(rev_w, "rev_w");
+  TORCH_CHECK(rev_indptr.numel() == num_nodes + 1, "rev_indptr must be [M+1]");
+  const int dim = grad_out.size(1);
+  auto gh = torch::empty({num_nodes, (long)dim}, grad_out.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_gather_mean_bwd_csr(
+      grad_out.data_ptr(), rev_indptr.data_ptr<long>(),
+      rev_src.data_ptr<long>(), rev_w.data_ptr<float>(), gh.data_ptr(),
+      (int)num_nodes, dim, is_bf16(grad_out), stream.stream());
+  return gh;
 }
 
 // Writes into caller-provided (h_out, c_out, gates_act) so the sequence loop
@@ -244,6 +266,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mcts_eval_plans", &mcts_eval_plans, "batch plan reward evaluation");
   m.def("gather_mean_fwd", &gather_mean_fwd, "weighted neighbor gather-mean");
   m.def("gather_mean_bwd", &gather_mean_bwd, "gather-mean backward");
+  m.def("gather_mean_bwd_csr", &gather_mean_bwd_csr,
+        "deterministic gather-mean backward over reverse CSR");
   m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "fused LSTM gate pointwise fwd");
   m.def("lstm_step_fused", &lstm_step_fused, "fully-fused MFMA LSTM step (bf16, H=256)");
   m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "fused LSTM gate pointwise bwd");

@@ -153,6 +153,59 @@ template __global__ void gather_mean_bwd_kernel<float>(
 template __global__ void gather_mean_bwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const long*, const float*, float*, int, int, int);
 
+// Deterministic backward over a precomputed reverse CSR: for node m,
+//   grad_h[m, :] = sum_e rev_w[e] * grad_out[rev_src[e], :]
+// (rev_w already contains w/denom).  No atomics, coalesced row writes; the
+// reverse index is built once per window batch on the host and reused by all
+// 28 layers and every training step on that batch.
+template <typename T>
+__global__ void gather_mean_bwd_csr_kernel(
+    const T* __restrict__ grad_out,      // [N, D]
+    const long* __restrict__ rev_indptr, // [M+1]
+    const long* __restrict__ rev_src,    // [E]
+    const float* __restrict__ rev_w,     // [E]
+    T* __restrict__ grad_h,              // [M, D]
+    int m_nodes, int dim) {
+  const int wave_in_block = threadIdx.x / NERRF_WAVE;
+  const int lane = threadIdx.x % NERRF_WAVE;
+  const int waves_per_block = blockDim.x / NERRF_WAVE;
+  for (int m = blockIdx.x * waves_per_block + wave_in_block; m < m_nodes;
+       m += gridDim.x * waves_per_block) {
+    const long e0 = rev_indptr[m];
+    const long e1 = rev_indptr[m + 1];
+    float acc[8];
+    const int cols = (dim + NERRF_WAVE - 1) / NERRF_WAVE;
+#pragma unroll
+    for (int cc = 0; cc < 8; ++cc) acc[cc] = 0.0f;
+    for (long base = e0; base < e1; base += NERRF_WAVE) {
+      // lanes cooperatively read up to 64 (src, w) pairs, then broadcast
+      const long e = base + lane;
+      const float w_lane = (e < e1) ? rev_w[e] : 0.0f;
+      const long s_lane = (e < e1) ? rev_src[e] : 0;
+      const int cnt = (int)((e1 - base < NERRF_WAVE) ? (e1 - base) : NERRF_WAVE);
+      for (int j = 0; j < cnt; ++j) {
+        const float wk = __shfl(w_lane, j, NERRF_WAVE);
+        const long src = __shfl(s_lane, j, NERRF_WAVE);
+        const T* row = grad_out + (long)src * dim;
+        for (int cc = 0; cc < cols; ++cc) {
+          const int d = lane + cc * NERRF_WAVE;
+          if (d < dim) acc[cc] = fmaf(wk, to_f32(row[d]), acc[cc]);
+        }
+      }
+    }
+    T* orow = grad_h + (long)m * dim;
+    for (int cc = 0; cc < cols; ++cc) {
+      const int d = lane + cc * NERRF_WAVE;
+      if (d < dim) orow[d] = from_f32<T>(acc[cc]);
+    }
+  }
+}
+
+template __global__ void gather_mean_bwd_csr_kernel<float>(
+    const float*, const long*, const long*, const float*, float*, int, int);
+template __global__ void gather_mean_bwd_csr_kernel<__hip_bfloat16>(
+    const __hip_bfloat16*, const long*, const long*, const float*, __hip_bfloat16*, int, int);
+
 // ---------------------------------------------------------------------------
 // host launchers
 // ---------------------------------------------------------------------------
@@ -180,6 +233,23 @@ void launch_gather_mean_fwd(const void* h, const long* idx, const float* w,
   } else {
     gather_mean_fwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)h, idx, w, (float*)out, n, dim, k);
+  }
+}
+
+void launch_gather_mean_bwd_csr(const void* gout, const long* rev_indptr,
+                                const long* rev_src, const float* rev_w,
+                                void* gh, int m_nodes, int dim, bool bf16,
+                                hipStream_t s) {
+  const int block = 256;
+  const int wpb = block / NERRF_WAVE;
+  const int grid = grid_for(m_nodes, wpb);
+  if (bf16) {
+    gather_mean_bwd_csr_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
+        (const __hip_bfloat16*)gout, rev_indptr, rev_src, rev_w,
+        (__hip_bfloat16*)gh, m_nodes, dim);
+  } else {
+    gather_mean_bwd_csr_kernel<float><<<grid, block, 0, s>>>(
+        (const float*)gout, rev_indptr, rev_src, rev_w, (float*)gh, m_nodes, dim);
   }
 }
 

@@ -41,7 +41,17 @@ class _LSTMSeqFn(torch.autograd.Function):
         steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
         if ext is not None:
             empty_mask = torch.empty(0, device=dev)
-            fused = dt == torch.bfloat16 and hdim == 256 and w_hh.is_contiguous()
+            # The fully-fused MFMA step (lstm_step_fused.hip) is numerically
+            # validated but measured slower than hipBLASLt-GEMM + fused
+            # pointwise at this shape (215us vs 48us per step: 1 block/CU and
+            # a single-buffered W slice expose full L2 latency); keep it
+            # opt-in until the pipelined variant lands.
+            import os
+
+            fused = (
+                os.environ.get("NERRF_FUSED_LSTM", "0") == "1"
+                and dt == torch.bfloat16 and hdim == 256 and w_hh.is_contiguous()
+            )
             if fused:
                 # fully-fused MFMA step: no separate GEMM, no gates HBM round trip
                 bias_c = bias.contiguous()

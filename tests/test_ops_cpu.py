@@ -223,3 +223,20 @@ def test_lstm_sequence_reverse_backward():
     auto = [p.grad for p in (xg, w_hh, bias)]
     for gm, ga, name in zip(manual, auto, ["xg", "w_hh", "bias"]):
         assert torch.allclose(gm, ga, atol=1e-5), f"grad {name}"
+
+
+def test_reverse_index_consistent_with_bwd_ref():
+    from nerrf_amd.graph.sampling import reverse_index
+
+    torch.manual_seed(5)
+    n, k, d = 29, 6, 9
+    idx = torch.randint(0, n, (n, k))
+    w = torch.rand(n, k) + 0.05
+    g = torch.randn(n, d)
+    expected = gather_mean_bwd_ref(g, idx, w, n)
+    rp, rs, rw = reverse_index(idx.numpy(), w.numpy())
+    out = torch.zeros(n, d)
+    for m in range(n):
+        for e in range(rp[m], rp[m + 1]):
+            out[m] += float(rw[e]) * g[int(rs[e])]
+    assert torch.allclose(out, expected, atol=1e-5)

@@ -325,3 +325,44 @@ def test_lstm_sequence_bf16_fused_path(gpu_device):
         outs.append(h)
     ref_out = torch.stack(outs)
     assert torch.allclose(out.float().cpu(), ref_out, atol=1e-1, rtol=1e-1)
+
+
+def test_gather_mean_bwd_csr_matches_ref(gpu_device):
+    from nerrf_amd.graph.sampling import reverse_index
+    from nerrf_amd.ops import gather_mean
+
+    torch.manual_seed(14)
+    n, k, d = 411, 16, 128
+    h = torch.randn(n, d, device=gpu_device, requires_grad=True)
+    idx = torch.randint(0, n, (n, k), device=gpu_device)
+    w = torch.rand(n, k, device=gpu_device) + 0.05
+    rp, rs, rw = reverse_index(idx.cpu().numpy(), w.cpu().numpy())
+    rev = (
+        torch.from_numpy(rp).to(gpu_device),
+        torch.from_numpy(rs).to(gpu_device),
+        torch.from_numpy(rw).to(gpu_device),
+    )
+    out = gather_mean(h, idx, w, rev)
+    g = torch.randn_like(out)
+    out.backward(g)
+    grad_ref = ref.gather_mean_bwd_ref(g.cpu(), idx.cpu(), w.cpu(), n)
+    assert torch.allclose(h.grad.cpu(), grad_ref, atol=1e-4, rtol=1e-4)
+
+
+def test_gather_rows_matches_indexing(gpu_device):
+    from nerrf_amd.ops import gather_rows
+
+    torch.manual_seed(15)
+    n, d, e = 200, 128, 5000
+    h = torch.randn(n, d, device=gpu_device, requires_grad=True)
+    idx = torch.randint(0, n, (e,), device=gpu_device)
+    out = gather_rows(h, idx)
+    g = torch.randn_like(out)
+    out.backward(g)
+    grad_fast = h.grad.clone()
+
+    h2 = h.detach().clone().requires_grad_(True)
+    out2 = h2[idx]
+    assert torch.allclose(out.detach(), out2.detach())
+    out2.backward(g)
+    assert torch.allclose(grad_fast, h2.grad, atol=1e-3, rtol=1e-3)
