@@ -37,7 +37,7 @@ class WindowBatch:
     y_seq: np.ndarray
     n_events: int
     seq_path_id: Optional[np.ndarray] = None  # [B] path id per sequence
-    rev_indptr: Optional[np.ndarray] = None  # reverse CSR for gather backward
+    rev_dst: Optional[np.ndarray] = None  # reverse index for gather backward
     rev_src: Optional[np.ndarray] = None
     rev_w: Optional[np.ndarray] = None
 
@@ -62,8 +62,8 @@ class WindowBatch:
             "y_seq": t(self.y_seq, torch.float32),
             "n_events": torch.tensor(self.n_events),
             "nbr_rev": None
-            if self.rev_indptr is None
-            else (t(self.rev_indptr), t(self.rev_src), t(self.rev_w, torch.float32)),
+            if self.rev_dst is None
+            else (t(self.rev_dst), t(self.rev_src), t(self.rev_w, torch.float32)),
         }
 
 
@@ -78,7 +78,7 @@ def window_to_batch(
     g = build_graph(events, window, y_ev)
     csr = to_csr(g.edge_index, g.num_nodes, g.edge_weight)
     nbr_idx, nbr_w = sample_fanout(csr, fanout, seed=seed)
-    rev_indptr, rev_src, rev_w = reverse_index(nbr_idx, nbr_w)
+    rev_dst, rev_src, rev_w = reverse_index(nbr_idx, nbr_w)
     seqs = build_sequences(events, y_ev, seq_len=seq_len)
     return WindowBatch(
         x=g.x,
@@ -94,7 +94,7 @@ def window_to_batch(
         y_seq=seqs.labels if seqs.labels is not None else np.zeros(len(seqs.lengths), dtype=np.float32),
         n_events=len(events),
         seq_path_id=seqs.file_path_id,
-        rev_indptr=rev_indptr,
+        rev_dst=rev_dst,
         rev_src=rev_src,
         rev_w=rev_w,
     )

@@ -87,12 +87,13 @@ def sample_fanout(
 
 
 def reverse_index(idx: np.ndarray, w: np.ndarray) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
-    """Reverse CSR of a sampled-fanout matrix for the deterministic backward.
+    """Reverse index of a sampled-fanout matrix for the backward kernel.
 
-    Returns (rev_indptr [N+1], rev_src [N*K], rev_w [N*K]) where for target
-    node m, entries rev_indptr[m]:rev_indptr[m+1] list the (source row n,
-    normalised weight w/denom) pairs that gathered from m.  Built once per
-    window batch; reused by every layer and training step on that batch.
+    Returns (rev_dst, rev_src, rev_w), each [N*K], sorted by destination:
+    entry e says "source row rev_src[e] gathered from node rev_dst[e] with
+    normalised weight rev_w[e] (= w/denom)".  Built once per window batch;
+    reused by every layer and training step on that batch.  The sort order
+    is what lets the GPU backward run as a segmented reduce.
     """
     n, k = idx.shape
     denom = np.maximum(w.sum(axis=1, keepdims=True), 1e-6)
@@ -100,7 +101,4 @@ def reverse_index(idx: np.ndarray, w: np.ndarray) -> Tuple[np.ndarray, np.ndarra
     flat_dst = idx.reshape(-1)
     src_n = np.repeat(np.arange(n, dtype=np.int64), k)
     order = np.argsort(flat_dst, kind="stable")
-    counts = np.bincount(flat_dst, minlength=n)
-    rev_indptr = np.zeros(n + 1, dtype=np.int64)
-    np.cumsum(counts, out=rev_indptr[1:])
-    return rev_indptr, src_n[order], wn[order]
+    return flat_dst[order], src_n[order], wn[order]

@@ -23,9 +23,9 @@ __all__ = ["gather_mean", "lstm_cell", "lstm_sequence", "native_available"]
 
 class _GatherMeanFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, h, idx, w, rev_indptr=None, rev_src=None, rev_w=None):
-        if rev_indptr is not None:
-            ctx.save_for_backward(idx, w, rev_indptr, rev_src, rev_w)
+    def forward(ctx, h, idx, w, rev_dst=None, rev_src=None, rev_w=None):
+        if rev_dst is not None:
+            ctx.save_for_backward(idx, w, rev_dst, rev_src, rev_w)
         else:
             ctx.save_for_backward(idx, w)
         ctx.num_nodes = h.shape[0]
@@ -57,8 +57,8 @@ def gather_mean(h: torch.Tensor, idx: torch.Tensor, w: torch.Tensor, rev=None) -
 
     h: [N, D] node features; idx: [N, K] int64; w: [N, K] float (treated as
     constants — causality weights are data, not parameters).
-    rev: optional (rev_indptr, rev_src, rev_w) reverse CSR (graph.sampling
-    .reverse_index) enabling the deterministic atomic-free backward kernel.
+    rev: optional (rev_dst, rev_src, rev_w) reverse index (graph.sampling
+    .reverse_index) enabling the load-balanced segmented-reduce backward.
     """
     if rev is not None:
         return _GatherMeanFn.apply(

@@ -18,7 +18,7 @@ void launch_gather_mean_fwd(const void*, const long*, const float*, void*, int,
 void launch_gather_mean_bwd(const void*, const long*, const float*, float*,
                             int, int, int, bool, hipStream_t);
 void launch_gather_mean_bwd_csr(const void*, const long*, const long*,
-                                const float*, void*, int, int, bool,
+                                const float*, float*, long, int, bool,
                                 hipStream_t);
 void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
@@ -84,22 +84,23 @@ torch::Tensor gather_mean_bwd(torch::Tensor grad_out, torch::Tensor idx,
 }
 
 torch::Tensor gather_mean_bwd_csr(torch::Tensor grad_out,
-                                  torch::Tensor rev_indptr,
+                                  torch::Tensor rev_dst,
                                   torch::Tensor rev_src, torch::Tensor rev_w,
                                   long num_nodes) {
   check_gpu_contig(grad_out, "grad_out");
-  check_gpu_contig(rev_indptr, "rev_indptr");
+  check_gpu_contig(rev_dst, "rev_dst");
   check_gpu_contig(rev_src, "rev_src");
   check_gpu_contig(rev_w, "rev_w");
-  TORCH_CHECK(rev_indptr.numel() == num_nodes + 1, "rev_indptr must be [M+1]");
   const int dim = grad_out.size(1);
-  auto gh = torch::empty({num_nodes, (long)dim}, grad_out.options());
+  const long n_entries = rev_dst.numel();
+  auto ws = torch::zeros({num_nodes, (long)dim},
+                         grad_out.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   nerrf::launch_gather_mean_bwd_csr(
-      grad_out.data_ptr(), rev_indptr.data_ptr<long>(),
-      rev_src.data_ptr<long>(), rev_w.data_ptr<float>(), gh.data_ptr(),
-      (int)num_nodes, dim, is_bf16(grad_out), stream.stream());
-  return gh;
+      grad_out.data_ptr(), rev_dst.data_ptr<long>(), rev_src.data_ptr<long>(),
+      rev_w.data_ptr<float>(), ws.data_ptr<float>(), n_entries, dim,
+      is_bf16(grad_out), stream.stream());
+  return ws.to(grad_out.scalar_type());
 }
 
 // Writes into caller-provided (h_out, c_out, gates_act) so the sequence loop

@@ -153,58 +153,79 @@ template __global__ void gather_mean_bwd_kernel<float>(
 template __global__ void gather_mean_bwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const long*, const float*, float*, int, int, int);
 
-// Deterministic backward over a precomputed reverse CSR: for node m,
-//   grad_h[m, :] = sum_e rev_w[e] * grad_out[rev_src[e], :]
-// (rev_w already contains w/denom).  No atomics, coalesced row writes; the
-// reverse index is built once per window batch on the host and reused by all
-// 28 layers and every training step on that batch.
+// Backward over the reverse index (entries sorted by destination node):
+//   grad_h[m, :] += sum_{e: rev_dst[e]==m} rev_w[e] * grad_out[rev_src[e], :]
+// (rev_w already contains w/denom).  Load-balanced segmented reduce: each
+// wave owns a fixed 256-entry tile regardless of per-node in-degree (a
+// wave-per-node schedule serialises on hub nodes — a popular process node
+// can carry 10^4-10^5 reverse entries), accumulating in registers while the
+// destination stays constant and flushing with fp32 atomics at segment
+// boundaries (rare: sorted entries change destination at most
+// nodes-per-tile times).
 template <typename T>
 __global__ void gather_mean_bwd_csr_kernel(
-    const T* __restrict__ grad_out,      // [N, D]
-    const long* __restrict__ rev_indptr, // [M+1]
-    const long* __restrict__ rev_src,    // [E]
-    const float* __restrict__ rev_w,     // [E]
-    T* __restrict__ grad_h,              // [M, D]
-    int m_nodes, int dim) {
+    const T* __restrict__ grad_out,   // [N, D]
+    const long* __restrict__ rev_dst, // [E] sorted
+    const long* __restrict__ rev_src, // [E]
+    const float* __restrict__ rev_w,  // [E]
+    float* __restrict__ grad_h,       // [M, D] fp32 workspace (zeroed)
+    long n_entries, int dim) {
+  const int ENT = 256;  // entries per wave tile
   const int wave_in_block = threadIdx.x / NERRF_WAVE;
   const int lane = threadIdx.x % NERRF_WAVE;
   const int waves_per_block = blockDim.x / NERRF_WAVE;
-  for (int m = blockIdx.x * waves_per_block + wave_in_block; m < m_nodes;
-       m += gridDim.x * waves_per_block) {
-    const long e0 = rev_indptr[m];
-    const long e1 = rev_indptr[m + 1];
+  const long n_tiles = (n_entries + ENT - 1) / ENT;
+  const int cols = (dim + NERRF_WAVE - 1) / NERRF_WAVE;
+  for (long tile = blockIdx.x * waves_per_block + wave_in_block; tile < n_tiles;
+       tile += (long)gridDim.x * waves_per_block) {
+    const long e0 = tile * ENT;
+    const long e1 = (e0 + ENT < n_entries) ? e0 + ENT : n_entries;
     float acc[8];
-    const int cols = (dim + NERRF_WAVE - 1) / NERRF_WAVE;
 #pragma unroll
     for (int cc = 0; cc < 8; ++cc) acc[cc] = 0.0f;
+    long cur_m = -1;
     for (long base = e0; base < e1; base += NERRF_WAVE) {
-      // lanes cooperatively read up to 64 (src, w) pairs, then broadcast
       const long e = base + lane;
       const float w_lane = (e < e1) ? rev_w[e] : 0.0f;
       const long s_lane = (e < e1) ? rev_src[e] : 0;
+      const long m_lane = (e < e1) ? rev_dst[e] : -1;
       const int cnt = (int)((e1 - base < NERRF_WAVE) ? (e1 - base) : NERRF_WAVE);
       for (int j = 0; j < cnt; ++j) {
+        const long m = __shfl(m_lane, j, NERRF_WAVE);
+        if (m != cur_m) {
+          if (cur_m >= 0) {
+            float* dst = grad_h + cur_m * dim;
+            for (int cc = 0; cc < cols; ++cc) {
+              const int d = lane + cc * NERRF_WAVE;
+              if (d < dim && acc[cc] != 0.0f) atomicAdd(dst + d, acc[cc]);
+              acc[cc] = 0.0f;
+            }
+          }
+          cur_m = m;
+        }
         const float wk = __shfl(w_lane, j, NERRF_WAVE);
         const long src = __shfl(s_lane, j, NERRF_WAVE);
-        const T* row = grad_out + (long)src * dim;
+        const T* row = grad_out + src * dim;
         for (int cc = 0; cc < cols; ++cc) {
           const int d = lane + cc * NERRF_WAVE;
           if (d < dim) acc[cc] = fmaf(wk, to_f32(row[d]), acc[cc]);
         }
       }
     }
-    T* orow = grad_h + (long)m * dim;
-    for (int cc = 0; cc < cols; ++cc) {
-      const int d = lane + cc * NERRF_WAVE;
-      if (d < dim) orow[d] = from_f32<T>(acc[cc]);
+    if (cur_m >= 0) {
+      float* dst = grad_h + cur_m * dim;
+      for (int cc = 0; cc < cols; ++cc) {
+        const int d = lane + cc * NERRF_WAVE;
+        if (d < dim && acc[cc] != 0.0f) atomicAdd(dst + d, acc[cc]);
+      }
     }
   }
 }
 
 template __global__ void gather_mean_bwd_csr_kernel<float>(
-    const float*, const long*, const long*, const float*, float*, int, int);
+    const float*, const long*, const long*, const float*, float*, long, int);
 template __global__ void gather_mean_bwd_csr_kernel<__hip_bfloat16>(
-    const __hip_bfloat16*, const long*, const long*, const float*, __hip_bfloat16*, int, int);
+    const __hip_bfloat16*, const long*, const long*, const float*, float*, long, int);
 
 // ---------------------------------------------------------------------------
 // host launchers
@@ -236,20 +257,20 @@ void launch_gather_mean_fwd(const void* h, const long* idx, const float* w,
   }
 }
 
-void launch_gather_mean_bwd_csr(const void* gout, const long* rev_indptr,
+void launch_gather_mean_bwd_csr(const void* gout, const long* rev_dst,
                                 const long* rev_src, const float* rev_w,
-                                void* gh, int m_nodes, int dim, bool bf16,
+                                float* gh, long n_entries, int dim, bool bf16,
                                 hipStream_t s) {
   const int block = 256;
   const int wpb = block / NERRF_WAVE;
-  const int grid = grid_for(m_nodes, wpb);
+  const long n_tiles = (n_entries + 255) / 256;
+  const int grid = grid_for(n_tiles, wpb);
   if (bf16) {
     gather_mean_bwd_csr_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
-        (const __hip_bfloat16*)gout, rev_indptr, rev_src, rev_w,
-        (__hip_bfloat16*)gh, m_nodes, dim);
+        (const __hip_bfloat16*)gout, rev_dst, rev_src, rev_w, gh, n_entries, dim);
   } else {
     gather_mean_bwd_csr_kernel<float><<<grid, block, 0, s>>>(
-        (const float*)gout, rev_indptr, rev_src, rev_w, (float*)gh, m_nodes, dim);
+        (const float*)gout, rev_dst, rev_src, rev_w, gh, n_entries, dim);
   }
 }
 

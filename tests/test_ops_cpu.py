@@ -234,9 +234,9 @@ def test_reverse_index_consistent_with_bwd_ref():
     w = torch.rand(n, k) + 0.05
     g = torch.randn(n, d)
     expected = gather_mean_bwd_ref(g, idx, w, n)
-    rp, rs, rw = reverse_index(idx.numpy(), w.numpy())
+    rd, rs, rw = reverse_index(idx.numpy(), w.numpy())
+    assert (np.diff(rd) >= 0).all()  # sorted by destination
     out = torch.zeros(n, d)
-    for m in range(n):
-        for e in range(rp[m], rp[m + 1]):
-            out[m] += float(rw[e]) * g[int(rs[e])]
+    for e in range(len(rd)):
+        out[int(rd[e])] += float(rw[e]) * g[int(rs[e])]
     assert torch.allclose(out, expected, atol=1e-5)

@@ -336,9 +336,9 @@ def test_gather_mean_bwd_csr_matches_ref(gpu_device):
     h = torch.randn(n, d, device=gpu_device, requires_grad=True)
     idx = torch.randint(0, n, (n, k), device=gpu_device)
     w = torch.rand(n, k, device=gpu_device) + 0.05
-    rp, rs, rw = reverse_index(idx.cpu().numpy(), w.cpu().numpy())
+    rd, rs, rw = reverse_index(idx.cpu().numpy(), w.cpu().numpy())
     rev = (
-        torch.from_numpy(rp).to(gpu_device),
+        torch.from_numpy(rd).to(gpu_device),
         torch.from_numpy(rs).to(gpu_device),
         torch.from_numpy(rw).to(gpu_device),
     )
