@@ -21,6 +21,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import numpy as np
@@ -75,6 +76,12 @@ def main() -> None:
     ap.add_argument("--scale", choices=["small", "full"], default="full")
     ap.add_argument("--windows", type=int, default=2, help="prebuilt windows per rank")
     ap.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
+    ap.add_argument(
+        "--graphs",
+        choices=["auto", "on", "off"],
+        default="auto",
+        help="hipGraph-capture the whole training step (auto: single-GPU only)",
+    )
     args = ap.parse_args()
 
     rank, world, local_rank = init_distributed()
@@ -94,22 +101,60 @@ def main() -> None:
     batches = [b.to_torch(device=device, dtype=dtype) for b in batches_np]
     events_per_step = [int(b.n_events) for b in batches_np]
 
+    use_graphs = args.graphs == "on" or (
+        args.graphs == "auto" and world == 1 and has_gpu
+    )
     model = NerrfJointModel(JointConfig(sage=SageConfig(), lstm=LSTMConfig())).to(
         device=device, dtype=dtype
     )
-    opt = torch.optim.AdamW(model.parameters(), lr=1e-3, weight_decay=1e-4, foreach=True)
+    opt = torch.optim.AdamW(
+        model.parameters(), lr=1e-3, weight_decay=1e-4, foreach=True,
+        capturable=use_graphs,
+    )
     reducer = GradAllReducer(model)
     reducer.broadcast_params(model)
 
-    def step(i: int) -> int:
-        b = batches[i % len(batches)]
+    def step_body(b) -> None:
         node_logit, edge_logit, seq_logit = model(b)
         losses = model.loss(node_logit, edge_logit, seq_logit, b)
         opt.zero_grad(set_to_none=False)
         losses["total"].backward()
         reducer.finalize()
         opt.step()
+
+    def step(i: int) -> int:
+        step_body(batches[i % len(batches)])
         return events_per_step[i % len(batches)]
+
+    if use_graphs:
+        # hipGraph-capture one full training step per prebuilt window shape;
+        # replay turns ~2.5k launches/step into one graph launch.
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for b in batches:
+                    for _ in range(2):
+                        step_body(b)
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            graphs = []
+            for b in batches:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    step_body(b)
+                graphs.append(g)
+
+            def step(i: int) -> int:  # noqa: F811 - graph replay path
+                graphs[i % len(graphs)].replay()
+                return events_per_step[i % len(batches)]
+
+            if rank == 0:
+                print(f"# hipGraph capture: {len(graphs)} step graphs", file=sys.stderr, flush=True)
+        except Exception as e:  # pragma: no cover
+            use_graphs = False
+            if rank == 0:
+                print(f"# hipGraph capture failed, eager fallback: {e}", file=sys.stderr, flush=True)
 
     import torch.distributed as dist
 
@@ -159,6 +204,7 @@ def main() -> None:
                 "seq_len": 100,
                 "window_s": 30,
                 "parallelism": f"dp{world}",
+                "hipgraph": use_graphs,
                 "scale": scale,
                 "events_per_window": int(np.mean(events_per_step)),
             },
