@@ -33,6 +33,11 @@ class NerrfJointModel(nn.Module):
         self.cfg = cfg or JointConfig()
         self.gnn = GraphSAGET(self.cfg.sage)
         self.lstm = BiLSTMDetector(self.cfg.lstm)
+        # buffer (not a per-call torch.tensor): loss() must stay
+        # hipGraph-capturable — scalar H2D copies are forbidden mid-capture
+        self.register_buffer(
+            "pos_weight_buf", torch.tensor(float(self.cfg.pos_weight)), persistent=False
+        )
 
     def forward(self, batch: Dict[str, torch.Tensor]):
         """batch keys: x, nbr_idx, nbr_w, edge_index, edge_weight, edge_ts,
@@ -59,8 +64,7 @@ class NerrfJointModel(nn.Module):
         batch: Dict[str, torch.Tensor],
     ) -> Dict[str, torch.Tensor]:
         cfg = self.cfg
-        dev = node_logit.device
-        pw = torch.tensor(cfg.pos_weight, device=dev, dtype=node_logit.dtype)
+        pw = self.pos_weight_buf
         zero = node_logit.new_zeros(())
         losses = {"node": zero, "edge": zero, "seq": zero}
         if batch.get("y_node") is not None:
