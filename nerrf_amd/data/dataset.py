@@ -13,7 +13,7 @@ import numpy as np
 import torch
 
 from ..graph.constructor import build_graph, sliding_windows
-from ..graph.sampling import reverse_index, sample_fanout, to_csr
+from ..graph.sampling import edge_reverse_index, reverse_index, sample_fanout, to_csr
 from .labels import event_labels
 from .sequences import build_sequences
 from .synth import AttackWindow, SynthConfig, generate
@@ -40,6 +40,8 @@ class WindowBatch:
     rev_dst: Optional[np.ndarray] = None  # reverse index for gather backward
     rev_src: Optional[np.ndarray] = None
     rev_w: Optional[np.ndarray] = None
+    e0_rev: Optional[tuple] = None  # edge endpoint reverse (dst, src, w)
+    e1_rev: Optional[tuple] = None
 
     def to_torch(self, device="cpu", dtype=torch.float32) -> Dict[str, torch.Tensor]:
         def t(a, dt=None):
@@ -64,6 +66,12 @@ class WindowBatch:
             "nbr_rev": None
             if self.rev_dst is None
             else (t(self.rev_dst), t(self.rev_src), t(self.rev_w, torch.float32)),
+            "edge_rev": None
+            if self.e0_rev is None
+            else (
+                tuple(t(a) for a in self.e0_rev),
+                tuple(t(a) for a in self.e1_rev),
+            ),
         }
 
 
@@ -79,6 +87,8 @@ def window_to_batch(
     csr = to_csr(g.edge_index, g.num_nodes, g.edge_weight)
     nbr_idx, nbr_w = sample_fanout(csr, fanout, seed=seed)
     rev_dst, rev_src, rev_w = reverse_index(nbr_idx, nbr_w)
+    e0_rev = edge_reverse_index(g.edge_index[0]) if g.num_edges else None
+    e1_rev = edge_reverse_index(g.edge_index[1]) if g.num_edges else None
     seqs = build_sequences(events, y_ev, seq_len=seq_len)
     return WindowBatch(
         x=g.x,
@@ -97,6 +107,8 @@ def window_to_batch(
         rev_dst=rev_dst,
         rev_src=rev_src,
         rev_w=rev_w,
+        e0_rev=e0_rev,
+        e1_rev=e1_rev,
     )
 
 

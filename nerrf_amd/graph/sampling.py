@@ -102,3 +102,14 @@ def reverse_index(idx: np.ndarray, w: np.ndarray) -> Tuple[np.ndarray, np.ndarra
     src_n = np.repeat(np.arange(n, dtype=np.int64), k)
     order = np.argsort(flat_dst, kind="stable")
     return flat_dst[order], src_n[order], wn[order]
+
+
+def edge_reverse_index(endpoints: np.ndarray) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Reverse index for an edge-endpoint gather (K=1, unit weights):
+    returns (rev_dst, rev_src, rev_w) sorted by destination node."""
+    order = np.argsort(endpoints, kind="stable")
+    return (
+        endpoints[order].astype(np.int64),
+        order.astype(np.int64),
+        np.ones(len(order), dtype=np.float32),
+    )

@@ -80,12 +80,15 @@ class GraphSAGET(nn.Module):
         edge_weight: torch.Tensor | None = None,  # [E]
         edge_ts: torch.Tensor | None = None,  # [E]
         nbr_rev=None,  # optional reverse CSR (sampling.reverse_index tensors)
+        edge_rev=None,  # optional (rev for edge_index[0], rev for edge_index[1])
     ):
         h = self.encode(x, nbr_idx, nbr_w, nbr_rev)
         node_logit = self.node_head(h).squeeze(-1)
         edge_logit = None
         if edge_index is not None and edge_index.numel():
-            hs, hd = gather_rows(h, edge_index[0]), gather_rows(h, edge_index[1])
+            er0 = edge_rev[0] if edge_rev is not None else None
+            er1 = edge_rev[1] if edge_rev is not None else None
+            hs, hd = gather_rows(h, edge_index[0], er0), gather_rows(h, edge_index[1], er1)
             ew = edge_weight if edge_weight is not None else torch.ones(
                 edge_index.shape[1], device=h.device, dtype=h.dtype
             )

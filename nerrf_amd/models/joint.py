@@ -50,6 +50,7 @@ class NerrfJointModel(nn.Module):
             batch.get("edge_weight"),
             batch.get("edge_ts"),
             batch.get("nbr_rev"),
+            batch.get("edge_rev"),
         )
         seq_logit = None
         if batch.get("seq_feats") is not None and batch["seq_feats"].shape[0] > 0:

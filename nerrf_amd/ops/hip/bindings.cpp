@@ -24,8 +24,8 @@ void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
                                void*, void*, long, int, bool, hipStream_t);
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
-                               const void*, const float*, void*, void*, void*,
-                               long, int, bool, hipStream_t);
+                               const void*, const void*, const float*, void*,
+                               void*, void*, long, int, bool, hipStream_t);
 void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
                             const void*, const float*, void*, void*, void*,
                             int, bool, hipStream_t);
@@ -137,10 +137,13 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
       gates_act.data_ptr(), batch, hdim, is_bf16(hg), stream.stream());
 }
 
-void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_c,
-                        torch::Tensor gates_act, torch::Tensor c_prev,
-                        torch::Tensor mask, torch::Tensor grad_gates,
-                        torch::Tensor grad_c_prev, torch::Tensor grad_h_pass) {
+// grad_out_t (may be empty): this timestep's dL/dh, folded in-kernel so the
+// sequence backward needs no separate elementwise add per step.
+void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
+                        torch::Tensor grad_c, torch::Tensor gates_act,
+                        torch::Tensor c_prev, torch::Tensor mask,
+                        torch::Tensor grad_gates, torch::Tensor grad_c_prev,
+                        torch::Tensor grad_h_pass) {
   check_gpu_contig(grad_h, "grad_h");
   check_gpu_contig(grad_c, "grad_c");
   check_gpu_contig(gates_act, "gates_act");
@@ -158,8 +161,10 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_c,
     mask_ptr = mf.data_ptr<float>();
   }
   auto stream = at::hip::getCurrentHIPStream();
+  const void* got = grad_out_t.numel() ? grad_out_t.data_ptr() : nullptr;
+  if (grad_out_t.numel()) check_gpu_contig(grad_out_t, "grad_out_t");
   nerrf::launch_lstm_pointwise_bwd(
-      grad_h.data_ptr(), grad_c.data_ptr(), gates_act.data_ptr(),
+      grad_h.data_ptr(), got, grad_c.data_ptr(), gates_act.data_ptr(),
       c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
       grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
       is_bf16(grad_h), stream.stream());

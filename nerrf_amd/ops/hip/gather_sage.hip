@@ -170,7 +170,7 @@ __global__ void gather_mean_bwd_csr_kernel(
     const float* __restrict__ rev_w,  // [E]
     float* __restrict__ grad_h,       // [M, D] fp32 workspace (zeroed)
     long n_entries, int dim) {
-  const int ENT = 256;  // entries per wave tile
+  const int ENT = 64;  // entries per wave tile (small => enough waves to fill 256 CUs)
   const int wave_in_block = threadIdx.x / NERRF_WAVE;
   const int lane = threadIdx.x % NERRF_WAVE;
   const int waves_per_block = blockDim.x / NERRF_WAVE;
@@ -263,7 +263,7 @@ void launch_gather_mean_bwd_csr(const void* gout, const long* rev_dst,
                                 hipStream_t s) {
   const int block = 256;
   const int wpb = block / NERRF_WAVE;
-  const long n_tiles = (n_entries + 255) / 256;
+  const long n_tiles = (n_entries + 63) / 64;
   const int grid = grid_for(n_tiles, wpb);
   if (bf16) {
     gather_mean_bwd_csr_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(

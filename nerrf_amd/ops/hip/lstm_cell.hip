@@ -61,7 +61,8 @@ __global__ void lstm_pointwise_fwd_kernel(
 
 template <typename T>
 __global__ void lstm_pointwise_bwd_kernel(
-    const T* __restrict__ grad_h,     // [B, H]
+    const T* __restrict__ grad_h,     // [B, H] recurrent grad
+    const T* __restrict__ grad_out_t, // [B, H] this step's output grad (or nullptr)
     const T* __restrict__ grad_c,     // [B, H]
     const T* __restrict__ gates_act,  // [B, 4H]
     const T* __restrict__ c_prev,     // [B, H]
@@ -83,7 +84,8 @@ __global__ void lstm_pointwise_bwd_kernel(
     const float cp = to_f32(c_prev[t]);
     const float m = (mask != nullptr) ? mask[b] : 1.0f;
     const float tcn = tanhf(f * cp + i * g);  // tanh of UNMASKED c_new
-    const float gh_in = to_f32(grad_h[t]);
+    float gh_in = to_f32(grad_h[t]);
+    if (grad_out_t != nullptr) gh_in += to_f32(grad_out_t[t]);
     const float gc_in = to_f32(grad_c[t]);
     const float gh = gh_in * m;
     const float gc = gc_in * m;
@@ -109,11 +111,11 @@ template __global__ void lstm_pointwise_fwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
     long, int);
 template __global__ void lstm_pointwise_bwd_kernel<float>(
-    const float*, const float*, const float*, const float*, const float*,
+    const float*, const float*, const float*, const float*, const float*, const float*,
     float*, float*, float*, long, int);
 template __global__ void lstm_pointwise_bwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
-    const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int);
+    const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int);
 
 // ---------------------------------------------------------------------------
 // host launchers
@@ -147,24 +149,25 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
   }
 }
 
-void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_c,
-                               const void* gates_act, const void* c_prev,
-                               const float* mask, void* grad_gates,
-                               void* grad_c_prev, void* grad_h_pass,
-                               long batch, int hdim, bool bf16, hipStream_t s) {
+void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
+                               const void* grad_c, const void* gates_act,
+                               const void* c_prev, const float* mask,
+                               void* grad_gates, void* grad_c_prev,
+                               void* grad_h_pass, long batch, int hdim,
+                               bool bf16, hipStream_t s) {
   const int block = 256;
   const int grid = grid_elems(batch * hdim, block);
   if (bf16) {
     lstm_pointwise_bwd_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
-        (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_c,
-        (const __hip_bfloat16*)gates_act, (const __hip_bfloat16*)c_prev, mask,
-        (__hip_bfloat16*)grad_gates, (__hip_bfloat16*)grad_c_prev,
-        (__hip_bfloat16*)grad_h_pass, batch, hdim);
+        (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_out_t,
+        (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
+        (const __hip_bfloat16*)c_prev, mask, (__hip_bfloat16*)grad_gates,
+        (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch, hdim);
   } else {
     lstm_pointwise_bwd_kernel<float><<<grid, block, 0, s>>>(
-        (const float*)grad_h, (const float*)grad_c, (const float*)gates_act,
-        (const float*)c_prev, mask, (float*)grad_gates, (float*)grad_c_prev,
-        (float*)grad_h_pass, batch, hdim);
+        (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
+        (const float*)gates_act, (const float*)c_prev, mask, (float*)grad_gates,
+        (float*)grad_c_prev, (float*)grad_h_pass, batch, hdim);
   }
 }
 

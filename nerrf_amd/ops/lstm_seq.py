@@ -115,7 +115,6 @@ class _LSTMSeqFn(torch.autograd.Function):
         steps = range(t_len) if reverse else range(t_len - 1, -1, -1)
         empty_mask = torch.empty(0, device=dev)
         for ti in steps:
-            grad_h = grad_h + grad_out[ti]
             # c/h input of step ti = previous step's output (or h0/c0 at start)
             first = (ti == t_len - 1) if reverse else (ti == 0)
             if first:
@@ -123,20 +122,23 @@ class _LSTMSeqFn(torch.autograd.Function):
             else:
                 c_in = c_all[ti + 1] if reverse else c_all[ti - 1]
             if ext is not None:
+                # grad_out[ti] is folded inside the kernel (no separate add)
                 ext.lstm_pointwise_bwd(
-                    grad_h.contiguous(), grad_c.contiguous(), gates_all[ti], c_in.contiguous(),
+                    grad_h.contiguous(), grad_out[ti], grad_c.contiguous(),
+                    gates_all[ti], c_in.contiguous(),
                     mask[ti] if mask is not None else empty_mask,
                     grad_gates_all[ti], grad_c_prev, grad_h_pass,
                 )
+                grad_h = torch.addmm(grad_h_pass, grad_gates_all[ti], w_hh)
             else:
                 gg, gcp, ghp = _ref.lstm_pointwise_bwd_ref(
-                    grad_h, grad_c, gates_all[ti], c_in,
+                    grad_h + grad_out[ti], grad_c, gates_all[ti], c_in,
                     mask[ti] if mask is not None else None,
                 )
                 grad_gates_all[ti] = gg
                 grad_c_prev = gcp
                 grad_h_pass = ghp
-            grad_h = torch.mm(grad_gates_all[ti], w_hh) + grad_h_pass
+                grad_h = torch.mm(grad_gates_all[ti], w_hh) + grad_h_pass
             grad_c, grad_c_prev = grad_c_prev, grad_c  # ping-pong buffers
 
         # weight grads as single large GEMMs over all timesteps
