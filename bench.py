@@ -79,8 +79,9 @@ def main() -> None:
     ap.add_argument(
         "--graphs",
         choices=["auto", "on", "off"],
-        default="auto",
-        help="hipGraph-capture the whole training step (auto: single-GPU only)",
+        default="off",
+        help="hipGraph-capture the whole training step (measured neutral at "
+        "full scale — the step is kernel-bound — so off by default)",
     )
     args = ap.parse_args()
 
