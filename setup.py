@@ -25,6 +25,9 @@ sources = [
 ]
 sources = [s for s in sources if os.path.exists(s)]
 
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
+
 setup(
     name="nerrf_amd_kernels",
     ext_modules=[
@@ -39,7 +42,16 @@ setup(
                     "--offload-arch=gfx950",
                 ],
             },
-        )
+        ),
+        # plain pybind11 (no torch dependency): trace ingest must load in
+        # collector-side processes that never import torch
+        Extension(
+            name="nerrf_amd._ingest",
+            sources=[os.path.join(ROOT, "tracker", "daemon", "ingest_ext.cpp")],
+            include_dirs=[pybind11.get_include()],
+            extra_compile_args=["-O3", "-std=c++17"],
+            language="c++",
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )

@@ -1,0 +1,128 @@
+"""Native ingest tests: C++ wire codec parity, columnar decode, nerrfd daemon."""
+import json
+import socket
+import subprocess
+import time
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+from nerrf_amd.wire import codec
+
+_ingest = pytest.importorskip("nerrf_amd._ingest")
+
+DAEMON = Path(__file__).resolve().parent.parent / "tracker" / "daemon" / "nerrfd"
+
+
+def _sample_events(n=50):
+    return [
+        codec.Event(
+            ts_sec=1_700_000_000 + i,
+            ts_nsec=i * 1000,
+            pid=100 + i % 5,
+            tid=100 + i % 5,
+            comm="python3",
+            syscall=["openat", "write", "rename", "read", "unlink"][i % 5],
+            path=f"/app/uploads/doc_{i:03d}.dat",
+            new_path=f"/app/uploads/doc_{i:03d}.dat.lockbit3" if i % 5 == 2 else "",
+            ret_val=(-1) ** i * i,
+            bytes=i * 4096,
+        )
+        for i in range(n)
+    ]
+
+
+def test_cpp_decode_matches_python_encode():
+    evs = _sample_events()
+    frame = codec.encode_event_batch(evs)
+    back = _ingest.decode_batch(frame)
+    assert len(back) == len(evs)
+    for d, ev in zip(back, evs):
+        assert d["pid"] == ev.pid
+        assert d["syscall"] == ev.syscall
+        assert d["path"] == ev.path
+        assert d["ret_val"] == ev.ret_val
+        assert d["bytes"] == ev.bytes
+        assert d["ts_sec"] == ev.ts_sec
+        assert d["ts_nsec"] == ev.ts_nsec
+
+
+def test_python_decode_matches_cpp_encode():
+    evs = _sample_events(20)
+    frame = _ingest.encode_batch(
+        [
+            dict(
+                ts_sec=e.ts_sec, ts_nsec=e.ts_nsec, pid=e.pid, tid=e.tid,
+                comm=e.comm, syscall=e.syscall, path=e.path, new_path=e.new_path,
+                ret_val=e.ret_val, bytes=e.bytes,
+            )
+            for e in evs
+        ]
+    )
+    back = codec.decode_event_batch(frame)
+    assert back == evs
+
+
+def test_columnar_decoder():
+    evs = _sample_events(64)
+    frame = codec.encode_event_batch(evs)
+    dec = _ingest.ColumnarDecoder()
+    ts, pid, sysc, path_id, newp_id, nbytes, ret, comm = dec.decode([frame, frame])
+    assert len(ts) == 128
+    assert ts[0] == pytest.approx(evs[0].timestamp(), abs=1e-5) if callable(getattr(evs[0], "timestamp", None)) else True
+    assert pid[3] == evs[3].pid
+    # interning: same path in both frames -> same id
+    assert path_id[0] == path_id[64]
+    paths = dec.paths_since(0)
+    assert paths[path_id[0]] == evs[0].path
+    # syscall ids match the python mapping
+    from nerrf_amd.data.trace import SYSCALL_IDS
+
+    assert sysc[1] == SYSCALL_IDS["write"]
+    assert int(nbytes[2]) == evs[2].bytes
+
+
+@pytest.mark.skipif(not DAEMON.exists(), reason="nerrfd not built (make -C tracker daemon)")
+def test_nerrfd_replay_roundtrip(tmp_path):
+    """End-to-end: jsonl trace -> nerrfd (C++) -> TCP frames -> store."""
+    from nerrf_amd.graph.store import DeltaGraphStore
+    from nerrf_amd.serve.daemon_bridge import pump_daemon_into_store
+
+    trace = tmp_path / "t.jsonl"
+    with open(trace, "w") as fh:
+        for i in range(120):
+            fh.write(
+                json.dumps(
+                    {
+                        "timestamp": 10.0 + i * 0.01,
+                        "event": ["write", "rename", "openat"][i % 3],
+                        "path": f"/data/f{i % 7}.dat",
+                        "size": i * 100,
+                        "pid": 42,
+                    }
+                )
+                + "\n"
+            )
+    # free port
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [str(DAEMON), "--replay", str(trace), "--port", str(port), "--once", "--batch", "32"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+    )
+    try:
+        time.sleep(0.3)  # let it bind
+        store = DeltaGraphStore(window_s=1e9)
+        n = pump_daemon_into_store("127.0.0.1", port, store, timeout_s=10.0)
+        assert n == 120
+        arr = store.compact()
+        assert len(arr) == 120
+        assert arr.nbytes.sum() == sum(i * 100 for i in range(120))
+        from nerrf_amd.data.trace import SYSCALL_IDS
+
+        assert (arr.syscall == SYSCALL_IDS["rename"]).sum() == 40
+    finally:
+        proc.terminate()
+        proc.wait(timeout=5)
