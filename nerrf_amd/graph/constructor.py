@@ -122,13 +122,16 @@ def _path_flags(
     return suspicious, note, recon, double_ext
 
 
-def build_graph(
+def build_graph_parts(
     events: EventArray,
-    window: Optional[AttackWindow] = None,
-    y_event: Optional[np.ndarray] = None,
     causality_tau_s: float = 10.0,
-) -> TemporalGraph:
-    """Build the dependency graph for one event window."""
+) -> dict:
+    """Identity + edge domain of graph construction (string/set work, host).
+
+    Returns the intermediates shared by the CPU feature path (build_graph)
+    and the GPU delta-compaction path (graph.gpu_store): per-event node ids,
+    node tables, aggregated edges, degrees, and path-pattern flags.
+    """
     n_ev = len(events)
     t0 = float(events.ts[0]) if n_ev else 0.0
     t1 = float(events.ts[-1]) if n_ev else 0.0
@@ -165,6 +168,102 @@ def build_graph(
     ev_file = np.where(events.path_id >= 0, root_to_file[path_root[np.clip(events.path_id, 0, None)]], -1)
     pid_sorted = np.argsort(upids)
     ev_proc = n_files + pid_sorted[np.searchsorted(upids[pid_sorted], events.pid)] if n_ev else np.empty(0, dtype=np.int64)
+
+    # degrees and peers are edge-domain: computed below with edges
+    return {
+        "events": events,
+        "n_ev": n_ev,
+        "t0": t0,
+        "t1": t1,
+        "span": span,
+        "ev_file": ev_file,
+        "ev_proc": ev_proc,
+        "n_files": n_files,
+        "n_procs": n_procs,
+        "n_nodes": n_nodes,
+        "path_root": path_root,
+        "root_to_file": root_to_file,
+        "touched_roots": touched_roots,
+        "upids": upids,
+    }
+
+
+def build_edges_and_flags(parts: dict, causality_tau_s: float = 10.0) -> dict:
+    """Edge aggregation, degrees/peers and path-pattern flags from parts."""
+    events: EventArray = parts["events"]
+    ev_file, ev_proc = parts["ev_file"], parts["ev_proc"]
+    n_nodes, n_files, n_procs = parts["n_nodes"], parts["n_files"], parts["n_procs"]
+    t0, t1, span = parts["t0"], parts["t1"], parts["span"]
+    sc = events.syscall
+
+    valid = (ev_file >= 0) & (ev_proc >= 0)
+    direction = np.where(np.isin(sc, _READ_LIKE), 1, 0)  # 0: proc->file, 1: file->proc
+    key = (ev_proc.astype(np.int64) * n_nodes + ev_file.astype(np.int64)) * 2 + direction
+    key = key[valid]
+    if key.size:
+        uk, inv = np.unique(key, return_inverse=True)
+        # causality confidence: recency-decayed count, saturating
+        rec = np.exp(-(t1 - events.ts[valid]) / causality_tau_s)
+        e_conf = np.bincount(inv, weights=rec)
+        e_last = np.zeros(len(uk))
+        np.maximum.at(e_last, inv, events.ts[valid])
+        dirs = uk % 2
+        pf = uk // 2
+        e_proc = (pf // n_nodes).astype(np.int64)
+        e_file = (pf % n_nodes).astype(np.int64)
+        src = np.where(dirs == 0, e_proc, e_file)
+        dst = np.where(dirs == 0, e_file, e_proc)
+        edge_index = np.stack([src, dst]).astype(np.int64)
+        edge_weight = (1.0 - np.exp(-e_conf)).astype(np.float32)  # saturate to (0,1)
+        edge_ts = ((e_last - t0) / span).astype(np.float32)
+    else:
+        edge_index = np.zeros((2, 0), dtype=np.int64)
+        edge_weight = np.zeros(0, dtype=np.float32)
+        edge_ts = np.zeros(0, dtype=np.float32)
+
+    peer = np.zeros(n_nodes, dtype=np.float64)
+    in_deg = np.zeros(n_nodes, dtype=np.float64)
+    out_deg = np.zeros(n_nodes, dtype=np.float64)
+    if edge_index.shape[1]:
+        np.add.at(peer, edge_index[0], 1.0)
+        np.add.at(peer, edge_index[1], 1.0)
+        np.add.at(out_deg, edge_index[0], 1.0)
+        np.add.at(in_deg, edge_index[1], 1.0)
+
+    suspicious, note, recon, double_ext = _path_flags(
+        events.paths, parts["path_root"], parts["root_to_file"], n_files
+    )
+    pad = np.zeros(n_procs, dtype=np.float32)
+    return {
+        "edge_index": edge_index,
+        "edge_weight": edge_weight,
+        "edge_ts": edge_ts,
+        "in_deg": in_deg,
+        "out_deg": out_deg,
+        "peer": peer,
+        "suspicious": np.concatenate([suspicious, pad]),
+        "note": np.concatenate([note, pad]),
+        "recon": np.concatenate([recon, pad]),
+        "double_ext": np.concatenate([double_ext, pad]),
+    }
+
+
+def build_graph(
+    events: EventArray,
+    window: Optional[AttackWindow] = None,
+    y_event: Optional[np.ndarray] = None,
+    causality_tau_s: float = 10.0,
+    parts: Optional[dict] = None,
+) -> TemporalGraph:
+    """Build the dependency graph for one event window."""
+    if parts is None:
+        parts = build_graph_parts(events, causality_tau_s)
+    n_ev = parts["n_ev"]
+    t0, t1, span = parts["t0"], parts["t1"], parts["span"]
+    ev_file, ev_proc = parts["ev_file"], parts["ev_proc"]
+    n_files, n_procs, n_nodes = parts["n_files"], parts["n_procs"], parts["n_nodes"]
+    path_root, root_to_file = parts["path_root"], parts["root_to_file"]
+    touched_roots, upids = parts["touched_roots"], parts["upids"]
 
     # ---- per-node counters (vectorised bincount) --------------------------
     def _count(mask: np.ndarray, ids: np.ndarray) -> np.ndarray:
@@ -212,55 +311,11 @@ def build_graph(
     t_first[~np.isfinite(t_first)] = t0
     t_last[~np.isfinite(t_last)] = t0
 
-    # ---- edges: (proc, file, direction) aggregation -----------------------
-    valid = (ev_file >= 0) & (ev_proc >= 0)
-    direction = np.where(np.isin(sc, _READ_LIKE), 1, 0)  # 0: proc->file, 1: file->proc
-    key = (ev_proc.astype(np.int64) * n_nodes + ev_file.astype(np.int64)) * 2 + direction
-    key = key[valid]
-    if key.size:
-        uk, inv = np.unique(key, return_inverse=True)
-        e_cnt = np.bincount(inv).astype(np.float64)
-        e_bytes = np.bincount(inv, weights=events.nbytes[valid].astype(np.float64))
-        # causality confidence: recency-decayed count, saturating
-        rec = np.exp(-(t1 - events.ts[valid]) / causality_tau_s)
-        e_conf = np.bincount(inv, weights=rec)
-        e_last = np.zeros(len(uk))
-        np.maximum.at(e_last, inv, events.ts[valid])
-        dirs = uk % 2
-        pf = uk // 2
-        e_proc = (pf // n_nodes).astype(np.int64)
-        e_file = (pf % n_nodes).astype(np.int64)
-        src = np.where(dirs == 0, e_proc, e_file)
-        dst = np.where(dirs == 0, e_file, e_proc)
-        edge_index = np.stack([src, dst]).astype(np.int64)
-        edge_weight = (1.0 - np.exp(-e_conf)).astype(np.float32)  # saturate to (0,1)
-        edge_ts = ((e_last - t0) / span).astype(np.float32)
-    else:
-        edge_index = np.zeros((2, 0), dtype=np.int64)
-        edge_weight = np.zeros(0, dtype=np.float32)
-        edge_ts = np.zeros(0, dtype=np.float32)
-        e_cnt = np.zeros(0)
-        e_bytes = np.zeros(0)
-
-    # distinct peers per node
-    peer = np.zeros(n_nodes, dtype=np.float64)
-    if edge_index.shape[1]:
-        np.add.at(peer, edge_index[0], 1.0)
-        np.add.at(peer, edge_index[1], 1.0)
-
-    in_deg = np.zeros(n_nodes, dtype=np.float64)
-    out_deg = np.zeros(n_nodes, dtype=np.float64)
-    if edge_index.shape[1]:
-        np.add.at(out_deg, edge_index[0], 1.0)
-        np.add.at(in_deg, edge_index[1], 1.0)
-
-    # ---- string-pattern indicator features --------------------------------
-    suspicious, note, recon, double_ext = _path_flags(events.paths, path_root, root_to_file, n_files)
-    pad = np.zeros(n_procs, dtype=np.float32)
-    suspicious = np.concatenate([suspicious, pad])
-    note = np.concatenate([note, pad])
-    recon = np.concatenate([recon, pad])
-    double_ext = np.concatenate([double_ext, pad])
+    # ---- edges + degrees + string flags (shared with the GPU path) --------
+    ed = build_edges_and_flags(parts, causality_tau_s)
+    edge_index, edge_weight, edge_ts = ed["edge_index"], ed["edge_weight"], ed["edge_ts"]
+    in_deg, out_deg, peer = ed["in_deg"], ed["out_deg"], ed["peer"]
+    suspicious, note, recon, double_ext = ed["suspicious"], ed["note"], ed["recon"], ed["double_ext"]
 
     # ---- assemble feature matrix ------------------------------------------
     x = np.zeros((n_nodes, NUM_NODE_FEATURES), dtype=np.float32)

@@ -29,6 +29,13 @@ void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
 void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
                             const void*, const float*, void*, void*, void*,
                             int, bool, hipStream_t);
+void launch_event_scatter(const long*, const long*, const signed char*,
+                          const float*, const int*, float*, int*, int*, long,
+                          hipStream_t);
+void launch_feature_assemble(const float*, const int*, const int*,
+                             const float*, const float*, const float*,
+                             const unsigned char*, const signed char*, float*,
+                             float, int, hipStream_t);
 }  // namespace nerrf
 
 namespace {
@@ -202,6 +209,38 @@ void lstm_step_fused(torch::Tensor h_prev, torch::Tensor w_hh, torch::Tensor xg,
       gates_act.data_ptr(), batch, raw, stream.stream());
 }
 
+// GPU delta compaction: event columns -> per-node accumulators -> x [M, 32].
+torch::Tensor event_features(torch::Tensor ev_file, torch::Tensor ev_proc,
+                             torch::Tensor syscall_id, torch::Tensor nbytes,
+                             torch::Tensor ts_ms, long m_nodes,
+                             torch::Tensor in_deg, torch::Tensor out_deg,
+                             torch::Tensor peer, torch::Tensor flags,
+                             torch::Tensor node_kind, double span_s) {
+  for (auto* t : {&ev_file, &ev_proc, &syscall_id, &nbytes, &ts_ms, &in_deg,
+                  &out_deg, &peer, &flags, &node_kind})
+    check_gpu_contig(*t, "event_features arg");
+  const long n_events = ev_file.numel();
+  auto opts_f = nbytes.options().dtype(torch::kFloat32);
+  auto opts_i = nbytes.options().dtype(torch::kInt32);
+  auto acc = torch::zeros({m_nodes, 13}, opts_f);
+  auto t_first = torch::full({m_nodes}, (long)INT_MAX, opts_i);
+  auto t_last = torch::full({m_nodes}, (long)INT_MIN, opts_i);
+  auto x = torch::empty({m_nodes, 32}, opts_f);
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_event_scatter(
+      ev_file.data_ptr<long>(), ev_proc.data_ptr<long>(),
+      syscall_id.data_ptr<signed char>(), nbytes.data_ptr<float>(),
+      ts_ms.data_ptr<int>(), acc.data_ptr<float>(), t_first.data_ptr<int>(),
+      t_last.data_ptr<int>(), n_events, stream.stream());
+  nerrf::launch_feature_assemble(
+      acc.data_ptr<float>(), t_first.data_ptr<int>(), t_last.data_ptr<int>(),
+      in_deg.data_ptr<float>(), out_deg.data_ptr<float>(),
+      peer.data_ptr<float>(), flags.data_ptr<unsigned char>(),
+      node_kind.data_ptr<signed char>(), x.data_ptr<float>(), (float)span_s,
+      (int)m_nodes, stream.stream());
+  return x;
+}
+
 nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
   nerrf::PlannerParamsDev p;
   p.n_groups = d["n_groups"].cast<int>();
@@ -268,6 +307,8 @@ torch::Tensor mcts_eval_plans(torch::Tensor gscore, torch::Tensor gmb,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("event_features", &event_features,
+        "GPU delta compaction: events -> per-node feature matrix");
   m.def("mcts_search", &mcts_search, "batched root-parallel MCTS");
   m.def("mcts_eval_plans", &mcts_eval_plans, "batch plan reward evaluation");
   m.def("gather_mean_fwd", &gather_mean_fwd, "weighted neighbor gather-mean");

@@ -1,0 +1,129 @@
+// Streaming delta-graph compaction — CDNA4 (gfx950).
+//
+// The GPU half of the 30 s delta edge store ("HBM-resident
+// ring of delta graphs, GPU compaction kernel"): raw event columns (staged
+// once per delta) are scatter-accumulated into per-node counters, then a
+// second kernel assembles the 32-dim feature matrix the GNN consumes —
+// exactly the formulas of graph/constructor.py (tests assert parity).
+// Identity resolution (path interning, rename union) is string-domain work
+// and stays on the host; everything numeric runs here.
+//
+// Accumulator layout per node (fp32 unless noted):
+//   [0..9]  syscall counts (ids 0..9)
+//   [10]    bytes_read   [11] bytes_write
+//   [12]    total events
+//   t_first / t_last: separate i32 millisecond arrays (atomicMin/Max).
+#include "common.h"
+
+namespace nerrf {
+
+#define NACC 13
+
+__global__ void event_scatter_kernel(
+    const long* __restrict__ ev_file,   // [E] file node id or -1
+    const long* __restrict__ ev_proc,   // [E] process node id or -1
+    const signed char* __restrict__ syscall_id,  // [E]
+    const float* __restrict__ nbytes,   // [E]
+    const int* __restrict__ ts_ms,      // [E] window-relative milliseconds
+    float* __restrict__ acc,            // [M, NACC] zeroed
+    int* __restrict__ t_first,          // [M] init INT_MAX
+    int* __restrict__ t_last,           // [M] init INT_MIN
+    long n_events) {
+  const long total = 2 * n_events;  // each event touches file node + proc node
+  for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long e = t >> 1;
+    const long node = (t & 1) ? ev_proc[e] : ev_file[e];
+    if (node < 0) continue;
+    const int sc = syscall_id[e];
+    float* a = acc + node * NACC;
+    if (sc >= 0 && sc < 10) atomicAdd(a + sc, 1.0f);
+    if (sc == 4) atomicAdd(a + 10, nbytes[e]);       // read
+    else if (sc == 2) atomicAdd(a + 11, nbytes[e]);  // write
+    atomicAdd(a + 12, 1.0f);
+    atomicMin(t_first + node, ts_ms[e]);
+    atomicMax(t_last + node, ts_ms[e]);
+  }
+}
+
+// Assemble x[M, 32] from accumulators + host-side identity/edge inputs.
+// Formula-for-formula the same as graph/constructor.py build_graph.
+__global__ void feature_assemble_kernel(
+    const float* __restrict__ acc,       // [M, NACC]
+    const int* __restrict__ t_first_ms,  // [M]
+    const int* __restrict__ t_last_ms,   // [M]
+    const float* __restrict__ in_deg,    // [M] (edge-domain, from host)
+    const float* __restrict__ out_deg,   // [M]
+    const float* __restrict__ peer,      // [M]
+    const unsigned char* __restrict__ flags,  // [M] bit0 susp, 1 note, 2 recon, 3 dbl_ext
+    const signed char* __restrict__ node_kind,  // [M] 0 proc, 1 file
+    float* __restrict__ x,               // [M, 32]
+    float span_s, int m_nodes) {
+  const int m = blockIdx.x * blockDim.x + threadIdx.x;
+  if (m >= m_nodes) return;
+  const float* a = acc + m * NACC;
+  const float cnt_read = a[4], cnt_write = a[2], cnt_rename = a[3];
+  const float cnt_unlink = a[5], cnt_open = a[1], cnt_exec = a[9];
+  const float bytes_read = a[10], bytes_write = a[11], cnt_total = a[12];
+  float tf = (t_first_ms[m] == INT_MAX) ? 0.0f : t_first_ms[m] * 1e-3f;
+  float tl = (t_last_ms[m] == INT_MIN) ? 0.0f : t_last_ms[m] * 1e-3f;
+  const float dur = fmaxf(tl - tf, 0.0f);
+  const unsigned char fl = flags[m];
+  float* o = x + m * 32;
+  o[0] = node_kind[m] == 0 ? 1.0f : 0.0f;
+  o[1] = node_kind[m] == 1 ? 1.0f : 0.0f;
+  o[2] = log1pf(in_deg[m]);
+  o[3] = log1pf(out_deg[m]);
+  o[4] = log1pf(cnt_read);
+  o[5] = log1pf(cnt_write);
+  o[6] = log1pf(cnt_rename);
+  o[7] = log1pf(cnt_unlink);
+  o[8] = log1pf(cnt_open);
+  o[9] = log1pf(bytes_read) / 16.0f;
+  o[10] = log1pf(bytes_write) / 16.0f;
+  o[11] = bytes_write / fmaxf(bytes_read + bytes_write, 1.0f);
+  o[12] = cnt_rename / fmaxf(cnt_write + cnt_rename, 1.0f);
+  o[13] = (fl & 1) ? 1.0f : 0.0f;
+  o[14] = (fl & 8) ? 1.0f : 0.0f;
+  o[15] = dur / span_s;
+  o[16] = tf / span_s;
+  o[17] = tl / span_s;
+  o[18] = log1pf(cnt_total / fmaxf(dur, 1.0f));
+  o[19] = log1pf(dur / fmaxf(cnt_total, 1.0f));
+  o[20] = (fl & 2) ? 1.0f : 0.0f;
+  o[21] = (fl & 4) ? 1.0f : 0.0f;
+  o[22] = (cnt_rename > 0.0f && cnt_unlink > 0.0f) ? 1.0f : 0.0f;
+  o[23] = log1pf(peer[m]);
+  o[24] = log1pf(cnt_total);
+  o[25] = log1pf(bytes_write / fmaxf(cnt_write, 1.0f)) / 16.0f;
+  o[26] = log1pf(cnt_exec);
+  o[27] = 0.0f; o[28] = 0.0f; o[29] = 0.0f; o[30] = 0.0f; o[31] = 0.0f;
+}
+
+void launch_event_scatter(const long* ev_file, const long* ev_proc,
+                          const signed char* syscall_id, const float* nbytes,
+                          const int* ts_ms, float* acc, int* t_first,
+                          int* t_last, long n_events, hipStream_t s) {
+  const int block = 256;
+  long blocks = (2 * n_events + block - 1) / block;
+  if (blocks > 8192) blocks = 8192;
+  if (blocks < 1) blocks = 1;
+  event_scatter_kernel<<<(int)blocks, block, 0, s>>>(
+      ev_file, ev_proc, syscall_id, nbytes, ts_ms, acc, t_first, t_last,
+      n_events);
+}
+
+void launch_feature_assemble(const float* acc, const int* t_first,
+                             const int* t_last, const float* in_deg,
+                             const float* out_deg, const float* peer,
+                             const unsigned char* flags,
+                             const signed char* node_kind, float* x,
+                             float span_s, int m_nodes, hipStream_t s) {
+  const int block = 256;
+  const int grid = (m_nodes + block - 1) / block;
+  feature_assemble_kernel<<<grid, block, 0, s>>>(
+      acc, t_first, t_last, in_deg, out_deg, peer, flags, node_kind, x,
+      span_s, m_nodes);
+}
+
+}  // namespace nerrf

@@ -366,3 +366,24 @@ def test_gather_rows_matches_indexing(gpu_device):
     assert torch.allclose(out.detach(), out2.detach())
     out2.backward(g)
     assert torch.allclose(grad_fast, h2.grad, atol=1e-3, rtol=1e-3)
+
+
+def test_gpu_window_graph_matches_cpu(gpu_device):
+    """GPU delta-compaction features == CPU build_graph features."""
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.graph.constructor import build_graph
+    from nerrf_amd.graph.gpu_store import gpu_window_graph
+
+    arr, _ = generate(SynthConfig(seed=21, duration_s=45, benign_rate_hz=400, n_victim_files=12))
+    g_cpu = build_graph(arr)
+    g_gpu = gpu_window_graph(arr, gpu_device)
+    assert g_gpu is not None
+    x_gpu = g_gpu["x"].cpu().numpy()
+    assert x_gpu.shape == g_cpu.x.shape
+    # ms-quantised timestamps: loose tolerance on the 4 time columns
+    for col in range(32):
+        atol = 5e-3 if col in (15, 16, 17, 18, 19) else 1e-4
+        assert np.allclose(x_gpu[:, col], g_cpu.x[:, col], atol=atol, rtol=1e-3), f"col {col}"
+    assert np.array_equal(g_gpu["edge_index"].cpu().numpy(), g_cpu.edge_index)
+    assert np.allclose(g_gpu["edge_weight"].cpu().numpy(), g_cpu.edge_weight, atol=1e-6)
+    assert np.array_equal(g_gpu["node_kind"].cpu().numpy(), g_cpu.node_kind)
