@@ -40,6 +40,7 @@ def build_bench_batches(rank: int, n_windows: int, scale: str):
     """Prebuild identically-shaped-per-rank window batches."""
     shapes = {
         # benign_rate, duration, n_files, n_victims
+        "tiny": (200.0, 31.0, 200, 12),  # CPU CI smoke only
         "small": (2_000.0, 60.0, 2_000, 48),
         "full": (20_000.0, 60.0, 16_000, 64),
     }[scale]
@@ -73,7 +74,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
-    ap.add_argument("--scale", choices=["small", "full"], default="full")
+    ap.add_argument("--scale", choices=["tiny", "small", "full"], default="full")
     ap.add_argument("--windows", type=int, default=2, help="prebuilt windows per rank")
     ap.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
     ap.add_argument(
@@ -96,7 +97,9 @@ def main() -> None:
     else:
         device = torch.device("cpu")
     dtype = torch.bfloat16 if (args.dtype == "bf16" and has_gpu) else torch.float32
-    scale = args.scale if has_gpu else "small"
+    scale = args.scale
+    if not has_gpu and scale == "full":
+        scale = "small"  # CPU hosts cannot time the full config meaningfully
 
     batches_np = build_bench_batches(rank, args.windows, scale)
     batches = [b.to_torch(device=device, dtype=dtype) for b in batches_np]
