@@ -29,19 +29,25 @@ def gpu_window_graph(
     device: torch.device,
     causality_tau_s: float = 10.0,
     dtype: torch.dtype = torch.float32,
+    parts: Optional[dict] = None,
+    ed: Optional[dict] = None,
 ) -> Optional[dict]:
     """Build one window's graph with GPU feature compaction.
 
     Returns dict(x, edge_index, edge_weight, edge_ts, node_kind, node_key)
-    on `device`, or None for an empty window.
+    on `device`, or None for an empty window.  `parts`/`ed` let a caller that
+    already did the host-side identity/edge work (e.g. the streaming engine)
+    skip recomputing it.
     """
     from ..ops.native import load_extension
 
     ext = load_extension(required=True)
-    parts = build_graph_parts(events, causality_tau_s)
+    if parts is None:
+        parts = build_graph_parts(events, causality_tau_s)
     if parts["n_nodes"] == 0:
         return None
-    ed = build_edges_and_flags(parts, causality_tau_s)
+    if ed is None:
+        ed = build_edges_and_flags(parts, causality_tau_s)
     n_nodes = parts["n_nodes"]
     n_files, n_procs = parts["n_files"], parts["n_procs"]
     t0, span = parts["t0"], parts["span"]

@@ -22,7 +22,6 @@ from typing import Dict, List, Optional
 import numpy as np
 import torch
 
-from ..data.dataset import window_to_batch
 from ..data.trace import EventArray
 from ..graph.store import DeltaGraphStore
 from ..models.joint import JointConfig, NerrfJointModel
@@ -97,11 +96,51 @@ class StreamingEngine:
         t_detect = time.time()
         if len(events) == 0:
             return Detection(False, t_detect, {}, {}, {}, window_events=0)
-        wb = window_to_batch(events, None, seed=self.scored_windows)
-        batch = wb.to_torch(device=self.device, dtype=self.dtype)
-        node_logit, _, seq_logit = self.model(
-            {**batch, "edge_index": batch["edge_index"], "seq_feats": batch["seq_feats"]}
+        # ---- graph build: host identity/edges once; GPU feature compaction
+        # on ROCm devices (HBM-resident x), CPU features otherwise ----------
+        import numpy as _np
+
+        from ..data.sequences import build_sequences
+        from ..graph.constructor import build_edges_and_flags, build_graph, build_graph_parts
+        from ..graph.sampling import sample_fanout, to_csr
+
+        parts = build_graph_parts(events)
+        ed = build_edges_and_flags(parts)
+        node_kind = _np.concatenate(
+            [_np.ones(parts["n_files"], dtype=_np.int8), _np.zeros(parts["n_procs"], dtype=_np.int8)]
         )
+        node_key = _np.concatenate(
+            [parts["touched_roots"], parts["upids"].astype(_np.int64)]
+        )
+        csr = to_csr(ed["edge_index"], parts["n_nodes"], ed["edge_weight"])
+        nbr_idx, nbr_w = sample_fanout(csr, 16, seed=self.scored_windows)
+        seqs = build_sequences(events, None)
+
+        if self.device.type == "cuda":
+            from ..graph.gpu_store import gpu_window_graph
+
+            gg = gpu_window_graph(events, self.device, parts=parts, ed=ed, dtype=self.dtype)
+            x = gg["x"]
+            edge_index = gg["edge_index"]
+            edge_weight = gg["edge_weight"]
+            edge_ts = gg["edge_ts"]
+        else:
+            g = build_graph(events, parts=parts)
+            x = torch.from_numpy(g.x).to(self.dtype)
+            edge_index = torch.from_numpy(g.edge_index)
+            edge_weight = torch.from_numpy(g.edge_weight)
+            edge_ts = torch.from_numpy(g.edge_ts)
+        batch = {
+            "x": x.to(self.device),
+            "nbr_idx": torch.from_numpy(nbr_idx).to(self.device),
+            "nbr_w": torch.from_numpy(nbr_w).to(self.device),
+            "edge_index": edge_index.to(self.device),
+            "edge_weight": edge_weight.to(self.device),
+            "edge_ts": edge_ts.to(self.device),
+            "seq_feats": torch.from_numpy(seqs.feats).to(self.device, self.dtype),
+            "seq_lengths": torch.from_numpy(seqs.lengths).to(self.device),
+        }
+        node_logit, _, seq_logit = self.model(batch)
         node_score = torch.sigmoid(node_logit.float()).cpu().numpy()
         seq_score = (
             torch.sigmoid(seq_logit.float()).cpu().numpy() if seq_logit is not None else None
@@ -109,24 +148,20 @@ class StreamingEngine:
         self.scored_windows += 1
         self.events_scored += len(events)
 
-        # map node/sequence scores back to live paths
-        from ..graph.constructor import build_graph  # node_key convention
-
-        g = build_graph(events)  # cheap relative to model; reuse keys
+        # ---- map node/sequence scores back to live paths -------------------
         file_scores: Dict[str, float] = {}
         file_mb: Dict[str, float] = {}
         proc_scores: Dict[int, float] = {}
-        n_files = int((g.node_kind == 1).sum())
-        for ni in range(g.num_nodes):
-            if g.node_kind[ni] == 1:
-                path = events.paths.lookup(int(g.node_key[ni]))
+        for ni in range(len(node_kind)):
+            if node_kind[ni] == 1:
+                path = events.paths.lookup(int(node_key[ni]))
                 s = float(node_score[ni]) if ni < len(node_score) else 0.0
                 file_scores[path] = max(file_scores.get(path, 0.0), s)
             else:
-                proc_scores[int(g.node_key[ni])] = float(node_score[ni]) if ni < len(node_score) else 0.0
-        if seq_score is not None and len(seq_score) == len(wb.seq_path_id):
+                proc_scores[int(node_key[ni])] = float(node_score[ni]) if ni < len(node_score) else 0.0
+        if seq_score is not None and len(seq_score) == len(seqs.file_path_id):
             for bi in range(len(seq_score)):
-                pid_ = int(wb.seq_path_id[bi])
+                pid_ = int(seqs.file_path_id[bi])
                 if pid_ >= 0:
                     path = events.paths.lookup(pid_)
                     file_scores[path] = max(file_scores.get(path, 0.0), float(seq_score[bi]))

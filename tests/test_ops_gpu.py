@@ -387,3 +387,25 @@ def test_gpu_window_graph_matches_cpu(gpu_device):
     assert np.array_equal(g_gpu["edge_index"].cpu().numpy(), g_cpu.edge_index)
     assert np.allclose(g_gpu["edge_weight"].cpu().numpy(), g_cpu.edge_weight, atol=1e-6)
     assert np.array_equal(g_gpu["node_kind"].cpu().numpy(), g_cpu.node_kind)
+
+
+def test_streaming_engine_gpu_scores(gpu_device):
+    """Engine scoring on GPU (delta compaction + model + planner kernels)."""
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.models.graphsage import SageConfig
+    from nerrf_amd.models.joint import JointConfig, NerrfJointModel
+    from nerrf_amd.models.lstm import LSTMConfig
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    model = NerrfJointModel(
+        JointConfig(sage=SageConfig(layers=4, hidden=128), lstm=LSTMConfig(hidden=256))
+    )
+    engine = StreamingEngine(model=model, device=str(gpu_device), dtype=torch.bfloat16)
+    engine.store.window_s = 1e9
+    arr, _ = generate(SynthConfig(seed=31, duration_s=40, benign_rate_hz=100, n_victim_files=8))
+    engine.ingest_events(arr)
+    det = engine.score_window()
+    assert det.alarm
+    assert det.window_events == len(arr)
+    plan = engine.plan(det, n_sims=512, use_gpu=True)
+    assert plan.simulations == 512

@@ -51,11 +51,12 @@ def test_tracker_sim_roundtrip():
     from nerrf_amd.serve.tracker_sim import TrackerSimServer
 
     arr, _ = generate(SynthConfig(seed=5, duration_s=20, benign_rate_hz=30))
-    server = TrackerSimServer(arr, batch_size=16)
+    # generous client buffer: this test checks fidelity, not drop semantics
+    server = TrackerSimServer(arr, batch_size=16, client_buffer=10_000)
     server.start()
     try:
         got = []
-        for batch in stream_events(server.address, timeout_s=10.0):
+        for batch in stream_events(server.address, timeout_s=15.0):
             got.extend(batch)
             if len(got) >= len(arr):
                 break
@@ -156,3 +157,40 @@ def test_cli_status_and_scenario(tmp_path, capsys):
     assert rc == 0
     rep = json.loads(capsys.readouterr().out)
     assert rep["recovered_ok"]
+
+
+def test_metrics_instrumentation(tmp_path):
+    from nerrf_amd.serve import metrics as M
+
+    engine = _small_engine(device="cpu")
+    M.instrument_engine(engine)
+    arr, _ = generate(SynthConfig(seed=8, duration_s=20, benign_rate_hz=30))
+    engine.store.window_s = 1e9
+    engine.ingest_events(arr)
+    det = engine.score_window()
+    plan = engine.plan(det, n_sims=64)
+    assert engine.scored_windows == 1
+    assert plan.simulations == 64
+
+
+def test_sweep_small(tmp_path):
+    from nerrf_amd.sweep import run_sweep
+
+    res = run_sweep(
+        [
+            "data.n_scenarios=1",
+            "data.benign_rate_hz=60",
+            "data.duration_s=45",
+            "optim.epochs=1",
+            "run.eval_holdout=1",
+            "run.log_every=1000",
+            f"run.checkpoint_dir={tmp_path}/sw",
+            "model.sage.layers=2",
+            "model.sage.hidden=24",
+            "model.lstm.hidden=16",
+        ],
+        space={"optim.lr": [1e-3, 3e-3]},
+        trials=2,
+    )
+    assert len(res) == 2
+    assert "node_auc" in res[0]["metrics"]
