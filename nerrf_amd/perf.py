@@ -22,14 +22,19 @@ def enable_tuned_gemms(tuning: bool = False) -> bool:
 
     if not torch.cuda.is_available() or not hasattr(torch.cuda, "tunable"):
         return False
+    # allow an operator-driven re-tune via the standard env knobs
+    tuning = tuning or os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") == "1"
     try:
         torch.cuda.tunable.enable(True)
         torch.cuda.tunable.tuning_enable(tuning)
         if tuning:
-            torch.cuda.tunable.set_filename(
+            fname = os.environ.get(
+                "PYTORCH_TUNABLEOP_FILENAME",
                 os.path.join(os.environ.get("TMPDIR", "/tmp"), "nerrf_tunableop_.csv"),
-                insert_device_ordinal=True,
             )
+            torch.cuda.tunable.set_filename(fname, insert_device_ordinal=True)
+            if _TABLE.exists():
+                torch.cuda.tunable.read_file(str(_TABLE))  # warm-start
         elif _TABLE.exists():
             torch.cuda.tunable.read_file(str(_TABLE))
         return True
