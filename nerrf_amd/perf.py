@@ -33,8 +33,8 @@ def enable_tuned_gemms(tuning: bool = False) -> bool:
                 os.path.join(os.environ.get("TMPDIR", "/tmp"), "nerrf_tunableop_.csv"),
             )
             torch.cuda.tunable.set_filename(fname, insert_device_ordinal=True)
-            if _TABLE.exists():
-                torch.cuda.tunable.read_file(str(_TABLE))  # warm-start
+            # no warm-start: shapes already present in a loaded table are
+            # never re-tuned, which would defeat a longer tuning budget
         elif _TABLE.exists():
             torch.cuda.tunable.read_file(str(_TABLE))
         return True
