@@ -22,10 +22,12 @@ void launch_gather_mean_bwd_csr(const void*, const long*, const long*,
                                 hipStream_t);
 void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
-                               void*, void*, long, int, bool, hipStream_t);
+                               void*, void*, long, int, long, long, bool,
+                               hipStream_t);
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
-                               void*, void*, long, int, bool, hipStream_t);
+                               void*, void*, long, int, long, bool,
+                               hipStream_t);
 void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
                             const void*, const float*, void*, void*, void*,
                             int, bool, hipStream_t);
@@ -50,6 +52,16 @@ bool is_bf16(const torch::Tensor& t) {
 void check_gpu_contig(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+// 2D tensor whose rows may be strided (last dim contiguous) — the dual-
+// direction LSTM reads gate slabs / writes hidden halves inside wider
+// concatenated buffers.
+long row_stride_checked(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.dim() == 2 && t.stride(1) == 1, name,
+              " must be 2D with contiguous rows");
+  return t.stride(0);
 }
 
 torch::Tensor gather_mean_fwd(torch::Tensor h, torch::Tensor idx,
@@ -117,10 +129,10 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
                         torch::Tensor mask, torch::Tensor h_out,
                         torch::Tensor c_out, torch::Tensor gates_act) {
   check_gpu_contig(hg, "hg");
-  check_gpu_contig(xg, "xg");
+  const long xg_stride = row_stride_checked(xg, "xg");
   check_gpu_contig(c_prev, "c_prev");
   check_gpu_contig(h_prev, "h_prev");
-  check_gpu_contig(h_out, "h_out");
+  const long hout_stride = row_stride_checked(h_out, "h_out");
   check_gpu_contig(c_out, "c_out");
   check_gpu_contig(gates_act, "gates_act");
   const long batch = c_prev.size(0);
@@ -141,7 +153,8 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
   nerrf::launch_lstm_pointwise_fwd(
       hg.data_ptr(), xg.data_ptr(), bc.data_ptr(), c_prev.data_ptr(),
       h_prev.data_ptr(), mask_ptr, h_out.data_ptr(), c_out.data_ptr(),
-      gates_act.data_ptr(), batch, hdim, is_bf16(hg), stream.stream());
+      gates_act.data_ptr(), batch, hdim, xg_stride, hout_stride, is_bf16(hg),
+      stream.stream());
 }
 
 // grad_out_t (may be empty): this timestep's dL/dh, folded in-kernel so the
@@ -169,12 +182,13 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
   }
   auto stream = at::hip::getCurrentHIPStream();
   const void* got = grad_out_t.numel() ? grad_out_t.data_ptr() : nullptr;
-  if (grad_out_t.numel()) check_gpu_contig(grad_out_t, "grad_out_t");
+  long gout_stride = hdim;
+  if (grad_out_t.numel()) gout_stride = row_stride_checked(grad_out_t, "grad_out_t");
   nerrf::launch_lstm_pointwise_bwd(
       grad_h.data_ptr(), got, grad_c.data_ptr(), gates_act.data_ptr(),
       c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
       grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
-      is_bf16(grad_h), stream.stream());
+      gout_stride, is_bf16(grad_h), stream.stream());
 }
 
 // Fully-fused MFMA step (bf16, H == 256). Writes into caller buffers.

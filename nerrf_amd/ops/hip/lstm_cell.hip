@@ -16,28 +16,33 @@
 
 namespace nerrf {
 
+// xg_stride: row stride (elements) of the xg slice — lets the sequence op
+// read gate slabs straight out of a [T, B, 2*4H] dual-direction projection.
+// hout_stride: row stride of h_out — lets both directions write straight
+// into the [T, B, 2H] concatenated output (no torch.cat afterwards).
 template <typename T>
 __global__ void lstm_pointwise_fwd_kernel(
     const T* __restrict__ hg,         // [B, 4H] = h_prev @ W_hh^T (beta=0 GEMM)
-    const T* __restrict__ xg,         // [B, 4H] = x_t @ W_ih^T (time-major slice)
+    const T* __restrict__ xg,         // [B(row-stride xg_stride), 4H]
     const T* __restrict__ bias,       // [4H]
     const T* __restrict__ c_prev,     // [B, H]
     const T* __restrict__ h_prev,     // [B, H]
     const float* __restrict__ mask,   // [B] or nullptr
-    T* __restrict__ h_out,            // [B, H]
+    T* __restrict__ h_out,            // [B(row-stride hout_stride), H]
     T* __restrict__ c_out,            // [B, H]
     T* __restrict__ gates_act,        // [B, 4H]
-    long batch, int hdim) {
+    long batch, int hdim, long xg_stride, long hout_stride) {
   const long total = batch * hdim;
   for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     const long b = t / hdim;
     const int d = (int)(t % hdim);
     const long g0 = b * 4 * hdim + d;
-    const float ip = to_f32(hg[g0]) + to_f32(xg[g0]) + to_f32(bias[d]);
-    const float fp = to_f32(hg[g0 + hdim]) + to_f32(xg[g0 + hdim]) + to_f32(bias[d + hdim]);
-    const float gp = to_f32(hg[g0 + 2 * hdim]) + to_f32(xg[g0 + 2 * hdim]) + to_f32(bias[d + 2 * hdim]);
-    const float op = to_f32(hg[g0 + 3 * hdim]) + to_f32(xg[g0 + 3 * hdim]) + to_f32(bias[d + 3 * hdim]);
+    const long x0 = b * xg_stride + d;
+    const float ip = to_f32(hg[g0]) + to_f32(xg[x0]) + to_f32(bias[d]);
+    const float fp = to_f32(hg[g0 + hdim]) + to_f32(xg[x0 + hdim]) + to_f32(bias[d + hdim]);
+    const float gp = to_f32(hg[g0 + 2 * hdim]) + to_f32(xg[x0 + 2 * hdim]) + to_f32(bias[d + 2 * hdim]);
+    const float op = to_f32(hg[g0 + 3 * hdim]) + to_f32(xg[x0 + 3 * hdim]) + to_f32(bias[d + 3 * hdim]);
     const float i = sigmoidf_(ip);
     const float f = sigmoidf_(fp);
     const float g = tanhf(gp);
@@ -51,7 +56,7 @@ __global__ void lstm_pointwise_fwd_kernel(
       hn = m * hn + (1.0f - m) * to_f32(h_prev[t]);
     }
     c_out[t] = from_f32<T>(cn);
-    h_out[t] = from_f32<T>(hn);
+    h_out[b * hout_stride + d] = from_f32<T>(hn);
     gates_act[g0] = from_f32<T>(i);
     gates_act[g0 + hdim] = from_f32<T>(f);
     gates_act[g0 + 2 * hdim] = from_f32<T>(g);
@@ -62,7 +67,7 @@ __global__ void lstm_pointwise_fwd_kernel(
 template <typename T>
 __global__ void lstm_pointwise_bwd_kernel(
     const T* __restrict__ grad_h,     // [B, H] recurrent grad
-    const T* __restrict__ grad_out_t, // [B, H] this step's output grad (or nullptr)
+    const T* __restrict__ grad_out_t, // [B(row-stride gout_stride), H] or nullptr
     const T* __restrict__ grad_c,     // [B, H]
     const T* __restrict__ gates_act,  // [B, 4H]
     const T* __restrict__ c_prev,     // [B, H]
@@ -70,7 +75,7 @@ __global__ void lstm_pointwise_bwd_kernel(
     T* __restrict__ grad_gates,       // [B, 4H]
     T* __restrict__ grad_c_prev,      // [B, H]
     T* __restrict__ grad_h_pass,      // [B, H]
-    long batch, int hdim) {
+    long batch, int hdim, long gout_stride) {
   const long total = batch * hdim;
   for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
@@ -85,7 +90,7 @@ __global__ void lstm_pointwise_bwd_kernel(
     const float m = (mask != nullptr) ? mask[b] : 1.0f;
     const float tcn = tanhf(f * cp + i * g);  // tanh of UNMASKED c_new
     float gh_in = to_f32(grad_h[t]);
-    if (grad_out_t != nullptr) gh_in += to_f32(grad_out_t[t]);
+    if (grad_out_t != nullptr) gh_in += to_f32(grad_out_t[b * gout_stride + d]);
     const float gc_in = to_f32(grad_c[t]);
     const float gh = gh_in * m;
     const float gc = gc_in * m;
@@ -105,17 +110,17 @@ __global__ void lstm_pointwise_bwd_kernel(
 
 template __global__ void lstm_pointwise_fwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
-    float*, float*, float*, long, int);
+    float*, float*, float*, long, int, long, long);
 template __global__ void lstm_pointwise_fwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
     const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
-    long, int);
+    long, int, long, long);
 template __global__ void lstm_pointwise_bwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
-    float*, float*, float*, long, int);
+    float*, float*, float*, long, int, long);
 template __global__ void lstm_pointwise_bwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
-    const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int);
+    const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long);
 
 // ---------------------------------------------------------------------------
 // host launchers
@@ -131,7 +136,8 @@ static inline int grid_elems(long total, int block) {
 void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
                                const void* c_prev, const void* h_prev,
                                const float* mask, void* h_out, void* c_out,
-                               void* gates_act, long batch, int hdim, bool bf16,
+                               void* gates_act, long batch, int hdim,
+                               long xg_stride, long hout_stride, bool bf16,
                                hipStream_t s) {
   const int block = 256;
   const int grid = grid_elems(batch * hdim, block);
@@ -140,12 +146,13 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
         (const __hip_bfloat16*)hg, (const __hip_bfloat16*)xg,
         (const __hip_bfloat16*)bias, (const __hip_bfloat16*)c_prev,
         (const __hip_bfloat16*)h_prev, mask, (__hip_bfloat16*)h_out,
-        (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim);
+        (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim,
+        xg_stride, hout_stride);
   } else {
     lstm_pointwise_fwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)hg, (const float*)xg, (const float*)bias,
         (const float*)c_prev, (const float*)h_prev, mask, (float*)h_out,
-        (float*)c_out, (float*)gates_act, batch, hdim);
+        (float*)c_out, (float*)gates_act, batch, hdim, xg_stride, hout_stride);
   }
 }
 
@@ -154,7 +161,7 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
                                const void* c_prev, const float* mask,
                                void* grad_gates, void* grad_c_prev,
                                void* grad_h_pass, long batch, int hdim,
-                               bool bf16, hipStream_t s) {
+                               long gout_stride, bool bf16, hipStream_t s) {
   const int block = 256;
   const int grid = grid_elems(batch * hdim, block);
   if (bf16) {
@@ -162,12 +169,13 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
         (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_out_t,
         (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
         (const __hip_bfloat16*)c_prev, mask, (__hip_bfloat16*)grad_gates,
-        (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch, hdim);
+        (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch, hdim,
+        gout_stride);
   } else {
     lstm_pointwise_bwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
         (const float*)gates_act, (const float*)c_prev, mask, (float*)grad_gates,
-        (float*)grad_c_prev, (float*)grad_h_pass, batch, hdim);
+        (float*)grad_c_prev, (float*)grad_h_pass, batch, hdim, gout_stride);
   }
 }
 
