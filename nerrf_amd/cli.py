@@ -65,7 +65,12 @@ def cmd_undo(args) -> int:
     from .serve.engine import StreamingEngine
 
     t0 = time.time()
-    engine = StreamingEngine(device=args.device)
+    model = None
+    if args.checkpoint:
+        from .serve.engine import load_model_from_checkpoint
+
+        model = load_model_from_checkpoint(args.checkpoint)
+    engine = StreamingEngine(model=model, device=args.device)
     if args.trace:
         engine.ingest_events(load_trace(args.trace))
     det = engine.score_window()
@@ -159,6 +164,7 @@ def main(argv=None) -> int:
     p.add_argument("--device", default="cpu")
     p.add_argument("--sims", type=int, default=1024)
     p.add_argument("--force", action="store_true")
+    p.add_argument("--checkpoint", default=None, help="trained model checkpoint dir")
 
     p = sub.add_parser("scenario", help="full e2e attack->detect->recover")
     p.add_argument("--dir", required=True)

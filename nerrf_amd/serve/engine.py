@@ -45,6 +45,33 @@ class Detection:
     window_events: int = 0
 
 
+PRETRAINED_DIR = "checkpoints/pretrained"
+
+
+def load_model_from_checkpoint(path: str) -> NerrfJointModel:
+    """Load a joint model from a checkpoint dir (manifest carries the config)."""
+    import json
+    from pathlib import Path
+
+    from ..checkpoint import load_checkpoint
+    from ..config import _from_dict
+    from ..models.graphsage import SageConfig
+    from ..models.lstm import LSTMConfig
+
+    manifest = json.loads((Path(path) / "checkpoint.json").read_text())
+    cfg_d = (manifest.get("config") or {}).get("model") or manifest.get("config")
+    cfg = JointConfig()
+    if cfg_d:
+        cfg = JointConfig(
+            sage=SageConfig(**cfg_d.get("sage", {})),
+            lstm=LSTMConfig(**cfg_d.get("lstm", {})),
+            **{k: v for k, v in cfg_d.items() if k not in ("sage", "lstm")},
+        )
+    model = NerrfJointModel(cfg)
+    load_checkpoint(path, model)
+    return model
+
+
 class StreamingEngine:
     def __init__(
         self,
