@@ -194,3 +194,55 @@ def test_sweep_small(tmp_path):
     )
     assert len(res) == 2
     assert "node_auc" in res[0]["metrics"]
+
+
+def test_store_concurrent_append_and_compact():
+    """Delta store under concurrent writers + compactor (thread safety)."""
+    import threading
+
+    store = DeltaGraphStore(window_s=1e9, delta_s=0.5)
+    errors = []
+
+    def writer(tid):
+        try:
+            for i in range(500):
+                store.append(ts=i * 0.01 + tid, pid=tid, syscall="write",
+                             path=f"/d/t{tid}_f{i % 11}", nbytes=i)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    def compactor():
+        try:
+            for _ in range(20):
+                arr = store.compact()
+                assert np.all(np.diff(arr.ts) >= 0)
+                time.sleep(0.005)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=writer, args=(t,)) for t in range(4)]
+    threads.append(threading.Thread(target=compactor))
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors
+    assert store.total_events == 2000
+    assert len(store.compact()) == 2000
+
+
+def test_codec_malformed_frames_fail_loudly():
+    """Truncated/garbage frames raise instead of returning corrupt events."""
+    from nerrf_amd.wire import codec
+
+    good = codec.encode_event_batch(
+        [codec.Event(pid=1, syscall="write", path="/a", bytes=10)]
+    )
+    with pytest.raises(ValueError):
+        codec.decode_event_batch(good[:-2] + b"\xff\xff\xff\xff\xff\xff\xff\xff\xff\xff\xff")
+    with pytest.raises(ValueError):
+        codec.decode_event_batch(b"\x0a\xff\xff\xff\xff\xff\xff\xff\xff\xff\xff")
+
+
+def test_grpc_bridge_main_importable():
+    from nerrf_amd.serve import daemon_bridge_main  # noqa: F401
