@@ -274,3 +274,51 @@ class StreamingEngine:
             decrypt=True,
             validate_in_sandbox=True,
         )
+
+    # ------------------------------------------------------------- multi-shard
+    def merge_detections(self, det: Detection, group=None) -> Detection:
+        """Merge this shard's Detection with every other shard's.
+
+        One engine shard runs per GPU (config 5); shards exchange compact
+        detection summaries (scores/indicators, not raw events) through
+        torch.distributed — gloo on CPU tests, RCCL over xGMI on an MI355X
+        node.  Alarm is the OR over shards; per-file scores merge by max.
+        """
+        import torch.distributed as dist
+
+        if not dist.is_initialized() or dist.get_world_size(group) == 1:
+            return det
+        world = dist.get_world_size(group)
+        payload = {
+            "alarm": det.alarm,
+            "t_detect": det.t_detect,
+            "file_scores": det.file_scores,
+            "file_mb": det.file_mb,
+            "proc_scores": det.proc_scores,
+            "encrypted_paths": det.encrypted_paths,
+            "indicators": det.indicators,
+            "window_events": det.window_events,
+        }
+        gathered: list = [None] * world
+        dist.all_gather_object(gathered, payload, group=group)
+        merged = Detection(
+            alarm=any(g["alarm"] for g in gathered),
+            t_detect=min(g["t_detect"] for g in gathered),
+            file_scores={},
+            file_mb={},
+            proc_scores={},
+        )
+        enc: list = []
+        for g in gathered:
+            for p, s in g["file_scores"].items():
+                merged.file_scores[p] = max(merged.file_scores.get(p, 0.0), s)
+            for p, mb in g["file_mb"].items():
+                merged.file_mb[p] = merged.file_mb.get(p, 0.0) + mb
+            for p, s in g["proc_scores"].items():
+                merged.proc_scores[p] = max(merged.proc_scores.get(p, 0.0), s)
+            enc.extend(g["encrypted_paths"])
+            for k, v in g["indicators"].items():
+                merged.indicators[k] = max(merged.indicators.get(k, 0.0), v)
+            merged.window_events += g["window_events"]
+        merged.encrypted_paths = sorted(set(enc))
+        return merged
