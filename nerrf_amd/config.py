@@ -26,6 +26,7 @@ class DataConfig:
     window_s: float = 30.0
     stride_s: float = 15.0
     attack_fraction: float = 0.6
+    scenario_kinds: tuple = ("lockbit", "supply_chain")
     fanout: int = 16
     seq_len: int = 100
     seed: int = 0

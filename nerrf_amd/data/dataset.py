@@ -122,8 +122,10 @@ def synth_window_batches(
     fanout: int = 16,
     seq_len: int = 100,
     base_seed: int = 0,
+    kinds: tuple = ("lockbit", "supply_chain"),
 ) -> List[WindowBatch]:
-    """Prebuild window batches from synthetic scenarios."""
+    """Prebuild window batches from synthetic scenarios (attack scenarios
+    alternate over `kinds` so the model trains on every family)."""
     batches: List[WindowBatch] = []
     for i in range(n_scenarios):
         cfg = SynthConfig(
@@ -132,6 +134,7 @@ def synth_window_batches(
             attack=(i % 100) < int(attack_fraction * 100),
             seed=base_seed + 7919 * i,
             attack_start_frac=0.2 + 0.5 * ((i * 13) % 10) / 10.0,
+            kind=kinds[i % len(kinds)],
         )
         arr, win = generate(cfg)
         for j, (t0, evw) in enumerate(sliding_windows(arr, window_s, stride_s)):

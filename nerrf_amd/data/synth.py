@@ -50,6 +50,7 @@ class SynthConfig:
     target_dir: str = "/app/uploads"
     encrypted_ext: str = ".lockbit3"
     seed: int = 0
+    kind: str = "lockbit"  # "lockbit" | "supply_chain"
 
 
 _BENIGN_DIRS = ["/var/www/html", "/home/svc/data", "/app/cache", "/var/log/app"]
@@ -65,6 +66,8 @@ def _interleave(cols: List[Tuple[np.ndarray, ...]]) -> Tuple[np.ndarray, ...]:
 
 
 def generate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
+    if cfg.kind == "supply_chain":
+        return generate_supply_chain(cfg)
     rng = np.random.default_rng(cfg.seed)
     paths = StringTable()
     comms = StringTable()
@@ -260,3 +263,80 @@ def generate_mixed_dataset(
         )
         out.append(generate(cfg))
     return out
+
+
+def generate_supply_chain(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
+    """Supply-chain compromise trace: trojanize dependency entrypoints, then
+    stage application data into one exfil blob.  No renames, no ransom note,
+    no suspicious extensions — detection must come from graph/sequence
+    anomalies (one process fanning over the whole dependency tree, data
+    reads feeding a single growing blob).  Mirrors
+    harness/supply_chain.py's on-disk attack as a pure event generator."""
+    rng = np.random.default_rng(cfg.seed)
+    base, _ = generate(SynthConfig(
+        duration_s=cfg.duration_s,
+        n_benign_procs=cfg.n_benign_procs,
+        n_benign_files=cfg.n_benign_files,
+        benign_rate_hz=cfg.benign_rate_hz,
+        attack=False,
+        seed=cfg.seed,
+    ))
+    if not cfg.attack:
+        return base, None
+    paths, comms = base.paths, base.comms
+    atk_pid = np.int64(7777)
+    atk_comm = comms.intern("postinstall")
+    t0 = cfg.attack_start_frac * cfg.duration_s
+    n_deps = max(cfg.n_victim_files, 4)
+    dep_ids = np.array(
+        [paths.intern(f"/srv/app/node_modules/dep_{j:03d}/index.js") for j in range(n_deps)],
+        dtype=np.int64,
+    )
+    n_data = max(n_deps // 2, 2)
+    data_ids = np.array(
+        [paths.intern(f"/srv/app/data/records_{j:02d}.db") for j in range(n_data)],
+        dtype=np.int64,
+    )
+    blob_id = paths.intern("/srv/app/.cache/telemetry.bin")
+
+    ts_l, sys_l, path_l, bytes_l = [], [], [], []
+    t = t0
+    for j in range(n_deps):  # trojanize: openat, read, write per dep
+        for sc, by in (("openat", 0), ("read", 4096), ("write", 4200)):
+            ts_l.append(t)
+            sys_l.append(SYSCALL_IDS[sc])
+            path_l.append(dep_ids[j])
+            bytes_l.append(by)
+            t += 0.01 + rng.uniform(0, 0.01)
+    chunk = cfg.chunk_kb * 1024
+    k = max(int(cfg.victim_file_mb * 1e6) // chunk, 1)
+    for j in range(n_data):  # exfil staging: read data, write blob
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["openat"]); path_l.append(data_ids[j]); bytes_l.append(0)
+        t += 0.005
+        for _ in range(k):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["read"]); path_l.append(data_ids[j]); bytes_l.append(chunk)
+            t += (chunk / 1e6) / cfg.encrypt_rate_mbps / 2
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["write"]); path_l.append(blob_id); bytes_l.append(chunk)
+            t += (chunk / 1e6) / cfg.encrypt_rate_mbps / 2
+    n_atk = len(ts_l)
+    atk_cols = (
+        np.asarray(ts_l),
+        np.full(n_atk, atk_pid),
+        np.asarray(sys_l, dtype=np.int8),
+        np.asarray(path_l, dtype=np.int64),
+        np.full(n_atk, -1, dtype=np.int64),
+        np.asarray(bytes_l, dtype=np.int64),
+        np.full(n_atk, atk_comm, dtype=np.int64),
+    )
+    benign_cols = (
+        base.ts, base.pid, base.syscall, base.path_id, base.new_path_id,
+        base.nbytes, base.comm_id,
+    )
+    ts_, pid_, sys_, path_, newp_, bytes_, comm_ = _interleave([benign_cols, atk_cols])
+    arr = EventArray(
+        ts=ts_, pid=pid_, syscall=sys_, path_id=path_, new_path_id=newp_,
+        nbytes=bytes_, ret_val=np.zeros(len(ts_), dtype=np.int64), comm_id=comm_,
+        paths=paths, comms=comms,
+    )
+    window = AttackWindow(t_start=float(t0), t_end=float(t) + 0.01, target_dir="/srv/app")
+    return arr, window

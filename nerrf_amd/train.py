@@ -106,6 +106,7 @@ def run_training(cfg: TrainConfig, resume: str | None = None) -> Dict[str, float
         fanout=d.fanout,
         seq_len=d.seq_len,
         base_seed=d.seed + 100000 * rank,
+        kinds=tuple(d.scenario_kinds),
     )
     holdout = synth_window_batches(
         n_scenarios=cfg.run.eval_holdout,
@@ -117,6 +118,7 @@ def run_training(cfg: TrainConfig, resume: str | None = None) -> Dict[str, float
         fanout=d.fanout,
         seq_len=d.seq_len,
         base_seed=d.seed + 999331,  # disjoint from every rank's train shard
+        kinds=tuple(d.scenario_kinds),
     )
 
     model = NerrfJointModel(cfg.model).to(device=device, dtype=dtype)
