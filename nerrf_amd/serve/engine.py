@@ -322,3 +322,56 @@ class StreamingEngine:
             merged.window_events += g["window_events"]
         merged.encrypted_paths = sorted(set(enc))
         return merged
+
+    # ----------------------------------------------------------------- monitor
+    def run_monitor(
+        self,
+        interval_s: float = 5.0,
+        max_iterations: Optional[int] = None,
+        on_alarm=None,
+        tracker_address: Optional[str] = None,
+        stop_event=None,
+        sims: int = 1024,
+    ):
+        """Continuous detection loop: ingest -> score -> (on alarm) plan.
+
+        The operational serve mode (config 5): every `interval_s` the engine
+        compacts the delta window, scores it, and on alarm produces a ranked
+        undo plan, invoking `on_alarm(detection, plan)` — the caller decides
+        whether to execute (`respond`).  Bounded by `max_iterations` /
+        `stop_event` for tests and supervised runs.  Yields per-iteration
+        status dicts.
+        """
+        import threading
+
+        it = 0
+        ingest_thread = None
+        if tracker_address is not None:
+            ingest_thread = threading.Thread(
+                target=self.ingest_from_tracker,
+                kwargs={"address": tracker_address, "timeout_s": None},
+                daemon=True,
+            )
+            ingest_thread.start()
+        while max_iterations is None or it < max_iterations:
+            if stop_event is not None and stop_event.is_set():
+                return
+            t0 = time.perf_counter()
+            det = self.score_window()
+            plan = None
+            if det.alarm:
+                plan = self.plan(det, n_sims=sims)
+                if on_alarm is not None:
+                    on_alarm(det, plan)
+            status = {
+                "iteration": it,
+                "window_events": det.window_events,
+                "alarm": det.alarm,
+                "score_s": time.perf_counter() - t0,
+                "plan": None if plan is None else plan.describe(self.planner_params.n_groups),
+            }
+            yield status
+            it += 1
+            elapsed = time.perf_counter() - t0
+            if max_iterations is None or it < max_iterations:
+                time.sleep(max(0.0, interval_s - elapsed))

@@ -246,3 +246,17 @@ def test_codec_malformed_frames_fail_loudly():
 
 def test_grpc_bridge_main_importable():
     from nerrf_amd.serve import daemon_bridge_main  # noqa: F401
+
+
+def test_monitor_loop_detects_and_plans(tmp_path):
+    engine = _small_engine(device="cpu")
+    engine.store.window_s = 1e9
+    arr, _ = generate(SynthConfig(seed=12, duration_s=30, benign_rate_hz=40, n_victim_files=6))
+    engine.ingest_events(arr)
+    alarms = []
+    statuses = list(
+        engine.run_monitor(interval_s=0.0, max_iterations=2, on_alarm=lambda d, p: alarms.append((d, p)))
+    )
+    assert len(statuses) == 2
+    assert statuses[0]["alarm"]
+    assert alarms and alarms[0][1].plan  # a non-empty undo plan was produced
