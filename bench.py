@@ -179,8 +179,10 @@ def main() -> None:
     elapsed = time.perf_counter() - t0
 
     # whole-job: max time over ranks, sum of events over ranks
-    t_tensor = torch.tensor([elapsed], dtype=torch.float64)
-    e_tensor = torch.tensor([float(events_done)], dtype=torch.float64)
+    # (tensors must live on the backend's device: RCCL reduces GPU tensors)
+    red_dev = device if (dist.is_initialized() and dist.get_backend() == "nccl") else "cpu"
+    t_tensor = torch.tensor([elapsed], dtype=torch.float64, device=red_dev)
+    e_tensor = torch.tensor([float(events_done)], dtype=torch.float64, device=red_dev)
     if dist.is_initialized():
         dist.all_reduce(t_tensor, op=dist.ReduceOp.MAX)
         dist.all_reduce(e_tensor, op=dist.ReduceOp.SUM)
