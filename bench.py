@@ -75,7 +75,11 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--scale", choices=["tiny", "small", "full"], default="full")
-    ap.add_argument("--windows", type=int, default=2, help="prebuilt windows per rank")
+    ap.add_argument("--windows", type=int, default=2, help="prebuilt window-batches per rank")
+    ap.add_argument(
+        "--batch-windows", type=int, default=2,
+        help="30s windows fused per training batch (disjoint graph union)",
+    )
     ap.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
     ap.add_argument(
         "--graphs",
@@ -101,7 +105,13 @@ def main() -> None:
     if not has_gpu and scale == "full":
         scale = "small"  # CPU hosts cannot time the full config meaningfully
 
-    batches_np = build_bench_batches(rank, args.windows, scale)
+    from nerrf_amd.data.dataset import collate_windows
+
+    raw_np = build_bench_batches(rank, args.windows * args.batch_windows, scale)
+    batches_np = [
+        collate_windows(raw_np[i : i + args.batch_windows])
+        for i in range(0, len(raw_np), args.batch_windows)
+    ]
     batches = [b.to_torch(device=device, dtype=dtype) for b in batches_np]
     events_per_step = [int(b.n_events) for b in batches_np]
 
@@ -206,7 +216,7 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "model": "GraphSAGE-T(28x128)+BiLSTM(2x256) joint",
-                "global_batch": world * 1,
+                "global_batch": world * args.batch_windows,
                 "seq_len": 100,
                 "window_s": 30,
                 "parallelism": f"dp{world}",

@@ -155,3 +155,73 @@ def iterate_epochs(
         order = rng.permutation(len(batches)) if shuffle else np.arange(len(batches))
         for i in order:
             yield batches[int(i)].to_torch(device=device, dtype=dtype)
+
+
+def collate_windows(batches: List[WindowBatch]) -> WindowBatch:
+    """Disjoint union of window batches into one training batch.
+
+    Graphs merge block-diagonally (node/edge ids offset, no cross-window
+    edges or samples — the model output over the union equals the
+    concatenation of per-window outputs), sequences concatenate along B.
+    Reverse indexes stay sorted because every window's node ids are offset
+    above the previous window's range.
+    """
+    if len(batches) == 1:
+        return batches[0]
+    node_off = 0
+    edge_off = 0
+    xs, nbr_i, nbr_w, ei, ew, ets, yn, ye = [], [], [], [], [], [], [], []
+    sf, sl, ysq = [], [], []
+    rds, rss, rws = [], [], []
+    e0d, e0s, e0w, e1d, e1s, e1w = [], [], [], [], [], []
+    n_events = 0
+    have_rev = all(b.rev_dst is not None for b in batches)
+    have_erev = all(b.e0_rev is not None for b in batches)
+    for b in batches:
+        n = b.x.shape[0]
+        e = b.edge_index.shape[1]
+        xs.append(b.x)
+        nbr_i.append(b.nbr_idx + node_off)
+        nbr_w.append(b.nbr_w)
+        ei.append(b.edge_index + node_off)
+        ew.append(b.edge_weight)
+        ets.append(b.edge_ts)
+        yn.append(b.y_node)
+        ye.append(b.y_edge)
+        sf.append(b.seq_feats)
+        sl.append(b.seq_lengths)
+        ysq.append(b.y_seq)
+        if have_rev:
+            rds.append(b.rev_dst + node_off)
+            rss.append(b.rev_src + node_off)
+            rws.append(b.rev_w)
+        if have_erev:
+            e0d.append(b.e0_rev[0] + node_off)
+            e0s.append(b.e0_rev[1] + edge_off)
+            e0w.append(b.e0_rev[2])
+            e1d.append(b.e1_rev[0] + node_off)
+            e1s.append(b.e1_rev[1] + edge_off)
+            e1w.append(b.e1_rev[2])
+        node_off += n
+        edge_off += e
+        n_events += b.n_events
+    return WindowBatch(
+        x=np.concatenate(xs),
+        nbr_idx=np.concatenate(nbr_i),
+        nbr_w=np.concatenate(nbr_w),
+        edge_index=np.concatenate(ei, axis=1),
+        edge_weight=np.concatenate(ew),
+        edge_ts=np.concatenate(ets),
+        y_node=np.concatenate(yn),
+        y_edge=np.concatenate(ye),
+        seq_feats=np.concatenate(sf),
+        seq_lengths=np.concatenate(sl),
+        y_seq=np.concatenate(ysq),
+        n_events=n_events,
+        seq_path_id=None,  # per-window path tables do not merge
+        rev_dst=np.concatenate(rds) if have_rev else None,
+        rev_src=np.concatenate(rss) if have_rev else None,
+        rev_w=np.concatenate(rws) if have_rev else None,
+        e0_rev=(np.concatenate(e0d), np.concatenate(e0s), np.concatenate(e0w)) if have_erev else None,
+        e1_rev=(np.concatenate(e1d), np.concatenate(e1s), np.concatenate(e1w)) if have_erev else None,
+    )
