@@ -61,7 +61,12 @@ def build_bench_batches(rank: int, n_windows: int, scale: str):
             seed=1234 + 7919 * (rank * 97 + i),
         )
         arr, win = generate(cfg)
+        min_events = int(rate * 30.0 * 0.8)  # full-rate windows only: the
+        # attack tail past the benign span yields sparse windows that would
+        # make step shapes uneven
         for t0, evw in sliding_windows(arr, window_s=30.0, stride_s=30.0):
+            if len(evw) < min_events:
+                continue
             batches.append(window_to_batch(evw, win, fanout=16, seq_len=100, seed=i))
             if len(batches) >= n_windows:
                 break
