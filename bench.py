@@ -82,7 +82,7 @@ def main() -> None:
     ap.add_argument("--scale", choices=["tiny", "small", "full"], default="full")
     ap.add_argument("--windows", type=int, default=2, help="prebuilt window-batches per rank")
     ap.add_argument(
-        "--batch-windows", type=int, default=2,
+        "--batch-windows", type=int, default=4,
         help="30s windows fused per training batch (disjoint graph union)",
     )
     ap.add_argument("--dtype", choices=["bf16", "fp32"], default="bf16")
