@@ -24,7 +24,7 @@ def _last_json_line(out: str) -> dict:
 def test_bench_single_process_contract():
     proc = subprocess.run(
         [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
-         "--scale", "tiny", "--windows", "1"],
+         "--scale", "tiny", "--windows", "1", "--batch-windows", "1"],
         cwd=ROOT, capture_output=True, text=True, timeout=900,
     )
     assert proc.returncode == 0, proc.stderr[-2000:]
@@ -50,7 +50,7 @@ def test_bench_torchrun_world2_gloo():
             "--nnodes=1", "--nproc-per-node", "2",
             "--master-addr", "127.0.0.1", "--master-port", "29517",
             "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
-            "--scale", "tiny", "--windows", "1",
+            "--scale", "tiny", "--windows", "1", "--batch-windows", "1",
         ],
         cwd=ROOT, capture_output=True, text=True, timeout=800, env=env,
     )
