@@ -141,19 +141,24 @@ class _LSTMSeqFn(torch.autograd.Function):
                 grad_h = torch.mm(grad_gates_all[ti], w_hh) + grad_h_pass
             grad_c, grad_c_prev = grad_c_prev, grad_c  # ping-pong buffers
 
-        # weight grads as single large GEMMs over all timesteps
-        # h input of step ti (time-major): shift h_all by one step
-        h_in_all = torch.empty_like(h_all)
-        if reverse:
-            h_in_all[t_len - 1] = h0
-            if t_len > 1:
-                h_in_all[:-1] = h_all[1:]
-        else:
-            h_in_all[0] = h0
-            if t_len > 1:
-                h_in_all[1:] = h_all[:-1]
+        # weight grads over all timesteps: the h input of step ti is h_all
+        # shifted by one step, whose bulk is a CONTIGUOUS prefix/suffix of
+        # h_all — two GEMMs (bulk + the h0 boundary row) instead of
+        # materialising a shifted copy of the whole history
         gg2 = grad_gates_all.reshape(t_len * batch, gdim)
-        grad_whh = torch.mm(gg2.t(), h_in_all.reshape(t_len * batch, hdim))
+        if t_len > 1:
+            if reverse:
+                bulk_gg = grad_gates_all[:-1].reshape((t_len - 1) * batch, gdim)
+                bulk_h = h_all[1:].reshape((t_len - 1) * batch, hdim)
+                edge_gg = grad_gates_all[t_len - 1]
+            else:
+                bulk_gg = grad_gates_all[1:].reshape((t_len - 1) * batch, gdim)
+                bulk_h = h_all[:-1].reshape((t_len - 1) * batch, hdim)
+                edge_gg = grad_gates_all[0]
+            grad_whh = torch.mm(bulk_gg.t(), bulk_h)
+            grad_whh = torch.addmm(grad_whh, edge_gg.t(), h0.to(dt))
+        else:
+            grad_whh = torch.mm(gg2.t(), h0.to(dt))
         grad_bias = gg2.sum(dim=0)
         grad_xg = grad_gates_all
         grad_h0 = grad_h
