@@ -143,3 +143,26 @@ def test_against_google_protobuf_if_available():
     theirs = parsed.SerializeToString()
     back = codec.decode_event(theirs)
     assert back == ev
+
+
+def test_codec_fuzz_random_bytes_no_silent_corruption():
+    """Random garbage either raises ValueError or decodes to events whose
+    re-encoding is stable (never crashes, never loops)."""
+    import numpy as np
+
+    rng = np.random.default_rng(0)
+    for i in range(200):
+        buf = bytes(rng.integers(0, 256, size=int(rng.integers(1, 80)), dtype=np.uint8))
+        try:
+            evs = codec.decode_event_batch(buf)
+        except ValueError:
+            continue
+        # decodable garbage must survive a re-encode/decode cycle
+        re_enc = codec.encode_event_batch(evs)
+        assert codec.decode_event_batch(re_enc) == evs
+
+
+def test_codec_large_values():
+    ev = codec.Event(pid=2**32 - 1, bytes=2**63 - 1, ret_val=-(2**62), uid=2**40)
+    back = codec.decode_event(codec.encode_event(ev))
+    assert back == ev
