@@ -20,14 +20,12 @@ from __future__ import annotations
 
 import argparse
 import json
-import os
 import sys
 import time
 
 import numpy as np
 import torch
 
-from nerrf_amd.data.dataset import synth_window_batches
 from nerrf_amd.models.graphsage import SageConfig
 from nerrf_amd.models.joint import JointConfig, NerrfJointModel
 from nerrf_amd.models.lstm import LSTMConfig

@@ -14,8 +14,6 @@ from typing import Any, List, Optional
 import yaml
 
 from .models.joint import JointConfig
-from .models.graphsage import SageConfig
-from .models.lstm import LSTMConfig
 
 
 @dataclass

@@ -14,7 +14,7 @@ millions of events/second (vectorised numpy, no per-event Python loop).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional, Tuple
 
 import numpy as np

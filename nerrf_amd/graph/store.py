@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import threading
 from collections import deque
-from typing import Deque, List, Optional, Tuple
+from typing import Deque, Optional
 
 import numpy as np
 

@@ -1,9 +1,10 @@
 """Fused-op public API with CPU-reference / CDNA4-native dispatch.
 
 Ops:
-  * gather_mean(h, idx, w)      — weighted neighbor aggregation (GraphSAGE-T)
+  * gather_mean(h, idx, w[, rev]) — weighted neighbor aggregation (GraphSAGE-T)
+  * gather_rows(h, idx[, rev])    — row gather with fast scatter backward
   * lstm_cell(xg, h, c, w_hh, b, mask) — fused LSTM recurrent step
-  * sage_layer_fused(...)       — inference-only fully fused SAGE layer (MFMA)
+  * lstm_sequence(...)            — whole-sequence fused LSTM (time-major)
 
 On CPU the pure-PyTorch reference runs; on ROCm devices the in-tree HIP
 extension is mandatory (missing extension => RuntimeError, never a silent
@@ -16,7 +17,7 @@ from typing import Optional, Tuple
 import torch
 
 from . import reference as _ref
-from .native import get_native, native_available
+from .native import get_native, native_available  # noqa: F401 (re-export)
 
 __all__ = ["gather_mean", "lstm_cell", "lstm_sequence", "native_available"]
 

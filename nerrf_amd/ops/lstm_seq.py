@@ -14,7 +14,7 @@ Everything is time-major [T, B, ...]; the model transposes once per sequence.
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

@@ -17,7 +17,7 @@ planner maximises  -(E[data_loss_MB] + 0.1 * downtime_s + fp_penalty_MB).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List
 
 import numpy as np

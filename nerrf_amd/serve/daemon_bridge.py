@@ -21,8 +21,6 @@ import threading
 from concurrent import futures
 from typing import Iterator, Optional
 
-import numpy as np
-
 from ..wire import codec
 
 

@@ -54,7 +54,6 @@ def load_model_from_checkpoint(path: str) -> NerrfJointModel:
     from pathlib import Path
 
     from ..checkpoint import load_checkpoint
-    from ..config import _from_dict
     from ..models.graphsage import SageConfig
     from ..models.lstm import LSTMConfig
 

@@ -15,7 +15,7 @@ from typing import Dict, List, Sequence
 
 import numpy as np
 
-from .config import load_config, to_dict
+from .config import load_config
 from .train import run_training
 
 DEFAULT_SPACE: Dict[str, Sequence] = {

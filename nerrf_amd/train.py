@@ -20,7 +20,7 @@ import numpy as np
 import torch
 
 from .checkpoint import load_checkpoint, save_checkpoint
-from .config import TrainConfig, load_config, to_dict
+from .config import TrainConfig, load_config
 from .data.dataset import WindowBatch, iterate_epochs, synth_window_batches
 from .data.trace import load_trace
 from .eval import detection_report
