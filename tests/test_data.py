@@ -94,3 +94,43 @@ def test_sequences_shapes_and_labels():
 def test_toy_trace_exists_and_loads():
     arr = load_trace("datasets/traces/toy_trace.csv")
     assert len(arr) > 100
+
+
+def test_event_array_time_window_and_slice():
+    arr, _ = generate(SynthConfig(seed=6, duration_s=40, benign_rate_hz=50))
+    w = arr.time_window(10.0, 20.0)
+    assert (w.ts >= 10.0).all() and (w.ts < 20.0).all()
+    s = arr.slice(5, 15)
+    assert len(s) == 10
+    assert s.paths is arr.paths  # tables shared, not copied
+
+
+def test_string_table_roundtrip():
+    from nerrf_amd.data.trace import StringTable
+
+    t = StringTable()
+    a = t.intern("/x/y")
+    b = t.intern("/x/z")
+    assert t.intern("/x/y") == a  # stable ids
+    assert t.lookup(b) == "/x/z"
+    assert t.get("/nope") is None
+    assert len(t) == 2
+
+
+def test_concat_event_arrays():
+    from nerrf_amd.data.trace import concat
+
+    a1, _ = generate(SynthConfig(seed=1, duration_s=10, benign_rate_hz=20, attack=False))
+    a2, _ = generate(SynthConfig(seed=2, duration_s=10, benign_rate_hz=20, attack=False))
+    c = concat([a1, a2])
+    assert len(c) == len(a1) + len(a2)
+    assert (c.ts[:-1] <= c.ts[1:]).all()
+
+
+def test_perf_noop_on_cpu():
+    import torch
+
+    from nerrf_amd.perf import enable_tuned_gemms
+
+    if not torch.cuda.is_available():
+        assert enable_tuned_gemms() is False
