@@ -10,7 +10,8 @@ from nerrf_amd.perf import enable_tuned_gemms
 
 enable_tuned_gemms()
 dev = "cuda:0" if torch.cuda.is_available() else "cpu"
-model = load_model_from_checkpoint("checkpoints/pretrained")
+_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+model = load_model_from_checkpoint(os.path.join(_ROOT, "checkpoints", "pretrained"))
 engine = StreamingEngine(model=model, device=dev, dtype=torch.bfloat16 if dev != "cpu" else torch.float32)
 engine.store.window_s = 1e9
 
