@@ -110,8 +110,26 @@ def cmd_scenario(args) -> int:
 
 
 def cmd_serve(args) -> int:
-    from .data.trace import load_trace
     from .serve.engine import StreamingEngine
+
+    if args.tracker:
+        # live mode: consume an existing Tracker/StreamEvents endpoint and
+        # monitor continuously (deploy/engine-deployment.yaml wiring)
+        model = None
+        if args.checkpoint:
+            from .serve.engine import load_model_from_checkpoint
+
+            model = load_model_from_checkpoint(args.checkpoint)
+        engine = StreamingEngine(model=model, device=args.device)
+        for status in engine.run_monitor(
+            interval_s=args.interval,
+            max_iterations=args.iterations,
+            tracker_address=args.tracker,
+        ):
+            print(json.dumps(status), flush=True)
+        return 0
+
+    from .data.trace import load_trace
     from .serve.tracker_sim import TrackerSimServer
 
     trace = load_trace(args.trace)
@@ -173,8 +191,12 @@ def main(argv=None) -> int:
     p.add_argument("--device", default="cpu")
     p.add_argument("--sims", type=int, default=512)
 
-    p = sub.add_parser("serve", help="tracker-sim + streaming engine demo")
-    p.add_argument("--trace", required=True)
+    p = sub.add_parser("serve", help="streaming engine (live tracker or trace demo)")
+    p.add_argument("--trace", default=None, help="trace file (demo mode)")
+    p.add_argument("--tracker", default=None, help="live Tracker/StreamEvents address")
+    p.add_argument("--interval", type=float, default=5.0)
+    p.add_argument("--iterations", type=int, default=None)
+    p.add_argument("--checkpoint", default=None)
     p.add_argument("--rate", type=float, default=0.0)
     p.add_argument("--max-events", type=int, default=None)
     p.add_argument("--timeout", type=float, default=15.0)

@@ -260,3 +260,22 @@ def test_monitor_loop_detects_and_plans(tmp_path):
     assert len(statuses) == 2
     assert statuses[0]["alarm"]
     assert alarms and alarms[0][1].plan  # a non-empty undo plan was produced
+
+
+def test_cli_serve_live_monitor_mode(capsys):
+    """`nerrf serve --tracker` consumes a live stream and monitors."""
+    from nerrf_amd.cli import main
+    from nerrf_amd.serve.tracker_sim import TrackerSimServer
+
+    arr, _ = generate(SynthConfig(seed=13, duration_s=25, benign_rate_hz=40, n_victim_files=5))
+    server = TrackerSimServer(arr, batch_size=64)
+    server.start()
+    try:
+        rc = main(["serve", "--tracker", server.address, "--interval", "1.5", "--iterations", "2"])
+        assert rc == 0
+        lines = [json.loads(l) for l in capsys.readouterr().out.strip().splitlines() if l.startswith("{")]
+        assert len(lines) == 2
+        assert lines[-1]["window_events"] > 0
+        assert lines[-1]["alarm"]  # .lockbit3 traffic in the stream
+    finally:
+        server.stop()

@@ -163,13 +163,27 @@ std::vector<Event> load_jsonl(const std::string& path) {
 
 #ifdef NERRF_HAVE_LIBBPF
 // ---- live source: eBPF ring buffer ----------------------------------------
+// capture timestamps are CLOCK_MONOTONIC (bpf_ktime_get_ns); convert to wall
+// clock with the boot-time offset computed once at startup (the upstream
+// tracker's contract: absolute UTC timestamps on the wire)
+static uint64_t mono_to_wall_offset_ns() {
+  timespec mono{}, wall{};
+  clock_gettime(CLOCK_MONOTONIC, &mono);
+  clock_gettime(CLOCK_REALTIME, &wall);
+  uint64_t m = uint64_t(mono.tv_sec) * 1000000000ull + mono.tv_nsec;
+  uint64_t w = uint64_t(wall.tv_sec) * 1000000000ull + wall.tv_nsec;
+  return w - m;
+}
+
 int handle_ringbuf_event(void* ctx, void* data, size_t size) {
   if (size < sizeof(nerrf_event)) return 0;
+  static const uint64_t boot_offset = mono_to_wall_offset_ns();
   auto* raw = static_cast<const nerrf_event*>(data);
   auto* out = static_cast<std::vector<Event>*>(ctx);
   Event ev;
-  ev.ts_sec = int64_t(raw->ts_ns / 1000000000ull);
-  ev.ts_nsec = int32_t(raw->ts_ns % 1000000000ull);
+  const uint64_t wall_ns = raw->ts_ns + boot_offset;
+  ev.ts_sec = int64_t(wall_ns / 1000000000ull);
+  ev.ts_nsec = int32_t(wall_ns % 1000000000ull);
   ev.pid = raw->pid;
   ev.tid = raw->tid;
   ev.comm = raw->comm;
