@@ -123,18 +123,24 @@ def synth_window_batches(
     seq_len: int = 100,
     base_seed: int = 0,
     kinds: tuple = ("lockbit", "supply_chain"),
+    benign_kinds: tuple = ("lockbit", "benign_rotate", "benign_backup"),
 ) -> List[WindowBatch]:
-    """Prebuild window batches from synthetic scenarios (attack scenarios
-    alternate over `kinds` so the model trains on every family)."""
+    """Prebuild window batches from synthetic scenarios.
+
+    Attack scenarios alternate over `kinds`; clean scenarios alternate over
+    `benign_kinds` (plain background plus the hard negatives — log rotation
+    and backup daemons — so the model learns to NOT fire on attack
+    lookalikes: the FP-undo < 5% target)."""
     batches: List[WindowBatch] = []
     for i in range(n_scenarios):
+        is_attack = (i % 100) < int(attack_fraction * 100)
         cfg = SynthConfig(
             duration_s=duration_s,
             benign_rate_hz=benign_rate_hz,
-            attack=(i % 100) < int(attack_fraction * 100),
+            attack=is_attack,
             seed=base_seed + 7919 * i,
             attack_start_frac=0.2 + 0.5 * ((i * 13) % 10) / 10.0,
-            kind=kinds[i % len(kinds)],
+            kind=kinds[i % len(kinds)] if is_attack else benign_kinds[i % len(benign_kinds)],
         )
         arr, win = generate(cfg)
         for j, (t0, evw) in enumerate(sliding_windows(arr, window_s, stride_s)):

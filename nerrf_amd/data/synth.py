@@ -50,7 +50,7 @@ class SynthConfig:
     target_dir: str = "/app/uploads"
     encrypted_ext: str = ".lockbit3"
     seed: int = 0
-    kind: str = "lockbit"  # "lockbit" | "supply_chain"
+    kind: str = "lockbit"  # lockbit | supply_chain | benign_rotate | benign_backup
 
 
 _BENIGN_DIRS = ["/var/www/html", "/home/svc/data", "/app/cache", "/var/log/app"]
@@ -68,6 +68,10 @@ def _interleave(cols: List[Tuple[np.ndarray, ...]]) -> Tuple[np.ndarray, ...]:
 def generate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
     if cfg.kind == "supply_chain":
         return generate_supply_chain(cfg)
+    if cfg.kind == "benign_rotate":
+        return generate_benign_rotate(cfg)
+    if cfg.kind == "benign_backup":
+        return generate_benign_backup(cfg)
     rng = np.random.default_rng(cfg.seed)
     paths = StringTable()
     comms = StringTable()
@@ -340,3 +344,97 @@ def generate_supply_chain(cfg: SynthConfig) -> Tuple[EventArray, Optional[Attack
     )
     window = AttackWindow(t_start=float(t0), t_end=float(t) + 0.01, target_dir="/srv/app")
     return arr, window
+
+
+def generate_benign_rotate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
+    """Hard negative: log rotation / housekeeping daemon.
+
+    Rename-heavy benign burst (x.log -> x.log.1, truncate + rewrite, gzip
+    side-writes) that stresses the write-to-rename and double-extension
+    indicators without any attack.  Ground truth: clean (window None)."""
+    rng = np.random.default_rng(cfg.seed)
+    base, _ = generate(SynthConfig(
+        duration_s=cfg.duration_s, n_benign_procs=cfg.n_benign_procs,
+        n_benign_files=cfg.n_benign_files, benign_rate_hz=cfg.benign_rate_hz,
+        attack=False, seed=cfg.seed,
+    ))
+    paths, comms = base.paths, base.comms
+    rot_pid = np.int64(888)
+    rot_comm = comms.intern("logrotate")
+    t0 = cfg.attack_start_frac * cfg.duration_s
+    n_logs = max(cfg.n_victim_files, 8)
+    ts_l, sys_l, path_l, newp_l, bytes_l = [], [], [], [], []
+    t = t0
+    for j in range(n_logs):
+        log_id = paths.intern(f"/var/log/svc/app_{j:03d}.log")
+        rot_id = paths.intern(f"/var/log/svc/app_{j:03d}.log.1")
+        gz_id = paths.intern(f"/var/log/svc/app_{j:03d}.log.1.gz")
+        # rename current -> .1
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["rename"]); path_l.append(log_id); newp_l.append(rot_id); bytes_l.append(0)
+        t += 0.02
+        # recreate + first write
+        for sc, by in (("openat", 0), ("write", 512)):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS[sc]); path_l.append(log_id); newp_l.append(-1); bytes_l.append(by)
+            t += 0.01
+        # compress the rotated file: read .1, write .gz, unlink .1
+        for _ in range(4):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["read"]); path_l.append(rot_id); newp_l.append(-1); bytes_l.append(65536)
+            t += 0.01
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["write"]); path_l.append(gz_id); newp_l.append(-1); bytes_l.append(16384)
+            t += 0.01
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["unlink"]); path_l.append(rot_id); newp_l.append(-1); bytes_l.append(0)
+        t += 0.02 + rng.uniform(0, 0.02)
+    n = len(ts_l)
+    rot_cols = (
+        np.asarray(ts_l), np.full(n, rot_pid), np.asarray(sys_l, dtype=np.int8),
+        np.asarray(path_l, dtype=np.int64), np.asarray(newp_l, dtype=np.int64),
+        np.asarray(bytes_l, dtype=np.int64), np.full(n, rot_comm, dtype=np.int64),
+    )
+    benign_cols = (base.ts, base.pid, base.syscall, base.path_id,
+                   base.new_path_id, base.nbytes, base.comm_id)
+    ts_, pid_, sys_, path_, newp_, bytes_, comm_ = _interleave([benign_cols, rot_cols])
+    arr = EventArray(ts=ts_, pid=pid_, syscall=sys_, path_id=path_, new_path_id=newp_,
+                     nbytes=bytes_, ret_val=np.zeros(len(ts_), dtype=np.int64),
+                     comm_id=comm_, paths=paths, comms=comms)
+    return arr, None
+
+
+def generate_benign_backup(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
+    """Hard negative: backup daemon — reads many data files, writes one big
+    archive (the supply-chain exfil lookalike).  Clean ground truth."""
+    base, _ = generate(SynthConfig(
+        duration_s=cfg.duration_s, n_benign_procs=cfg.n_benign_procs,
+        n_benign_files=cfg.n_benign_files, benign_rate_hz=cfg.benign_rate_hz,
+        attack=False, seed=cfg.seed,
+    ))
+    paths, comms = base.paths, base.comms
+    bk_pid = np.int64(999)
+    bk_comm = comms.intern("backupd")
+    t0 = cfg.attack_start_frac * cfg.duration_s
+    n_files = max(cfg.n_victim_files, 8)
+    archive = paths.intern("/backup/daily/archive.tar")
+    chunk = cfg.chunk_kb * 1024
+    ts_l, sys_l, path_l, bytes_l = [], [], [], []
+    t = t0
+    for j in range(n_files):
+        src = paths.intern(f"/srv/data/records_{j:03d}.db")
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["openat"]); path_l.append(src); bytes_l.append(0)
+        t += 0.005
+        for _ in range(3):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["read"]); path_l.append(src); bytes_l.append(chunk)
+            t += 0.02
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["write"]); path_l.append(archive); bytes_l.append(chunk)
+            t += 0.02
+    n = len(ts_l)
+    bk_cols = (
+        np.asarray(ts_l), np.full(n, bk_pid), np.asarray(sys_l, dtype=np.int8),
+        np.asarray(path_l, dtype=np.int64), np.full(n, -1, dtype=np.int64),
+        np.asarray(bytes_l, dtype=np.int64), np.full(n, bk_comm, dtype=np.int64),
+    )
+    benign_cols = (base.ts, base.pid, base.syscall, base.path_id,
+                   base.new_path_id, base.nbytes, base.comm_id)
+    ts_, pid_, sys_, path_, newp_, bytes_, comm_ = _interleave([benign_cols, bk_cols])
+    arr = EventArray(ts=ts_, pid=pid_, syscall=sys_, path_id=path_, new_path_id=newp_,
+                     nbytes=bytes_, ret_val=np.zeros(len(ts_), dtype=np.int64),
+                     comm_id=comm_, paths=paths, comms=comms)
+    return arr, None

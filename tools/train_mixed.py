@@ -7,7 +7,7 @@ from nerrf_amd.data.dataset import synth_window_batches
 from nerrf_amd.serve.engine import load_model_from_checkpoint
 
 cfg = load_config(None, [
-    "optim.dtype=bfloat16", "optim.epochs=3", "data.n_scenarios=10",
+    "optim.dtype=bfloat16", "optim.epochs=3", "data.n_scenarios=14",
     "run.eval_holdout=4", "run.checkpoint_dir=gpurun_out/ckpt_mixed",
     "run.log_every=100",
 ])
@@ -18,3 +18,16 @@ for kind in ("lockbit", "supply_chain"):
                               kinds=(kind,))
     rep = evaluate(model, hb, "cuda", torch.bfloat16)
     print(f"family={kind}: " + json.dumps({k: round(float(v), 4) for k, v in rep.items() if "auc" in k or k.endswith("f1")}))
+# hard negatives: max anomaly score the model assigns on clean lookalikes
+import numpy as np
+for kind in ("benign_rotate", "benign_backup"):
+    hb = synth_window_batches(n_scenarios=2, attack_fraction=0.0, base_seed=777000,
+                              benign_kinds=(kind,))
+    mx = []
+    for b in hb:
+        tb = b.to_torch("cuda", torch.bfloat16)
+        nl, _, sl = model(tb)
+        mx.append(float(torch.sigmoid(nl.float()).max()))
+        if sl is not None and sl.numel():
+            mx.append(float(torch.sigmoid(sl.float()).max()))
+    print(f"negative={kind}: max_score={max(mx):.4f}")
