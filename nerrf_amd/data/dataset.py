@@ -133,7 +133,10 @@ def synth_window_batches(
     lookalikes: the FP-undo < 5% target)."""
     batches: List[WindowBatch] = []
     for i in range(n_scenarios):
-        is_attack = (i % 100) < int(attack_fraction * 100)
+        # interleave attack/benign at the requested ratio over ANY
+        # scenario count (i % 100 < frac*100 made every scenario an
+        # attack for n < 100)
+        is_attack = (i % 10) < round(attack_fraction * 10)
         cfg = SynthConfig(
             duration_s=duration_s,
             benign_rate_hz=benign_rate_hz,
