@@ -134,9 +134,9 @@ def synth_window_batches(
     batches: List[WindowBatch] = []
     for i in range(n_scenarios):
         # interleave attack/benign at the requested ratio over ANY
-        # scenario count (i % 100 < frac*100 made every scenario an
-        # attack for n < 100)
-        is_attack = (i % 10) < round(attack_fraction * 10)
+        # scenario count, spread so small n still gets both classes
+        _rank = [0, 2, 4, 6, 8, 1, 3, 5, 7, 9]  # rank of i%10 in a strided order
+        is_attack = _rank[i % 10] < round(attack_fraction * 10)
         cfg = SynthConfig(
             duration_s=duration_s,
             benign_rate_hz=benign_rate_hz,

@@ -7,7 +7,8 @@ from nerrf_amd.data.dataset import synth_window_batches
 from nerrf_amd.serve.engine import load_model_from_checkpoint
 
 cfg = load_config(None, [
-    "optim.dtype=bfloat16", "optim.epochs=3", "data.n_scenarios=14",
+    "optim.dtype=bfloat16", "optim.epochs=4", "data.n_scenarios=20",
+    "data.attack_fraction=0.5",
     "run.eval_holdout=4", "run.checkpoint_dir=gpurun_out/ckpt_mixed",
     "run.log_every=100",
 ])
