@@ -92,3 +92,31 @@ def test_sharded_detection_merge():
     assert results[1]["merged_alarm"] is True
     assert results[0]["merged_events"] == results[1]["merged_events"]
     assert results[0]["merged_enc"] > 0
+
+
+def test_fp_undo_zero_on_hard_negatives(tmp_path):
+    """FP-undo target: on benign lookalikes (log rotation, backup daemon)
+    the response loop takes ZERO destructive actions — even when the model
+    raises scores, there is nothing the executor will touch and the sandbox
+    gate refuses an empty restore."""
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.serve.engine import StreamingEngine, load_model_from_checkpoint
+
+    model = load_model_from_checkpoint("checkpoints/pretrained")
+    for kind in ("benign_rotate", "benign_backup"):
+        engine = StreamingEngine(model=model, device="cpu")
+        engine.store.window_s = 1e9
+        arr, _ = generate(SynthConfig(seed=91, duration_s=40, benign_rate_hz=50, kind=kind))
+        engine.ingest_events(arr)
+        det = engine.score_window()
+        # rule indicators must stay silent on these
+        assert det.indicators["suspicious_ext_count"] == 0
+        assert det.indicators["ransom_note"] == 0
+        # a victim dir with only clean files: respond() must not alter it
+        victim = tmp_path / kind
+        victim.mkdir()
+        (victim / "a.dat").write_bytes(b"x" * 64)
+        plan = engine.plan(det, n_sims=256)
+        res = engine.respond(det, plan, str(victim))
+        assert res.files_restored == 0
+        assert (victim / "a.dat").read_bytes() == b"x" * 64  # untouched
