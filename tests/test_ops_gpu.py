@@ -409,3 +409,52 @@ def test_streaming_engine_gpu_scores(gpu_device):
     assert det.window_events == len(arr)
     plan = engine.plan(det, n_sims=512, use_gpu=True)
     assert plan.simulations == 512
+
+
+def test_gather_mean_max_fanout_k64(gpu_device):
+    """K=64 is the wave-resident bound — exercise it exactly."""
+    torch.manual_seed(40)
+    n, k, d = 130, 64, 128
+    h = torch.randn(n, d, device=gpu_device)
+    idx = torch.randint(0, n, (n, k), device=gpu_device)
+    w = torch.rand(n, k, device=gpu_device) + 0.01
+    out = gather_mean(h, idx, w)
+    out_ref = ref.gather_mean_ref(h.cpu(), idx.cpu(), w.cpu())
+    assert torch.allclose(out.cpu(), out_ref, atol=1e-4, rtol=1e-4)
+
+
+def test_gather_mean_fanout_over_64_rejected(gpu_device):
+    h = torch.randn(8, 32, device=gpu_device)
+    idx = torch.randint(0, 8, (8, 65), device=gpu_device)
+    w = torch.rand(8, 65, device=gpu_device)
+    with pytest.raises(RuntimeError):
+        gather_mean(h, idx, w)
+
+
+def test_mcts_search_16_groups(gpu_device):
+    """Upper bound of the planner's group space (MAX_GROUPS=16)."""
+    from nerrf_amd.planner.mcts import run_mcts_gpu
+    from nerrf_amd.planner.rewards import PlannerParams, build_state
+
+    rng = np.random.default_rng(3)
+    st = build_state(rng.random(80), rng.random(80) * 3, 0.9, 40.0, n_groups=16)
+    res = run_mcts_gpu(st, PlannerParams(n_groups=16), n_sims=512, device=str(gpu_device))
+    assert res.simulations == 512
+    assert len(res.ranked_actions) == 18
+
+
+def test_lstm_sequence_t1_edge(gpu_device):
+    """Single-timestep sequences (boundary of the bwd two-GEMM split)."""
+    from nerrf_amd.ops import lstm_sequence
+
+    torch.manual_seed(41)
+    t, b, hd = 1, 17, 32
+    xg = torch.randn(t, b, 4 * hd, device=gpu_device, requires_grad=True)
+    h0 = torch.randn(b, hd, device=gpu_device)
+    c0 = torch.randn(b, hd, device=gpu_device)
+    w_hh = (torch.randn(4 * hd, hd, device=gpu_device) * 0.2).requires_grad_(True)
+    bias = torch.randn(4 * hd, device=gpu_device, requires_grad=True)
+    out = lstm_sequence(xg, h0, c0, w_hh, bias, torch.ones(t, b, device=gpu_device))
+    out.sum().backward()
+    assert torch.isfinite(xg.grad).all()
+    assert torch.isfinite(w_hh.grad).all()
