@@ -126,3 +126,63 @@ def test_nerrfd_replay_roundtrip(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=5)
+
+
+@pytest.mark.skipif(not DAEMON.exists(), reason="nerrfd not built")
+def test_full_pipeline_daemon_bridge_engine(tmp_path):
+    """Whole chain: C++ nerrfd (replay) -> TCP frames -> gRPC bridge ->
+    engine ingest -> detection + plan.  The deployment wiring, in-process."""
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.data.trace import write_csv
+    from nerrf_amd.models.graphsage import SageConfig
+    from nerrf_amd.models.joint import JointConfig, NerrfJointModel
+    from nerrf_amd.models.lstm import LSTMConfig
+    from nerrf_amd.serve.daemon_bridge import GrpcBridge
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    # attack trace -> jsonl for the daemon
+    arr, _ = generate(SynthConfig(seed=19, duration_s=30, benign_rate_hz=40, n_victim_files=6))
+    trace = tmp_path / "t.jsonl"
+    with open(trace, "w") as fh:
+        from nerrf_amd.data.trace import SYSCALL_NAMES
+
+        for i in range(len(arr)):
+            rec = {
+                "timestamp": float(arr.ts[i]),
+                "event": SYSCALL_NAMES.get(int(arr.syscall[i]), "unknown"),
+                "path": arr.paths.lookup(int(arr.path_id[i])) if arr.path_id[i] >= 0 else "",
+                "size": int(arr.nbytes[i]),
+                "pid": int(arr.pid[i]),
+            }
+            if arr.new_path_id[i] >= 0:
+                rec["new_path"] = arr.paths.lookup(int(arr.new_path_id[i]))
+            fh.write(json.dumps(rec) + "\n")
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [str(DAEMON), "--replay", str(trace), "--port", str(port), "--batch", "64"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+    )
+    bridge = None
+    try:
+        time.sleep(0.3)
+        bridge = GrpcBridge("127.0.0.1", port)
+        bridge.start()
+        model = NerrfJointModel(
+            JointConfig(sage=SageConfig(layers=3, hidden=32), lstm=LSTMConfig(hidden=32))
+        )
+        engine = StreamingEngine(model=model, device="cpu")
+        engine.store.window_s = 1e9
+        n = engine.ingest_from_tracker(bridge.address, max_events=len(arr), timeout_s=15.0)
+        assert n >= len(arr) * 0.9
+        det = engine.score_window()
+        assert det.alarm
+        plan = engine.plan(det, n_sims=128)
+        assert plan.plan  # a concrete undo plan came out the far end
+    finally:
+        if bridge is not None:
+            bridge.stop()
+        proc.terminate()
+        proc.wait(timeout=5)
