@@ -105,11 +105,13 @@ def pump_daemon_into_store(
 class GrpcBridge:
     """gRPC Tracker/StreamEvents server backed by a live nerrfd stream."""
 
-    def __init__(self, daemon_host: str, daemon_port: int, address: str = "127.0.0.1:0"):
+    def __init__(self, daemon_host: str, daemon_port: int, address: str = "127.0.0.1:0",
+                 wait_for_client: bool = True):
         import grpc
 
         self.daemon_host = daemon_host
         self.daemon_port = daemon_port
+        self.wait_for_client = wait_for_client
         self._stop = threading.Event()
         self._clients: list[queue.Queue] = []
         self._lock = threading.Lock()
@@ -141,6 +143,16 @@ class GrpcBridge:
         self._server.stop(grace)
 
     def _pump(self) -> None:
+        if self.wait_for_client:
+            # nerrfd replays as soon as ITS first client (this pump) connects;
+            # hold off until a gRPC consumer is attached so nothing is lost
+            import time as _time
+
+            while not self._stop.is_set():
+                with self._lock:
+                    if self._clients:
+                        break
+                _time.sleep(0.01)
         for frame in frames_from_daemon(self.daemon_host, self.daemon_port, timeout_s=30.0):
             if self._stop.is_set():
                 return
