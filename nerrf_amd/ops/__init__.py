@@ -161,3 +161,34 @@ def lstm_cell(
 
 
 from .lstm_seq import lstm_sequence  # noqa: E402  (re-export)
+
+
+def sage_encode_fused(gnn, x, nbr_idx, nbr_w):
+    """Inference-only fused GraphSAGE-T encode (bf16, hidden=128 on GPU).
+
+    Runs the input projection in hipBLASLt, then each of the 28 layers as one
+    fused MFMA kernel (gather + dual GEMM + GELU + LayerNorm + residual —
+    ops/hip/sage_fused.hip).  Falls back to the module path when the shape /
+    dtype / device contract is not met.
+    """
+    h = gnn.input_proj(x)
+    if not (
+        h.is_cuda
+        and h.dtype == torch.bfloat16
+        and h.shape[1] == 128
+        and nbr_idx.shape[1] <= 64
+    ):
+        for layer in gnn.layers:
+            h = layer(h, nbr_idx, nbr_w)
+        return h
+    ext = get_native(h)
+    idx = nbr_idx.contiguous()
+    w = nbr_w.detach().to(torch.float32).contiguous()
+    h = h.contiguous()
+    for layer in gnn.layers:
+        h = ext.sage_layer_fwd(
+            h, idx, w,
+            layer.w_self.weight, layer.w_nbr.weight, layer.w_nbr.bias,
+            layer.norm.weight, layer.norm.bias,
+        )
+    return h

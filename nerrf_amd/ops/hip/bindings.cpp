@@ -38,6 +38,9 @@ void launch_feature_assemble(const float*, const int*, const int*,
                              const float*, const float*, const float*,
                              const unsigned char*, const signed char*, float*,
                              float, int, hipStream_t);
+void launch_sage_layer_fwd(const void*, const long*, const float*, const void*,
+                           const void*, const void*, const void*, const void*,
+                           void*, int, int, hipStream_t);
 }  // namespace nerrf
 
 namespace {
@@ -255,6 +258,38 @@ torch::Tensor event_features(torch::Tensor ev_file, torch::Tensor ev_proc,
   return x;
 }
 
+// Fused GraphSAGE-T layer forward (inference): gather + dual MFMA GEMM +
+// GELU + LayerNorm + residual in one launch.  bf16, D=128 only.
+torch::Tensor sage_layer_fwd(torch::Tensor h, torch::Tensor nbr_idx,
+                             torch::Tensor nbr_w, torch::Tensor w_self,
+                             torch::Tensor w_nbr, torch::Tensor bias,
+                             torch::Tensor gamma, torch::Tensor beta) {
+  for (auto* t : {&h, &w_self, &w_nbr}) {
+    check_gpu_contig(*t, "sage arg");
+    TORCH_CHECK(t->scalar_type() == torch::kBFloat16, "sage_layer_fwd is bf16-only");
+  }
+  check_gpu_contig(nbr_idx, "nbr_idx");
+  TORCH_CHECK(h.size(1) == 128 && w_self.size(0) == 128 && w_self.size(1) == 128 &&
+                  w_nbr.size(0) == 128 && w_nbr.size(1) == 128,
+              "sage_layer_fwd requires D=128");
+  const int n = h.size(0);
+  const int k = nbr_idx.size(1);
+  TORCH_CHECK(k <= 64, "fanout must be <= 64");
+  auto wf = nbr_w.scalar_type() == torch::kFloat32 ? nbr_w.contiguous()
+                                                   : nbr_w.to(torch::kFloat32).contiguous();
+  auto bc = bias.contiguous().to(torch::kBFloat16);
+  auto gc = gamma.contiguous().to(torch::kBFloat16);
+  auto be = beta.contiguous().to(torch::kBFloat16);
+  auto out = torch::empty_like(h);
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_sage_layer_fwd(h.data_ptr(), nbr_idx.data_ptr<long>(),
+                               wf.data_ptr<float>(), w_self.data_ptr(),
+                               w_nbr.data_ptr(), bc.data_ptr(), gc.data_ptr(),
+                               be.data_ptr(), out.data_ptr(), n, k,
+                               stream.stream());
+  return out;
+}
+
 nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
   nerrf::PlannerParamsDev p;
   p.n_groups = d["n_groups"].cast<int>();
@@ -321,6 +356,8 @@ torch::Tensor mcts_eval_plans(torch::Tensor gscore, torch::Tensor gmb,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sage_layer_fwd", &sage_layer_fwd,
+        "fused GraphSAGE-T layer forward (MFMA, inference)");
   m.def("event_features", &event_features,
         "GPU delta compaction: events -> per-node feature matrix");
   m.def("mcts_search", &mcts_search, "batched root-parallel MCTS");
