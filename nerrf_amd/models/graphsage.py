@@ -66,6 +66,15 @@ class GraphSAGET(nn.Module):
         )
 
     def encode(self, x: torch.Tensor, nbr_idx: torch.Tensor, nbr_w: torch.Tensor, rev=None) -> torch.Tensor:
+        if (
+            not self.training
+            and getattr(self, "use_fused_inference", False)
+            and not torch.is_grad_enabled()
+        ):
+            # serving path: each layer = ONE MFMA kernel (ops/hip/sage_fused.hip)
+            from ..ops import sage_encode_fused
+
+            return sage_encode_fused(self, x, nbr_idx, nbr_w)
         h = self.input_proj(x)
         for layer in self.layers:
             h = layer(h, nbr_idx, nbr_w, rev)

@@ -86,6 +86,9 @@ class StreamingEngine:
         self.device = torch.device(device)
         self.dtype = dtype
         self.model = (model or NerrfJointModel(JointConfig())).to(self.device, self.dtype).eval()
+        if self.device.type == "cuda" and self.dtype == torch.bfloat16:
+            # one fused MFMA kernel per GNN layer on the scoring path
+            self.model.gnn.use_fused_inference = True
         self.store = DeltaGraphStore(window_s=window_s)
         self.alarm_threshold = alarm_threshold
         self.planner_params = planner_params or PlannerParams()
