@@ -129,6 +129,9 @@ def cmd_serve(args) -> int:
             print(json.dumps(status), flush=True)
         return 0
 
+    if not args.trace:
+        print(json.dumps({"error": "serve needs --tracker <addr> or --trace <file>"}))
+        return 2
     from .data.trace import load_trace
     from .serve.tracker_sim import TrackerSimServer
 
