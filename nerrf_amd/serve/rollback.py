@@ -59,7 +59,7 @@ def _restore_dir(
     restored = failed = 0
     nbytes = 0
     details: List[str] = []
-    for enc in sorted(directory.glob(f"*{encrypted_ext}")):
+    for enc in sorted(directory.rglob(f"*{encrypted_ext}")):
         try:
             size = enc.stat().st_size
             if decrypt:
@@ -91,9 +91,15 @@ def sandbox_validate(
         if failed:
             return False
         if manifest is not None:
-            remapped = {
-                str(staging / Path(p).name): digest for p, digest in manifest.items()
-            }
+            # map manifest entries into the staging tree by path relative to
+            # the live directory (supports nested victim layouts)
+            remapped = {}
+            for p, digest in manifest.items():
+                try:
+                    rel = Path(p).relative_to(directory)
+                except ValueError:
+                    rel = Path(Path(p).name)
+                remapped[str(staging / rel)] = digest
             checks = verify_manifest(remapped)
             return all(checks.values())
         return restored > 0

@@ -279,3 +279,18 @@ def test_cli_serve_live_monitor_mode(capsys):
         assert lines[-1]["alarm"]  # .lockbit3 traffic in the stream
     finally:
         server.stop()
+
+
+def test_rollback_nested_directories(tmp_path):
+    """Encrypted files in subdirectories are found and restored (rglob)."""
+    from nerrf_amd.harness.attack_sim import run_attack, seed_files, verify_manifest
+
+    m1 = seed_files(tmp_path / "a", n_files=3, file_kb=4, seed=5)
+    m2 = seed_files(tmp_path / "a" / "deep" / "b", n_files=2, file_kb=4, seed=6)
+    run_attack(tmp_path / "a")
+    run_attack(tmp_path / "a" / "deep" / "b")
+    manifest = {**m1, **m2}
+    res = execute_rollback(tmp_path / "a", manifest=manifest)
+    assert res.files_restored == 5
+    assert res.sha256_ok is True
+    assert all(verify_manifest(manifest).values())
