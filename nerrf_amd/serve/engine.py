@@ -194,11 +194,17 @@ class StreamingEngine:
                 if pid_ >= 0:
                     path = events.paths.lookup(pid_)
                     file_scores[path] = max(file_scores.get(path, 0.0), float(seq_score[bi]))
-        # bytes per file (window-local)
-        for i in range(len(events)):
-            if events.path_id[i] >= 0:
-                p = events.paths.lookup(int(events.path_id[i]))
-                file_mb[p] = file_mb.get(p, 0.0) + float(events.nbytes[i]) / 1e6
+        # bytes per file (window-local) — vectorised: a per-event Python loop
+        # here costs ~0.4 s per 600k-event window, which dominates the host
+        # path once the model forward is on the GPU
+        valid = events.path_id >= 0
+        mb_by_id = np.bincount(
+            events.path_id[valid],
+            weights=events.nbytes[valid].astype(np.float64),
+            minlength=len(events.paths.strings),
+        ) / 1e6
+        for i in np.nonzero(mb_by_id)[0]:
+            file_mb[events.paths.lookup(int(i))] = float(mb_by_id[i])
 
         # ---- rule indicators --------------------------------------------
         sc = events.syscall
