@@ -96,3 +96,38 @@ def test_empty_trace():
     g = build_graph(arr)
     assert g.num_nodes == 0
     assert g.num_edges == 0
+
+
+def test_rename_components_match_serial_union_find():
+    """The scipy connected-components root assignment reproduces the serial
+    min-parent union-find (_UnionFind) on a randomized rename graph."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable, SYSCALL_IDS
+    from nerrf_amd.graph.constructor import _UnionFind, build_graph_parts
+
+    rng = np.random.default_rng(7)
+    n_paths = 300
+    paths = StringTable()
+    names = [f"/d/f{i}" for i in range(n_paths)]
+    b = EventArrayBuilder(paths, StringTable())
+    pairs = []
+    t = 0.0
+    for _ in range(180):
+        i, j = rng.integers(0, n_paths, size=2)
+        pairs.append((int(i), int(j)))
+        b.add(ts=t, pid=10, syscall="rename", path=names[i], new_path=names[j])
+        t += 0.01
+    # a few touches so every path id exists in the table
+    for i in range(n_paths):
+        b.add(ts=t, pid=10, syscall="write", path=names[i], nbytes=1)
+        t += 0.001
+    ev = b.build()
+    parts = build_graph_parts(ev)
+
+    uf = _UnionFind(len(ev.paths))
+    ren = (ev.syscall == SYSCALL_IDS["rename"]) & (ev.new_path_id >= 0)
+    for a, c in zip(ev.path_id[ren], ev.new_path_id[ren]):
+        uf.union(int(a), int(c))
+    expect = np.array([uf.find(i) for i in range(len(ev.paths))])
+    assert np.array_equal(parts["path_root"], expect)
