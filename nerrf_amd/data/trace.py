@@ -103,6 +103,8 @@ class EventArray:
         return int(self.ts.shape[0])
 
     def sort_by_time(self) -> "EventArray":
+        if len(self.ts) == 0 or bool(np.all(self.ts[1:] >= self.ts[:-1])):
+            return self  # streaming deltas arrive near-sorted; skip the sort
         order = np.argsort(self.ts, kind="stable")
         return EventArray(
             ts=self.ts[order],
