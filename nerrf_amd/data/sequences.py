@@ -13,7 +13,6 @@ serving path's host time.
 """
 from __future__ import annotations
 
-import re
 from dataclasses import dataclass
 from typing import Optional
 
@@ -23,8 +22,6 @@ from .trace import EventArray
 
 SEQ_LEN = 100
 NUM_SEQ_FEATURES = 16
-
-_SUSPICIOUS_EXT = re.compile(r"\.(lockbit\w*|encrypted|locked|crypt\w*)$", re.IGNORECASE)
 
 
 @dataclass
@@ -42,10 +39,9 @@ def build_sequences(
     min_events: int = 2,
 ) -> SequenceBatch:
     """Group events by file, keep the trailing `seq_len` per file."""
-    sus = np.zeros(len(events.paths), dtype=np.float32)
-    for i, s in enumerate(events.paths.strings):
-        if _SUSPICIOUS_EXT.search(s):
-            sus[i] = 1.0
+    from ..graph.constructor import _string_flag_bits
+
+    sus = (_string_flag_bits(events.paths) & 1).astype(np.float32)
 
     idx = np.nonzero(events.path_id >= 0)[0]
     if idx.size == 0:
@@ -111,3 +107,84 @@ def build_sequences(
         np.maximum.at(labels, seq_of, y_event[ev])
 
     return SequenceBatch(feats=feats, lengths=lengths, file_path_id=fids, labels=labels)
+
+
+def build_sequences_torch(
+    events: EventArray,
+    device,
+    seq_len: int = SEQ_LEN,
+    min_events: int = 2,
+    dtype=None,
+):
+    """GPU sequence assembly: same semantics as build_sequences, in torch.
+
+    The serving host path spent ~57 ms per 600k-event window on the numpy
+    build; on the GPU the grouping is a radix argsort plus a handful of
+    flat scatters (~2 ms), and the 25 MB feature tensor never crosses PCIe.
+    Returns (feats [B,T,E] on device, lengths [B] cpu, file_path_id [B] cpu).
+    """
+    import torch
+
+    from ..graph.constructor import _string_flag_bits
+
+    dev = torch.device(device)
+    sus_np = (_string_flag_bits(events.paths) & 1).astype(np.float32)
+
+    def empty():
+        f = torch.zeros(0, seq_len, NUM_SEQ_FEATURES, device=dev, dtype=dtype or torch.float32)
+        z = torch.zeros(0, dtype=torch.int64)
+        return f, z, z
+
+    if not len(events):
+        return empty()
+    t_path = torch.from_numpy(events.path_id).to(dev, non_blocking=True)
+    t_newp = torch.from_numpy(events.new_path_id).to(dev, non_blocking=True)
+    t_sc = torch.from_numpy(events.syscall).to(dev, non_blocking=True)
+    t_nb = torch.from_numpy(np.ascontiguousarray(events.nbytes, dtype=np.float32)).to(dev, non_blocking=True)
+    t_ts = torch.from_numpy(events.ts).to(dev, non_blocking=True)  # float64: epoch ts
+    t_pid = torch.from_numpy(events.pid).to(dev, non_blocking=True)
+    t_sus = torch.from_numpy(sus_np).to(dev, non_blocking=True)
+
+    idx = (t_path >= 0).nonzero(as_tuple=True)[0]
+    if idx.numel() == 0:
+        return empty()
+    pids = t_path[idx]
+    order = torch.argsort(pids, stable=True)
+    idx = idx[order]
+    pids = pids[order]
+    n = idx.numel()
+    new_grp = torch.ones(n, dtype=torch.bool, device=dev)
+    new_grp[1:] = pids[1:] != pids[:-1]
+    starts = new_grp.nonzero(as_tuple=True)[0]
+    ends = torch.cat([starts[1:], torch.tensor([n], device=dev)])
+    sizes = ends - starts
+    keep = sizes >= min_events
+    starts, ends, sizes = starts[keep], ends[keep], sizes[keep]
+    b = int(starts.numel())
+    if b == 0:
+        return empty()
+    lengths = sizes.clamp(max=seq_len)
+    sel_start = ends - lengths
+    seq_of = torch.repeat_interleave(torch.arange(b, device=dev), lengths)
+    cum = torch.cumsum(lengths, 0) - lengths
+    pos = torch.arange(seq_of.numel(), device=dev) - cum[seq_of]
+    flat = sel_start[seq_of] + pos
+    ev = idx[flat]
+    prev = idx[(flat - 1).clamp(min=0)]
+
+    feats = torch.zeros(b * seq_len * NUM_SEQ_FEATURES, device=dev, dtype=torch.float32)
+    base = (seq_of * seq_len + pos) * NUM_SEQ_FEATURES
+    feats[base + t_sc[ev].long().clamp(0, 9)] = 1.0
+    feats[base + 10] = torch.log1p(t_nb[ev]) / 16.0
+    ts_ev = t_ts[ev]
+    dt = ts_ev - torch.where(pos > 0, t_ts[prev], ts_ev)
+    feats[base + 11] = torch.log1p(dt).float()
+    feats[base + 12] = t_sus[t_path[ev]]
+    np_ids = t_newp[ev]
+    feats[base + 13] = torch.where(np_ids >= 0, t_sus[np_ids.clamp(min=0)], torch.zeros((), device=dev))
+    feats[base + 14] = ((pos > 0) & (t_pid[ev] == t_pid[prev])).float()
+    feats = feats.view(b, seq_len, NUM_SEQ_FEATURES)
+    if dtype is not None:
+        feats = feats.to(dtype)
+    fids = t_path[idx[sel_start]].cpu()
+    return feats, lengths.cpu(), fids

@@ -134,3 +134,33 @@ def test_perf_noop_on_cpu():
 
     if not torch.cuda.is_available():
         assert enable_tuned_gemms() is False
+
+
+def test_build_sequences_torch_matches_numpy():
+    """The torch (GPU-path) sequence builder reproduces the numpy build
+    bit-for-bit on CPU tensors: same grouping, lengths, ids and features."""
+    import numpy as np
+    import torch
+
+    from nerrf_amd.data.sequences import build_sequences, build_sequences_torch
+    from nerrf_amd.data.synth import SynthConfig, generate
+
+    arr, _ = generate(SynthConfig(duration_s=6.0, benign_rate_hz=2000.0,
+                                  n_benign_files=150, seed=11))
+    ref = build_sequences(arr, None)
+    feats, lengths, fids = build_sequences_torch(arr, device="cpu")
+    assert feats.shape == ref.feats.shape
+    assert np.array_equal(lengths.numpy(), ref.lengths)
+    assert np.array_equal(fids.numpy(), ref.file_path_id)
+    torch.testing.assert_close(feats, torch.from_numpy(ref.feats), rtol=1e-6, atol=1e-6)
+
+
+def test_build_sequences_torch_empty():
+    import torch
+
+    from nerrf_amd.data.sequences import build_sequences_torch
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+
+    ev = EventArrayBuilder(StringTable(), StringTable()).build()
+    feats, lengths, fids = build_sequences_torch(ev, device="cpu")
+    assert feats.shape[0] == 0 and lengths.numel() == 0

@@ -504,3 +504,22 @@ def test_sage_encode_fused_full_stack(gpu_device):
     assert float(rel.mean()) < 0.05
     corr = torch.corrcoef(torch.stack([ref.flatten(), fus.flatten()]))[0, 1]
     assert float(corr) > 0.999
+
+
+@pytest.mark.gpu
+def test_build_sequences_torch_gpu_matches_numpy(gpu_device):
+    """On-device sequence assembly (serving path) matches the numpy build."""
+    import numpy as np
+
+    from nerrf_amd.data.sequences import build_sequences, build_sequences_torch
+    from nerrf_amd.data.synth import SynthConfig, generate
+
+    arr, _ = generate(SynthConfig(duration_s=8.0, benign_rate_hz=3000.0,
+                                  n_benign_files=300, seed=21))
+    ref = build_sequences(arr, None)
+    feats, lengths, fids = build_sequences_torch(arr, device=gpu_device)
+    assert np.array_equal(lengths.numpy(), ref.lengths)
+    assert np.array_equal(fids.numpy(), ref.file_path_id)
+    torch.testing.assert_close(
+        feats.cpu(), torch.from_numpy(ref.feats), rtol=1e-5, atol=1e-5
+    )
