@@ -81,8 +81,17 @@ class DeltaGraphStore:
             "comm_id": np.concatenate([d.comm_id for d in deltas]),
         }
         if now is not None:
-            keep = cols["ts"] >= now - self.window_s
-            cols = {k: v[keep] for k, v in cols.items()}
+            cutoff = now - self.window_s
+            ts = cols["ts"]
+            if len(ts) and bool(np.all(ts[1:] >= ts[:-1])):
+                # time-ordered stream: the window filter is a suffix slice
+                # (zero-copy views) instead of eight boolean gathers
+                lo = int(np.searchsorted(ts, cutoff, side="left"))
+                if lo:
+                    cols = {k: v[lo:] for k, v in cols.items()}
+            else:
+                keep = ts >= cutoff
+                cols = {k: v[keep] for k, v in cols.items()}
         arr = EventArray(paths=self.paths, comms=self.comms, **cols)
         return arr.sort_by_time()
 
