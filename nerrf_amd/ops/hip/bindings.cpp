@@ -137,7 +137,8 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
   check_gpu_contig(h_prev, "h_prev");
   const long hout_stride = row_stride_checked(h_out, "h_out");
   check_gpu_contig(c_out, "c_out");
-  check_gpu_contig(gates_act, "gates_act");
+  const bool want_gates = gates_act.numel() > 0;
+  if (want_gates) check_gpu_contig(gates_act, "gates_act");
   const long batch = c_prev.size(0);
   const int hdim = c_prev.size(1);
   TORCH_CHECK(hg.size(1) == 4 * hdim && xg.size(1) == 4 * hdim,
@@ -156,8 +157,8 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
   nerrf::launch_lstm_pointwise_fwd(
       hg.data_ptr(), xg.data_ptr(), bc.data_ptr(), c_prev.data_ptr(),
       h_prev.data_ptr(), mask_ptr, h_out.data_ptr(), c_out.data_ptr(),
-      gates_act.data_ptr(), batch, hdim, xg_stride, hout_stride, is_bf16(hg),
-      stream.stream());
+      want_gates ? gates_act.data_ptr() : nullptr, batch, hdim, xg_stride,
+      hout_stride, is_bf16(hg), stream.stream());
 }
 
 // grad_out_t (may be empty): this timestep's dL/dh, folded in-kernel so the

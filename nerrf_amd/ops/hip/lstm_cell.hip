@@ -30,7 +30,7 @@ __global__ void lstm_pointwise_fwd_kernel(
     const float* __restrict__ mask,   // [B] or nullptr
     T* __restrict__ h_out,            // [B(row-stride hout_stride), H]
     T* __restrict__ c_out,            // [B, H]
-    T* __restrict__ gates_act,        // [B, 4H]
+    T* __restrict__ gates_act,        // [B, 4H] or nullptr (inference)
     long batch, int hdim, long xg_stride, long hout_stride) {
   const long total = batch * hdim;
   for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
@@ -57,10 +57,12 @@ __global__ void lstm_pointwise_fwd_kernel(
     }
     c_out[t] = from_f32<T>(cn);
     h_out[b * hout_stride + d] = from_f32<T>(hn);
-    gates_act[g0] = from_f32<T>(i);
-    gates_act[g0 + hdim] = from_f32<T>(f);
-    gates_act[g0 + 2 * hdim] = from_f32<T>(g);
-    gates_act[g0 + 3 * hdim] = from_f32<T>(o);
+    if (gates_act != nullptr) {  // inference skips the backward-only store
+      gates_act[g0] = from_f32<T>(i);
+      gates_act[g0 + hdim] = from_f32<T>(f);
+      gates_act[g0 + 2 * hdim] = from_f32<T>(g);
+      gates_act[g0 + 3 * hdim] = from_f32<T>(o);
+    }
   }
 }
 

@@ -24,7 +24,7 @@ from .native import get_native
 
 class _LSTMSeqFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, xg, h0, c0, w_hh, bias, mask, reverse: bool):
+    def forward(ctx, xg, h0, c0, w_hh, bias, mask, reverse: bool, infer: bool = False):
         """xg: [T, B, 4H] (time-major input projection); mask: [T, B] or None.
 
         Returns h_all: [T, B, H] (time-major hidden states).
@@ -33,9 +33,16 @@ class _LSTMSeqFn(torch.autograd.Function):
         hdim = gdim // 4
         dev, dt = xg.device, xg.dtype
         ext = get_native(xg)
+        # inference (no_grad at the lstm_sequence call site) skips the
+        # backward-only activated-gates store: 32 MB of HBM writes per step
+        # at serving batch sizes.  (grad mode is always off inside
+        # Function.forward, so the flag is computed by the wrapper.)
         h_all = torch.empty(t_len, batch, hdim, device=dev, dtype=dt)
         c_all = torch.empty(t_len, batch, hdim, device=dev, dtype=dt)
-        gates_all = torch.empty(t_len, batch, gdim, device=dev, dtype=dt)
+        gates_all = torch.empty(
+            0 if infer else t_len, batch, gdim, device=dev, dtype=dt
+        )
+        g_none = torch.empty(0, device=dev, dtype=dt)
         h = h0.contiguous()
         c = c0.contiguous()
         steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
@@ -57,11 +64,17 @@ class _LSTMSeqFn(torch.autograd.Function):
                 bias_c = bias.contiguous()
                 # k-slice-contiguous weight tiling (see lstm_step_fused.hip)
                 w_tiled = w_hh.reshape(gdim, hdim // 32, 32).permute(1, 0, 2).contiguous()
+                # the fused kernel always writes gates; reuse one scratch
+                # slab in inference instead of a [T, B, 4H] history
+                g_scratch = (
+                    torch.empty(batch, gdim, device=dev, dtype=dt) if infer else None
+                )
                 for ti in steps:
                     ext.lstm_step_fused(
                         h, w_tiled, xg[ti], bias_c, c,
                         mask[ti] if mask is not None else empty_mask,
-                        h_all[ti], c_all[ti], gates_all[ti], False,
+                        h_all[ti], c_all[ti],
+                        g_scratch if infer else gates_all[ti], False,
                     )
                     h = h_all[ti]
                     c = c_all[ti]
@@ -73,7 +86,7 @@ class _LSTMSeqFn(torch.autograd.Function):
                     ext.lstm_pointwise_fwd(
                         hg, xg[ti], bias, c, h,
                         mask[ti] if mask is not None else empty_mask,
-                        h_all[ti], c_all[ti], gates_all[ti],
+                        h_all[ti], c_all[ti], g_none if infer else gates_all[ti],
                     )
                     h = h_all[ti]
                     c = c_all[ti]
@@ -85,7 +98,8 @@ class _LSTMSeqFn(torch.autograd.Function):
                 )
                 h_all[ti] = h_new
                 c_all[ti] = c_new
-                gates_all[ti] = g_act
+                if not infer:
+                    gates_all[ti] = g_act
                 h, c = h_new, c_new
         ctx.save_for_backward(
             gates_all, h_all, c_all, h0, c0, w_hh,
@@ -163,7 +177,7 @@ class _LSTMSeqFn(torch.autograd.Function):
         grad_xg = grad_gates_all
         grad_h0 = grad_h
         grad_c0 = grad_c
-        return grad_xg, grad_h0, grad_c0, grad_whh, grad_bias, None, None
+        return grad_xg, grad_h0, grad_c0, grad_whh, grad_bias, None, None, None
 
 
 def lstm_sequence(
@@ -176,4 +190,8 @@ def lstm_sequence(
     reverse: bool = False,
 ) -> torch.Tensor:
     m = mask.detach().contiguous().to(torch.float32) if mask is not None else None
-    return _LSTMSeqFn.apply(xg.contiguous(), h0, c0, w_hh, bias, m, reverse)
+    infer = not torch.is_grad_enabled() or not (
+        xg.requires_grad or h0.requires_grad or c0.requires_grad
+        or w_hh.requires_grad or bias.requires_grad
+    )
+    return _LSTMSeqFn.apply(xg.contiguous(), h0, c0, w_hh, bias, m, reverse, infer)

@@ -523,3 +523,22 @@ def test_build_sequences_torch_gpu_matches_numpy(gpu_device):
     torch.testing.assert_close(
         feats.cpu(), torch.from_numpy(ref.feats), rtol=1e-5, atol=1e-5
     )
+
+
+@pytest.mark.gpu
+def test_lstm_sequence_infer_matches_train_forward(gpu_device):
+    """The inference fast path (gates store skipped) produces the same h."""
+    from nerrf_amd.ops.lstm_seq import lstm_sequence
+
+    torch.manual_seed(3)
+    t_len, b, h = 7, 64, 256
+    xg = torch.randn(t_len, b, 4 * h, device=gpu_device, dtype=torch.bfloat16)
+    h0 = torch.zeros(b, h, device=gpu_device, dtype=torch.bfloat16)
+    c0 = torch.zeros_like(h0)
+    w = torch.randn(4 * h, h, device=gpu_device, dtype=torch.bfloat16) * 0.05
+    bias = torch.randn(4 * h, device=gpu_device, dtype=torch.bfloat16) * 0.05
+    wg = w.clone().requires_grad_(True)
+    out_train = lstm_sequence(xg, h0, c0, wg, bias)
+    with torch.no_grad():
+        out_infer = lstm_sequence(xg, h0, c0, w, bias)
+    torch.testing.assert_close(out_infer, out_train.detach(), rtol=0, atol=0)
