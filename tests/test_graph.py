@@ -131,3 +131,55 @@ def test_rename_components_match_serial_union_find():
         uf.union(int(a), int(c))
     expect = np.array([uf.find(i) for i in range(len(ev.paths))])
     assert np.array_equal(parts["path_root"], expect)
+
+
+def test_edge_dedup_radix_fallback_matches_bincount():
+    """Large key spaces take the LSD-radix branch; results must match the
+    bincount branch (and the original np.unique semantics)."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+
+    rng = np.random.default_rng(3)
+    n_procs_target, n_files_target = 3000, 3000  # 2*3000*3000 = 18M > 2**24
+    b = EventArrayBuilder(StringTable(), StringTable())
+    t = 0.0
+    for _ in range(20000):
+        p = int(rng.integers(0, n_procs_target))
+        f = int(rng.integers(0, n_files_target))
+        b.add(ts=t, pid=1000 + p, syscall="write" if rng.random() < 0.7 else "read",
+              path=f"/data/f{f}", nbytes=64)
+        t += 1e-4
+    ev = b.build()
+    parts = build_graph_parts(ev)
+    assert 2 * parts["n_procs"] * parts["n_files"] >= (1 << 24)
+    ed = build_edges_and_flags(parts)
+
+    # independent reference: np.unique on the same compact key
+    sc = ev.syscall
+    from nerrf_amd.data.trace import SYSCALL_IDS
+
+    read_like = np.isin(sc, [SYSCALL_IDS["read"], SYSCALL_IDS["exec"]]).astype(np.int64)
+    n_files = parts["n_files"]
+    valid = (parts["ev_file"] >= 0) & (parts["ev_proc"] >= 0)
+    key = (((parts["ev_proc"] - n_files) * n_files + parts["ev_file"]) * 2 + read_like)[valid]
+    uk, inv = np.unique(key, return_inverse=True)
+    rec = np.exp(-(parts["t1"] - ev.ts[valid]) / 10.0)
+    conf_ref = np.bincount(inv, weights=rec)
+    assert ed["edge_index"].shape[1] == len(uk)
+    w_ref = (1.0 - np.exp(-conf_ref)).astype(np.float32)
+    assert np.allclose(np.sort(ed["edge_weight"]), np.sort(w_ref), atol=1e-6)
+
+
+def test_store_compact_unsorted_stream_filter():
+    """Out-of-order events still filter correctly (boolean-mask branch)."""
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    st = DeltaGraphStore(window_s=10.0, delta_s=2.0)
+    # deliberately jittered timestamps
+    for i, ts in enumerate([100.0, 99.5, 101.0, 104.9, 103.0, 112.0, 111.5]):
+        st.append(ts=ts, pid=5, syscall="write", path=f"/f{i}", nbytes=1)
+    ev = st.compact(now=112.0)
+    kept = sorted(float(t) for t in ev.ts)
+    assert kept == [103.0, 104.9, 111.5, 112.0]  # window [102, 112]
