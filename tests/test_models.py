@@ -121,3 +121,29 @@ def test_collate_windows_blockdiag_equivalence():
     # per-window re-eval at the union's padding
     tmax = union.seq_feats.shape[1]
     assert su.shape[0] == s0.shape[0] + s1.shape[0]
+
+
+def test_bilstm_matches_torch_nn_lstm():
+    """Weight-copied parity against torch.nn.LSTM (bidirectional, 2 layers)
+    on full-length sequences — guards the fused recurrence end-to-end."""
+    torch.manual_seed(0)
+    b, t, e, h = 5, 12, 16, 32
+    m = BiLSTMDetector(LSTMConfig(in_dim=e, hidden=h, layers=2, seq_len=t))
+    ref = torch.nn.LSTM(e, h, num_layers=2, bidirectional=True, batch_first=False)
+    with torch.no_grad():
+        for li, layer in enumerate(m.dirs):
+            for di, d in enumerate(layer):
+                sfx = f"_l{li}" + ("_reverse" if di else "")
+                getattr(ref, f"weight_ih{sfx}").copy_(d.w_ih)
+                getattr(ref, f"weight_hh{sfx}").copy_(d.w_hh)
+                getattr(ref, f"bias_ih{sfx}").copy_(d.b)
+                getattr(ref, f"bias_hh{sfx}").zero_()
+    x = torch.randn(b, t, e)
+    lengths = torch.full((b,), t, dtype=torch.int64)
+    out_ref, _ = ref(x.transpose(0, 1))  # [T, B, 2H]
+    # reproduce the head input from the reference states
+    h_fwd_ref = out_ref[-1, :, :h]
+    h_bwd_ref = out_ref[0, :, h:]
+    logits_ref = m.head(torch.cat([h_fwd_ref, h_bwd_ref], dim=-1)).squeeze(-1)
+    logits = m(x, lengths)
+    torch.testing.assert_close(logits, logits_ref, rtol=1e-4, atol=1e-5)
