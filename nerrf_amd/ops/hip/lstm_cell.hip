@@ -110,6 +110,159 @@ __global__ void lstm_pointwise_bwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// vectorised variants: V adjacent hidden elements per thread so every global
+// access is a 8/16-byte load (scalar bf16 is a measured 2-2.5x loss on CDNA4).
+// Used whenever hdim % V == 0 (H=256 in production).
+// ---------------------------------------------------------------------------
+
+template <typename T, int V>
+struct alignas(sizeof(T) * V) VecT {
+  T v[V];
+};
+
+template <typename T, int V>
+__global__ void lstm_pointwise_fwd_vec_kernel(
+    const T* __restrict__ hg, const T* __restrict__ xg,
+    const T* __restrict__ bias, const T* __restrict__ c_prev,
+    const T* __restrict__ h_prev, const float* __restrict__ mask,
+    T* __restrict__ h_out, T* __restrict__ c_out, T* __restrict__ gates_act,
+    long batch, int hdim, long xg_stride, long hout_stride) {
+  using VT = VecT<T, V>;
+  const int hv = hdim / V;
+  const long total = batch * hv;
+  for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long b = t / hv;
+    const int dv = (int)(t % hv) * V;
+    const long g0 = b * 4 * hdim + dv;
+    const long x0 = b * xg_stride + dv;
+    const long c0 = b * (long)hdim + dv;
+    const VT hg_i = *reinterpret_cast<const VT*>(hg + g0);
+    const VT hg_f = *reinterpret_cast<const VT*>(hg + g0 + hdim);
+    const VT hg_g = *reinterpret_cast<const VT*>(hg + g0 + 2 * hdim);
+    const VT hg_o = *reinterpret_cast<const VT*>(hg + g0 + 3 * hdim);
+    const VT xg_i = *reinterpret_cast<const VT*>(xg + x0);
+    const VT xg_f = *reinterpret_cast<const VT*>(xg + x0 + hdim);
+    const VT xg_g = *reinterpret_cast<const VT*>(xg + x0 + 2 * hdim);
+    const VT xg_o = *reinterpret_cast<const VT*>(xg + x0 + 3 * hdim);
+    const VT b_i = *reinterpret_cast<const VT*>(bias + dv);
+    const VT b_f = *reinterpret_cast<const VT*>(bias + dv + hdim);
+    const VT b_g = *reinterpret_cast<const VT*>(bias + dv + 2 * hdim);
+    const VT b_o = *reinterpret_cast<const VT*>(bias + dv + 3 * hdim);
+    const VT cp_v = *reinterpret_cast<const VT*>(c_prev + c0);
+    VT hp_v;
+    const float m = (mask != nullptr) ? mask[b] : 1.0f;
+    if (mask != nullptr) hp_v = *reinterpret_cast<const VT*>(h_prev + c0);
+    VT ho_v, co_v, ga_i, ga_f, ga_g, ga_o;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const float i = sigmoidf_(to_f32(hg_i.v[j]) + to_f32(xg_i.v[j]) + to_f32(b_i.v[j]));
+      const float f = sigmoidf_(to_f32(hg_f.v[j]) + to_f32(xg_f.v[j]) + to_f32(b_f.v[j]));
+      const float g = tanhf(to_f32(hg_g.v[j]) + to_f32(xg_g.v[j]) + to_f32(b_g.v[j]));
+      const float o = sigmoidf_(to_f32(hg_o.v[j]) + to_f32(xg_o.v[j]) + to_f32(b_o.v[j]));
+      const float cp = to_f32(cp_v.v[j]);
+      float cn = f * cp + i * g;
+      float hn = o * tanhf(cn);
+      if (mask != nullptr) {
+        cn = m * cn + (1.0f - m) * cp;
+        hn = m * hn + (1.0f - m) * to_f32(hp_v.v[j]);
+      }
+      co_v.v[j] = from_f32<T>(cn);
+      ho_v.v[j] = from_f32<T>(hn);
+      ga_i.v[j] = from_f32<T>(i);
+      ga_f.v[j] = from_f32<T>(f);
+      ga_g.v[j] = from_f32<T>(g);
+      ga_o.v[j] = from_f32<T>(o);
+    }
+    *reinterpret_cast<VT*>(c_out + c0) = co_v;
+    *reinterpret_cast<VT*>(h_out + b * hout_stride + dv) = ho_v;
+    if (gates_act != nullptr) {
+      *reinterpret_cast<VT*>(gates_act + g0) = ga_i;
+      *reinterpret_cast<VT*>(gates_act + g0 + hdim) = ga_f;
+      *reinterpret_cast<VT*>(gates_act + g0 + 2 * hdim) = ga_g;
+      *reinterpret_cast<VT*>(gates_act + g0 + 3 * hdim) = ga_o;
+    }
+  }
+}
+
+template <typename T, int V>
+__global__ void lstm_pointwise_bwd_vec_kernel(
+    const T* __restrict__ grad_h, const T* __restrict__ grad_out_t,
+    const T* __restrict__ grad_c, const T* __restrict__ gates_act,
+    const T* __restrict__ c_prev, const float* __restrict__ mask,
+    T* __restrict__ grad_gates, T* __restrict__ grad_c_prev,
+    T* __restrict__ grad_h_pass, long batch, int hdim, long gout_stride) {
+  using VT = VecT<T, V>;
+  const int hv = hdim / V;
+  const long total = batch * hv;
+  for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long b = t / hv;
+    const int dv = (int)(t % hv) * V;
+    const long g0 = b * 4 * hdim + dv;
+    const long c0 = b * (long)hdim + dv;
+    const VT ga_i = *reinterpret_cast<const VT*>(gates_act + g0);
+    const VT ga_f = *reinterpret_cast<const VT*>(gates_act + g0 + hdim);
+    const VT ga_g = *reinterpret_cast<const VT*>(gates_act + g0 + 2 * hdim);
+    const VT ga_o = *reinterpret_cast<const VT*>(gates_act + g0 + 3 * hdim);
+    const VT cp_v = *reinterpret_cast<const VT*>(c_prev + c0);
+    const VT gh_v = *reinterpret_cast<const VT*>(grad_h + c0);
+    const VT gc_v = *reinterpret_cast<const VT*>(grad_c + c0);
+    VT go_v;
+    if (grad_out_t != nullptr)
+      go_v = *reinterpret_cast<const VT*>(grad_out_t + b * gout_stride + dv);
+    const float m = (mask != nullptr) ? mask[b] : 1.0f;
+    VT gg_i, gg_f, gg_g, gg_o, gcp_v, ghp_v;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const float i = to_f32(ga_i.v[j]);
+      const float f = to_f32(ga_f.v[j]);
+      const float g = to_f32(ga_g.v[j]);
+      const float o = to_f32(ga_o.v[j]);
+      const float cp = to_f32(cp_v.v[j]);
+      const float tcn = tanhf(f * cp + i * g);
+      float gh_in = to_f32(gh_v.v[j]);
+      if (grad_out_t != nullptr) gh_in += to_f32(go_v.v[j]);
+      const float gc_in = to_f32(gc_v.v[j]);
+      const float gh = gh_in * m;
+      const float gc = gc_in * m;
+      const float d_o = gh * tcn;
+      const float d_c = gc + gh * o * (1.0f - tcn * tcn);
+      const float d_i = d_c * g;
+      const float d_f = d_c * cp;
+      const float d_g = d_c * i;
+      gcp_v.v[j] = from_f32<T>(d_c * f + gc_in * (1.0f - m));
+      ghp_v.v[j] = from_f32<T>(gh_in * (1.0f - m));
+      gg_i.v[j] = from_f32<T>(d_i * i * (1.0f - i));
+      gg_f.v[j] = from_f32<T>(d_f * f * (1.0f - f));
+      gg_g.v[j] = from_f32<T>(d_g * (1.0f - g * g));
+      gg_o.v[j] = from_f32<T>(d_o * o * (1.0f - o));
+    }
+    *reinterpret_cast<VT*>(grad_c_prev + c0) = gcp_v;
+    *reinterpret_cast<VT*>(grad_h_pass + c0) = ghp_v;
+    *reinterpret_cast<VT*>(grad_gates + g0) = gg_i;
+    *reinterpret_cast<VT*>(grad_gates + g0 + hdim) = gg_f;
+    *reinterpret_cast<VT*>(grad_gates + g0 + 2 * hdim) = gg_g;
+    *reinterpret_cast<VT*>(grad_gates + g0 + 3 * hdim) = gg_o;
+  }
+}
+
+template __global__ void lstm_pointwise_fwd_vec_kernel<__hip_bfloat16, 8>(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long, long);
+template __global__ void lstm_pointwise_fwd_vec_kernel<float, 4>(
+    const float*, const float*, const float*, const float*, const float*,
+    const float*, float*, float*, float*, long, int, long, long);
+template __global__ void lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8>(
+    const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*,
+    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long);
+template __global__ void lstm_pointwise_bwd_vec_kernel<float, 4>(
+    const float*, const float*, const float*, const float*, const float*,
+    const float*, float*, float*, float*, long, int, long);
+
 template __global__ void lstm_pointwise_fwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
     float*, float*, float*, long, int, long, long);
@@ -142,8 +295,21 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
                                long xg_stride, long hout_stride, bool bf16,
                                hipStream_t s) {
   const int block = 256;
-  const int grid = grid_elems(batch * hdim, block);
+  // vectorised path (16-B global accesses) whenever the layout allows
+  const int v = bf16 ? 8 : 4;
+  const bool vec = hdim % v == 0 && xg_stride % v == 0 && hout_stride % v == 0;
   if (bf16) {
+    if (vec) {
+      const int grid = grid_elems(batch * (hdim / v), block);
+      lstm_pointwise_fwd_vec_kernel<__hip_bfloat16, 8><<<grid, block, 0, s>>>(
+          (const __hip_bfloat16*)hg, (const __hip_bfloat16*)xg,
+          (const __hip_bfloat16*)bias, (const __hip_bfloat16*)c_prev,
+          (const __hip_bfloat16*)h_prev, mask, (__hip_bfloat16*)h_out,
+          (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim,
+          xg_stride, hout_stride);
+      return;
+    }
+    const int grid = grid_elems(batch * hdim, block);
     lstm_pointwise_fwd_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)hg, (const __hip_bfloat16*)xg,
         (const __hip_bfloat16*)bias, (const __hip_bfloat16*)c_prev,
@@ -151,6 +317,16 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
         (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim,
         xg_stride, hout_stride);
   } else {
+    if (vec) {
+      const int grid = grid_elems(batch * (hdim / v), block);
+      lstm_pointwise_fwd_vec_kernel<float, 4><<<grid, block, 0, s>>>(
+          (const float*)hg, (const float*)xg, (const float*)bias,
+          (const float*)c_prev, (const float*)h_prev, mask, (float*)h_out,
+          (float*)c_out, (float*)gates_act, batch, hdim, xg_stride,
+          hout_stride);
+      return;
+    }
+    const int grid = grid_elems(batch * hdim, block);
     lstm_pointwise_fwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)hg, (const float*)xg, (const float*)bias,
         (const float*)c_prev, (const float*)h_prev, mask, (float*)h_out,
@@ -165,8 +341,20 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
                                void* grad_h_pass, long batch, int hdim,
                                long gout_stride, bool bf16, hipStream_t s) {
   const int block = 256;
-  const int grid = grid_elems(batch * hdim, block);
+  const int v = bf16 ? 8 : 4;
+  const bool vec = hdim % v == 0 && gout_stride % v == 0;
   if (bf16) {
+    if (vec) {
+      const int grid = grid_elems(batch * (hdim / v), block);
+      lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8><<<grid, block, 0, s>>>(
+          (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_out_t,
+          (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
+          (const __hip_bfloat16*)c_prev, mask, (__hip_bfloat16*)grad_gates,
+          (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch,
+          hdim, gout_stride);
+      return;
+    }
+    const int grid = grid_elems(batch * hdim, block);
     lstm_pointwise_bwd_kernel<__hip_bfloat16><<<grid, block, 0, s>>>(
         (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_out_t,
         (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
@@ -174,6 +362,16 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
         (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch, hdim,
         gout_stride);
   } else {
+    if (vec) {
+      const int grid = grid_elems(batch * (hdim / v), block);
+      lstm_pointwise_bwd_vec_kernel<float, 4><<<grid, block, 0, s>>>(
+          (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
+          (const float*)gates_act, (const float*)c_prev, mask,
+          (float*)grad_gates, (float*)grad_c_prev, (float*)grad_h_pass, batch,
+          hdim, gout_stride);
+      return;
+    }
+    const int grid = grid_elems(batch * hdim, block);
     lstm_pointwise_bwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
         (const float*)gates_act, (const float*)c_prev, mask, (float*)grad_gates,
