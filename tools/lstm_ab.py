@@ -5,7 +5,7 @@ from nerrf_amd.ops.native import load_extension
 
 ext = load_extension(required=True)
 dev = "cuda:0"
-B, H = 16000, 256
+B, H = (int(sys.argv[1]) if len(sys.argv) > 1 else 16000), 256
 torch.manual_seed(0)
 h = (torch.randn(B, H, device=dev) * 0.3).to(torch.bfloat16)
 c = (torch.randn(B, H, device=dev) * 0.3).to(torch.bfloat16)
