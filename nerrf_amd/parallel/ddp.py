@@ -43,7 +43,14 @@ def init_distributed(backend: Optional[str] = None) -> tuple[int, int, int]:
 
 
 class GradAllReducer:
-    """Bucketed async all-reduce of gradients, overlapped with backward."""
+    """Bucketed async all-reduce of gradients, overlapped with backward.
+
+    Contract: exactly ONE backward() per finalize().  A bucket is reduced
+    the moment its last grad lands, so accumulating a second backward before
+    finalize() would reduce the first pass's grads early; use a no-hook
+    accumulation phase (or call finalize() per micro-batch) if micro-batching
+    is ever added.
+    """
 
     def __init__(
         self,
