@@ -63,7 +63,10 @@ def gpu_window_graph(
     )
 
     def dev(a, dt=None):
-        t = torch.from_numpy(np.ascontiguousarray(a))
+        if isinstance(a, torch.Tensor):  # already staged on-device (merge_window)
+            t = a
+        else:
+            t = torch.from_numpy(np.ascontiguousarray(a))
         if dt is not None:
             t = t.to(dt)
         return t.to(device, non_blocking=True)

@@ -65,11 +65,19 @@ class DeltaGraphStore:
 
     def compact(self, now: Optional[float] = None) -> EventArray:
         """Merge the window's deltas into one time-sorted columnar batch."""
+        return self.compact_with_deltas(now)[0]
+
+    def compact_with_deltas(
+        self, now: Optional[float] = None
+    ) -> tuple[EventArray, list]:
+        """compact() plus the exact delta list it merged (one lock scope, so
+        a concurrent ingest thread cannot skew the pair — the incremental
+        scorer needs summaries over precisely the compacted deltas)."""
         with self._lock:
             self._seal_locked()
             deltas = list(self._deltas)
         if not deltas:
-            return EventArrayBuilder(self.paths, self.comms).build()
+            return EventArrayBuilder(self.paths, self.comms).build(), []
         cols = {
             "ts": np.concatenate([d.ts for d in deltas]),
             "pid": np.concatenate([d.pid for d in deltas]),
@@ -93,7 +101,7 @@ class DeltaGraphStore:
                 keep = ts >= cutoff
                 cols = {k: v[keep] for k, v in cols.items()}
         arr = EventArray(paths=self.paths, comms=self.comms, **cols)
-        return arr.sort_by_time()
+        return arr.sort_by_time(), deltas
 
     @property
     def window_event_count(self) -> int:

@@ -90,6 +90,9 @@ class StreamingEngine:
             # one fused MFMA kernel per GNN layer on the scoring path
             self.model.gnn.use_fused_inference = True
         self.store = DeltaGraphStore(window_s=window_s)
+        from ..graph.incremental import IncrementalWindowState
+
+        self._inc_state = IncrementalWindowState()
         self.alarm_threshold = alarm_threshold
         self.planner_params = planner_params or PlannerParams()
         self.shard_id = shard_id
@@ -121,7 +124,12 @@ class StreamingEngine:
     # ------------------------------------------------------------------- score
     @torch.no_grad()
     def score_window(self, now: Optional[float] = None) -> Detection:
-        events = self.store.compact(now)
+        # incremental mode (the production `now=None` tick: window == whole
+        # delta set) merges cached per-delta summaries instead of re-scanning
+        # every event; an explicit `now` trims at event granularity, which
+        # invalidates the summaries, so that path takes the full rebuild
+        events, window_deltas = self.store.compact_with_deltas(now)
+        use_incremental = now is None and len(window_deltas) > 0
         t_detect = time.time()
         if len(events) == 0:
             return Detection(False, t_detect, {}, {}, {}, window_events=0)
@@ -133,8 +141,17 @@ class StreamingEngine:
         from ..graph.constructor import build_edges_and_flags, build_graph, build_graph_parts
         from ..graph.sampling import sample_fanout, to_csr
 
-        parts = build_graph_parts(events)
-        ed = build_edges_and_flags(parts)
+        if use_incremental:
+            from ..graph.incremental import merge_window
+
+            sums = self._inc_state.summaries(window_deltas)
+            parts, ed = merge_window(
+                events, sums,
+                device=self.device if self.device.type == "cuda" else None,
+            )
+        else:
+            parts = build_graph_parts(events)
+            ed = build_edges_and_flags(parts)
         node_kind = _np.concatenate(
             [_np.ones(parts["n_files"], dtype=_np.int8), _np.zeros(parts["n_procs"], dtype=_np.int8)]
         )

@@ -183,3 +183,44 @@ def test_store_compact_unsorted_stream_filter():
     ev = st.compact(now=112.0)
     kept = sorted(float(t) for t in ev.ts)
     assert kept == [103.0, 104.9, 111.5, 112.0]  # window [102, 112]
+
+
+def test_incremental_merge_matches_full_rebuild():
+    """Per-delta summary merge reproduces build_graph_parts +
+    build_edges_and_flags on the same window (ids/order exact, weights fp)."""
+    import numpy as np
+
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    arr, _ = generate(SynthConfig(duration_s=20.0, benign_rate_hz=3000.0,
+                                  n_benign_files=400, seed=13))
+    eng = StreamingEngine(device="cpu")
+    eng.ingest_events(arr)
+    ev = eng.store.compact(None)
+    ref_parts = build_graph_parts(ev)
+    ref_ed = build_edges_and_flags(ref_parts)
+
+    with eng.store._lock:
+        deltas = list(eng.store._deltas)
+    state = IncrementalWindowState()
+    sums = state.summaries(deltas)
+    parts, ed = merge_window(ev, sums)
+
+    assert parts["n_files"] == ref_parts["n_files"]
+    assert parts["n_procs"] == ref_parts["n_procs"]
+    assert np.array_equal(parts["ev_file"], ref_parts["ev_file"])
+    assert np.array_equal(parts["ev_proc"], ref_parts["ev_proc"])
+    assert np.array_equal(parts["touched_roots"], ref_parts["touched_roots"])
+    assert np.array_equal(parts["path_root"], ref_parts["path_root"])
+    assert np.array_equal(ed["edge_index"], ref_ed["edge_index"])
+    assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-5)
+    assert np.allclose(ed["edge_ts"], ref_ed["edge_ts"], atol=1e-6)
+    for k in ("in_deg", "out_deg", "peer", "suspicious", "note", "recon", "double_ext"):
+        assert np.array_equal(ed[k], ref_ed[k]), k
+
+    # summary cache: second call reuses every entry (no recompute)
+    sums2 = state.summaries(deltas)
+    assert all(a is b for a, b in zip(sums, sums2))
