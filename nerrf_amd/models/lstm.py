@@ -79,14 +79,22 @@ class BiLSTMDetector(nn.Module):
 
     def forward(self, feats: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
         """feats: [B, T, E]; lengths: [B] -> per-sequence logit [B]."""
+        from ..ops.lstm_seq import lstm_bilayer
+
         b, t, _ = feats.shape
         ar = torch.arange(t, device=feats.device)
         mask = (ar.unsqueeze(1) < lengths.unsqueeze(0)).to(feats.dtype)  # [T, B]
         h = feats.transpose(0, 1).contiguous()  # time-major [T, B, E]
         for layer in self.dirs:
-            fwd = layer[0](h, mask)
-            bwd = layer[1](h, mask)
-            h = torch.cat([fwd, bwd], dim=-1)  # [T, B, 2H]
+            f, r = layer[0], layer[1]
+            # both directions write one [T, B, 2H] buffer through the
+            # kernels' row-stride args — no torch.cat on the hot path
+            flat = h.reshape(t * b, -1)
+            xg_f = torch.matmul(flat, f.w_ih.t()).reshape(t, b, 4 * f.hidden)
+            xg_b = torch.matmul(flat, r.w_ih.t()).reshape(t, b, 4 * r.hidden)
+            h0 = h.new_zeros(b, f.hidden)
+            c0 = h.new_zeros(b, f.hidden)
+            h = lstm_bilayer(xg_f, xg_b, h0, c0, f.w_hh, f.b, r.w_hh, r.b, mask)
         # forward state at t=len-1, backward state at t=0
         idx = (lengths.clamp(min=1) - 1).view(1, b, 1).expand(1, b, self.cfg.hidden)
         h_fwd = h[:, :, : self.cfg.hidden].gather(0, idx).squeeze(0)  # [B, H]

@@ -32,3 +32,13 @@ print(f"wall span: {span:.1f} ms   kernel-busy: {busy/1e6:.1f} ms   gap: {span -
 print(f"{'kernel':<72}{'count':>8}{'total ms':>12}{'avg us':>10}")
 for name, (cnt, ms) in sorted(agg.items(), key=lambda kv: -kv[1][1])[:30]:
     print(f"{name[:72]:<72}{cnt:>8}{ms:>12.2f}{ms/cnt*1000:>10.1f}")
+
+# RAWDUMP: top raw kernel names (un-shortened) for identification
+raw = {}
+for st, en, name in rows:
+    a = raw.setdefault(name, [0, 0.0])
+    a[0] += 1
+    a[1] += (en - st) / 1e6
+print("\n-- top raw names --")
+for name, (c, tot) in sorted(raw.items(), key=lambda kv: -kv[1][1])[:18]:
+    print(f"{tot:9.2f} ms  x{c:<6d} {name[:150]}")
