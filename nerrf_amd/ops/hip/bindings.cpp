@@ -22,8 +22,8 @@ void launch_gather_mean_bwd_csr(const void*, const long*, const long*,
                                 hipStream_t);
 void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
-                               void*, void*, long, int, long, long, bool,
-                               hipStream_t);
+                               void*, void*, long, int, long, long, long,
+                               bool, hipStream_t);
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
                                void*, void*, long, int, long, bool,
@@ -134,7 +134,7 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
   check_gpu_contig(hg, "hg");
   const long xg_stride = row_stride_checked(xg, "xg");
   check_gpu_contig(c_prev, "c_prev");
-  check_gpu_contig(h_prev, "h_prev");
+  const long hprev_stride = row_stride_checked(h_prev, "h_prev");
   const long hout_stride = row_stride_checked(h_out, "h_out");
   check_gpu_contig(c_out, "c_out");
   const bool want_gates = gates_act.numel() > 0;
@@ -158,7 +158,7 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
       hg.data_ptr(), xg.data_ptr(), bc.data_ptr(), c_prev.data_ptr(),
       h_prev.data_ptr(), mask_ptr, h_out.data_ptr(), c_out.data_ptr(),
       want_gates ? gates_act.data_ptr() : nullptr, batch, hdim, xg_stride,
-      hout_stride, is_bf16(hg), stream.stream());
+      hout_stride, hprev_stride, is_bf16(hg), stream.stream());
 }
 
 // grad_out_t (may be empty): this timestep's dL/dh, folded in-kernel so the

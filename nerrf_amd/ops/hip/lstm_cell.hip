@@ -26,12 +26,13 @@ __global__ void lstm_pointwise_fwd_kernel(
     const T* __restrict__ xg,         // [B(row-stride xg_stride), 4H]
     const T* __restrict__ bias,       // [4H]
     const T* __restrict__ c_prev,     // [B, H]
-    const T* __restrict__ h_prev,     // [B, H]
+    const T* __restrict__ h_prev,     // [B(row-stride hprev_stride), H]
     const float* __restrict__ mask,   // [B] or nullptr
     T* __restrict__ h_out,            // [B(row-stride hout_stride), H]
     T* __restrict__ c_out,            // [B, H]
     T* __restrict__ gates_act,        // [B, 4H] or nullptr (inference)
-    long batch, int hdim, long xg_stride, long hout_stride) {
+    long batch, int hdim, long xg_stride, long hout_stride,
+    long hprev_stride) {
   const long total = batch * hdim;
   for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
@@ -53,7 +54,7 @@ __global__ void lstm_pointwise_fwd_kernel(
     if (mask != nullptr) {
       const float m = mask[b];
       cn = m * cn + (1.0f - m) * cp;
-      hn = m * hn + (1.0f - m) * to_f32(h_prev[t]);
+      hn = m * hn + (1.0f - m) * to_f32(h_prev[b * hprev_stride + d]);
     }
     c_out[t] = from_f32<T>(cn);
     h_out[b * hout_stride + d] = from_f32<T>(hn);
@@ -127,7 +128,8 @@ __global__ void lstm_pointwise_fwd_vec_kernel(
     const T* __restrict__ bias, const T* __restrict__ c_prev,
     const T* __restrict__ h_prev, const float* __restrict__ mask,
     T* __restrict__ h_out, T* __restrict__ c_out, T* __restrict__ gates_act,
-    long batch, int hdim, long xg_stride, long hout_stride) {
+    long batch, int hdim, long xg_stride, long hout_stride,
+    long hprev_stride) {
   using VT = VecT<T, V>;
   const int hv = hdim / V;
   const long total = batch * hv;
@@ -153,7 +155,8 @@ __global__ void lstm_pointwise_fwd_vec_kernel(
     const VT cp_v = *reinterpret_cast<const VT*>(c_prev + c0);
     VT hp_v;
     const float m = (mask != nullptr) ? mask[b] : 1.0f;
-    if (mask != nullptr) hp_v = *reinterpret_cast<const VT*>(h_prev + c0);
+    if (mask != nullptr)
+      hp_v = *reinterpret_cast<const VT*>(h_prev + b * hprev_stride + dv);
     VT ho_v, co_v, ga_i, ga_f, ga_g, ga_o;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
@@ -251,10 +254,11 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
 template __global__ void lstm_pointwise_fwd_vec_kernel<__hip_bfloat16, 8>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
-    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long, long);
+    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long, long,
+    long);
 template __global__ void lstm_pointwise_fwd_vec_kernel<float, 4>(
     const float*, const float*, const float*, const float*, const float*,
-    const float*, float*, float*, float*, long, int, long, long);
+    const float*, float*, float*, float*, long, int, long, long, long);
 template __global__ void lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
@@ -265,11 +269,11 @@ template __global__ void lstm_pointwise_bwd_vec_kernel<float, 4>(
 
 template __global__ void lstm_pointwise_fwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
-    float*, float*, float*, long, int, long, long);
+    float*, float*, float*, long, int, long, long, long);
 template __global__ void lstm_pointwise_fwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
     const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*,
-    long, int, long, long);
+    long, int, long, long, long);
 template __global__ void lstm_pointwise_bwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
     float*, float*, float*, long, int, long);
@@ -292,12 +296,13 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
                                const void* c_prev, const void* h_prev,
                                const float* mask, void* h_out, void* c_out,
                                void* gates_act, long batch, int hdim,
-                               long xg_stride, long hout_stride, bool bf16,
-                               hipStream_t s) {
+                               long xg_stride, long hout_stride,
+                               long hprev_stride, bool bf16, hipStream_t s) {
   const int block = 256;
   // vectorised path (16-B global accesses) whenever the layout allows
   const int v = bf16 ? 8 : 4;
-  const bool vec = hdim % v == 0 && xg_stride % v == 0 && hout_stride % v == 0;
+  const bool vec = hdim % v == 0 && xg_stride % v == 0 &&
+                   hout_stride % v == 0 && hprev_stride % v == 0;
   if (bf16) {
     if (vec) {
       const int grid = grid_elems(batch * (hdim / v), block);
@@ -306,7 +311,7 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
           (const __hip_bfloat16*)bias, (const __hip_bfloat16*)c_prev,
           (const __hip_bfloat16*)h_prev, mask, (__hip_bfloat16*)h_out,
           (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim,
-          xg_stride, hout_stride);
+          xg_stride, hout_stride, hprev_stride);
       return;
     }
     const int grid = grid_elems(batch * hdim, block);
@@ -315,7 +320,7 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
         (const __hip_bfloat16*)bias, (const __hip_bfloat16*)c_prev,
         (const __hip_bfloat16*)h_prev, mask, (__hip_bfloat16*)h_out,
         (__hip_bfloat16*)c_out, (__hip_bfloat16*)gates_act, batch, hdim,
-        xg_stride, hout_stride);
+        xg_stride, hout_stride, hprev_stride);
   } else {
     if (vec) {
       const int grid = grid_elems(batch * (hdim / v), block);
@@ -323,14 +328,15 @@ void launch_lstm_pointwise_fwd(const void* hg, const void* xg, const void* bias,
           (const float*)hg, (const float*)xg, (const float*)bias,
           (const float*)c_prev, (const float*)h_prev, mask, (float*)h_out,
           (float*)c_out, (float*)gates_act, batch, hdim, xg_stride,
-          hout_stride);
+          hout_stride, hprev_stride);
       return;
     }
     const int grid = grid_elems(batch * hdim, block);
     lstm_pointwise_fwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)hg, (const float*)xg, (const float*)bias,
         (const float*)c_prev, (const float*)h_prev, mask, (float*)h_out,
-        (float*)c_out, (float*)gates_act, batch, hdim, xg_stride, hout_stride);
+        (float*)c_out, (float*)gates_act, batch, hdim, xg_stride, hout_stride,
+        hprev_stride);
   }
 }
 
