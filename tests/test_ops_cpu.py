@@ -240,3 +240,48 @@ def test_reverse_index_consistent_with_bwd_ref():
     for e in range(len(rd)):
         out[int(rd[e])] += float(rw[e]) * g[int(rs[e])]
     assert torch.allclose(out, expected, atol=1e-5)
+
+
+def test_lstm_bilayer_matches_two_directions():
+    """lstm_bilayer == cat(lstm_sequence fwd, lstm_sequence bwd) — outputs
+    and every gradient (xg, weights, biases)."""
+    from nerrf_amd.ops.lstm_seq import lstm_bilayer, lstm_sequence
+
+    torch.manual_seed(11)
+    t, b, h = 9, 7, 16
+    mk = (torch.rand(t, b) > 0.25).float()
+
+    def inputs():
+        torch.manual_seed(12)
+        xf = (torch.randn(t, b, 4 * h) * 0.3).requires_grad_(True)
+        xb = (torch.randn(t, b, 4 * h) * 0.3).requires_grad_(True)
+        wf = (torch.randn(4 * h, h) * 0.2).requires_grad_(True)
+        wb = (torch.randn(4 * h, h) * 0.2).requires_grad_(True)
+        bf = torch.randn(4 * h).requires_grad_(True)
+        bb = torch.randn(4 * h).requires_grad_(True)
+        return xf, xb, wf, wb, bf, bb
+
+    h0 = torch.zeros(b, h)
+    c0 = torch.zeros(b, h)
+
+    xf1, xb1, wf1, wb1, bf1, bb1 = inputs()
+    out_ref = torch.cat(
+        [
+            lstm_sequence(xf1, h0, c0, wf1, bf1, mk, reverse=False),
+            lstm_sequence(xb1, h0, c0, wb1, bb1, mk, reverse=True),
+        ],
+        dim=-1,
+    )
+    gseed = torch.Generator().manual_seed(3)
+    g = torch.randn(out_ref.shape, generator=gseed)
+    out_ref.backward(g)
+
+    xf2, xb2, wf2, wb2, bf2, bb2 = inputs()
+    out = lstm_bilayer(xf2, xb2, h0, c0, wf2, bf2, wb2, bb2, mk)
+    torch.testing.assert_close(out, out_ref)
+    out.backward(g)
+    for a, bT, name in [
+        (xf1, xf2, "xg_f"), (xb1, xb2, "xg_b"), (wf1, wf2, "w_f"),
+        (wb1, wb2, "w_b"), (bf1, bf2, "b_f"), (bb1, bb2, "b_b"),
+    ]:
+        torch.testing.assert_close(a.grad, bT.grad, msg=name)
