@@ -224,3 +224,68 @@ def test_incremental_merge_matches_full_rebuild():
     # summary cache: second call reuses every entry (no recompute)
     sums2 = state.summaries(deltas)
     assert all(a is b for a, b in zip(sums, sums2))
+
+
+def test_incremental_rename_chain_across_deltas():
+    """A rename chain a->b (delta 1), b->c (delta 2) must union {a,b,c} in
+    the merged window exactly like the full rebuild."""
+    import numpy as np
+
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    st = DeltaGraphStore(window_s=100.0, delta_s=2.0)
+    st.append(ts=0.1, pid=7, syscall="write", path="/d/a", nbytes=8)
+    st.append(ts=0.2, pid=7, syscall="rename", path="/d/a", new_path="/d/b")
+    # force a new delta, then continue the chain
+    st.append(ts=3.0, pid=7, syscall="write", path="/d/b", nbytes=8)
+    st.append(ts=3.1, pid=7, syscall="rename", path="/d/b", new_path="/d/c")
+    st.append(ts=6.0, pid=9, syscall="read", path="/d/c", nbytes=4)
+    ev, deltas = st.compact_with_deltas(None)
+    assert len(deltas) >= 2
+
+    ref_parts = build_graph_parts(ev)
+    ref_ed = build_edges_and_flags(ref_parts)
+    parts, ed = merge_window(ev, IncrementalWindowState().summaries(deltas))
+    # one merged file node for {a, b, c}
+    assert parts["n_files"] == ref_parts["n_files"] == 1
+    assert np.array_equal(parts["path_root"], ref_parts["path_root"])
+    assert np.array_equal(ed["edge_index"], ref_ed["edge_index"])
+    assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-6)
+
+
+def test_incremental_random_streams_match_full():
+    """Property check: random multi-delta streams, incremental == full."""
+    import numpy as np
+
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    rng = np.random.default_rng(99)
+    names = [f"/x/f{i}" for i in range(40)]
+    calls = ["write", "read", "openat", "unlink", "chmod", "exec"]
+    for trial in range(3):
+        st = DeltaGraphStore(window_s=1000.0, delta_s=1.0)
+        t = 0.0
+        for _ in range(800):
+            t += float(rng.random() * 0.02)
+            if rng.random() < 0.1:
+                i, j = rng.integers(0, len(names), 2)
+                st.append(ts=t, pid=int(10 + rng.integers(0, 5)), syscall="rename",
+                          path=names[i], new_path=names[j])
+            else:
+                st.append(ts=t, pid=int(10 + rng.integers(0, 5)),
+                          syscall=str(rng.choice(calls)), path=str(rng.choice(names)),
+                          nbytes=int(rng.integers(0, 4096)))
+        ev, deltas = st.compact_with_deltas(None)
+        ref_parts = build_graph_parts(ev)
+        ref_ed = build_edges_and_flags(ref_parts)
+        parts, ed = merge_window(ev, IncrementalWindowState().summaries(deltas))
+        assert np.array_equal(parts["path_root"], ref_parts["path_root"]), trial
+        assert np.array_equal(parts["ev_file"], ref_parts["ev_file"]), trial
+        assert np.array_equal(parts["ev_proc"], ref_parts["ev_proc"]), trial
+        assert np.array_equal(ed["edge_index"], ref_ed["edge_index"]), trial
+        assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-6), trial
+        assert np.array_equal(ed["suspicious"], ref_ed["suspicious"]), trial
