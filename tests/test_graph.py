@@ -289,3 +289,42 @@ def test_incremental_random_streams_match_full():
         assert np.array_equal(ed["edge_index"], ref_ed["edge_index"]), trial
         assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-6), trial
         assert np.array_equal(ed["suspicious"], ref_ed["suspicious"]), trial
+
+
+def test_bulk_append_array_matches_scalar_append():
+    """store.append_array == per-event append: same delta boundaries, same
+    compacted window, same string interning."""
+    import numpy as np
+
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.data.trace import SYSCALL_NAMES
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    arr, _ = generate(SynthConfig(duration_s=12.0, benign_rate_hz=600.0,
+                                  n_benign_files=60, seed=4))
+    a = DeltaGraphStore(window_s=1000.0, delta_s=2.0)
+    for i in range(len(arr)):
+        a.append(
+            ts=float(arr.ts[i]), pid=int(arr.pid[i]),
+            syscall=SYSCALL_NAMES.get(int(arr.syscall[i]), "unknown"),
+            path=arr.paths.lookup(int(arr.path_id[i])) if arr.path_id[i] >= 0 else "",
+            new_path=arr.paths.lookup(int(arr.new_path_id[i])) if arr.new_path_id[i] >= 0 else "",
+            nbytes=int(arr.nbytes[i]),
+            comm=arr.comms.lookup(int(arr.comm_id[i])) if arr.comm_id[i] >= 0 else "",
+        )
+    b = DeltaGraphStore(window_s=1000.0, delta_s=2.0)
+    b.append_array(arr)
+
+    ea, da = a.compact_with_deltas(None)
+    eb, db = b.compact_with_deltas(None)
+    assert [len(d) for d in da] == [len(d) for d in db]
+    assert np.array_equal(ea.ts, eb.ts)
+    assert np.array_equal(ea.syscall, eb.syscall)
+    assert np.array_equal(ea.nbytes, eb.nbytes)
+    # same strings behind the (possibly differently-ordered) ids
+    pa = [ea.paths.lookup(int(i)) if i >= 0 else "" for i in ea.path_id[:500]]
+    pb = [eb.paths.lookup(int(i)) if i >= 0 else "" for i in eb.path_id[:500]]
+    assert pa == pb
+    ca = [ea.comms.lookup(int(i)) if i >= 0 else "" for i in ea.comm_id[:500]]
+    cb = [eb.comms.lookup(int(i)) if i >= 0 else "" for i in eb.comm_id[:500]]
+    assert ca == cb
