@@ -9,6 +9,10 @@ evicted.
 The store is columnar end-to-end so the GPU path stages each delta as a
 handful of contiguous copies into HBM (288 GB/GPU leaves the whole window
 resident).
+
+Threading: one ingest thread per store (appends intern into the shared
+string tables); compaction/scoring may run concurrently from another thread
+(the (events, deltas) snapshot is taken under the lock).
 """
 from __future__ import annotations
 
@@ -58,19 +62,21 @@ class DeltaGraphStore:
         n = len(arr)
         if n == 0:
             return
+        same_tables = arr.paths is self.paths and arr.comms is self.comms
         order = None
         if not bool(np.all(arr.ts[1:] >= arr.ts[:-1])):
             order = np.argsort(arr.ts, kind="stable")
         with self._lock:
             self._seal_locked()
             # intern this array's string tables into the store-wide ones
-            # (O(unique strings), not O(events))
+            # (O(unique strings), not O(events)); identity when the array
+            # was already built against the store's tables
             plut = (
                 np.fromiter(
                     (self.paths.intern(s) for s in arr.paths.strings),
                     dtype=np.int64, count=len(arr.paths),
                 )
-                if len(arr.paths)
+                if len(arr.paths) and not same_tables
                 else np.empty(0, np.int64)
             )
             clut = (
@@ -78,7 +84,7 @@ class DeltaGraphStore:
                     (self.comms.intern(s) for s in arr.comms.strings),
                     dtype=np.int64, count=len(arr.comms),
                 )
-                if len(arr.comms)
+                if len(arr.comms) and not same_tables
                 else np.empty(0, np.int64)
             )
 
@@ -111,6 +117,16 @@ class DeltaGraphStore:
             self._evict_locked(float(ts[-1]))
 
     def append_wire_batch(self, events) -> None:
+        if len(events) >= 64:
+            # columnar fast path for replay-sized frames (live broadcast
+            # frames are 1-event; the scalar path is cheaper there)
+            b = EventArrayBuilder(self.paths, self.comms)
+            for ev in events:
+                b.add(ts=ev.timestamp, pid=ev.pid, syscall=ev.syscall,
+                      path=ev.path, new_path=ev.new_path, nbytes=ev.bytes,
+                      ret_val=ev.ret_val, comm=ev.comm)
+            self.append_array(b.build(sort=False))
+            return
         for ev in events:
             self.append(
                 ts=ev.timestamp, pid=ev.pid, syscall=ev.syscall, path=ev.path,
