@@ -195,22 +195,26 @@ class StreamingEngine:
         self.events_scored += len(events)
 
         # ---- map node/sequence scores back to live paths -------------------
-        file_scores: Dict[str, float] = {}
+        # file nodes are the first n_files entries (unique path roots), so
+        # the dict builds are straight zips — no per-node function calls
+        strings = events.paths.strings
+        n_files = int(parts["n_files"])
+        ns = node_score
+        if len(ns) < len(node_kind):
+            ns = _np.concatenate([ns, _np.zeros(len(node_kind) - len(ns))])
+        file_scores: Dict[str, float] = dict(
+            zip((strings[k] for k in node_key[:n_files].tolist()), ns[:n_files].tolist())
+        )
+        proc_scores: Dict[int, float] = dict(
+            zip(node_key[n_files:].tolist(), ns[n_files : len(node_kind)].tolist())
+        )
         file_mb: Dict[str, float] = {}
-        proc_scores: Dict[int, float] = {}
-        for ni in range(len(node_kind)):
-            if node_kind[ni] == 1:
-                path = events.paths.lookup(int(node_key[ni]))
-                s = float(node_score[ni]) if ni < len(node_score) else 0.0
-                file_scores[path] = max(file_scores.get(path, 0.0), s)
-            else:
-                proc_scores[int(node_key[ni])] = float(node_score[ni]) if ni < len(node_score) else 0.0
         if seq_score is not None and len(seq_score) == len(seq_fids):
-            for bi in range(len(seq_score)):
-                pid_ = int(seq_fids[bi])
+            for pid_, sc in zip(seq_fids.tolist(), seq_score.tolist()):
                 if pid_ >= 0:
-                    path = events.paths.lookup(pid_)
-                    file_scores[path] = max(file_scores.get(path, 0.0), float(seq_score[bi]))
+                    path = strings[pid_]
+                    if sc > file_scores.get(path, 0.0):
+                        file_scores[path] = sc
         # bytes per file (window-local) — vectorised: a per-event Python loop
         # here costs ~0.4 s per 600k-event window, which dominates the host
         # path once the model forward is on the GPU
@@ -230,8 +234,13 @@ class StreamingEngine:
         writes = int((sc == SYSCALL_IDS["write"]).sum())
         renames = int((sc == SYSCALL_IDS["rename"]).sum())
         w2r = renames / max(writes + renames, 1)
-        encrypted_paths = [p for p in events.paths.strings if _SUSPICIOUS_EXT.search(p)]
-        note = any(_RANSOM_NOTE.search(p) for p in events.paths.strings)
+        # cached per-path regex bits (grow-only table => only new strings
+        # are ever scanned; a fresh regex pass here cost ~5 ms/tick)
+        from ..graph.constructor import _string_flag_bits
+
+        bits = _string_flag_bits(events.paths)
+        encrypted_paths = [strings[i] for i in _np.nonzero(bits & 1)[0].tolist()]
+        note = bool((bits & 2).any())
         indicators = {
             "write_to_rename": w2r,
             "suspicious_ext_count": float(len(encrypted_paths)),
