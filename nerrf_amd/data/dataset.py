@@ -123,7 +123,7 @@ def synth_window_batches(
     seq_len: int = 100,
     base_seed: int = 0,
     kinds: tuple = ("lockbit", "supply_chain"),
-    benign_kinds: tuple = ("lockbit", "benign_rotate", "benign_backup"),
+    benign_kinds: tuple = ("lockbit", "benign_rotate", "benign_backup", "benign_build"),
 ) -> List[WindowBatch]:
     """Prebuild window batches from synthetic scenarios.
 

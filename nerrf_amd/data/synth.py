@@ -50,7 +50,7 @@ class SynthConfig:
     target_dir: str = "/app/uploads"
     encrypted_ext: str = ".lockbit3"
     seed: int = 0
-    kind: str = "lockbit"  # lockbit | supply_chain | benign_rotate | benign_backup
+    kind: str = "lockbit"  # lockbit | supply_chain | benign_rotate | benign_backup | benign_build
 
 
 _BENIGN_DIRS = ["/var/www/html", "/home/svc/data", "/app/cache", "/var/log/app"]
@@ -72,6 +72,8 @@ def generate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
         return generate_benign_rotate(cfg)
     if cfg.kind == "benign_backup":
         return generate_benign_backup(cfg)
+    if cfg.kind == "benign_build":
+        return generate_benign_build(cfg)
     rng = np.random.default_rng(cfg.seed)
     paths = StringTable()
     comms = StringTable()
@@ -434,6 +436,73 @@ def generate_benign_backup(cfg: SynthConfig) -> Tuple[EventArray, Optional[Attac
     benign_cols = (base.ts, base.pid, base.syscall, base.path_id,
                    base.new_path_id, base.nbytes, base.comm_id)
     ts_, pid_, sys_, path_, newp_, bytes_, comm_ = _interleave([benign_cols, bk_cols])
+    arr = EventArray(ts=ts_, pid=pid_, syscall=sys_, path_id=path_, new_path_id=newp_,
+                     nbytes=bytes_, ret_val=np.zeros(len(ts_), dtype=np.int64),
+                     comm_id=comm_, paths=paths, comms=comms)
+    return arr, None
+
+
+def generate_benign_build(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
+    """Hard negative: compiler walking a dependency tree.
+
+    One toolchain process fans reads over many sources/headers and writes
+    one object per translation unit, with exec bursts (cc1/as/ld) — the
+    degree/recon-burst profile of a supply-chain compromise touching a
+    dependency tree, but clean ground truth (NEXT.md detection-realism
+    item: "compiler-touching-dependency-tree negatives")."""
+    rng = np.random.default_rng(cfg.seed)
+    base, _ = generate(SynthConfig(
+        duration_s=cfg.duration_s, n_benign_procs=cfg.n_benign_procs,
+        n_benign_files=cfg.n_benign_files, benign_rate_hz=cfg.benign_rate_hz,
+        attack=False, seed=cfg.seed,
+    ))
+    paths, comms = base.paths, base.comms
+    cc_pid = np.int64(777)
+    cc_comm = comms.intern("cc1plus")
+    t0 = cfg.attack_start_frac * cfg.duration_s
+    n_units = max(cfg.n_victim_files, 8)
+    headers = [paths.intern(f"/src/include/h_{k:03d}.h") for k in range(16)]
+    tool_bins = [paths.intern(p) for p in ("/usr/bin/cc1plus", "/usr/bin/as", "/usr/bin/ld")]
+    ts_l, sys_l, path_l, bytes_l = [], [], [], []
+    t = t0
+    for j in range(n_units):
+        src = paths.intern(f"/src/module/unit_{j:03d}.cc")
+        obj = paths.intern(f"/src/build/unit_{j:03d}.o")
+        # exec burst: toolchain binaries
+        for b in tool_bins[:2]:
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["exec"]); path_l.append(b); bytes_l.append(0)
+            t += 0.002
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["openat"]); path_l.append(src); bytes_l.append(0)
+        t += 0.003
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["read"]); path_l.append(src); bytes_l.append(32768)
+        t += 0.004
+        # dependency-tree fan: read a random subset of shared headers
+        for h in rng.choice(len(headers), size=6, replace=False):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["read"]); path_l.append(headers[int(h)]); bytes_l.append(8192)
+            t += 0.002
+        for _ in range(2):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["write"]); path_l.append(obj); bytes_l.append(65536)
+            t += 0.004
+        t += float(rng.uniform(0, 0.01))
+    # link step: read all objects, write one binary
+    out_bin = paths.intern("/src/build/app.bin")
+    ts_l.append(t); sys_l.append(SYSCALL_IDS["exec"]); path_l.append(tool_bins[2]); bytes_l.append(0)
+    t += 0.005
+    for j in range(n_units):
+        obj = paths.intern(f"/src/build/unit_{j:03d}.o")
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["read"]); path_l.append(obj); bytes_l.append(65536)
+        t += 0.002
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["write"]); path_l.append(out_bin); bytes_l.append(65536)
+        t += 0.002
+    n = len(ts_l)
+    cc_cols = (
+        np.asarray(ts_l), np.full(n, cc_pid), np.asarray(sys_l, dtype=np.int8),
+        np.asarray(path_l, dtype=np.int64), np.full(n, -1, dtype=np.int64),
+        np.asarray(bytes_l, dtype=np.int64), np.full(n, cc_comm, dtype=np.int64),
+    )
+    benign_cols = (base.ts, base.pid, base.syscall, base.path_id,
+                   base.new_path_id, base.nbytes, base.comm_id)
+    ts_, pid_, sys_, path_, newp_, bytes_, comm_ = _interleave([benign_cols, cc_cols])
     arr = EventArray(ts=ts_, pid=pid_, syscall=sys_, path_id=path_, new_path_id=newp_,
                      nbytes=bytes_, ret_val=np.zeros(len(ts_), dtype=np.int64),
                      comm_id=comm_, paths=paths, comms=comms)

@@ -95,7 +95,8 @@ def test_sharded_detection_merge():
 
 
 def test_fp_undo_zero_on_hard_negatives(tmp_path):
-    """FP-undo target: on benign lookalikes (log rotation, backup daemon)
+    """FP-undo target: on benign lookalikes (log rotation, backup daemon,
+    compiler walking a dependency tree)
     the response loop takes ZERO destructive actions — even when the model
     raises scores, there is nothing the executor will touch and the sandbox
     gate refuses an empty restore."""
@@ -103,7 +104,7 @@ def test_fp_undo_zero_on_hard_negatives(tmp_path):
     from nerrf_amd.serve.engine import StreamingEngine, load_model_from_checkpoint
 
     model = load_model_from_checkpoint("checkpoints/pretrained")
-    for kind in ("benign_rotate", "benign_backup"):
+    for kind in ("benign_rotate", "benign_backup", "benign_build"):
         engine = StreamingEngine(model=model, device="cpu")
         engine.store.window_s = 1e9
         arr, _ = generate(SynthConfig(seed=91, duration_s=40, benign_rate_hz=50, kind=kind))
