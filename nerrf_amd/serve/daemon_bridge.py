@@ -69,30 +69,35 @@ def pump_daemon_into_store(
         decoder = None
         have_native = False
 
-    from ..data.trace import SYSCALL_NAMES
+    import numpy as np
+
+    from ..data.trace import EventArray
 
     n = 0
     for frame in frames_from_daemon(host, port, timeout_s=timeout_s):
         if have_native:
             ts, pid, sysc, path_id, newp_id, nbytes, ret, comm = decoder.decode([frame])
-            new_paths = decoder.paths_since(len(store.paths))
-            for s in new_paths:
+            # the decoder's id space and the store tables are kept in
+            # lockstep (this pump is the store's sole feeder), so the
+            # columns drop straight into an EventArray — no per-event
+            # string round trips
+            for s in decoder.paths_since(len(store.paths)):
                 store.paths.intern(s)
-            new_comms = decoder.comms_since(len(store.comms))
-            for s in new_comms:
+            for s in decoder.comms_since(len(store.comms)):
                 store.comms.intern(s)
-            for i in range(len(ts)):
-                store.append(
-                    ts=float(ts[i]),
-                    pid=int(pid[i]),
-                    syscall=SYSCALL_NAMES.get(int(sysc[i]), "unknown"),
-                    path=store.paths.lookup(int(path_id[i])) if path_id[i] >= 0 else "",
-                    new_path=store.paths.lookup(int(newp_id[i])) if newp_id[i] >= 0 else "",
-                    nbytes=int(nbytes[i]),
-                    ret_val=int(ret[i]),
-                    comm=store.comms.lookup(int(comm[i])) if comm[i] >= 0 else "",
-                )
-            n += len(ts)
+            arr = EventArray(
+                paths=store.paths, comms=store.comms,
+                ts=np.asarray(ts, dtype=np.float64),
+                pid=np.asarray(pid, dtype=np.int64),
+                syscall=np.asarray(sysc, dtype=np.int8),
+                path_id=np.asarray(path_id, dtype=np.int64),
+                new_path_id=np.asarray(newp_id, dtype=np.int64),
+                nbytes=np.asarray(nbytes, dtype=np.int64),
+                ret_val=np.asarray(ret, dtype=np.int64),
+                comm_id=np.asarray(comm, dtype=np.int64),
+            )
+            store.append_array(arr)
+            n += len(arr)
         else:
             events = codec.decode_event_batch(frame)
             store.append_wire_batch(events)
