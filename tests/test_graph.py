@@ -328,3 +328,26 @@ def test_bulk_append_array_matches_scalar_append():
     ca = [ea.comms.lookup(int(i)) if i >= 0 else "" for i in ea.comm_id[:500]]
     cb = [eb.comms.lookup(int(i)) if i >= 0 else "" for i in eb.comm_id[:500]]
     assert ca == cb
+
+
+def test_append_array_unsorted_and_eviction():
+    """Bulk ingest sorts jittered timestamps and evicts out-of-window deltas
+    like the scalar path."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    b = EventArrayBuilder(StringTable(), StringTable())
+    ts = [5.0, 1.0, 3.0, 2.0, 4.0, 50.0, 49.0, 51.0]
+    for i, t in enumerate(ts):
+        b.add(ts=t, pid=3, syscall="write", path=f"/u/f{i}", nbytes=10)
+    st = DeltaGraphStore(window_s=10.0, delta_s=2.0)
+    st.append_array(b.build(sort=False))
+    ev = st.compact(None)
+    kept = [float(x) for x in ev.ts]
+    # everything before 51-10=41 was evicted at ingest time
+    assert kept == [49.0, 50.0, 51.0]
+    assert st.evicted_events == 5
+    # string identity survived the remap
+    assert all(ev.paths.lookup(int(i)).startswith("/u/f") for i in ev.path_id)
