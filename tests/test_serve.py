@@ -294,3 +294,29 @@ def test_rollback_nested_directories(tmp_path):
     assert res.files_restored == 5
     assert res.sha256_ok is True
     assert all(verify_manifest(manifest).values())
+
+
+def test_monitor_long_run_bounded_state():
+    """Over many windows with eviction, the store keeps a bounded delta ring
+    and the incremental summary cache prunes to the live window."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    eng = StreamingEngine(device="cpu", window_s=8.0)
+    rng = np.random.default_rng(0)
+    t = 1000.0
+    for chunk in range(30):
+        b = EventArrayBuilder(eng.store.paths, eng.store.comms)
+        for _ in range(300):
+            t += 0.01
+            b.add(ts=t, pid=int(50 + rng.integers(0, 4)), syscall="write",
+                  path=f"/w/f{int(rng.integers(0, 50))}", nbytes=128)
+        eng.store.append_array(b.build(sort=False))
+        det = eng.score_window()
+        assert det.window_events > 0
+        # delta ring bounded by window/delta_s (+ slack for partial chunks)
+        assert len(eng.store._deltas) <= int(8.0 / eng.store.delta_s) + 3
+        assert len(eng._inc_state._cache) <= len(eng.store._deltas) + 1
+    assert eng.store.evicted_events > 0
