@@ -320,3 +320,24 @@ def test_monitor_long_run_bounded_state():
         assert len(eng.store._deltas) <= int(8.0 / eng.store.delta_s) + 3
         assert len(eng._inc_state._cache) <= len(eng.store._deltas) + 1
     assert eng.store.evicted_events > 0
+
+
+def test_cli_simulate_then_undo(tmp_path, capsys):
+    """`nerrf simulate --seed-files` then `nerrf undo --trace` restores the
+    victim directory byte-exactly (manifest-gated)."""
+    from nerrf_amd.cli import main
+    from nerrf_amd.harness.attack_sim import verify_manifest
+
+    d = tmp_path / "victim"
+    trace = tmp_path / "attack.jsonl"
+    rc = main(["simulate", "--dir", str(d), "--seed-files", "--n-files", "5",
+               "--file-kb", "4", "--trace-out", str(trace)])
+    assert rc == 0
+    capsys.readouterr()
+    rc = main(["undo", "--dir", str(d), "--trace", str(trace), "--sims", "64"])
+    out = json.loads(capsys.readouterr().out)
+    assert rc == 0
+    assert out["alarm"] is True
+    assert out["sha256_ok"] is True
+    manifest = json.loads((d / ".nerrf_manifest.json").read_text())
+    assert all(verify_manifest(manifest).values())
