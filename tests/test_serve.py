@@ -341,3 +341,24 @@ def test_cli_simulate_then_undo(tmp_path, capsys):
     assert out["sha256_ok"] is True
     manifest = json.loads((d / ".nerrf_manifest.json").read_text())
     assert all(verify_manifest(manifest).values())
+
+
+def test_rollback_fails_closed_on_unverifiable_manifest(tmp_path):
+    """A manifest entry that can never verify (missing file) makes the
+    sandbox gate reject the WHOLE plan: nothing is touched on the live
+    tree, and the encrypted files remain recoverable."""
+    from nerrf_amd.harness.attack_sim import run_attack, seed_files
+
+    manifest = seed_files(tmp_path, n_files=3, file_kb=4, seed=2)
+    run_attack(tmp_path)
+    manifest[str(tmp_path / "ghost.dat")] = "0" * 64
+    res = execute_rollback(tmp_path, manifest=manifest)
+    assert res.files_restored == 0
+    assert res.sandbox_validated is False and res.sha256_ok is False
+    assert any("sandbox gate" in d for d in res.details)
+    # encrypted artifacts untouched => a corrected manifest can still recover
+    assert len(list(tmp_path.glob("*.lockbit3"))) == 3
+    res2 = execute_rollback(
+        tmp_path, manifest={k: v for k, v in manifest.items() if "ghost" not in k}
+    )
+    assert res2.files_restored == 3 and res2.sha256_ok is True
