@@ -95,3 +95,46 @@ def test_roc_auc_against_sklearn():
     # with heavy ties
     s_t = np.round(s, 1)
     assert abs(roc_auc(y, s_t) - roc_auc_score(y, s_t)) < 1e-9
+
+
+def _worker_unused_head(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from nerrf_amd.parallel.ddp import GradAllReducer
+
+        class TwoHeads(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.trunk = torch.nn.Linear(4, 8)
+                self.used = torch.nn.Linear(8, 1)
+                self.unused = torch.nn.Linear(8, 1)
+
+            def forward(self, x):
+                return self.used(self.trunk(x))
+
+        torch.manual_seed(3)
+        model = TwoHeads()
+        reducer = GradAllReducer(model, bucket_bytes=64)  # many buckets
+        x = torch.full((2, 4), float(rank + 1))
+        model(x).sum().backward()
+        # the unused head's bucket hooks never fire; finalize must still
+        # all-reduce it (as zeros) without deadlocking
+        reducer.finalize()
+        results[rank] = {
+            "used": model.used.weight.grad.numpy().copy(),
+            "unused": model.unused.weight.grad.numpy().copy(),
+        }
+    finally:
+        dist.destroy_process_group()
+
+
+def test_grad_allreduce_unused_head_no_deadlock():
+    port = _find_free_port()
+    mgr = mp.Manager()
+    results = mgr.dict()
+    mp.spawn(_worker_unused_head, args=(2, port, results), nprocs=2, join=True)
+    assert np.allclose(results[0]["used"], results[1]["used"], atol=1e-6)
+    assert np.allclose(results[0]["unused"], 0.0)
+    assert np.allclose(results[1]["unused"], 0.0)
