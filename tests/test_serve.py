@@ -362,3 +362,15 @@ def test_rollback_fails_closed_on_unverifiable_manifest(tmp_path):
         tmp_path, manifest={k: v for k, v in manifest.items() if "ghost" not in k}
     )
     assert res2.files_restored == 3 and res2.sha256_ok is True
+
+
+def test_cli_serve_trace_demo(tmp_path, capsys):
+    """`nerrf serve --trace` demo: tracker-sim stream -> engine -> verdict."""
+    from nerrf_amd.cli import main
+
+    rc = main(["serve", "--trace", "datasets/traces/toy_trace.csv",
+               "--timeout", "10"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    verdict = json.loads(out[out.index("{"):])
+    assert verdict["events_ingested"] > 0
