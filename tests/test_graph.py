@@ -1,5 +1,6 @@
 """Graph constructor + sampling tests."""
 import numpy as np
+import pytest
 
 from nerrf_amd.data.labels import event_labels
 from nerrf_amd.data.synth import SynthConfig, generate
@@ -351,3 +352,30 @@ def test_append_array_unsorted_and_eviction():
     assert st.evicted_events == 5
     # string identity survived the remap
     assert all(ev.paths.lookup(int(i)).startswith("/u/f") for i in ev.path_id)
+
+
+@pytest.mark.parametrize("kind", ["supply_chain", "benign_rotate", "benign_backup", "benign_build"])
+def test_incremental_merge_all_scenario_kinds(kind):
+    """Incremental == full across every synthetic scenario family."""
+    import numpy as np
+
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    arr, _ = generate(SynthConfig(kind=kind, duration_s=18.0, benign_rate_hz=800.0,
+                                  n_benign_files=120, seed=31))
+    eng = StreamingEngine(device="cpu")
+    eng.store.window_s = 1e9
+    eng.ingest_events(arr)
+    ev, deltas = eng.store.compact_with_deltas(None)
+    ref_parts = build_graph_parts(ev)
+    ref_ed = build_edges_and_flags(ref_parts)
+    parts, ed = merge_window(ev, IncrementalWindowState().summaries(deltas))
+    assert np.array_equal(parts["ev_file"], ref_parts["ev_file"]), kind
+    assert np.array_equal(parts["ev_proc"], ref_parts["ev_proc"]), kind
+    assert np.array_equal(ed["edge_index"], ref_ed["edge_index"]), kind
+    assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-5), kind
+    for k in ("suspicious", "note", "recon", "double_ext", "in_deg", "out_deg"):
+        assert np.array_equal(ed[k], ref_ed[k]), (kind, k)
