@@ -157,6 +157,44 @@ def cmd_serve(args) -> int:
     return 0
 
 
+def cmd_eval(args) -> int:
+    """Reproduce the detection-quality table: per-family ROC-AUC / F1 on
+    held-out synthetic scenarios, plus max model scores on the benign
+    hard negatives (spec targets: ROC-AUC >= 0.90, seq F1 >= 0.95)."""
+    import torch
+
+    from .data.dataset import synth_window_batches
+    from .serve.engine import load_model_from_checkpoint
+    from .train import evaluate
+
+    device = args.device
+    dtype = torch.bfloat16 if (device != "cpu" and args.dtype == "bf16") else torch.float32
+    model = load_model_from_checkpoint(args.checkpoint).to(device, dtype).eval()
+    out = {}
+    for kind in args.families.split(","):
+        hb = synth_window_batches(n_scenarios=args.scenarios, attack_fraction=0.67,
+                                  base_seed=555000, kinds=(kind,))
+        rep = evaluate(model, hb, device, dtype)
+        out[kind] = {k: round(float(v), 4) for k, v in rep.items()
+                     if "auc" in k or k.endswith("f1")}
+    for kind in args.negatives.split(","):
+        if not kind:
+            continue
+        hb = synth_window_batches(n_scenarios=2, attack_fraction=0.0,
+                                  base_seed=777000, benign_kinds=(kind,))
+        mx = 0.0
+        with torch.no_grad():
+            for b in hb:
+                tb = b.to_torch(device, dtype)
+                nl, _, sl = model(tb)
+                mx = max(mx, float(torch.sigmoid(nl.float()).max()))
+                if sl is not None and sl.numel():
+                    mx = max(mx, float(torch.sigmoid(sl.float()).max()))
+        out[f"negative:{kind}"] = {"max_model_score": round(mx, 4)}
+    print(json.dumps(out, indent=2))
+    return 0
+
+
 def cmd_train(args, extra) -> int:
     from .train import main as train_main
 
@@ -205,6 +243,14 @@ def main(argv=None) -> int:
     p.add_argument("--timeout", type=float, default=15.0)
     p.add_argument("--device", default="cpu")
 
+    p = sub.add_parser("eval", help="detection-quality report on synthetic scenarios")
+    p.add_argument("--checkpoint", default="checkpoints/pretrained")
+    p.add_argument("--families", default="lockbit,supply_chain")
+    p.add_argument("--negatives", default="benign_rotate,benign_backup,benign_build")
+    p.add_argument("--scenarios", type=int, default=3)
+    p.add_argument("--device", default="cpu")
+    p.add_argument("--dtype", default="bf16")
+
     sub.add_parser("train", help="training entrypoint (args passed through)")
 
     if argv is None:
@@ -218,6 +264,7 @@ def main(argv=None) -> int:
         "undo": cmd_undo,
         "scenario": cmd_scenario,
         "serve": cmd_serve,
+        "eval": cmd_eval,
     }[args.cmd](args)
 
 

@@ -374,3 +374,15 @@ def test_cli_serve_trace_demo(tmp_path, capsys):
     out = capsys.readouterr().out
     verdict = json.loads(out[out.index("{"):])
     assert verdict["events_ingested"] > 0
+
+
+def test_cli_eval_report(capsys):
+    """`nerrf eval` reproduces the detection-quality table."""
+    from nerrf_amd.cli import main
+
+    rc = main(["eval", "--scenarios", "1", "--families", "lockbit",
+               "--negatives", "benign_rotate"])
+    assert rc == 0
+    rep = json.loads(capsys.readouterr().out)
+    assert rep["lockbit"]["node_auc"] >= 0.9
+    assert "negative:benign_rotate" in rep
