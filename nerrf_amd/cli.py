@@ -115,12 +115,16 @@ def cmd_serve(args) -> int:
     if args.tracker:
         # live mode: consume an existing Tracker/StreamEvents endpoint and
         # monitor continuously (deploy/engine-deployment.yaml wiring)
+        import torch
+
         model = None
         if args.checkpoint:
             from .serve.engine import load_model_from_checkpoint
 
             model = load_model_from_checkpoint(args.checkpoint)
-        engine = StreamingEngine(model=model, device=args.device)
+        # bf16 on GPU enables the fused-MFMA scoring path
+        dtype = torch.bfloat16 if args.device != "cpu" else torch.float32
+        engine = StreamingEngine(model=model, device=args.device, dtype=dtype)
         for status in engine.run_monitor(
             interval_s=args.interval,
             max_iterations=args.iterations,
