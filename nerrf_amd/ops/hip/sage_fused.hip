@@ -6,7 +6,9 @@
 //   out   = h + LayerNorm(GELU(z)) * gamma + beta           (residual)
 // — the inference path of models/graphsage.SageLayer in a single kernel
 // (eager: gather + 2 hipBLASLt GEMMs + ~5 elementwise/reduce launches).
-// Inference-only: training keeps the autograd path.
+// Inference-only: training keeps the autograd path.  (Model spec:
+// reference docs architecture.mdx:49-53 — GraphSAGE-T, 28 layers; the
+// kernel design itself has no upstream counterpart.)
 //
 // Geometry: 8 waves (512 threads); wave w owns rows [(w>>1)*16, +16) and the
 // column half (w&1)*64 of the 64x128 output tile -> per GEMM 4 accumulators
