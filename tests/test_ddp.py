@@ -138,3 +138,24 @@ def test_grad_allreduce_unused_head_no_deadlock():
     assert np.allclose(results[0]["used"], results[1]["used"], atol=1e-6)
     assert np.allclose(results[0]["unused"], 0.0)
     assert np.allclose(results[1]["unused"], 0.0)
+
+
+def test_roc_auc_degenerate_classes():
+    import math
+
+    assert math.isnan(roc_auc(np.zeros(5), np.random.rand(5)))
+    assert math.isnan(roc_auc(np.ones(5), np.random.rand(5)))
+
+
+def test_precision_recall_edge_cases():
+    from nerrf_amd.eval import best_f1, precision_recall_f1
+
+    y = np.array([0, 0, 1, 1])
+    s = np.array([0.1, 0.2, 0.8, 0.9])
+    m = precision_recall_f1(y, s, 0.5)
+    assert m["precision"] == 1.0 and m["recall"] == 1.0 and m["f1"] == 1.0
+    # no predictions above threshold
+    m = precision_recall_f1(y, s, 0.95)
+    assert m["precision"] == 0.0 and m["recall"] == 0.0 and m["f1"] == 0.0
+    b = best_f1(y, s)
+    assert b["f1"] == 1.0
