@@ -216,7 +216,10 @@ def load_jsonl(path: str | Path) -> EventArray:
             line = line.strip()
             if not line:
                 continue
-            rec = json.loads(line)
+            try:
+                rec = json.loads(line)
+            except json.JSONDecodeError:
+                continue  # tolerate truncated/corrupt lines in field traces
             builder.add(
                 ts=float(rec.get("timestamp", 0.0)),
                 pid=int(rec.get("pid", 0)),

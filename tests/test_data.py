@@ -164,3 +164,16 @@ def test_build_sequences_torch_empty():
     ev = EventArrayBuilder(StringTable(), StringTable()).build()
     feats, lengths, fids = build_sequences_torch(ev, device="cpu")
     assert feats.shape[0] == 0 and lengths.numel() == 0
+
+
+def test_load_jsonl_tolerates_corrupt_lines(tmp_path):
+    p = tmp_path / "t.jsonl"
+    p.write_text(
+        '{"timestamp": 1.0, "event": "write", "path": "/a", "size": 4, "pid": 9}\n'
+        "{this is not json\n"
+        '{"timestamp": 2.0, "event": "read", "path": "/a", "size": 2, "pid": 9}\n'
+    )
+    from nerrf_amd.data.trace import load_trace
+
+    arr = load_trace(p)
+    assert len(arr) == 2
