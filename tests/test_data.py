@@ -177,3 +177,22 @@ def test_load_jsonl_tolerates_corrupt_lines(tmp_path):
 
     arr = load_trace(p)
     assert len(arr) == 2
+
+
+def test_write_csv_load_csv_roundtrip(tmp_path):
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.data.trace import load_trace, write_csv
+
+    arr, _ = generate(SynthConfig(duration_s=4.0, benign_rate_hz=200.0,
+                                  n_benign_files=30, attack=False, seed=6))
+    p = tmp_path / "rt.csv"
+    write_csv(p, arr)
+    back = load_trace(p)
+    assert len(back) == len(arr)
+    assert np.allclose(back.ts, arr.ts, atol=1e-5)
+    assert np.array_equal(back.syscall, arr.syscall)
+    assert np.array_equal(back.nbytes, arr.nbytes)
+    # path strings survive (ids may be renumbered)
+    a = [arr.paths.lookup(int(i)) if i >= 0 else "" for i in arr.path_id[:200]]
+    b = [back.paths.lookup(int(i)) if i >= 0 else "" for i in back.path_id[:200]]
+    assert a == b
