@@ -50,7 +50,7 @@ class SynthConfig:
     target_dir: str = "/app/uploads"
     encrypted_ext: str = ".lockbit3"
     seed: int = 0
-    kind: str = "lockbit"  # lockbit | supply_chain | benign_rotate | benign_backup | benign_build
+    kind: str = "lockbit"  # lockbit | supply_chain | supply_chain_net | benign_rotate | benign_backup | benign_build
 
 
 _BENIGN_DIRS = ["/var/www/html", "/home/svc/data", "/app/cache", "/var/log/app"]
@@ -66,7 +66,7 @@ def _interleave(cols: List[Tuple[np.ndarray, ...]]) -> Tuple[np.ndarray, ...]:
 
 
 def generate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
-    if cfg.kind == "supply_chain":
+    if cfg.kind in ("supply_chain", "supply_chain_net"):
         return generate_supply_chain(cfg)
     if cfg.kind == "benign_rotate":
         return generate_benign_rotate(cfg)
@@ -324,6 +324,15 @@ def generate_supply_chain(cfg: SynthConfig) -> Tuple[EventArray, Optional[Attack
             t += (chunk / 1e6) / cfg.encrypt_rate_mbps / 2
             ts_l.append(t); sys_l.append(SYSCALL_IDS["write"]); path_l.append(blob_id); bytes_l.append(chunk)
             t += (chunk / 1e6) / cfg.encrypt_rate_mbps / 2
+    if cfg.kind == "supply_chain_net":
+        # network egress: ship the staged blob to an unlisted destination —
+        # the socket node (kind 2) + destination-allowlist flag light up
+        dest_id = paths.intern("tcp://203.0.113.37:443")
+        ts_l.append(t); sys_l.append(SYSCALL_IDS["connect"]); path_l.append(dest_id); bytes_l.append(0)
+        t += 0.005
+        for _ in range(k * n_data):
+            ts_l.append(t); sys_l.append(SYSCALL_IDS["sendto"]); path_l.append(dest_id); bytes_l.append(chunk)
+            t += (chunk / 1e6) / cfg.encrypt_rate_mbps
     n_atk = len(ts_l)
     atk_cols = (
         np.asarray(ts_l),

@@ -34,6 +34,11 @@ SYSCALL_IDS: Dict[str, int] = {
     "close": 7,
     "mkdir": 8,
     "exec": 9,
+    # network egress (graph spec lists socket nodes: reference docs
+    # architecture.mdx:36-43); destinations enter the path domain as
+    # "tcp://host:port" strings
+    "connect": 10,
+    "sendto": 11,
 }
 SYSCALL_NAMES = {v: k for k, v in SYSCALL_IDS.items()}
 UNKNOWN_SYSCALL = 0

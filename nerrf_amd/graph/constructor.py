@@ -35,6 +35,13 @@ _RANSOM_NOTE = re.compile(r"(^|/)(README|HOW_TO|RESTORE)[-_]", re.IGNORECASE)
 _RECON_BIN = re.compile(r"^/(usr/)?s?bin/")
 _DOUBLE_EXT = re.compile(r"\.\w{1,5}\.\w{1,9}$")
 
+# socket-destination policy: egress to a destination NOT matching the
+# allowlist marks the socket node suspicious (the "destination allowlist"
+# feature channel — threat-model.md hard-negative discussion).  Deployment
+# overrides this module constant from config.
+_SOCKET_PREFIXES = ("tcp://", "udp://")
+DEST_ALLOWLIST = ("tcp://10.", "tcp://192.168.", "tcp://backup.", "udp://10.")
+
 
 class _UnionFind:
     def __init__(self, n: int) -> None:
@@ -120,6 +127,10 @@ def _string_flag_bits(paths) -> np.ndarray:
             b |= 4
         if _DOUBLE_EXT.search(s):
             b |= 8
+        if s.startswith(_SOCKET_PREFIXES):
+            b |= 16  # socket node
+            if not s.startswith(DEST_ALLOWLIST):
+                b |= 32  # egress to an unlisted destination
         bits[path_idx] = b
     paths._nerrf_flag_cache = bits
     return bits
@@ -142,11 +153,27 @@ def _path_flags(
     note = np.zeros(n_files, dtype=np.float32)
     recon = np.zeros(n_files, dtype=np.float32)
     double_ext = np.zeros(n_files, dtype=np.float32)
-    np.maximum.at(suspicious, nodes, (b & 1).astype(np.float32))
+    # unlisted socket destinations share the suspicious channel (no GPU
+    # kernel signature change; the flag is host-computed either way)
+    np.maximum.at(suspicious, nodes, (((b & 1) | ((b >> 5) & 1)) != 0).astype(np.float32))
     np.maximum.at(note, nodes, ((b >> 1) & 1).astype(np.float32))
     np.maximum.at(recon, nodes, ((b >> 2) & 1).astype(np.float32))
     np.maximum.at(double_ext, nodes, ((b >> 3) & 1).astype(np.float32))
     return suspicious, note, recon, double_ext
+
+
+def file_node_kinds(paths, path_root: np.ndarray, touched_roots: np.ndarray) -> np.ndarray:
+    """Kind per path-domain node: 1 = file, 2 = socket destination.
+
+    Spec: nodes = processes/files/sockets (reference architecture.mdx:36-43).
+    Socket destinations live in the path string domain ("tcp://host:port"),
+    so they ride the same interning/rename/feature machinery; the kind
+    one-hot (is_process, is_file) naturally encodes sockets as (0, 0) in
+    both the CPU and GPU feature paths.
+    """
+    bits = _string_flag_bits(paths)
+    is_sock = (bits[touched_roots] & 16) != 0 if len(touched_roots) else np.zeros(0, bool)
+    return np.where(is_sock, np.int8(2), np.int8(1)).astype(np.int8)
 
 
 def build_graph_parts(
@@ -410,7 +437,8 @@ def build_graph(
     # ---- assemble feature matrix ------------------------------------------
     x = np.zeros((n_nodes, NUM_NODE_FEATURES), dtype=np.float32)
     node_kind = np.concatenate(
-        [np.ones(n_files, dtype=np.int8), np.zeros(n_procs, dtype=np.int8)]
+        [file_node_kinds(events.paths, path_root, touched_roots),
+         np.zeros(n_procs, dtype=np.int8)]
     )
     dur = np.maximum(t_last - t_first, 0.0)
     x[:, 0] = node_kind == 0  # is_process

@@ -20,7 +20,7 @@ from typing import Optional
 import numpy as np
 import torch
 
-from .constructor import build_edges_and_flags, build_graph_parts
+from .constructor import build_edges_and_flags, build_graph_parts, file_node_kinds
 from ..data.trace import EventArray
 
 
@@ -53,7 +53,8 @@ def gpu_window_graph(
     t0, span = parts["t0"], parts["span"]
 
     node_kind = np.concatenate(
-        [np.ones(n_files, dtype=np.int8), np.zeros(n_procs, dtype=np.int8)]
+        [file_node_kinds(events.paths, parts["path_root"], parts["touched_roots"]),
+         np.zeros(n_procs, dtype=np.int8)]
     )
     flags = (
         ed["suspicious"].astype(np.uint8)

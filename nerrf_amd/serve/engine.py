@@ -41,6 +41,7 @@ class Detection:
     file_mb: Dict[str, float]
     proc_scores: Dict[int, float]
     encrypted_paths: List[str] = field(default_factory=list)
+    exfil_destinations: List[str] = field(default_factory=list)
     indicators: Dict[str, float] = field(default_factory=dict)
     window_events: int = 0
 
@@ -142,8 +143,11 @@ class StreamingEngine:
         else:
             parts = build_graph_parts(events)
             ed = build_edges_and_flags(parts)
+        from ..graph.constructor import file_node_kinds
+
         node_kind = _np.concatenate(
-            [_np.ones(parts["n_files"], dtype=_np.int8), _np.zeros(parts["n_procs"], dtype=_np.int8)]
+            [file_node_kinds(events.paths, parts["path_root"], parts["touched_roots"]),
+             _np.zeros(parts["n_procs"], dtype=_np.int8)]
         )
         node_key = _np.concatenate(
             [parts["touched_roots"], parts["upids"].astype(_np.int64)]
@@ -241,14 +245,19 @@ class StreamingEngine:
         bits = _string_flag_bits(events.paths)
         encrypted_paths = [strings[i] for i in _np.nonzero(bits & 1)[0].tolist()]
         note = bool((bits & 2).any())
+        # socket egress to destinations outside the allowlist (the policy
+        # channel syscall data alone cannot provide — threat-model.md)
+        exfil_dests = [strings[i] for i in _np.nonzero(bits & 32)[0].tolist()]
         indicators = {
             "write_to_rename": w2r,
             "suspicious_ext_count": float(len(encrypted_paths)),
             "ransom_note": float(note),
+            "exfil_dest_count": float(len(exfil_dests)),
         }
         ind_score = min(
             1.0,
-            0.6 * float(len(encrypted_paths) > 0) + 0.3 * float(note) + 0.4 * float(w2r > 0.1),
+            0.6 * float(len(encrypted_paths) > 0) + 0.3 * float(note)
+            + 0.4 * float(w2r > 0.1) + 0.7 * float(len(exfil_dests) > 0),
         )
         # boost file scores for files with suspicious aliases
         for p in encrypted_paths:
@@ -270,6 +279,7 @@ class StreamingEngine:
             file_mb=file_mb,
             proc_scores=proc_scores,
             encrypted_paths=encrypted_paths,
+            exfil_destinations=exfil_dests,
             indicators=indicators,
             window_events=len(events),
         )
@@ -330,6 +340,7 @@ class StreamingEngine:
             "file_mb": det.file_mb,
             "proc_scores": det.proc_scores,
             "encrypted_paths": det.encrypted_paths,
+            "exfil_destinations": det.exfil_destinations,
             "indicators": det.indicators,
             "window_events": det.window_events,
         }
@@ -351,6 +362,9 @@ class StreamingEngine:
             for p, s in g["proc_scores"].items():
                 merged.proc_scores[p] = max(merged.proc_scores.get(p, 0.0), s)
             enc.extend(g["encrypted_paths"])
+            merged.exfil_destinations = sorted(
+                set(merged.exfil_destinations) | set(g.get("exfil_destinations", []))
+            )
             for k, v in g["indicators"].items():
                 merged.indicators[k] = max(merged.indicators.get(k, 0.0), v)
             merged.window_events += g["window_events"]

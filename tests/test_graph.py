@@ -354,7 +354,7 @@ def test_append_array_unsorted_and_eviction():
     assert all(ev.paths.lookup(int(i)).startswith("/u/f") for i in ev.path_id)
 
 
-@pytest.mark.parametrize("kind", ["supply_chain", "benign_rotate", "benign_backup", "benign_build"])
+@pytest.mark.parametrize("kind", ["supply_chain", "supply_chain_net", "benign_rotate", "benign_backup", "benign_build"])
 def test_incremental_merge_all_scenario_kinds(kind):
     """Incremental == full across every synthetic scenario family."""
     import numpy as np

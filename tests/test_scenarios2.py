@@ -121,3 +121,60 @@ def test_fp_undo_zero_on_hard_negatives(tmp_path):
         res = engine.respond(det, plan, str(victim))
         assert res.files_restored == 0
         assert (victim / "a.dat").read_bytes() == b"x" * 64  # untouched
+
+
+def test_socket_nodes_and_exfil_destination_alarm():
+    """supply_chain_net: the exfil socket becomes a kind-2 node, the
+    destination-allowlist indicator fires, and the engine alarms even
+    without a trained model (defence in depth)."""
+    import numpy as np
+
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.graph.constructor import build_graph
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    arr, w = generate(SynthConfig(kind="supply_chain_net", duration_s=20.0,
+                                  benign_rate_hz=120.0, seed=5))
+    g = build_graph(arr, window=w)
+    sock = np.nonzero(g.node_kind == 2)[0]
+    assert len(sock) == 1
+    assert g.x[sock, 13] == 1.0  # unlisted destination -> suspicious channel
+    assert g.x[sock, 0] == 0.0 and g.x[sock, 1] == 0.0  # one-hot: neither
+
+    eng = StreamingEngine(device="cpu")
+    eng.store.window_s = 1e9
+    eng.ingest_events(arr)
+    det = eng.score_window()
+    assert det.indicators["exfil_dest_count"] == 1.0
+    assert det.exfil_destinations == ["tcp://203.0.113.37:443"]
+    assert det.alarm
+
+
+def test_allowlisted_destination_not_flagged():
+    """Egress to an allowlisted destination (e.g. the backup server) does
+    not set the suspicious channel or the exfil indicator."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+    from nerrf_amd.graph.constructor import build_graph
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    b = EventArrayBuilder(StringTable(), StringTable())
+    t = 0.0
+    for i in range(40):
+        b.add(ts=t, pid=42, syscall="read", path=f"/srv/data/f{i}.db", nbytes=4096)
+        t += 0.01
+        b.add(ts=t, pid=42, syscall="sendto", path="tcp://10.0.0.9:873", nbytes=4096)
+        t += 0.01
+    arr = b.build()
+    g = build_graph(arr)
+    sock = np.nonzero(g.node_kind == 2)[0]
+    assert len(sock) == 1
+    assert g.x[sock, 13] == 0.0  # allowlisted -> clean
+
+    eng = StreamingEngine(device="cpu")
+    eng.store.window_s = 1e9
+    eng.ingest_events(arr)
+    det = eng.score_window()
+    assert det.indicators["exfil_dest_count"] == 0.0
+    assert det.exfil_destinations == []
