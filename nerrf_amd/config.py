@@ -111,7 +111,16 @@ def _apply_override(cfg: Any, dotted: str, raw: str) -> None:
         val = yaml.safe_load(raw)
     except yaml.YAMLError:
         pass
-    if cur is not None and val is not None and not isinstance(val, type(cur)):
+    if isinstance(cur, (tuple, list)) and val is not None:
+        # sequence-typed leaf: accept "(a,b,c)", "a,b,c" or YAML "[a, b]"
+        # (a bare type(cur)(str) would split the string into characters)
+        if isinstance(val, str):
+            val = type(cur)(
+                x.strip() for x in val.strip("()[]").split(",") if x.strip()
+            )
+        else:
+            val = type(cur)(val if isinstance(val, (list, tuple)) else [val])
+    elif cur is not None and val is not None and not isinstance(val, type(cur)):
         try:
             val = type(cur)(val)
         except (TypeError, ValueError):

@@ -123,3 +123,14 @@ def test_build_edges_and_flags_matches_build_graph():
     assert np.allclose(ed["edge_weight"], g.edge_weight)
     assert np.allclose(np.log1p(ed["in_deg"]), g.x[:, 2])
     assert np.allclose(ed["suspicious"], g.x[:, 13])
+
+
+def test_config_sequence_override_forms():
+    """Sequence-typed overrides parse as item tuples, never char-splits."""
+    for ov in (
+        "data.scenario_kinds=(lockbit,supply_chain)",
+        "data.scenario_kinds=lockbit,supply_chain",
+        "data.scenario_kinds=[lockbit, supply_chain]",
+    ):
+        cfg = load_config(None, [ov])
+        assert cfg.data.scenario_kinds == ("lockbit", "supply_chain"), ov

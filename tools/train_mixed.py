@@ -9,19 +9,24 @@ from nerrf_amd.serve.engine import load_model_from_checkpoint
 cfg = load_config(None, [
     "optim.dtype=bfloat16", "optim.epochs=4", "data.n_scenarios=20",
     "data.attack_fraction=0.5",
+    # round-2 mix: add the net-exfil supply-chain variant; benign hard
+    # negatives (rotate/backup/build) come from dataset defaults.
+    "data.scenario_kinds=(lockbit,supply_chain,supply_chain_net)",
     "run.eval_holdout=4", "run.checkpoint_dir=gpurun_out/ckpt_mixed",
     "run.log_every=100",
 ])
+# NOTE (round-1 CPU finding): watch holdout seq F1 per epoch — the sequence
+# head overfit by epoch 3 on a smaller CPU run; keep the best-epoch ckpt.
 run_training(cfg)
 model = load_model_from_checkpoint("gpurun_out/ckpt_mixed").to("cuda", torch.bfloat16)
-for kind in ("lockbit", "supply_chain"):
+for kind in ("lockbit", "supply_chain", "supply_chain_net"):
     hb = synth_window_batches(n_scenarios=3, attack_fraction=0.67, base_seed=555000,
                               kinds=(kind,))
     rep = evaluate(model, hb, "cuda", torch.bfloat16)
     print(f"family={kind}: " + json.dumps({k: round(float(v), 4) for k, v in rep.items() if "auc" in k or k.endswith("f1")}))
 # hard negatives: max anomaly score the model assigns on clean lookalikes
 import numpy as np
-for kind in ("benign_rotate", "benign_backup"):
+for kind in ("benign_rotate", "benign_backup", "benign_build"):
     hb = synth_window_batches(n_scenarios=2, attack_fraction=0.0, base_seed=777000,
                               benign_kinds=(kind,))
     mx = []
