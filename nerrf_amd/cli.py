@@ -249,7 +249,7 @@ def main(argv=None) -> int:
 
     p = sub.add_parser("eval", help="detection-quality report on synthetic scenarios")
     p.add_argument("--checkpoint", default="checkpoints/pretrained")
-    p.add_argument("--families", default="lockbit,supply_chain")
+    p.add_argument("--families", default="lockbit,supply_chain,supply_chain_net")
     p.add_argument("--negatives", default="benign_rotate,benign_backup,benign_build")
     p.add_argument("--scenarios", type=int, default=3)
     p.add_argument("--device", default="cpu")
