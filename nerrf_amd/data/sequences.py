@@ -90,6 +90,8 @@ def build_sequences(
     # assignment iterates element-wise and was ~6x slower)
     base = (seq_of * seq_len + pos) * NUM_SEQ_FEATURES
     sc = events.syscall[ev].astype(np.int64)
+    # one-hot channels 0..9; network ids (connect/sendto, 10+) share channel
+    # 9 until the per-event feature map widens (same clip in the torch path)
     flatf[base + np.clip(sc, 0, 9)] = 1.0
     flatf[base + 10] = np.log1p(events.nbytes[ev]) / 16.0
     ts = events.ts[ev]
@@ -174,7 +176,7 @@ def build_sequences_torch(
 
     feats = torch.zeros(b * seq_len * NUM_SEQ_FEATURES, device=dev, dtype=torch.float32)
     base = (seq_of * seq_len + pos) * NUM_SEQ_FEATURES
-    feats[base + t_sc[ev].long().clamp(0, 9)] = 1.0
+    feats[base + t_sc[ev].long().clamp(0, 9)] = 1.0  # same clip as numpy path
     feats[base + 10] = torch.log1p(t_nb[ev]) / 16.0
     ts_ev = t_ts[ev]
     dt = ts_ev - torch.where(pos > 0, t_ts[prev], ts_ev)
