@@ -186,3 +186,26 @@ def test_full_pipeline_daemon_bridge_engine(tmp_path):
             bridge.stop()
         proc.terminate()
         proc.wait(timeout=5)
+
+
+def test_native_decoder_rejects_crafted_length_overflow():
+    """A WT_LEN varint of ~2^64 must raise cleanly (no wrap-around, no
+    hang): the decoder parses frames from network clients."""
+    import pytest
+
+    ingest = pytest.importorskip("nerrf_amd._ingest")
+    dec = ingest.ColumnarDecoder()
+    # EventBatch field 1 (event), WT_LEN, length = 2^64 - 1
+    evil = bytes([0x0A]) + b"\xff\xff\xff\xff\xff\xff\xff\xff\xff\x01"
+    with pytest.raises(Exception):
+        dec.decode([evil])
+    # nested: valid batch envelope, event payload with a huge string length
+    inner = bytes([0x32]) + b"\xff\xff\xff\xff\xff\xff\xff\xff\xff\x01"  # field 6 path
+    frame = bytes([0x0A, len(inner)]) + inner
+    with pytest.raises(Exception):
+        dec.decode([frame])
+    # and a huge skip length on an unknown field
+    inner2 = bytes([0xFA, 0x01]) + b"\xff\xff\xff\xff\xff\xff\xff\xff\xff\x01"  # field 31 WT_LEN
+    frame2 = bytes([0x0A, len(inner2)]) + inner2
+    with pytest.raises(Exception):
+        dec.decode([frame2])

@@ -101,14 +101,18 @@ inline void skip_field(const uint8_t* buf, size_t len, size_t& pos, int wt) {
       read_varint(buf, len, pos);
       break;
     case WT_I64:
+      if (len - pos < 8) throw std::runtime_error("truncated i64");
       pos += 8;
       break;
     case WT_LEN: {
       uint64_t n = read_varint(buf, len, pos);
+      // overflow-safe bound: a crafted 64-bit length must not wrap pos
+      if (n > len - pos) throw std::runtime_error("truncated skip");
       pos += n;
       break;
     }
     case WT_I32:
+      if (len - pos < 4) throw std::runtime_error("truncated i32");
       pos += 4;
       break;
     default:
@@ -184,7 +188,7 @@ inline Event decode_event(const uint8_t* buf, size_t len) {
     int field = int(key >> 3), wt = int(key & 7);
     if (wt == WT_LEN) {
       uint64_t n = read_varint(buf, len, pos);
-      if (pos + n > len) throw std::runtime_error("truncated field");
+      if (n > len - pos) throw std::runtime_error("truncated field");
       const char* p = reinterpret_cast<const char*>(buf + pos);
       switch (field) {
         case 1:
@@ -238,7 +242,7 @@ inline std::vector<Event> decode_event_batch(const uint8_t* buf, size_t len) {
     int field = int(key >> 3), wt = int(key & 7);
     if (field == 1 && wt == WT_LEN) {
       uint64_t n = read_varint(buf, len, pos);
-      if (pos + n > len) throw std::runtime_error("truncated event");
+      if (n > len - pos) throw std::runtime_error("truncated event");
       events.push_back(decode_event(buf + pos, n));
       pos += n;
     } else {
