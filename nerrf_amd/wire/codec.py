@@ -90,11 +90,17 @@ def _skip_field(buf: bytes, pos: int, wire_type: int) -> int:
     if wire_type == _VARINT:
         _, pos = _read_varint(buf, pos)
     elif wire_type == _I64:
+        if len(buf) - pos < 8:
+            raise ValueError("truncated i64")
         pos += 8
     elif wire_type == _LEN:
         n, pos = _read_varint(buf, pos)
+        if n > len(buf) - pos:
+            raise ValueError("truncated skip")
         pos += n
     elif wire_type == _I32:
+        if len(buf) - pos < 4:
+            raise ValueError("truncated i32")
         pos += 4
     else:
         raise ValueError(f"unsupported wire type {wire_type}")
@@ -191,6 +197,8 @@ def decode_event(buf: bytes) -> Event:
         fnum, wtype = key >> 3, key & 7
         if wtype == _LEN:
             ln, pos = _read_varint(buf, pos)
+            if ln > n - pos:
+                raise ValueError("truncated length-delimited field")
             payload = buf[pos : pos + ln]
             pos += ln
             if fnum == 1:
@@ -247,6 +255,8 @@ def decode_event_batch(buf: bytes) -> List[Event]:
         fnum, wtype = key >> 3, key & 7
         if fnum == 1 and wtype == _LEN:
             ln, pos = _read_varint(buf, pos)
+            if ln > n - pos:
+                raise ValueError("truncated event")
             events.append(decode_event(buf[pos : pos + ln]))
             pos += ln
         else:

@@ -166,3 +166,15 @@ def test_codec_large_values():
     ev = codec.Event(pid=2**32 - 1, bytes=2**63 - 1, ret_val=-(2**62), uid=2**40)
     back = codec.decode_event(codec.encode_event(ev))
     assert back == ev
+
+
+def test_truncated_length_fields_raise():
+    """Crafted/truncated length-delimited fields raise instead of silently
+    decoding short payloads (matches the C++ decoder's behavior)."""
+    evil = bytes([0x0A]) + b"\xff\xff\xff\xff\xff\xff\xff\xff\xff\x01"
+    with pytest.raises(ValueError):
+        codec.decode_event_batch(evil)
+    inner = bytes([0x32, 0x20]) + b"hi"  # field 6 claims 32 bytes, has 2
+    frame = bytes([0x0A, len(inner)]) + inner
+    with pytest.raises(ValueError):
+        codec.decode_event_batch(frame)
