@@ -178,3 +178,46 @@ def test_truncated_length_fields_raise():
     frame = bytes([0x0A, len(inner)]) + inner
     with pytest.raises(ValueError):
         codec.decode_event_batch(frame)
+
+
+def test_event_roundtrip_property():
+    """Property: arbitrary Events roundtrip byte-exactly through the
+    hand-written codec, and the native columnar decoder agrees."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    text = st.text(max_size=40)
+    u32 = st.integers(min_value=0, max_value=2**32 - 1)
+    u64 = st.integers(min_value=0, max_value=2**63 - 1)
+
+    @settings(max_examples=60, deadline=None)
+    @given(
+        sec=st.integers(min_value=0, max_value=2**40),
+        nsec=st.integers(min_value=0, max_value=999_999_999),
+        pid=u32, tid=u32, comm=text, syscall=text, path=text,
+        new_path=text, ret_val=st.integers(min_value=-2**40, max_value=2**40),
+        nbytes=u64,
+    )
+    def check(sec, nsec, pid, tid, comm, syscall, path, new_path, ret_val, nbytes):
+        ev = codec.Event(ts_sec=sec, ts_nsec=nsec, pid=pid, tid=tid,
+                         comm=comm, syscall=syscall, path=path,
+                         new_path=new_path, ret_val=ret_val, bytes=nbytes)
+        frame = codec.encode_event_batch([ev])
+        back = codec.decode_event_batch(frame)
+        assert len(back) == 1
+        b = back[0]
+        assert (b.ts_sec, b.ts_nsec, b.pid, b.tid) == (sec, nsec, pid, tid)
+        assert (b.comm, b.syscall, b.path, b.new_path) == (comm, syscall, path, new_path)
+        assert (b.ret_val, b.bytes) == (ret_val, nbytes)
+        try:
+            from nerrf_amd import _ingest
+        except ImportError:
+            return
+        dec = _ingest.ColumnarDecoder()
+        ts, pids, sysc, path_id, newp_id, nb, ret, _ = dec.decode([frame])
+        assert len(ts) == 1
+        assert int(pids[0]) == pid
+        assert int(nb[0]) == nbytes
+        assert int(ret[0]) == ret_val
+
+    check()
