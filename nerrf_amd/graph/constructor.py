@@ -2,9 +2,10 @@
 
 Behavioral spec (reference docs architecture.mdx:36-43, threat-model.mdx:179-188):
   * sliding window of 30-60 s over the event stream,
-  * nodes = processes and files, files deduped across renames (the upstream
-    spec dedups by inode; traces without inodes get rename-union instead,
-    which is the same equivalence for the attack pattern),
+  * nodes = processes, files and socket destinations (files deduped across
+    renames — the upstream spec dedups by inode; traces without inodes get
+    rename-union instead, which is the same equivalence for the attack
+    pattern; sockets enter the path domain as "tcp://host:port"),
   * edge weight = causality confidence (recency-decayed interaction count),
   * node features: in/out degree, temporal deltas, byte-count ratios,
     extension-pattern confidence, read/write/rename counters, plus the
@@ -69,7 +70,7 @@ class TemporalGraph:
     edge_index: np.ndarray  # [2, E] int64 (src, dst) in node ids
     edge_weight: np.ndarray  # [E] float32 causality confidence
     edge_ts: np.ndarray  # [E] float32 last-interaction time (window-relative)
-    node_kind: np.ndarray  # [N] int8: 0 = process, 1 = file
+    node_kind: np.ndarray  # [N] int8: 0 = process, 1 = file, 2 = socket
     node_key: np.ndarray  # [N] int64: pid for processes, path root id for files
     y_node: Optional[np.ndarray] = None  # [N] float32 anomaly ground truth
     y_edge: Optional[np.ndarray] = None  # [E] float32
