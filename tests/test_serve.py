@@ -386,3 +386,43 @@ def test_cli_eval_report(capsys):
     rep = json.loads(capsys.readouterr().out)
     assert rep["lockbit"]["node_auc"] >= 0.9
     assert "negative:benign_rotate" in rep
+
+
+def test_store_concurrent_bulk_ingest_and_scoring():
+    """Bulk append_array from one thread while another compacts+scores:
+    no exceptions, consistent snapshots.  (The producer is bounded — an
+    unthrottled Python-loop producer just starves the scorer of the GIL,
+    which is a test artifact, not a store property.)"""
+    import threading
+    import time as _time
+
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    eng = StreamingEngine(device="cpu", window_s=50.0)
+    errors = []
+
+    def writer():
+        t = 0.0
+        try:
+            for _ in range(40):
+                b = EventArrayBuilder(StringTable(), StringTable())
+                for i in range(400):
+                    t += 0.001
+                    b.add(ts=t, pid=7, syscall="write", path=f"/c/f{i % 60}", nbytes=32)
+                eng.store.append_array(b.build(sort=False))
+                _time.sleep(0.005)
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    th = threading.Thread(target=writer)
+    th.start()
+    try:
+        for _ in range(6):
+            det = eng.score_window()
+            assert det.window_events >= 0
+    finally:
+        th.join(timeout=30)
+    assert not errors
+    det = eng.score_window()
+    assert det.window_events == 40 * 400
