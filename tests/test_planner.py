@@ -86,3 +86,37 @@ def test_build_state_buckets():
     assert st.group_mb.sum() == pytest.approx(64.0)
     # score-ordered buckets: group 0 hottest
     assert st.group_score[0] > st.group_score[-1]
+
+
+def test_reward_kill_stops_ongoing_encryption():
+    """While the attack process lives, remaining clean data keeps getting
+    encrypted at attack_rate; an early kill caps the loss."""
+    import numpy as np
+
+    from nerrf_amd.planner.rewards import (
+        A_KILL, A_STOP, PlannerParams, PlannerState, simulate_plan,
+    )
+
+    st = PlannerState(
+        group_score=np.array([0.9]), group_mb=np.array([10.0]),
+        group_files=np.array([5.0]), proc_score=0.95,
+        remaining_clean_mb=100.0,
+    )
+    p = PlannerParams(n_groups=1)
+    r_idle = simulate_plan(st, [A_STOP], p)
+    r_kill = simulate_plan(st, [A_KILL, A_STOP], p)
+    assert r_kill > r_idle  # killing beats watching the encryption continue
+
+
+def test_empty_state_plans_nothing():
+    import numpy as np
+
+    from nerrf_amd.planner.mcts import run_mcts
+    from nerrf_amd.planner.rewards import PlannerState
+
+    st = PlannerState(
+        group_score=np.zeros(4), group_mb=np.zeros(4),
+        group_files=np.zeros(4), proc_score=0.0, remaining_clean_mb=0.0,
+    )
+    res = run_mcts(st, n_sims=128)
+    assert res.plan == []
