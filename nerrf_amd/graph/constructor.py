@@ -44,6 +44,45 @@ _SOCKET_PREFIXES = ("tcp://", "udp://")
 DEST_ALLOWLIST = ("tcp://10.", "tcp://192.168.", "tcp://backup.", "udp://10.")
 
 
+def aggregate_sparse_keys(
+    key: np.ndarray, wsum: np.ndarray, ts_v: np.ndarray, key_space: int
+) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Group-by over integer keys -> (unique_keys, sum(wsum), max(ts_v)).
+
+    Dense bincount when the key space is small (three O(n) passes); LSD
+    radix sort otherwise so a window with many procs*files never allocates
+    key_space-sized arrays (multi-GB at ~2k procs x ~4M paths).
+    """
+    if not key.size:
+        e = np.empty(0, np.int64)
+        return e, np.empty(0, np.float64), np.empty(0, np.float64)
+    if key_space < (1 << 24):
+        cnt = np.bincount(key, minlength=key_space)
+        uk = np.nonzero(cnt)[0]
+        agg_w = np.bincount(key, weights=wsum, minlength=key_space)[uk]
+        last_all = np.full(key_space, -np.inf)
+        np.maximum.at(last_all, key, ts_v)
+        return uk, agg_w, last_all[uk]
+    # LSD radix by 16-bit digits: numpy's stable argsort is radix only for
+    # <= 16-bit dtypes
+    order = np.arange(len(key), dtype=np.int64)
+    shift = 0
+    kmax = int(key.max())
+    while kmax >> shift:
+        digit = ((key >> shift) & 0xFFFF).astype(np.uint16)
+        order = order[np.argsort(digit[order], kind="stable")]
+        shift += 16
+    ks = key[order]
+    new_grp = np.empty(len(ks), dtype=bool)
+    new_grp[0] = True
+    np.not_equal(ks[1:], ks[:-1], out=new_grp[1:])
+    starts = np.nonzero(new_grp)[0]
+    uk = ks[starts]
+    agg_w = np.add.reduceat(wsum[order], starts)
+    agg_t = np.maximum.reduceat(ts_v[order], starts)
+    return uk, agg_w, agg_t
+
+
 class _UnionFind:
     def __init__(self, n: int) -> None:
         self.parent = np.arange(n, dtype=np.int64)
@@ -300,33 +339,10 @@ def build_edges_and_flags(parts: dict, causality_tau_s: float = 10.0) -> dict:
         ts_v = events.ts[valid]
         # causality confidence: recency-decayed count, saturating
         rec = np.exp(-(t1 - ts_v) / causality_tau_s)
-        if key_space < (1 << 24):
-            # sort-free dedup + aggregation: three O(n) bincount-class passes
-            # (np.unique's int64 sort cost ~40 ms per 600k-event window)
-            cnt = np.bincount(key, minlength=key_space)
-            uk = np.nonzero(cnt)[0]
-            e_conf = np.bincount(key, weights=rec, minlength=key_space)[uk]
-            last_all = np.full(key_space, -np.inf)
-            np.maximum.at(last_all, key, ts_v)
-            e_last = last_all[uk]
-        else:
-            # LSD radix by 16-bit digits: numpy's stable argsort is radix
-            # only for <= 16-bit dtypes
-            order = np.arange(len(key), dtype=np.int64)
-            shift = 0
-            kmax = int(key.max())
-            while kmax >> shift:
-                digit = ((key >> shift) & 0xFFFF).astype(np.uint16)
-                order = order[np.argsort(digit[order], kind="stable")]
-                shift += 16
-            ks = key[order]
-            new_grp = np.empty(len(ks), dtype=bool)
-            new_grp[0] = True
-            np.not_equal(ks[1:], ks[:-1], out=new_grp[1:])
-            starts = np.nonzero(new_grp)[0]
-            uk = ks[starts]
-            e_conf = np.add.reduceat(rec[order], starts)
-            e_last = np.maximum.reduceat(ts_v[order], starts)
+        # sort-free dedup + aggregation when the key space is small (three
+        # O(n) bincount-class passes; np.unique's int64 sort cost ~40 ms per
+        # 600k-event window), radix fallback otherwise
+        uk, e_conf, e_last = aggregate_sparse_keys(key, rec, ts_v, key_space)
         dirs = uk % 2
         pf = uk // 2
         e_proc = n_files + (pf // n_files).astype(np.int64)

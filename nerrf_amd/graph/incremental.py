@@ -30,7 +30,7 @@ from typing import Dict, List, Tuple
 import numpy as np
 
 from ..data.trace import SYSCALL_IDS, EventArray
-from .constructor import _READ_LIKE, _path_flags
+from .constructor import _READ_LIKE, _path_flags, aggregate_sparse_keys
 
 _SCALE_GUARD = 60.0  # max (t1 - t_d0)/tau before a partial sum underflows
 
@@ -91,12 +91,10 @@ def summarize_delta(d: EventArray, causality_tau_s: float = 10.0) -> DeltaSummar
         key = (pk * len(ua) + ak) * 2 + dirs
         kspace = 2 * len(up) * len(ua)
         rec = np.exp((ts_v - s.t0) / causality_tau_s)
-        cnt = np.bincount(key, minlength=kspace)
-        uk = np.nonzero(cnt)[0]
-        s.key_wsum = np.bincount(key, weights=rec, minlength=kspace)[uk]
-        last = np.full(kspace, -np.inf)
-        np.maximum.at(last, key, ts_v)
-        s.key_tmax = last[uk]
+        # guarded aggregation (bincount / radix) — same size guard as the
+        # full rebuild, so a delta with many procs*paths never allocates
+        # kspace-sized dense arrays
+        uk, s.key_wsum, s.key_tmax = aggregate_sparse_keys(key, rec, ts_v, kspace)
         kd = uk % 2
         kp = uk // 2
         s.key_pid = up[kp // len(ua)]
@@ -266,12 +264,10 @@ def merge_window(
             tmaxs = np.concatenate(tmax_l)
             key = (procs * n_files + files) * 2 + dirs
             kspace = 2 * n_procs * n_files
-            cnt = np.bincount(key, minlength=kspace)
-            uk = np.nonzero(cnt)[0]
-            e_conf = np.bincount(key, weights=wsums, minlength=kspace)[uk]
-            last = np.full(kspace, -np.inf)
-            np.maximum.at(last, key, tmaxs)
-            e_last = last[uk]
+            # guarded aggregation mirroring the full rebuild's key_space
+            # check: the production run_monitor tick must never
+            # allocate kspace-dense arrays for a wide window
+            uk, e_conf, e_last = aggregate_sparse_keys(key, wsums, tmaxs, kspace)
             kd = uk % 2
             kp = uk // 2
             e_proc = n_files + kp // n_files

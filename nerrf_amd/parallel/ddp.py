@@ -77,6 +77,8 @@ class GradAllReducer:
             self.buckets.append(cur)
         self._bucket_of = {}
         self._ready = [0] * len(self.buckets)
+        self._filled = [False] * len(self.buckets)
+        self._next_launch = 0
         self._flat: List[Optional[torch.Tensor]] = [None] * len(self.buckets)
         self._works: List = []
         if self.enabled:
@@ -89,7 +91,19 @@ class GradAllReducer:
         bi = self._bucket_of[p]
         self._ready[bi] += 1
         if self._ready[bi] == len(self.buckets[bi]):
-            self._launch(bi)
+            self._filled[bi] = True
+            self._drain()
+
+    def _drain(self) -> None:
+        # launch strictly in bucket-index order: with data-dependent heads
+        # (a rank whose window has no sequences skips that head), arrival
+        # order can differ across ranks, and out-of-order collectives on
+        # one communicator hang or mismatch.  A filled bucket
+        # waits until every lower-index bucket has launched; stragglers are
+        # zero-filled and launched in order at finalize().
+        while self._next_launch < len(self.buckets) and self._filled[self._next_launch]:
+            self._launch(self._next_launch)
+            self._next_launch += 1
 
     def _launch(self, bi: int) -> None:
         bucket = self.buckets[bi]
@@ -103,13 +117,15 @@ class GradAllReducer:
         """Wait on all pending reduces and write averaged grads back."""
         if not self.enabled:
             return
-        # buckets whose hooks never all fired (e.g. a head unused this step)
-        for bi, bucket in enumerate(self.buckets):
-            if self._flat[bi] is None:
-                for p in bucket:
+        # buckets whose hooks never all fired (e.g. a head unused this step):
+        # zero-fill and launch in the same fixed index order as _drain
+        for bi in range(self._next_launch, len(self.buckets)):
+            if not self._filled[bi]:
+                for p in self.buckets[bi]:
                     if p.grad is None:
                         p.grad = torch.zeros_like(p)
-                self._launch(bi)
+                self._filled[bi] = True
+            self._drain()
         for work, bi in self._works:
             work.wait()
             flat = self._flat[bi]
@@ -120,6 +136,8 @@ class GradAllReducer:
             self._flat[bi] = None
         self._works.clear()
         self._ready = [0] * len(self.buckets)
+        self._filled = [False] * len(self.buckets)
+        self._next_launch = 0
 
     def broadcast_params(self, model: torch.nn.Module, src: int = 0) -> None:
         if not self.enabled:

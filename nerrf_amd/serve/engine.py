@@ -243,6 +243,18 @@ class StreamingEngine:
         from ..graph.constructor import _string_flag_bits
 
         bits = _string_flag_bits(events.paths)
+        # mask to path ids actually referenced by THIS window: the store's
+        # StringTable is global and grow-only, so unmasked bits would latch
+        # every indicator permanently once a suspicious string is ever
+        # interned (alarm-every-tick after remediation — ADVICE r1)
+        present = _np.zeros(len(bits), dtype=bool)
+        _pv = events.path_id[events.path_id >= 0]
+        _nv = events.new_path_id[events.new_path_id >= 0]
+        if len(_pv):
+            present[_pv] = True
+        if len(_nv):
+            present[_nv] = True
+        bits = _np.where(present, bits, 0)
         encrypted_paths = [strings[i] for i in _np.nonzero(bits & 1)[0].tolist()]
         note = bool((bits & 2).any())
         # socket egress to destinations outside the allowlist (the policy

@@ -157,8 +157,14 @@ def decode_timestamp(buf: bytes) -> Tuple[int, int]:
         fnum, wtype = key >> 3, key & 7
         if fnum == 1 and wtype == _VARINT:
             sec, pos = _read_varint(buf, pos)
+            # protobuf int64: two's-complement interpretation of the varint
+            if sec >= 1 << 63:
+                sec -= 1 << 64
         elif fnum == 2 and wtype == _VARINT:
             nsec, pos = _read_varint(buf, pos)
+            # protobuf int32: truncate to the low 32 bits, signed — matches
+            # the C++ codec and the google runtime on sign-extended negatives
+            nsec = ((nsec & 0xFFFFFFFF) ^ 0x80000000) - 0x80000000
         else:
             pos = _skip_field(buf, pos, wtype)
     return sec, nsec

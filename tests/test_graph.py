@@ -379,3 +379,44 @@ def test_incremental_merge_all_scenario_kinds(kind):
     assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-5), kind
     for k in ("suspicious", "note", "recon", "double_ext", "in_deg", "out_deg"):
         assert np.array_equal(ed[k], ref_ed[k]), (kind, k)
+
+
+def test_incremental_merge_radix_guard_wide_window():
+    """merge_window on a window whose key space exceeds 1<<24 must take the
+    guarded radix path (no kspace-dense allocations on the
+    production tick) and still match the full rebuild."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+
+    rng = np.random.default_rng(17)
+    paths_tab, comm_tab = StringTable(), StringTable()
+    deltas = []
+    t = 0.0
+    for _ in range(4):  # 4 sealed deltas sharing the global string tables
+        b = EventArrayBuilder(paths_tab, comm_tab)
+        for _ in range(6000):
+            p = int(rng.integers(0, 3000))
+            f = int(rng.integers(0, 3000))
+            b.add(ts=t, pid=1000 + p,
+                  syscall="write" if rng.random() < 0.7 else "read",
+                  path=f"/data/f{f}", nbytes=64)
+            t += 1e-4
+        deltas.append(b.build())
+
+    # window = concatenation of the deltas
+    from nerrf_amd.data.trace import concat
+
+    ev = concat(deltas)
+    parts_ref = build_graph_parts(ev)
+    assert 2 * parts_ref["n_procs"] * parts_ref["n_files"] >= (1 << 24)
+    ed_ref = build_edges_and_flags(parts_ref)
+
+    parts, ed = merge_window(ev, IncrementalWindowState().summaries(deltas))
+    assert parts["n_files"] == parts_ref["n_files"]
+    assert parts["n_procs"] == parts_ref["n_procs"]
+    assert np.array_equal(ed["edge_index"], ed_ref["edge_index"])
+    assert np.allclose(ed["edge_weight"], ed_ref["edge_weight"], atol=1e-5)
+    assert np.allclose(ed["edge_ts"], ed_ref["edge_ts"], atol=1e-6)

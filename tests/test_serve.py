@@ -426,3 +426,28 @@ def test_store_concurrent_bulk_ingest_and_scoring():
     assert not errors
     det = eng.score_window()
     assert det.window_events == 40 * 400
+
+
+def test_indicators_do_not_latch_after_window_eviction():
+    """The string table is global and grow-only; once an attack's paths are
+    interned, later clean windows must NOT keep raising the path-pattern
+    indicators (flag bits have to be masked to the paths the
+    current window actually references)."""
+    attack, _ = generate(SynthConfig(seed=2, duration_s=40, benign_rate_hz=40, n_victim_files=8))
+    benign, _ = generate(SynthConfig(seed=9, duration_s=30, benign_rate_hz=50, attack=False))
+    benign.ts = benign.ts + 1000.0  # long after the attack window
+
+    engine = _small_engine(device="cpu", window_s=30.0)
+    engine.ingest_events(attack)
+    det_attack = engine.score_window(now=float(attack.ts.max()))
+    assert det_attack.indicators["suspicious_ext_count"] > 0
+
+    engine.ingest_events(benign)
+    det_clean = engine.score_window(now=float(benign.ts.max()))
+    # attack strings are still interned in the global table, but the clean
+    # window references none of them
+    assert det_clean.indicators["suspicious_ext_count"] == 0
+    assert det_clean.indicators["ransom_note"] == 0
+    assert det_clean.indicators["exfil_dest_count"] == 0
+    assert det_clean.encrypted_paths == []
+    assert not det_clean.alarm

@@ -221,3 +221,17 @@ def test_event_roundtrip_property():
         assert int(ret[0]) == ret_val
 
     check()
+
+
+def test_timestamp_negative_nanos_roundtrip():
+    """Negative nanos (invalid in a well-formed Timestamp but representable):
+    both codecs sign-extend to 64-bit on encode and truncate to int32 on
+    decode, matching the google runtime (ADVICE r1 low finding)."""
+    enc = codec.encode_timestamp(5, -7)
+    sec, nsec = codec.decode_timestamp(enc)
+    assert (sec, nsec) == (5, -7)
+    # sign-extension => 10-byte varint for the nanos payload
+    assert len(enc) > 6
+    # negative seconds too (protobuf int64 two's complement)
+    sec, nsec = codec.decode_timestamp(codec.encode_timestamp(-3, 1))
+    assert (sec, nsec) == (-3, 1)

@@ -132,7 +132,10 @@ inline std::string encode_timestamp(int64_t sec, int32_t nsec) {
   }
   if (nsec) {
     write_tag(out, 2, WT_VARINT);
-    write_varint(out, uint64_t(uint32_t(nsec)));
+    // sign-extend to 64 bits like protobuf/the Python codec do: negative
+    // nanos (invalid in a well-formed Timestamp but representable) must
+    // round-trip identically in both codecs, not truncate to 32 bits
+    write_varint(out, uint64_t(int64_t(nsec)));
   }
   return out;
 }
