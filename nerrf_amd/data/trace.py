@@ -15,6 +15,7 @@ from __future__ import annotations
 
 import csv
 import json
+import re
 from dataclasses import dataclass
 from pathlib import Path
 from typing import Dict, Iterable, List, Optional, Sequence
@@ -44,6 +45,10 @@ SYSCALL_NAMES = {v: k for k, v in SYSCALL_IDS.items()}
 UNKNOWN_SYSCALL = 0
 
 # Event names used by the upstream jsonl artifacts -> canonical syscall.
+# The second block is the recorded m0/m1 simulator vocabulary
+# (reference sim_lockbit_m1.py TRACE: records, present verbatim in
+# /root/reference/benchmarks/m{0,1}/results/*_trace.jsonl): seed/encrypt
+# phases are write flows, the recon enumerations are exec-like.
 _JSONL_EVENT_MAP = {
     "open": "openat",
     "openat": "openat",
@@ -57,7 +62,45 @@ _JSONL_EVENT_MAP = {
     "recon": "exec",
     "exec": "exec",
     "chmod": "chmod",
+    # recorded-artifact vocabulary
+    "file_created": "write",
+    "file_encrypt_start": "write",
+    "file_encrypt_complete": "rename",  # path is the NEW .lockbit3 name
+    "ransom_note_created": "write",
+    "process_enum": "exec",
+    "network_enum": "exec",
+    "user_enum": "exec",
+    "disk_enum": "exec",
+    "mount_enum": "exec",
+    "lateral_movement_start": "exec",
+    "lateral_movement_complete": "exec",
 }
+
+# the recorded simulator renames <stem>.dat -> <stem>.lockbit3 and logs only
+# the new name on file_encrypt_complete (sim_lockbit_m1.py encrypt loop);
+# reconstruct the rename pair so rename-union and the extension indicators
+# see the same graph a live tracker would have produced
+_ENCRYPTED_SUFFIX = re.compile(r"\.(lockbit\w*|encrypted|locked|crypt\w*)$", re.IGNORECASE)
+
+
+def _coerce_ts(v) -> float:
+    """Accept epoch floats or ISO-8601 strings (the recorded artifacts use
+    naive-UTC ISO, e.g. '2025-08-30T14:07:06.542871')."""
+    if isinstance(v, (int, float)):
+        return float(v)
+    s = str(v).strip()
+    try:
+        return float(s)
+    except ValueError:
+        pass
+    from datetime import datetime, timezone
+
+    if s.endswith("Z"):
+        s = s[:-1]
+    dt = datetime.fromisoformat(s)
+    if dt.tzinfo is None:
+        dt = dt.replace(tzinfo=timezone.utc)
+    return dt.timestamp()
 
 
 class StringTable:
@@ -225,12 +268,21 @@ def load_jsonl(path: str | Path) -> EventArray:
                 rec = json.loads(line)
             except json.JSONDecodeError:
                 continue  # tolerate truncated/corrupt lines in field traces
+            name = str(rec.get("event", ""))
+            path = str(rec.get("path", ""))
+            new_path = str(rec.get("new_path", ""))
+            if name == "file_encrypt_complete" and not new_path:
+                m = _ENCRYPTED_SUFFIX.search(path)
+                if m:
+                    # reconstruct the logged-new-name-only rename pair
+                    new_path = path
+                    path = path[: m.start()] + ".dat"
             builder.add(
-                ts=float(rec.get("timestamp", 0.0)),
+                ts=_coerce_ts(rec.get("timestamp", 0.0)),
                 pid=int(rec.get("pid", 0)),
-                syscall=_normalise_event_name(str(rec.get("event", ""))),
-                path=str(rec.get("path", "")),
-                new_path=str(rec.get("new_path", "")),
+                syscall=_normalise_event_name(name),
+                path=path,
+                new_path=new_path,
                 nbytes=int(rec.get("size", rec.get("bytes", 0)) or 0),
             )
     return builder.build()
@@ -242,7 +294,7 @@ def load_csv(path: str | Path) -> EventArray:
     with open(path, "r", encoding="utf-8", newline="") as fh:
         for rec in csv.DictReader(fh):
             builder.add(
-                ts=float(rec.get("timestamp", 0.0) or 0.0),
+                ts=_coerce_ts(rec.get("timestamp", 0.0) or 0.0),
                 pid=int(rec.get("pid", 0) or 0),
                 syscall=_normalise_event_name(str(rec.get("event", ""))),
                 path=str(rec.get("path", "")),
