@@ -31,6 +31,12 @@ void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
 void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
                             const void*, const float*, void*, void*, void*,
                             int, bool, hipStream_t);
+void launch_lstm_rec_fwd(const void*, const void*, const void*, const void*,
+                         const void*, const float*, void*, void*, void*, int,
+                         long, long, long, hipStream_t);
+void launch_lstm_rec_bwd(const void*, const void*, const void*, const void*,
+                         const void*, const void*, const float*, void*, void*,
+                         void*, int, long, hipStream_t);
 void launch_event_scatter(const long*, const long*, const signed char*,
                           const float*, const int*, float*, int*, int*, long,
                           hipStream_t);
@@ -193,6 +199,86 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
       c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
       grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
       gout_stride, is_bf16(grad_h), stream.stream());
+}
+
+// Fused recurrent step (bf16, H == 256): h_prev @ W_hh^T + LSTM pointwise
+// in one launch — the [B, 4H] pre-activation slab never touches HBM.
+void lstm_rec_fwd(torch::Tensor h_prev, torch::Tensor w_hh, torch::Tensor xg,
+                  torch::Tensor bias, torch::Tensor c_prev, torch::Tensor mask,
+                  torch::Tensor h_out, torch::Tensor c_out,
+                  torch::Tensor gates_act) {
+  const long hprev_stride = row_stride_checked(h_prev, "h_prev");
+  const long xg_stride = row_stride_checked(xg, "xg");
+  const long hout_stride = row_stride_checked(h_out, "h_out");
+  check_gpu_contig(w_hh, "w_hh");
+  check_gpu_contig(c_prev, "c_prev");
+  check_gpu_contig(c_out, "c_out");
+  const bool want_gates = gates_act.numel() > 0;
+  if (want_gates) check_gpu_contig(gates_act, "gates_act");
+  for (auto* t : {&h_prev, &w_hh, &xg, &c_prev, &h_out, &c_out}) {
+    TORCH_CHECK(t->scalar_type() == torch::kBFloat16,
+                "lstm_rec_fwd is bf16-only");
+  }
+  const long batch = c_prev.size(0);
+  TORCH_CHECK(c_prev.size(1) == 256 && h_prev.size(1) == 256 &&
+                  xg.size(1) == 1024 && w_hh.size(0) == 1024 &&
+                  w_hh.size(1) == 256,
+              "lstm_rec_fwd requires H=256");
+  const float* mask_ptr = nullptr;
+  torch::Tensor mf;
+  if (mask.numel() > 0) {
+    mf = mask.scalar_type() == torch::kFloat32 ? mask.contiguous()
+                                               : mask.to(torch::kFloat32).contiguous();
+    mask_ptr = mf.data_ptr<float>();
+  }
+  auto bc = bias.contiguous();
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_lstm_rec_fwd(
+      h_prev.data_ptr(), w_hh.data_ptr(), xg.data_ptr(), bc.data_ptr(),
+      c_prev.data_ptr(), mask_ptr, h_out.data_ptr(), c_out.data_ptr(),
+      want_gates ? gates_act.data_ptr() : nullptr, (int)batch, hprev_stride,
+      xg_stride, hout_stride, stream.stream());
+}
+
+// Fused recurrent backward (bf16, H == 256): gate grads + grad_c_prev +
+// grad_h = ghp + grad_gates @ W_hh in one launch.
+void lstm_rec_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
+                  torch::Tensor grad_c, torch::Tensor gates_act,
+                  torch::Tensor c_prev, torch::Tensor w_hh_t,
+                  torch::Tensor mask, torch::Tensor grad_gates,
+                  torch::Tensor grad_c_prev, torch::Tensor grad_h_out) {
+  check_gpu_contig(grad_h, "grad_h");
+  check_gpu_contig(grad_c, "grad_c");
+  check_gpu_contig(gates_act, "gates_act");
+  check_gpu_contig(c_prev, "c_prev");
+  check_gpu_contig(w_hh_t, "w_hh_t");
+  check_gpu_contig(grad_gates, "grad_gates");
+  check_gpu_contig(grad_c_prev, "grad_c_prev");
+  check_gpu_contig(grad_h_out, "grad_h_out");
+  for (auto* t : {&grad_h, &grad_c, &gates_act, &c_prev, &w_hh_t}) {
+    TORCH_CHECK(t->scalar_type() == torch::kBFloat16,
+                "lstm_rec_bwd is bf16-only");
+  }
+  const long batch = c_prev.size(0);
+  TORCH_CHECK(c_prev.size(1) == 256 && gates_act.size(1) == 1024 &&
+                  w_hh_t.size(0) == 256 && w_hh_t.size(1) == 1024,
+              "lstm_rec_bwd requires H=256 and w_hh_t = W_hh^T contiguous");
+  const float* mask_ptr = nullptr;
+  torch::Tensor mf;
+  if (mask.numel() > 0) {
+    mf = mask.scalar_type() == torch::kFloat32 ? mask.contiguous()
+                                               : mask.to(torch::kFloat32).contiguous();
+    mask_ptr = mf.data_ptr<float>();
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  const void* got = grad_out_t.numel() ? grad_out_t.data_ptr() : nullptr;
+  long gout_stride = 256;
+  if (grad_out_t.numel()) gout_stride = row_stride_checked(grad_out_t, "grad_out_t");
+  nerrf::launch_lstm_rec_bwd(
+      grad_h.data_ptr(), got, grad_c.data_ptr(), gates_act.data_ptr(),
+      c_prev.data_ptr(), w_hh_t.data_ptr(), mask_ptr, grad_gates.data_ptr(),
+      grad_c_prev.data_ptr(), grad_h_out.data_ptr(), (int)batch, gout_stride,
+      stream.stream());
 }
 
 // Fully-fused MFMA step (bf16, H == 256). Writes into caller buffers.
@@ -368,6 +454,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gather_mean_bwd_csr", &gather_mean_bwd_csr,
         "deterministic gather-mean backward over reverse CSR");
   m.def("lstm_pointwise_fwd", &lstm_pointwise_fwd, "fused LSTM gate pointwise fwd");
+  m.def("lstm_rec_fwd", &lstm_rec_fwd,
+        "fused recurrent GEMM + LSTM pointwise fwd (bf16, H=256)");
+  m.def("lstm_rec_bwd", &lstm_rec_bwd,
+        "fused LSTM gate grads + grad_h GEMM bwd (bf16, H=256)");
   m.def("lstm_step_fused", &lstm_step_fused, "fully-fused MFMA LSTM step (bf16, H=256)");
   m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "fused LSTM gate pointwise bwd");
 }

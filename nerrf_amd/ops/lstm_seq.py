@@ -47,6 +47,26 @@ def _dir_forward(ext, xg, h0, c0, w_hh, bias, mask, reverse, infer, h_all):
     steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
     if ext is not None:
         empty_mask = torch.empty(0, device=dev)
+        # Default bf16/H=256 path: the fused recurrent step
+        # (lstm_rec_fused.hip) keeps the [B, 4H] pre-activation slab out of
+        # HBM entirely — one launch per timestep instead of GEMM+pointwise.
+        rec_fused = (
+            hasattr(ext, "lstm_rec_fwd")
+            and os.environ.get("NERRF_REC_FUSED", "1") == "1"
+            and dt == torch.bfloat16 and hdim == 256 and w_hh.is_contiguous()
+        )
+        if rec_fused:
+            bias_c = bias.contiguous()
+            for ti in steps:
+                ext.lstm_rec_fwd(
+                    h, w_hh, xg[ti], bias_c, c,
+                    mask[ti] if mask is not None else empty_mask,
+                    h_all[ti], c_all[ti],
+                    g_none if infer else gates_all[ti],
+                )
+                h = h_all[ti]
+                c = c_all[ti]
+            return c_all, gates_all
         # The fully-fused MFMA step (lstm_step_fused.hip) is numerically
         # validated but measured slower than hipBLASLt-GEMM + fused
         # pointwise (profiles/PROFILES.md has the ladder); opt-in, and it
@@ -120,6 +140,44 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
     # iterate in the opposite order of forward
     steps = range(t_len) if reverse else range(t_len - 1, -1, -1)
     empty_mask = torch.empty(0, device=dev)
+    rec_fused = (
+        ext is not None
+        and hasattr(ext, "lstm_rec_bwd")
+        and os.environ.get("NERRF_REC_FUSED", "1") == "1"
+        and dt == torch.bfloat16 and hdim == 256
+    )
+    if rec_fused:
+        # fused gate grads + in-launch grad_h GEMM (lstm_rec_fused.hip):
+        # grad_gates goes to HBM once, the addmm disappears
+        w_hh_t = w_hh.t().contiguous()
+        grad_h_next = torch.empty(batch, hdim, device=dev, dtype=dt)
+        for ti in steps:
+            first = (ti == t_len - 1) if reverse else (ti == 0)
+            c_in = c0 if first else (c_all[ti + 1] if reverse else c_all[ti - 1])
+            ext.lstm_rec_bwd(
+                grad_h.contiguous(), grad_out[ti], grad_c.contiguous(),
+                gates_all[ti], c_in.contiguous(), w_hh_t,
+                mask[ti] if mask is not None else empty_mask,
+                grad_gates_all[ti], grad_c_prev, grad_h_next,
+            )
+            grad_h, grad_h_next = grad_h_next, grad_h
+            grad_c, grad_c_prev = grad_c_prev, grad_c
+        gg2 = grad_gates_all.reshape(t_len * batch, gdim)
+        if t_len > 1:
+            if reverse:
+                bulk_gg = grad_gates_all[:-1].reshape((t_len - 1) * batch, gdim)
+                bulk_h = h_flat[batch:]
+                edge_gg = grad_gates_all[t_len - 1]
+            else:
+                bulk_gg = grad_gates_all[1:].reshape((t_len - 1) * batch, gdim)
+                bulk_h = h_flat[: (t_len - 1) * batch]
+                edge_gg = grad_gates_all[0]
+            grad_whh = torch.mm(bulk_gg.t(), bulk_h)
+            grad_whh = torch.addmm(grad_whh, edge_gg.t(), h0.to(dt))
+        else:
+            grad_whh = torch.mm(gg2.t(), h0.to(dt))
+        grad_bias = gg2.sum(dim=0)
+        return grad_gates_all, grad_h, grad_c, grad_whh, grad_bias
     for ti in steps:
         # c/h input of step ti = previous step's output (or h0/c0 at start)
         first = (ti == t_len - 1) if reverse else (ti == 0)

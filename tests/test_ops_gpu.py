@@ -582,3 +582,109 @@ def test_lstm_bilayer_gpu_matches_two_directions(gpu_device):
     for have, p, name in zip(got, (xf, xb, wf, wb, bf, bb),
                              ["xg_f", "xg_b", "w_f", "w_b", "b_f", "b_b"]):
         torch.testing.assert_close(have, p.grad, rtol=0, atol=0, msg=name)
+
+
+def _lstm_cpu_ref_seq(xg, h0, c0, w_hh, bias, mask, rev):
+    from nerrf_amd.ops.reference import lstm_pointwise_fwd_ref
+
+    t = xg.shape[0]
+    h, c = h0.cpu().float(), c0.cpu().float()
+    outs = [None] * t
+    steps = range(t - 1, -1, -1) if rev else range(t)
+    for ti in steps:
+        gp = torch.addmm(bias.cpu().float(), h, w_hh.cpu().float().t()) + xg[ti].cpu().float()
+        h, c, _ = lstm_pointwise_fwd_ref(gp, c, h, mask[ti].cpu().float())
+        outs[ti] = h
+    return torch.stack(outs)
+
+
+def test_lstm_rec_fused_fwd_matches_split_and_ref(gpu_device, monkeypatch):
+    """The fused recurrent step (lstm_rec_fused.hip) vs the split
+    GEMM+pointwise path and the fp32 CPU reference, H=256 bf16."""
+    from nerrf_amd.ops import lstm_sequence
+
+    torch.manual_seed(21)
+    t, b, hd = 5, 200, 256  # b NOT divisible by 64: tail-block guard
+    xg = (torch.randn(t, b, 4 * hd, device=gpu_device) * 0.5).to(torch.bfloat16)
+    h0 = (torch.randn(b, hd, device=gpu_device) * 0.5).to(torch.bfloat16)
+    c0 = (torch.randn(b, hd, device=gpu_device) * 0.5).to(torch.bfloat16)
+    w_hh = ((torch.randn(4 * hd, hd, device=gpu_device)) * 0.05).to(torch.bfloat16)
+    bias = torch.randn(4 * hd, device=gpu_device).to(torch.bfloat16)
+    mask = (torch.rand(t, b, device=gpu_device) > 0.2).float()
+
+    for rev in (False, True):
+        monkeypatch.setenv("NERRF_REC_FUSED", "1")
+        out_fused = lstm_sequence(xg, h0, c0, w_hh, bias, mask, reverse=rev)
+        monkeypatch.setenv("NERRF_REC_FUSED", "0")
+        out_split = lstm_sequence(xg, h0, c0, w_hh, bias, mask, reverse=rev)
+        ref = _lstm_cpu_ref_seq(xg, h0, c0, w_hh, bias, mask, rev)
+        # fused vs split: both bf16 paths, tiny accumulation-order diff
+        assert torch.allclose(out_fused.float().cpu(), out_split.float().cpu(),
+                              atol=3e-2, rtol=3e-2), f"fused vs split rev={rev}"
+        assert torch.allclose(out_fused.float().cpu(), ref, atol=6e-2,
+                              rtol=6e-2), f"fused vs fp32 ref rev={rev}"
+
+
+def test_lstm_rec_fused_backward_matches_split(gpu_device, monkeypatch):
+    from nerrf_amd.ops import lstm_sequence
+
+    torch.manual_seed(22)
+    t, b, hd = 4, 130, 256
+    xg32 = torch.randn(t, b, 4 * hd, device=gpu_device) * 0.5
+    w32 = torch.randn(4 * hd, hd, device=gpu_device) * 0.05
+    b32 = torch.randn(4 * hd, device=gpu_device) * 0.5
+    mask = (torch.rand(t, b, device=gpu_device) > 0.15).float()
+    h0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    c0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    g_out = torch.randn(t, b, hd, device=gpu_device).to(torch.bfloat16)
+
+    grads = {}
+    for tag in ("1", "0"):
+        monkeypatch.setenv("NERRF_REC_FUSED", tag)
+        xg = xg32.to(torch.bfloat16).requires_grad_(True)
+        w = w32.to(torch.bfloat16).requires_grad_(True)
+        bb = b32.to(torch.bfloat16).requires_grad_(True)
+        out = lstm_sequence(xg, h0, c0, w, bb, mask, reverse=False)
+        out.backward(g_out)
+        grads[tag] = [p.grad.float().cpu() for p in (xg, w, bb)]
+    for gf, gs, name in zip(grads["1"], grads["0"], ["xg", "w_hh", "bias"]):
+        scale = gs.abs().max().clamp(min=1.0)
+        assert torch.allclose(gf / scale, gs / scale, atol=4e-2), (
+            f"grad {name}: max diff {(gf - gs).abs().max()}"
+        )
+
+
+def test_lstm_bilayer_rec_fused_strided_slabs(gpu_device, monkeypatch):
+    """Bidirectional layer (strided [T,B,2H] slabs) through the fused
+    recurrent kernels, fwd + bwd vs the split path."""
+    from nerrf_amd.ops.lstm_seq import lstm_bilayer
+
+    torch.manual_seed(23)
+    t, b, hd = 6, 96, 256
+    xf32 = torch.randn(t, b, 4 * hd, device=gpu_device) * 0.5
+    xb32 = torch.randn(t, b, 4 * hd, device=gpu_device) * 0.5
+    wf32 = torch.randn(4 * hd, hd, device=gpu_device) * 0.05
+    wb32 = torch.randn(4 * hd, hd, device=gpu_device) * 0.05
+    bf32 = torch.randn(4 * hd, device=gpu_device) * 0.5
+    bb32 = torch.randn(4 * hd, device=gpu_device) * 0.5
+    mask = (torch.rand(t, b, device=gpu_device) > 0.2).float()
+    h0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    c0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    g_out = torch.randn(t, b, 2 * hd, device=gpu_device).to(torch.bfloat16)
+
+    res = {}
+    for tag in ("1", "0"):
+        monkeypatch.setenv("NERRF_REC_FUSED", tag)
+        args = [x.to(torch.bfloat16).requires_grad_(True)
+                for x in (xf32, xb32, wf32, bf32, wb32, bb32)]
+        xf, xb, wf, bf, wb, bb = args
+        out = lstm_bilayer(xf, xb, h0, c0, wf, bf, wb, bb, mask)
+        out.backward(g_out)
+        res[tag] = ([out.detach().float().cpu()] +
+                    [p.grad.float().cpu() for p in args])
+    names = ["out", "gxf", "gxb", "gwf", "gbf", "gwb", "gbb"]
+    for a, s, name in zip(res["1"], res["0"], names):
+        scale = s.abs().max().clamp(min=1.0)
+        assert torch.allclose(a / scale, s / scale, atol=4e-2), (
+            f"{name}: max diff {(a - s).abs().max()}"
+        )
