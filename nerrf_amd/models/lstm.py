@@ -85,13 +85,17 @@ class BiLSTMDetector(nn.Module):
         ar = torch.arange(t, device=feats.device)
         mask = (ar.unsqueeze(1) < lengths.unsqueeze(0)).to(feats.dtype)  # [T, B]
         h = feats.transpose(0, 1).contiguous()  # time-major [T, B, E]
+        from ..ops.proj import dual_projection
+
         for layer in self.dirs:
             f, r = layer[0], layer[1]
             # both directions write one [T, B, 2H] buffer through the
-            # kernels' row-stride args — no torch.cat on the hot path
+            # kernels' row-stride args — no torch.cat on the hot path;
+            # the input projections share one A pass (ops/proj.py)
             flat = h.reshape(t * b, -1)
-            xg_f = torch.matmul(flat, f.w_ih.t()).reshape(t, b, 4 * f.hidden)
-            xg_b = torch.matmul(flat, r.w_ih.t()).reshape(t, b, 4 * r.hidden)
+            xg2_f, xg2_b = dual_projection(flat, f.w_ih, r.w_ih)
+            xg_f = xg2_f.reshape(t, b, 4 * f.hidden)
+            xg_b = xg2_b.reshape(t, b, 4 * r.hidden)
             h0 = h.new_zeros(b, f.hidden)
             c0 = h.new_zeros(b, f.hidden)
             h = lstm_bilayer(xg_f, xg_b, h0, c0, f.w_hh, f.b, r.w_hh, r.b, mask)

@@ -37,6 +37,10 @@ void launch_lstm_rec_fwd(const void*, const void*, const void*, const void*,
 void launch_lstm_rec_bwd(const void*, const void*, const void*, const void*,
                          const void*, const void*, const float*, void*, void*,
                          void*, int, long, hipStream_t);
+void launch_proj_fwd_dual(const void*, const void*, const void*, void*, void*,
+                          long, long, hipStream_t);
+void launch_proj_dgrad_dual(const void*, const void*, const void*, const void*,
+                            void*, long, hipStream_t);
 void launch_event_scatter(const long*, const long*, const signed char*,
                           const float*, const int*, float*, int*, int*, long,
                           hipStream_t);
@@ -313,6 +317,54 @@ void lstm_step_fused(torch::Tensor h_prev, torch::Tensor w_hh, torch::Tensor xg,
       gates_act.data_ptr(), batch, raw, stream.stream());
 }
 
+// Dual-direction LSTM input projection: c1 = a @ w1^T, c2 = a @ w2^T in one
+// launch (A staged in LDS once; K=512, N=1024 per direction, bf16).
+void proj_fwd_dual(torch::Tensor a, torch::Tensor w1, torch::Tensor w2,
+                   torch::Tensor c1, torch::Tensor c2) {
+  const long a_stride = row_stride_checked(a, "a");
+  check_gpu_contig(w1, "w1");
+  check_gpu_contig(c1, "c1");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16, "proj_fwd_dual is bf16-only");
+  TORCH_CHECK(a.size(1) == 512 && w1.size(0) == 1024 && w1.size(1) == 512 &&
+                  c1.size(1) == 1024,
+              "proj_fwd_dual requires [M,512] x [1024,512]");
+  const bool dual = w2.numel() > 0;
+  if (dual) {
+    check_gpu_contig(w2, "w2");
+    check_gpu_contig(c2, "c2");
+    TORCH_CHECK(w2.sizes() == w1.sizes() && c2.sizes() == c1.sizes(),
+                "dual shapes must match");
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_proj_fwd_dual(
+      a.data_ptr(), w1.data_ptr(), dual ? w2.data_ptr() : nullptr,
+      c1.data_ptr(), dual ? c2.data_ptr() : nullptr, a.size(0), a_stride,
+      stream.stream());
+}
+
+// Dual-direction input-projection dgrad: c = a1 @ w1t^T + a2 @ w2t^T
+// (w*t = W^T contiguous [512, 1024]).
+void proj_dgrad_dual(torch::Tensor a1, torch::Tensor a2, torch::Tensor w1t,
+                     torch::Tensor w2t, torch::Tensor c) {
+  check_gpu_contig(a1, "a1");
+  check_gpu_contig(w1t, "w1t");
+  check_gpu_contig(c, "c");
+  TORCH_CHECK(a1.scalar_type() == torch::kBFloat16, "proj_dgrad_dual is bf16-only");
+  TORCH_CHECK(a1.size(1) == 1024 && w1t.size(0) == 512 && w1t.size(1) == 1024 &&
+                  c.size(1) == 512,
+              "proj_dgrad_dual requires [M,1024] x [512,1024]");
+  const bool dual = a2.numel() > 0;
+  if (dual) {
+    check_gpu_contig(a2, "a2");
+    check_gpu_contig(w2t, "w2t");
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_proj_dgrad_dual(
+      a1.data_ptr(), dual ? a2.data_ptr() : nullptr, w1t.data_ptr(),
+      dual ? w2t.data_ptr() : nullptr, c.data_ptr(), a1.size(0),
+      stream.stream());
+}
+
 // GPU delta compaction: event columns -> per-node accumulators -> x [M, 32].
 torch::Tensor event_features(torch::Tensor ev_file, torch::Tensor ev_proc,
                              torch::Tensor syscall_id, torch::Tensor nbytes,
@@ -458,6 +510,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused recurrent GEMM + LSTM pointwise fwd (bf16, H=256)");
   m.def("lstm_rec_bwd", &lstm_rec_bwd,
         "fused LSTM gate grads + grad_h GEMM bwd (bf16, H=256)");
+  m.def("proj_fwd_dual", &proj_fwd_dual,
+        "dual-direction LSTM input projection (A read once)");
+  m.def("proj_dgrad_dual", &proj_dgrad_dual,
+        "dual-direction input-projection dgrad (summed)");
   m.def("lstm_step_fused", &lstm_step_fused, "fully-fused MFMA LSTM step (bf16, H=256)");
   m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "fused LSTM gate pointwise bwd");
 }

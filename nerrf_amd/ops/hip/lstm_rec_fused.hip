@@ -71,8 +71,8 @@ __global__ void lstm_rec_fwd_kernel(
     __hip_bfloat16* __restrict__ gates_act,     // [B, 1024] or nullptr
     int batch, long hprev_stride, long xg_stride, long hout_stride) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* h_lds = smem;                         // 32 KB, swizzled
-  char* xg_lds = smem + REC_BM * REC_HROW_B;  // 128 KB, linear
+  char* h_lds = smem;                           // 32 KB, swizzled
+  char* gpre_lds = smem + REC_BM * REC_HROW_B;  // 128 KB, linear bf16
 
   const int row0 = blockIdx.x * REC_BM;
   const int tid = threadIdx.x;
@@ -98,110 +98,135 @@ __global__ void lstm_rec_fwd_kernel(
   }
   __syncthreads();
 
-  // ---- MFMA phase: pre[64,1024] tile, wave-local gate column groups ------
-  // acc[gate][row-frag][col-frag]
-  rf32x4 acc[4][4][2];
-#pragma unroll
-  for (int g = 0; g < 4; ++g)
+  // ---- MFMA phase, gate-major: one gate's accumulators (8 f32x4) live at
+  // a time, dumped to LDS before the next gate — the all-gates-resident
+  // variant held 128 acc + fragment regs and spilled 44 VGPRs into the
+  // MFMA loop.  A fragments re-read per gate from LDS (cheap; 16 KB/wave)
+  // and the LDS dump of gate g overlaps the next gate's loads. ----------
+  const int frag_col = lane & 15;
+  const int kchunk = (lane >> 4) * 8;  // element offset within k-step
+#pragma unroll 1
+  for (int g = 0; g < 4; ++g) {
+    rf32x4 acc[4][2];
 #pragma unroll
     for (int rf = 0; rf < 4; ++rf)
 #pragma unroll
-      for (int cf = 0; cf < 2; ++cf) acc[g][rf][cf] = rf32x4{0.f, 0.f, 0.f, 0.f};
-
-  const int frag_col = lane & 15;
-  const int kchunk = (lane >> 4) * 8;  // element offset within k-step
+      for (int cf = 0; cf < 2; ++cf) acc[rf][cf] = rf32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-  for (int ks = 0; ks < REC_H / 32; ++ks) {
-    // xg prefetch: register-staged into LDS (2 x 16 B per lane per k-step;
-    // wave w stages tile rows [8w, 8w+8), 128 chunks each)
-#pragma unroll
-    for (int q = 0; q < 2; ++q) {
-      const int chunk = (ks * 2 + q) * 64 + lane;  // 0..1023 per wave
-      const int xrow = wave * 8 + (chunk >> 7);
-      const int xbyte = (chunk & 127) * 16;
-      uint4 v = make_uint4(0, 0, 0, 0);
-      const long grow = (long)(row0 + xrow);
-      if (grow < batch)
-        v = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const char*>(xg + grow * xg_stride) + xbyte);
-      *reinterpret_cast<uint4*>(xg_lds + xrow * REC_GROW_B + xbyte) = v;
-    }
-    // B fragments from L2 (w_hh is ~0.5 MB, shared by every block)
-    rbf16x8 bfr[4][2];
-#pragma unroll
-    for (int g = 0; g < 4; ++g)
+    for (int ks = 0; ks < REC_H / 32; ++ks) {
+      // B fragments from L2 (w_hh is ~0.5 MB, shared by every block)
+      rbf16x8 bfr[2];
 #pragma unroll
       for (int cf = 0; cf < 2; ++cf) {
         const int orow = g * REC_H + j0 + cf * 16 + frag_col;
-        bfr[g][cf] = *reinterpret_cast<const rbf16x8*>(
+        bfr[cf] = *reinterpret_cast<const rbf16x8*>(
             w_hh + (long)orow * REC_H + ks * 32 + kchunk);
       }
-    // A fragments from the swizzled h tile
-    rbf16x8 afr[4];
 #pragma unroll
-    for (int rf = 0; rf < 4; ++rf) {
-      const int arow = rf * 16 + frag_col;
-      afr[rf] = *reinterpret_cast<const rbf16x8*>(
-          h_lds + rec_swz(arow * REC_HROW_B, arow, ks * 64 + (lane >> 4) * 16));
+      for (int rf = 0; rf < 4; ++rf) {
+        const int arow = rf * 16 + frag_col;
+        const rbf16x8 afr = *reinterpret_cast<const rbf16x8*>(
+            h_lds +
+            rec_swz(arow * REC_HROW_B, arow, ks * 64 + (lane >> 4) * 16));
+#pragma unroll
+        for (int cf = 0; cf < 2; ++cf)
+          acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr[cf], acc[rf][cf], 0, 0, 0);
+      }
     }
+    // dump this gate's pre-activations to LDS (bf16) so the pointwise
+    // phase can do 16-B row-major global I/O — scalar 2-B global access
+    // is the 2-2.5x CDNA4 loss the v1 epilogue measured (Guideline 13)
 #pragma unroll
-    for (int g = 0; g < 4; ++g)
+    for (int cf = 0; cf < 2; ++cf) {
+      const int j = j0 + cf * 16 + frag_col;
 #pragma unroll
       for (int rf = 0; rf < 4; ++rf)
 #pragma unroll
-        for (int cf = 0; cf < 2; ++cf)
-          acc[g][rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[rf], bfr[g][cf], acc[g][rf][cf], 0, 0, 0);
+        for (int r = 0; r < 4; ++r) {
+          const int row = rf * 16 + (lane >> 4) * 4 + r;
+          *reinterpret_cast<__hip_bfloat16*>(
+              gpre_lds + row * REC_GROW_B + (g * REC_H + j) * 2) =
+              __float2bfloat16(acc[rf][cf][r]);
+        }
+    }
   }
-  __syncthreads();  // xg tile complete
+  __syncthreads();
 
-  // ---- epilogue: wave-local LSTM pointwise + stores ----------------------
+  // ---- pointwise phase: thread -> (row, 8-col chunk); all global traffic
+  // 16-B and row-contiguous (2 rows x 512 B per wave-instruction).
+  // unroll 1: a 4x unroll keeps ~64 VT8 registers live and spills ---------
+#pragma unroll 1
+  for (int p = 0; p < 4; ++p) {
+    const int row = p * 16 + (tid >> 5);  // 0..63
+    const int ch = (tid & 31) * 8;        // hidden-unit chunk base (8 els)
+    const long grow = (long)(row0 + row);
+    if (grow >= batch) continue;
+    const char* pre_p = gpre_lds + row * REC_GROW_B;
+    const rbf16x8 p_i = *reinterpret_cast<const rbf16x8*>(pre_p + ch * 2);
+    const rbf16x8 p_f =
+        *reinterpret_cast<const rbf16x8*>(pre_p + (REC_H + ch) * 2);
+    const rbf16x8 p_g =
+        *reinterpret_cast<const rbf16x8*>(pre_p + (2 * REC_H + ch) * 2);
+    const rbf16x8 p_o =
+        *reinterpret_cast<const rbf16x8*>(pre_p + (3 * REC_H + ch) * 2);
+    const __hip_bfloat16* xrow = xg + grow * xg_stride;
+    const rbf16x8 x_i = *reinterpret_cast<const rbf16x8*>(xrow + ch);
+    const rbf16x8 x_f = *reinterpret_cast<const rbf16x8*>(xrow + REC_H + ch);
+    const rbf16x8 x_g =
+        *reinterpret_cast<const rbf16x8*>(xrow + 2 * REC_H + ch);
+    const rbf16x8 x_o =
+        *reinterpret_cast<const rbf16x8*>(xrow + 3 * REC_H + ch);
+    const rbf16x8 b_i = *reinterpret_cast<const rbf16x8*>(bias + ch);
+    const rbf16x8 b_f = *reinterpret_cast<const rbf16x8*>(bias + REC_H + ch);
+    const rbf16x8 b_g =
+        *reinterpret_cast<const rbf16x8*>(bias + 2 * REC_H + ch);
+    const rbf16x8 b_o =
+        *reinterpret_cast<const rbf16x8*>(bias + 3 * REC_H + ch);
+    const rbf16x8 cp_v =
+        *reinterpret_cast<const rbf16x8*>(c_prev + grow * REC_H + ch);
+    rbf16x8 hp_v{};
+    const float m = (mask != nullptr) ? mask[grow] : 1.0f;
+    if (mask != nullptr)
+      hp_v = *reinterpret_cast<const rbf16x8*>(h_prev + grow * hprev_stride +
+                                               ch);
+    rbf16x8 ho_v, co_v, ga_i, ga_f, ga_g, ga_o;
 #pragma unroll
-  for (int cf = 0; cf < 2; ++cf) {
-    const int j = j0 + cf * 16 + frag_col;  // hidden unit (= column in H)
-    const float b_i = __bfloat162float(bias[j]);
-    const float b_f = __bfloat162float(bias[REC_H + j]);
-    const float b_g = __bfloat162float(bias[2 * REC_H + j]);
-    const float b_o = __bfloat162float(bias[3 * REC_H + j]);
-#pragma unroll
-    for (int rf = 0; rf < 4; ++rf) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = rf * 16 + (lane >> 4) * 4 + r;
-        const long grow = (long)(row0 + row);
-        if (grow >= batch) continue;
-        const char* xrow_p = xg_lds + row * REC_GROW_B;
-        const float x_i = __bfloat162float(
-            *reinterpret_cast<const __hip_bfloat16*>(xrow_p + j * 2));
-        const float x_f = __bfloat162float(
-            *reinterpret_cast<const __hip_bfloat16*>(xrow_p + (REC_H + j) * 2));
-        const float x_g = __bfloat162float(*reinterpret_cast<const __hip_bfloat16*>(
-            xrow_p + (2 * REC_H + j) * 2));
-        const float x_o = __bfloat162float(*reinterpret_cast<const __hip_bfloat16*>(
-            xrow_p + (3 * REC_H + j) * 2));
-        const float i = sigmoidf_(acc[0][rf][cf][r] + x_i + b_i);
-        const float f = sigmoidf_(acc[1][rf][cf][r] + x_f + b_f);
-        const float g = tanhf(acc[2][rf][cf][r] + x_g + b_g);
-        const float o = sigmoidf_(acc[3][rf][cf][r] + x_o + b_o);
-        const float cp = __bfloat162float(c_prev[grow * REC_H + j]);
-        float cn = f * cp + i * g;
-        float hn = o * tanhf(cn);
-        if (mask != nullptr) {
-          const float m = mask[grow];
-          cn = m * cn + (1.0f - m) * cp;
-          hn = m * hn +
-               (1.0f - m) * __bfloat162float(h_prev[grow * hprev_stride + j]);
-        }
-        c_out[grow * REC_H + j] = __float2bfloat16(cn);
-        h_out[grow * hout_stride + j] = __float2bfloat16(hn);
-        if (gates_act != nullptr) {
-          __hip_bfloat16* gp = gates_act + grow * REC_G + j;
-          gp[0] = __float2bfloat16(i);
-          gp[REC_H] = __float2bfloat16(f);
-          gp[2 * REC_H] = __float2bfloat16(g);
-          gp[3 * REC_H] = __float2bfloat16(o);
-        }
+    for (int e = 0; e < 8; ++e) {
+      const float i = sigmoidf_(__bfloat162float(p_i[e]) +
+                                __bfloat162float(x_i[e]) +
+                                __bfloat162float(b_i[e]));
+      const float f = sigmoidf_(__bfloat162float(p_f[e]) +
+                                __bfloat162float(x_f[e]) +
+                                __bfloat162float(b_f[e]));
+      const float g = tanhf(__bfloat162float(p_g[e]) +
+                            __bfloat162float(x_g[e]) +
+                            __bfloat162float(b_g[e]));
+      const float o = sigmoidf_(__bfloat162float(p_o[e]) +
+                                __bfloat162float(x_o[e]) +
+                                __bfloat162float(b_o[e]));
+      const float cp = __bfloat162float(cp_v[e]);
+      float cn = f * cp + i * g;
+      float hn = o * tanhf(cn);
+      if (mask != nullptr) {
+        cn = m * cn + (1.0f - m) * cp;
+        hn = m * hn + (1.0f - m) * __bfloat162float(hp_v[e]);
       }
+      co_v[e] = (__bf16)cn;
+      ho_v[e] = (__bf16)hn;
+      ga_i[e] = (__bf16)i;
+      ga_f[e] = (__bf16)f;
+      ga_g[e] = (__bf16)g;
+      ga_o[e] = (__bf16)o;
+    }
+    *reinterpret_cast<rbf16x8*>(c_out + grow * REC_H + ch) = co_v;
+    *reinterpret_cast<rbf16x8*>(h_out + grow * hout_stride + ch) = ho_v;
+    if (gates_act != nullptr) {
+      __hip_bfloat16* gp = gates_act + grow * REC_G + ch;
+      *reinterpret_cast<rbf16x8*>(gp) = ga_i;
+      *reinterpret_cast<rbf16x8*>(gp + REC_H) = ga_f;
+      *reinterpret_cast<rbf16x8*>(gp + 2 * REC_H) = ga_g;
+      *reinterpret_cast<rbf16x8*>(gp + 3 * REC_H) = ga_o;
     }
   }
 }
@@ -233,72 +258,74 @@ __global__ void lstm_rec_bwd_kernel(
   const int lane = tid % NERRF_WAVE;
   const int j0 = wave * 32;
 
-  // ---- phase 1: gate gradients, lane -> one batch row of the j slice -----
-  {
-    const int row = lane;  // 0..63
+  // ---- phase 1: gate gradients; per pass a wave covers 16 rows x 4
+  // 8-unit chunks of its j slice (16 rows x 64 B per instruction — the
+  // lane->row map of v1 made every load 2048-B-strided / uncoalesced) ----
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int row = p * 16 + (lane >> 2);  // 0..63
+    const int cc = lane & 3;
+    const int jc = j0 + cc * 8;
     const long grow = (long)(row0 + row);
     const bool live = grow < batch;
     const float m = (live && mask != nullptr) ? mask[grow] : 1.0f;
-#pragma unroll
-    for (int cc = 0; cc < 4; ++cc) {
-      const int jc = j0 + cc * 8;  // 8 hidden units per chunk
-      rbf16x8 ga_i{}, ga_f{}, ga_g{}, ga_o{}, cp_v{}, gh_v{}, gc_v{}, go_v{};
-      if (live) {
-        const __hip_bfloat16* gp = gates_act + grow * REC_G + jc;
-        ga_i = *reinterpret_cast<const rbf16x8*>(gp);
-        ga_f = *reinterpret_cast<const rbf16x8*>(gp + REC_H);
-        ga_g = *reinterpret_cast<const rbf16x8*>(gp + 2 * REC_H);
-        ga_o = *reinterpret_cast<const rbf16x8*>(gp + 3 * REC_H);
-        cp_v = *reinterpret_cast<const rbf16x8*>(c_prev + grow * REC_H + jc);
-        gh_v = *reinterpret_cast<const rbf16x8*>(grad_h + grow * REC_H + jc);
-        gc_v = *reinterpret_cast<const rbf16x8*>(grad_c + grow * REC_H + jc);
-        if (grad_out_t != nullptr)
-          go_v = *reinterpret_cast<const rbf16x8*>(grad_out_t +
-                                                   grow * gout_stride + jc);
-      }
-      rbf16x8 gg_i, gg_f, gg_g, gg_o, gcp_v, ghp_v;
-#pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const float i = __bfloat162float(ga_i[e]);
-        const float f = __bfloat162float(ga_f[e]);
-        const float g = __bfloat162float(ga_g[e]);
-        const float o = __bfloat162float(ga_o[e]);
-        const float cp = __bfloat162float(cp_v[e]);
-        const float tcn = tanhf(f * cp + i * g);
-        float gh_in = __bfloat162float(gh_v[e]);
-        if (grad_out_t != nullptr) gh_in += __bfloat162float(go_v[e]);
-        const float gc_in = __bfloat162float(gc_v[e]);
-        const float gh = gh_in * m;
-        const float gc = gc_in * m;
-        const float d_o = gh * tcn;
-        const float d_c = gc + gh * o * (1.0f - tcn * tcn);
-        gcp_v[e] = (__bf16)(d_c * f + gc_in * (1.0f - m));
-        ghp_v[e] = (__bf16)(gh_in * (1.0f - m));
-        gg_i[e] = (__bf16)(d_c * g * i * (1.0f - i));
-        gg_f[e] = (__bf16)(d_c * cp * f * (1.0f - f));
-        gg_g[e] = (__bf16)(d_c * i * (1.0f - g * g));
-        gg_o[e] = (__bf16)(d_o * o * (1.0f - o));
-      }
-      // grad_gates to HBM (the weight-grad GEMMs read it) AND to LDS (the
-      // in-launch grad_h GEMM reads it as swizzled A fragments)
-      if (live) {
-        __hip_bfloat16* op = grad_gates + grow * REC_G + jc;
-        *reinterpret_cast<rbf16x8*>(op) = gg_i;
-        *reinterpret_cast<rbf16x8*>(op + REC_H) = gg_f;
-        *reinterpret_cast<rbf16x8*>(op + 2 * REC_H) = gg_g;
-        *reinterpret_cast<rbf16x8*>(op + 3 * REC_H) = gg_o;
-        *reinterpret_cast<rbf16x8*>(grad_c_prev + grow * REC_H + jc) = gcp_v;
-      }
-      *reinterpret_cast<rbf16x8*>(
-          gg_lds + rec_swz(row * REC_GROW_B, row, jc * 2)) = gg_i;
-      *reinterpret_cast<rbf16x8*>(
-          gg_lds + rec_swz(row * REC_GROW_B, row, (REC_H + jc) * 2)) = gg_f;
-      *reinterpret_cast<rbf16x8*>(
-          gg_lds + rec_swz(row * REC_GROW_B, row, (2 * REC_H + jc) * 2)) = gg_g;
-      *reinterpret_cast<rbf16x8*>(
-          gg_lds + rec_swz(row * REC_GROW_B, row, (3 * REC_H + jc) * 2)) = gg_o;
-      *reinterpret_cast<rbf16x8*>(ghp_lds + row * REC_HROW_B + jc * 2) = ghp_v;
+    rbf16x8 ga_i{}, ga_f{}, ga_g{}, ga_o{}, cp_v{}, gh_v{}, gc_v{}, go_v{};
+    if (live) {
+      const __hip_bfloat16* gp = gates_act + grow * REC_G + jc;
+      ga_i = *reinterpret_cast<const rbf16x8*>(gp);
+      ga_f = *reinterpret_cast<const rbf16x8*>(gp + REC_H);
+      ga_g = *reinterpret_cast<const rbf16x8*>(gp + 2 * REC_H);
+      ga_o = *reinterpret_cast<const rbf16x8*>(gp + 3 * REC_H);
+      cp_v = *reinterpret_cast<const rbf16x8*>(c_prev + grow * REC_H + jc);
+      gh_v = *reinterpret_cast<const rbf16x8*>(grad_h + grow * REC_H + jc);
+      gc_v = *reinterpret_cast<const rbf16x8*>(grad_c + grow * REC_H + jc);
+      if (grad_out_t != nullptr)
+        go_v = *reinterpret_cast<const rbf16x8*>(grad_out_t +
+                                                 grow * gout_stride + jc);
     }
+    rbf16x8 gg_i, gg_f, gg_g, gg_o, gcp_v, ghp_v;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float i = __bfloat162float(ga_i[e]);
+      const float f = __bfloat162float(ga_f[e]);
+      const float g = __bfloat162float(ga_g[e]);
+      const float o = __bfloat162float(ga_o[e]);
+      const float cp = __bfloat162float(cp_v[e]);
+      const float tcn = tanhf(f * cp + i * g);
+      float gh_in = __bfloat162float(gh_v[e]);
+      if (grad_out_t != nullptr) gh_in += __bfloat162float(go_v[e]);
+      const float gc_in = __bfloat162float(gc_v[e]);
+      const float gh = gh_in * m;
+      const float gc = gc_in * m;
+      const float d_o = gh * tcn;
+      const float d_c = gc + gh * o * (1.0f - tcn * tcn);
+      gcp_v[e] = (__bf16)(d_c * f + gc_in * (1.0f - m));
+      ghp_v[e] = (__bf16)(gh_in * (1.0f - m));
+      gg_i[e] = (__bf16)(d_c * g * i * (1.0f - i));
+      gg_f[e] = (__bf16)(d_c * cp * f * (1.0f - f));
+      gg_g[e] = (__bf16)(d_c * i * (1.0f - g * g));
+      gg_o[e] = (__bf16)(d_o * o * (1.0f - o));
+    }
+    // grad_gates to HBM (the weight-grad GEMMs read it) AND to LDS (the
+    // in-launch grad_h GEMM reads it as swizzled A fragments)
+    if (live) {
+      __hip_bfloat16* op = grad_gates + grow * REC_G + jc;
+      *reinterpret_cast<rbf16x8*>(op) = gg_i;
+      *reinterpret_cast<rbf16x8*>(op + REC_H) = gg_f;
+      *reinterpret_cast<rbf16x8*>(op + 2 * REC_H) = gg_g;
+      *reinterpret_cast<rbf16x8*>(op + 3 * REC_H) = gg_o;
+      *reinterpret_cast<rbf16x8*>(grad_c_prev + grow * REC_H + jc) = gcp_v;
+    }
+    *reinterpret_cast<rbf16x8*>(
+        gg_lds + rec_swz(row * REC_GROW_B, row, jc * 2)) = gg_i;
+    *reinterpret_cast<rbf16x8*>(
+        gg_lds + rec_swz(row * REC_GROW_B, row, (REC_H + jc) * 2)) = gg_f;
+    *reinterpret_cast<rbf16x8*>(
+        gg_lds + rec_swz(row * REC_GROW_B, row, (2 * REC_H + jc) * 2)) = gg_g;
+    *reinterpret_cast<rbf16x8*>(
+        gg_lds + rec_swz(row * REC_GROW_B, row, (3 * REC_H + jc) * 2)) = gg_o;
+    *reinterpret_cast<rbf16x8*>(ghp_lds + row * REC_HROW_B +
+                                ((jc * 2) ^ ((row & 7u) << 4))) = ghp_v;
   }
   __syncthreads();
 
@@ -348,8 +375,10 @@ __global__ void lstm_rec_bwd_kernel(
         const long grow = (long)(row0 + row);
         if (grow >= batch) continue;
         const float ghp = __bfloat162float(
-            *reinterpret_cast<const __hip_bfloat16*>(ghp_lds +
-                                                     row * REC_HROW_B + j * 2));
+            *reinterpret_cast<const __hip_bfloat16*>(
+                ghp_lds + row * REC_HROW_B +
+                (((unsigned)(j * 2) & ~15u) ^ ((row & 7u) << 4)) +
+                ((j * 2) & 15)));
         grad_h_out[grow * REC_H + j] = __float2bfloat16(acc[rf][cf][r] + ghp);
       }
     }

@@ -688,3 +688,65 @@ def test_lstm_bilayer_rec_fused_strided_slabs(gpu_device, monkeypatch):
         assert torch.allclose(a / scale, s / scale, atol=4e-2), (
             f"{name}: max diff {(a - s).abs().max()}"
         )
+
+
+def test_proj_fwd_dual_matches_matmul(gpu_device):
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(31)
+    m = 1000  # not a multiple of 128: tail guard
+    x = (torch.randn(m, 512, device=gpu_device) * 0.3).to(torch.bfloat16)
+    w1 = (torch.randn(1024, 512, device=gpu_device) * 0.05).to(torch.bfloat16)
+    w2 = (torch.randn(1024, 512, device=gpu_device) * 0.05).to(torch.bfloat16)
+    c1 = torch.empty(m, 1024, device=gpu_device, dtype=torch.bfloat16)
+    c2 = torch.empty_like(c1)
+    ext.proj_fwd_dual(x, w1, w2, c1, c2)
+    r1 = (x.float() @ w1.float().t())
+    r2 = (x.float() @ w2.float().t())
+    assert torch.allclose(c1.float(), r1, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(c2.float(), r2, atol=3e-2, rtol=3e-2)
+
+
+def test_proj_dgrad_dual_matches_matmul(gpu_device):
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(32)
+    m = 700
+    a1 = (torch.randn(m, 1024, device=gpu_device) * 0.3).to(torch.bfloat16)
+    a2 = (torch.randn(m, 1024, device=gpu_device) * 0.3).to(torch.bfloat16)
+    w1 = (torch.randn(1024, 512, device=gpu_device) * 0.05).to(torch.bfloat16)
+    w2 = (torch.randn(1024, 512, device=gpu_device) * 0.05).to(torch.bfloat16)
+    c = torch.empty(m, 512, device=gpu_device, dtype=torch.bfloat16)
+    ext.proj_dgrad_dual(a1, a2, w1.t().contiguous(), w2.t().contiguous(), c)
+    ref = a1.float() @ w1.float() + a2.float() @ w2.float()
+    assert torch.allclose(c.float(), ref, atol=8e-2, rtol=4e-2)
+
+
+def test_dual_projection_autograd_matches_matmul(gpu_device):
+    from nerrf_amd.ops.proj import dual_projection
+
+    torch.manual_seed(33)
+    m = 513
+    x32 = torch.randn(m, 512, device=gpu_device) * 0.3
+    wf32 = torch.randn(1024, 512, device=gpu_device) * 0.05
+    wb32 = torch.randn(1024, 512, device=gpu_device) * 0.05
+    gf = (torch.randn(m, 1024, device=gpu_device) * 0.3).to(torch.bfloat16)
+    gb = (torch.randn(m, 1024, device=gpu_device) * 0.3).to(torch.bfloat16)
+
+    res = {}
+    for use in ("1", "0"):
+        import os
+        os.environ["NERRF_STREAM_PROJ"] = use
+        x = x32.to(torch.bfloat16).requires_grad_(True)
+        wf = wf32.to(torch.bfloat16).requires_grad_(True)
+        wb = wb32.to(torch.bfloat16).requires_grad_(True)
+        c1, c2 = dual_projection(x, wf, wb)
+        (c1 * gf.float().to(c1.dtype)).sum().backward(retain_graph=False)
+        res[use] = [c1.detach().float().cpu(), c2.detach().float().cpu(),
+                    x.grad.float().cpu(), wf.grad.float().cpu()]
+    os.environ["NERRF_STREAM_PROJ"] = "1"
+    for a, b, name in zip(res["1"], res["0"], ["c1", "c2", "gx", "gwf"]):
+        scale = b.abs().max().clamp(min=1.0)
+        assert torch.allclose(a / scale, b / scale, atol=3e-2), name
