@@ -41,6 +41,8 @@ void launch_proj_fwd_dual(const void*, const void*, const void*, void*, void*,
                           long, long, hipStream_t);
 void launch_proj_dgrad_dual(const void*, const void*, const void*, const void*,
                             void*, long, hipStream_t);
+void launch_proj_wgrad(const void*, const void*, const void*, float*, float*,
+                       long, int, hipStream_t);
 void launch_event_scatter(const long*, const long*, const signed char*,
                           const float*, const int*, float*, int*, int*, long,
                           hipStream_t);
@@ -365,6 +367,28 @@ void proj_dgrad_dual(torch::Tensor a1, torch::Tensor a2, torch::Tensor w1t,
       stream.stream());
 }
 
+// Dual-direction input-projection weight grads: dW_d = g_d^T @ x for both
+// directions in one launch (transposed LDS tiles, f32 atomic partials).
+std::vector<torch::Tensor> proj_wgrad(torch::Tensor g1, torch::Tensor g2,
+                                      torch::Tensor x, long n_mchunks) {
+  check_gpu_contig(g1, "g1");
+  check_gpu_contig(x, "x");
+  TORCH_CHECK(g1.scalar_type() == torch::kBFloat16, "proj_wgrad is bf16-only");
+  TORCH_CHECK(g1.size(1) == 1024 && x.size(1) == 512,
+              "proj_wgrad requires g [M,1024], x [M,512]");
+  const bool dual = g2.numel() > 0;
+  if (dual) check_gpu_contig(g2, "g2");
+  auto dw1 = torch::zeros({1024, 512}, x.options().dtype(torch::kFloat32));
+  auto dw2 = dual ? torch::zeros({1024, 512}, x.options().dtype(torch::kFloat32))
+                  : torch::empty({0}, x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_proj_wgrad(g1.data_ptr(), dual ? g2.data_ptr() : nullptr,
+                           x.data_ptr(), dw1.data_ptr<float>(),
+                           dual ? dw2.data_ptr<float>() : nullptr, g1.size(0),
+                           (int)n_mchunks, stream.stream());
+  return {dw1, dw2};
+}
+
 // GPU delta compaction: event columns -> per-node accumulators -> x [M, 32].
 torch::Tensor event_features(torch::Tensor ev_file, torch::Tensor ev_proc,
                              torch::Tensor syscall_id, torch::Tensor nbytes,
@@ -514,6 +538,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "dual-direction LSTM input projection (A read once)");
   m.def("proj_dgrad_dual", &proj_dgrad_dual,
         "dual-direction input-projection dgrad (summed)");
+  m.def("proj_wgrad", &proj_wgrad,
+        "dual-direction input-projection weight grads (f32 partials)");
   m.def("lstm_step_fused", &lstm_step_fused, "fully-fused MFMA LSTM step (bf16, H=256)");
   m.def("lstm_pointwise_bwd", &lstm_pointwise_bwd, "fused LSTM gate pointwise bwd");
 }

@@ -83,17 +83,18 @@ __global__ void lstm_rec_fwd_kernel(
   // ---- stage h tile: 64 rows x 512 B, swizzled ---------------------------
   {
     const int r = tid >> 3;               // 0..63
-    const int c0 = (tid & 7) * 4;         // 4 x 16-B chunks per thread
-    const long grow = (long)(row0 + r);
+    const int c0 = tid & 7;               // interleaved: consecutive lanes
+    const long grow = (long)(row0 + r);   // read consecutive 16-B chunks
 #pragma unroll
     for (int cc = 0; cc < 4; ++cc) {
+      const int chunk = c0 + cc * 8;
       uint4 v = make_uint4(0, 0, 0, 0);
       if (grow < batch)
         v = *reinterpret_cast<const uint4*>(
             reinterpret_cast<const char*>(h_prev + grow * hprev_stride) +
-            (c0 + cc) * 16);
+            chunk * 16);
       *reinterpret_cast<uint4*>(
-          h_lds + rec_swz(r * REC_HROW_B, r, (c0 + cc) * 16)) = v;
+          h_lds + rec_swz(r * REC_HROW_B, r, chunk * 16)) = v;
     }
   }
   __syncthreads();

@@ -68,16 +68,16 @@ __global__ void proj_fwd_dual_kernel(
   // ---- stage the A strip once: 128 rows x 1024 B -------------------------
   {
     const int r = tid >> 2;              // 0..127
-    const int c0 = (tid & 3) * 16;       // 16 x 16-B chunks per thread
-    const long grow = row0 + r;
+    const int c0 = tid & 3;              // interleaved: lanes 0..3 read the
+    const long grow = row0 + r;          // same row's consecutive chunks
 #pragma unroll
     for (int cc = 0; cc < 16; ++cc) {
+      const int chunk = c0 + cc * 4;
       uint4 v = make_uint4(0, 0, 0, 0);
       if (grow < m_rows)
         v = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const char*>(a + grow * a_stride) +
-            (c0 + cc) * 16);
-      *reinterpret_cast<uint4*>(a_lds + sg_swz(r, (c0 + cc) * 16, ROW_B)) = v;
+            reinterpret_cast<const char*>(a + grow * a_stride) + chunk * 16);
+      *reinterpret_cast<uint4*>(a_lds + sg_swz(r, chunk * 16, ROW_B)) = v;
     }
   }
   __syncthreads();
@@ -97,26 +97,39 @@ __global__ void proj_fwd_dual_kernel(
       for (int rf = 0; rf < 8; ++rf)
 #pragma unroll
         for (int cf = 0; cf < 2; ++cf) acc[rf][cf] = sf32x4{0.f, 0.f, 0.f, 0.f};
-#pragma unroll 4
+      // software-pipelined K loop: B fragments for step ks+1 issue before
+      // step ks's MFMAs (the naive loop waited the full L2 latency at the
+      // top of every step — measured ~35% MFMA util), and all 8 A
+      // fragments load into distinct registers so hipcc can partial-wait
+      const char* wrow0 = reinterpret_cast<const char*>(w) +
+                          (long)(col0 + frag_col) * ROW_B + kchunk_b;
+      const char* wrow1 = wrow0 + 16 * ROW_B;
+      sbf16x8 b_cur[2], b_nxt[2];
+      b_cur[0] = *reinterpret_cast<const sbf16x8*>(wrow0);
+      b_cur[1] = *reinterpret_cast<const sbf16x8*>(wrow1);
+#pragma unroll 2
       for (int ks = 0; ks < K / 32; ++ks) {
-        sbf16x8 bfr[2];
-#pragma unroll
-        for (int cf = 0; cf < 2; ++cf) {
-          const int orow = col0 + cf * 16 + frag_col;
-          bfr[cf] = *reinterpret_cast<const sbf16x8*>(
-              reinterpret_cast<const char*>(w) + (long)orow * ROW_B +
-              ks * 64 + kchunk_b);
+        if (ks + 1 < K / 32) {
+          b_nxt[0] = *reinterpret_cast<const sbf16x8*>(wrow0 + (ks + 1) * 64);
+          b_nxt[1] = *reinterpret_cast<const sbf16x8*>(wrow1 + (ks + 1) * 64);
         }
+        sbf16x8 afr[8];
 #pragma unroll
-        for (int rf = 0; rf < 8; ++rf) {
-          const int arow = rf * 16 + frag_col;
-          const sbf16x8 afr = *reinterpret_cast<const sbf16x8*>(
-              a_lds + sg_swz(arow, ks * 64 + kchunk_b, ROW_B));
+        for (int rf = 0; rf < 8; ++rf)
+          afr[rf] = *reinterpret_cast<const sbf16x8*>(
+              a_lds + sg_swz(rf * 16 + frag_col, ks * 64 + kchunk_b, ROW_B));
+        // force {8 ds_read batch -> 16 MFMA} grouping: the default schedule
+        // collapses afr to one register quad with lgkmcnt(0) per read
+        __builtin_amdgcn_sched_group_barrier(0x100, 8, 0);   // DS_READ x8
+        __builtin_amdgcn_sched_group_barrier(0x8, 16, 0);    // MFMA x16
+#pragma unroll
+        for (int rf = 0; rf < 8; ++rf)
 #pragma unroll
           for (int cf = 0; cf < 2; ++cf)
             acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afr, bfr[cf], acc[rf][cf], 0, 0, 0);
-        }
+                afr[rf], b_cur[cf], acc[rf][cf], 0, 0, 0);
+        b_cur[0] = b_nxt[0];
+        b_cur[1] = b_nxt[1];
       }
       // ---- store this wave's [128, 32] slice -----------------------------
 #pragma unroll
@@ -166,9 +179,10 @@ __global__ void proj_dgrad_dual_kernel(
   const int n_dirs = (a2 != nullptr) ? 2 : 1;
   const int n_chunks = n_dirs * (K / BK);  // 8 or 16 A chunks
 
-  // chunk staging: thread -> (row, 16-B piece); 32 KB per chunk
+  // chunk staging: thread -> (row, 16-B piece), pieces interleaved so
+  // consecutive lanes read consecutive 16 B; 32 KB per chunk
   const int st_r = tid >> 2;         // 0..127
-  const int st_c = (tid & 3) * 4;    // 4 x 16-B pieces
+  const int st_c = tid & 3;
 
   auto chunk_src = [&](int ch, int piece) -> const char* {
     const __hip_bfloat16* a = (ch < K / BK || n_dirs == 1) ? a1 : a2;
@@ -182,11 +196,11 @@ __global__ void proj_dgrad_dual_kernel(
   {
 #pragma unroll
     for (int cc = 0; cc < 4; ++cc) {
-      const char* src = chunk_src(0, st_c + cc);
+      const char* src = chunk_src(0, st_c + cc * 4);
       uint4 v = make_uint4(0, 0, 0, 0);
       if (src != nullptr) v = *reinterpret_cast<const uint4*>(src);
       *reinterpret_cast<uint4*>(a_buf(0) +
-                                sg_swz(st_r, (st_c + cc) * 16, AROW_B)) = v;
+                                sg_swz(st_r, (st_c + cc * 4) * 16, AROW_B)) = v;
     }
   }
   __syncthreads();
@@ -208,7 +222,7 @@ __global__ void proj_dgrad_dual_kernel(
     if (has_next) {
 #pragma unroll
       for (int cc = 0; cc < 4; ++cc) {
-        const char* src = chunk_src(ch + 1, st_c + cc);
+        const char* src = chunk_src(ch + 1, st_c + cc * 4);
         pre[cc] = make_uint4(0, 0, 0, 0);
         if (src != nullptr) pre[cc] = *reinterpret_cast<const uint4*>(src);
       }
@@ -216,33 +230,35 @@ __global__ void proj_dgrad_dual_kernel(
     const char* cur = a_buf(ch & 1);
     const __hip_bfloat16* wt = (ch < K / BK || n_dirs == 1) ? w1t : w2t;
     const int kc_b = (ch % (K / BK)) * BK * 2;  // byte offset in the W^T row
+    // pipelined like proj_fwd: next step's B fragments in flight under
+    // this step's MFMAs, A fragments batched into distinct registers
+    const char* wt0 = reinterpret_cast<const char*>(wt) +
+                      (long)(col0 + frag_col) * WROW_B + kc_b + kchunk_b;
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       sbf16x8 bfr[4];
 #pragma unroll
-      for (int cf = 0; cf < 4; ++cf) {
-        const int orow = col0 + cf * 16 + frag_col;
-        bfr[cf] = *reinterpret_cast<const sbf16x8*>(
-            reinterpret_cast<const char*>(wt) + (long)orow * WROW_B + kc_b +
-            ks * 64 + kchunk_b);
-      }
+      for (int cf = 0; cf < 4; ++cf)
+        bfr[cf] = *reinterpret_cast<const sbf16x8*>(wt0 + (long)cf * 16 * WROW_B +
+                                                    ks * 64);
+      sbf16x8 afr[8];
 #pragma unroll
-      for (int rf = 0; rf < 8; ++rf) {
-        const int arow = rf * 16 + frag_col;
-        const sbf16x8 afr = *reinterpret_cast<const sbf16x8*>(
-            cur + sg_swz(arow, ks * 64 + kchunk_b, AROW_B));
+      for (int rf = 0; rf < 8; ++rf)
+        afr[rf] = *reinterpret_cast<const sbf16x8*>(
+            cur + sg_swz(rf * 16 + frag_col, ks * 64 + kchunk_b, AROW_B));
+#pragma unroll
+      for (int rf = 0; rf < 8; ++rf)
 #pragma unroll
         for (int cf = 0; cf < 4; ++cf)
           acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr, bfr[cf], acc[rf][cf], 0, 0, 0);
-      }
+              afr[rf], bfr[cf], acc[rf][cf], 0, 0, 0);
     }
     if (has_next) {
       __syncthreads();  // everyone done reading buf[(ch+1)&1] last round
 #pragma unroll
       for (int cc = 0; cc < 4; ++cc)
         *reinterpret_cast<uint4*>(a_buf((ch + 1) & 1) +
-                                  sg_swz(st_r, (st_c + cc) * 16, AROW_B)) =
+                                  sg_swz(st_r, (st_c + cc * 4) * 16, AROW_B)) =
             pre[cc];
       __syncthreads();
     }
@@ -288,6 +304,167 @@ void launch_proj_dgrad_dual(const void* a1, const void* a2, const void* w1t,
       (const __hip_bfloat16*)a1, (const __hip_bfloat16*)a2,
       (const __hip_bfloat16*)w1t, (const __hip_bfloat16*)w2t,
       (__hip_bfloat16*)c, m_rows);
+}
+
+// ---------------------------------------------------------------------------
+// proj_wgrad: dW[1024, 512] += dXg[M, 1024]^T @ X[M, 512]  (f32 accumulate)
+//
+// The weight-grad reduction runs over the huge M axis: both operands need
+// m-contiguous MFMA fragments, i.e. transposed tiles.  Each block owns one
+// 256x256 output tile and an M chunk; tiles are staged transposed through
+// LDS with an in-register 4x8 micro-transpose (4 row-loads of 16 B ->
+// 8 column writes of 8 B), 80-B row stride for bank-conflict-free
+// ds_read_b128 fragments.  Partials land in a f32 workspace by device
+// atomicAdd (output is only 2 MB; the GEMM dwarfs the atomic traffic).
+// Grid: (m_chunks) x (tiles of both directions), consecutive ids share the
+// chunk so an XCD-contiguous dispatch reuses the operand panels in L2.
+// ---------------------------------------------------------------------------
+
+// stage a [32 m][256 col] global panel as a transposed [256 col][32 m]
+// bf16 LDS tile (row stride 80 B).  256 threads; returns nothing staged for
+// out-of-range rows (zeros).
+__device__ __forceinline__ void wg_stage_t(
+    const __hip_bfloat16* __restrict__ src, long src_stride, long m0,
+    long m_rows, int col0, char* dst, int task) {
+  const int mg = task >> 5;       // 0..7: group of 4 m rows
+  const int cg = task & 31;       // 0..31: group of 8 columns
+  uint4 in[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const long m = m0 + mg * 4 + i;
+    in[i] = make_uint4(0, 0, 0, 0);
+    if (m < m_rows)
+      in[i] = *reinterpret_cast<const uint4*>(src + m * src_stride + col0 +
+                                              cg * 8);
+  }
+  // 4x8 micro-transpose: out[j] = {in[0].h[j], in[1].h[j], in[2].h[j], in[3].h[j]}
+  const ushort* h0 = reinterpret_cast<const ushort*>(&in[0]);
+  const ushort* h1 = reinterpret_cast<const ushort*>(&in[1]);
+  const ushort* h2 = reinterpret_cast<const ushort*>(&in[2]);
+  const ushort* h3 = reinterpret_cast<const ushort*>(&in[3]);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    uint2 o;
+    o.x = (unsigned)h0[j] | ((unsigned)h1[j] << 16);
+    o.y = (unsigned)h2[j] | ((unsigned)h3[j] << 16);
+    *reinterpret_cast<uint2*>(dst + (cg * 8 + j) * 80 + mg * 8) = o;
+  }
+}
+
+__launch_bounds__(512)
+__global__ void proj_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ g1,  // [M, 1024] grad_xg dir 0
+    const __hip_bfloat16* __restrict__ g2,  // [M, 1024] dir 1 or nullptr
+    const __hip_bfloat16* __restrict__ x,   // [M, 512]
+    float* __restrict__ dw1,                // [1024, 512] f32 (zeroed)
+    float* __restrict__ dw2,                // [1024, 512] f32
+    long m_rows, int n_mchunks) {
+  constexpr int TILE_BYTES = 256 * 80;  // 20 KB per transposed tile
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto ta = [&](int i) -> char* { return smem + i * TILE_BYTES; };          // A^T dbuf
+  auto tb = [&](int i) -> char* { return smem + (2 + i) * TILE_BYTES; };    // X^T dbuf
+
+  // grid: blockIdx.x = chunk * n_tiles + tile; tile enumerates
+  // (dir, n0/256, k0/256) = dir*8 + nt*2 + kt
+  const int n_dirs = (g2 != nullptr) ? 2 : 1;
+  const int n_tiles = n_dirs * 8;
+  const int chunk = blockIdx.x / n_tiles;
+  const int tile = blockIdx.x % n_tiles;
+  const int dir = tile >> 3;
+  const int nt = (tile >> 1) & 3;
+  const int kt = tile & 1;
+  const __hip_bfloat16* g = dir ? g2 : g1;
+  float* dw = dir ? dw2 : dw1;
+  const int n0 = nt * 256;
+  const int k0 = kt * 256;
+
+  const long rows_per_chunk =
+      ((m_rows + (long)n_mchunks - 1) / n_mchunks + 31) & ~31L;
+  const long m_lo = (long)chunk * rows_per_chunk;
+  const long m_hi = min(m_lo + rows_per_chunk, m_rows);
+  if (m_lo >= m_hi) return;
+
+  const int tid = threadIdx.x;
+  const int lane = tid % NERRF_WAVE;
+  const int wave = tid / NERRF_WAVE;
+  const int nrow0 = wave * 32;  // wave's 32 output rows (n of dW)
+
+  // prologue: stage first 32-m panel of both operands
+  if (tid < 256)
+    wg_stage_t(g, 1024, m_lo, m_hi, n0, ta(0), tid);
+  else
+    wg_stage_t(x, 512, m_lo, m_hi, k0, tb(0), tid - 256);
+  __syncthreads();
+
+  sf32x4 acc[2][16];
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int cf = 0; cf < 16; ++cf) acc[rf][cf] = sf32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int frag_col = lane & 15;
+  const int mchunk_b = (lane >> 4) * 16;
+  const long n_steps = (m_hi - m_lo + 31) / 32;
+
+  for (long st = 0; st < n_steps; ++st) {
+    const int cur = (int)(st & 1);
+    // compute on cur while the other half of the block... (single phase:
+    // stage next AFTER compute, standard 2-buffer with two barriers)
+    const char* a_t = ta(cur);
+    const char* b_t = tb(cur);
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      sbf16x8 bfr[8];
+#pragma unroll
+      for (int c8 = 0; c8 < 8; ++c8) {
+        const int krow = half * 128 + c8 * 16 + frag_col;
+        bfr[c8] = *reinterpret_cast<const sbf16x8*>(b_t + krow * 80 + mchunk_b);
+      }
+#pragma unroll
+      for (int rf = 0; rf < 2; ++rf) {
+        const int arow = nrow0 + rf * 16 + frag_col;
+        const sbf16x8 afr =
+            *reinterpret_cast<const sbf16x8*>(a_t + arow * 80 + mchunk_b);
+#pragma unroll
+        for (int c8 = 0; c8 < 8; ++c8)
+          acc[rf][half * 8 + c8] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, bfr[c8], acc[rf][half * 8 + c8], 0, 0, 0);
+      }
+    }
+    if (st + 1 < n_steps) {
+      __syncthreads();  // everyone done with the buffer we are overwriting
+      const int nxt = (int)((st + 1) & 1);
+      if (tid < 256)
+        wg_stage_t(g, 1024, m_lo + (st + 1) * 32, m_hi, n0, ta(nxt), tid);
+      else
+        wg_stage_t(x, 512, m_lo + (st + 1) * 32, m_hi, k0, tb(nxt), tid - 256);
+      __syncthreads();
+    }
+  }
+
+  // epilogue: device-scope atomic accumulate into the f32 workspace
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int cf = 0; cf < 16; ++cf) {
+      const int col = k0 + cf * 16 + frag_col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = n0 + nrow0 + rf * 16 + (lane >> 4) * 4 + r;
+        atomicAdd(dw + (long)row * 512 + col, acc[rf][cf][r]);
+      }
+    }
+}
+
+void launch_proj_wgrad(const void* g1, const void* g2, const void* x,
+                       float* dw1, float* dw2, long m_rows, int n_mchunks,
+                       hipStream_t s) {
+  const int n_tiles = (g2 != nullptr) ? 16 : 8;
+  const int grid = n_mchunks * n_tiles;
+  const size_t lds = 4 * 256 * 80;  // 80 KB
+  proj_wgrad_kernel<<<grid, 512, lds, s>>>(
+      (const __hip_bfloat16*)g1, (const __hip_bfloat16*)g2,
+      (const __hip_bfloat16*)x, dw1, dw2, m_rows, n_mchunks);
 }
 
 }  // namespace nerrf

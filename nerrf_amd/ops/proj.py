@@ -68,8 +68,22 @@ class _DualProjFn(torch.autograd.Function):
                 )
             else:
                 gx = torch.matmul(g1, w_f) + torch.matmul(g2, w_b)
-        gwf = torch.matmul(g1.t(), x) if ctx.needs_input_grad[1] else None
-        gwb = torch.matmul(g2.t(), x) if ctx.needs_input_grad[2] else None
+        gwf = gwb = None
+        if ctx.needs_input_grad[1] or ctx.needs_input_grad[2]:
+            if (
+                ext is not None
+                and hasattr(ext, "proj_wgrad")
+                and _use_stream(x, w_f, w_b)
+                and os.environ.get("NERRF_STREAM_WGRAD", "1") == "1"
+            ):
+                # chunk count sized so the grid is a few block-waves deep
+                n_chunks = max(16, min(256, x.shape[0] // 8192))
+                dw1, dw2 = ext.proj_wgrad(g1, g2, x, n_chunks)
+                gwf = dw1.to(x.dtype)
+                gwb = dw2.to(x.dtype)
+            else:
+                gwf = torch.matmul(g1.t(), x) if ctx.needs_input_grad[1] else None
+                gwb = torch.matmul(g2.t(), x) if ctx.needs_input_grad[2] else None
         return gx, gwf, gwb
 
 

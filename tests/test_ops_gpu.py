@@ -750,3 +750,20 @@ def test_dual_projection_autograd_matches_matmul(gpu_device):
     for a, b, name in zip(res["1"], res["0"], ["c1", "c2", "gx", "gwf"]):
         scale = b.abs().max().clamp(min=1.0)
         assert torch.allclose(a / scale, b / scale, atol=3e-2), name
+
+
+def test_proj_wgrad_matches_matmul(gpu_device):
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(34)
+    m = 5000  # exercises several m-chunks and the 32-row tail
+    g1 = (torch.randn(m, 1024, device=gpu_device) * 0.3).to(torch.bfloat16)
+    g2 = (torch.randn(m, 1024, device=gpu_device) * 0.3).to(torch.bfloat16)
+    x = (torch.randn(m, 512, device=gpu_device) * 0.3).to(torch.bfloat16)
+    dw1, dw2 = ext.proj_wgrad(g1, g2, x, 16)
+    r1 = g1.float().t() @ x.float()
+    r2 = g2.float().t() @ x.float()
+    # f32 accumulate over bf16 products; tolerance scales with K=m
+    assert torch.allclose(dw1, r1, atol=0.5, rtol=2e-2)
+    assert torch.allclose(dw2, r2, atol=0.5, rtol=2e-2)

@@ -41,6 +41,20 @@ def dgrad_stream():
     ext.proj_dgrad_dual(g1, g2, w1t, w2t, dx)
 
 
+dwf32 = torch.empty(1024, 512, device=dev, dtype=torch.bfloat16)
+
+
+def wgrad_blas():
+    global dwf32
+    a = torch.mm(g1.t(), x)
+    b = torch.mm(g2.t(), x)
+    dwf32 = a
+
+
+def wgrad_stream():
+    ext.proj_wgrad(g1, g2, x, max(16, min(256, M // 8192)))
+
+
 def tm(fn, n=20):
     for _ in range(3):
         fn()
@@ -53,6 +67,8 @@ def tm(fn, n=20):
 for rnd in range(3):
     fb = tm(fwd_blas); fs = tm(fwd_stream)
     db = tm(dgrad_blas); ds = tm(dgrad_stream)
+    wb = tm(wgrad_blas, n=10); ws = tm(wgrad_stream, n=10)
+    print(f"  wgrad blas {wb:.2f} ms  stream {ws:.2f} ms ({wb/ws:.2f}x)")
     # effective bytes: fwd reads A once + writes 2C (+W); dgrad reads 2A + writes C
     fwd_gb = (M * 512 * 2 + 2 * M * 1024 * 2) / 1e9
     dg_gb = (2 * M * 1024 * 2 + M * 512 * 2) / 1e9
