@@ -44,7 +44,7 @@ typedef float rf32x4 __attribute__((ext_vector_type(4)));
 
 #define REC_H 256
 #define REC_G 1024
-#define REC_BM 32
+#define REC_BM 64
 #define REC_HROW_B 512   // h tile row bytes (256 * 2)
 #define REC_GROW_B 2048  // gate tile row bytes (1024 * 2)
 
@@ -70,14 +70,9 @@ __global__ void lstm_rec_fwd_kernel(
     __hip_bfloat16* __restrict__ c_out,         // [B, 256]
     __hip_bfloat16* __restrict__ gates_act,     // [B, 1024] or nullptr
     int batch, long hprev_stride, long xg_stride, long hout_stride) {
-  // v3 geometry: BM = 32 rows, NO h LDS tile (A fragments load straight
-  // from global — the tile is L1/L2-hot since all 8 waves read the same
-  // 16 KB), LDS = the bf16 pre-activation stage only (64 KB) -> 2 blocks
-  // per CU.  v2 at BM = 64 with 160 KB LDS ran 1 block/CU and its memory
-  // phases were latency-bound at ~2.5 TB/s (the chip needs ~9 KB of
-  // loads in flight per CU; 8 waves cannot post that).
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* gpre_lds = smem;  // [32][2048 B] linear bf16
+  char* h_lds = smem;                           // 32 KB, swizzled
+  char* gpre_lds = smem + REC_BM * REC_HROW_B;  // 128 KB, linear bf16
 
   const int row0 = blockIdx.x * REC_BM;
   const int tid = threadIdx.x;
@@ -85,20 +80,42 @@ __global__ void lstm_rec_fwd_kernel(
   const int lane = tid % NERRF_WAVE;
   const int j0 = wave * 32;  // this wave's hidden-unit slice
 
-  const int frag_col = lane & 15;
-  const int kchunk = (lane >> 4) * 8;
-  const int tail = (row0 + REC_BM > batch);
+  // ---- stage h tile: 64 rows x 512 B, swizzled ---------------------------
+  {
+    const int r = tid >> 3;               // 0..63
+    const int c0 = tid & 7;               // interleaved: consecutive lanes
+    const long grow = (long)(row0 + r);   // read consecutive 16-B chunks
+#pragma unroll
+    for (int cc = 0; cc < 4; ++cc) {
+      const int chunk = c0 + cc * 8;
+      uint4 v = make_uint4(0, 0, 0, 0);
+      if (grow < batch)
+        v = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(h_prev + grow * hprev_stride) +
+            chunk * 16);
+      *reinterpret_cast<uint4*>(
+          h_lds + rec_swz(r * REC_HROW_B, r, chunk * 16)) = v;
+    }
+  }
+  __syncthreads();
 
-  // ---- MFMA phase, gate-major --------------------------------------------
+  // ---- MFMA phase, gate-major: one gate's accumulators (8 f32x4) live at
+  // a time, dumped to LDS before the next gate — the all-gates-resident
+  // variant held 128 acc + fragment regs and spilled 44 VGPRs into the
+  // MFMA loop.  A fragments re-read per gate from LDS (cheap; 16 KB/wave)
+  // and the LDS dump of gate g overlaps the next gate's loads. ----------
+  const int frag_col = lane & 15;
+  const int kchunk = (lane >> 4) * 8;  // element offset within k-step
 #pragma unroll 1
   for (int g = 0; g < 4; ++g) {
-    rf32x4 acc[2][2];
+    rf32x4 acc[4][2];
 #pragma unroll
-    for (int rf = 0; rf < 2; ++rf)
+    for (int rf = 0; rf < 4; ++rf)
 #pragma unroll
       for (int cf = 0; cf < 2; ++cf) acc[rf][cf] = rf32x4{0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int ks = 0; ks < REC_H / 32; ++ks) {
+      // B fragments from L2 (w_hh is ~0.5 MB, shared by every block)
       rbf16x8 bfr[2];
 #pragma unroll
       for (int cf = 0; cf < 2; ++cf) {
@@ -106,27 +123,26 @@ __global__ void lstm_rec_fwd_kernel(
         bfr[cf] = *reinterpret_cast<const rbf16x8*>(
             w_hh + (long)orow * REC_H + ks * 32 + kchunk);
       }
-      rbf16x8 afr[2];
 #pragma unroll
-      for (int rf = 0; rf < 2; ++rf) {
-        long arow = (long)(row0 + rf * 16 + frag_col);
-        if (tail && arow >= batch) arow = batch - 1;  // clamp, rows unused
-        afr[rf] = *reinterpret_cast<const rbf16x8*>(
-            h_prev + arow * hprev_stride + ks * 32 + kchunk);
-      }
-#pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
+      for (int rf = 0; rf < 4; ++rf) {
+        const int arow = rf * 16 + frag_col;
+        const rbf16x8 afr = *reinterpret_cast<const rbf16x8*>(
+            h_lds +
+            rec_swz(arow * REC_HROW_B, arow, ks * 64 + (lane >> 4) * 16));
 #pragma unroll
         for (int cf = 0; cf < 2; ++cf)
           acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[rf], bfr[cf], acc[rf][cf], 0, 0, 0);
+              afr, bfr[cf], acc[rf][cf], 0, 0, 0);
+      }
     }
-    // stage this gate's pre-activations (bf16) for the pointwise phase
+    // dump this gate's pre-activations to LDS (bf16) so the pointwise
+    // phase can do 16-B row-major global I/O — scalar 2-B global access
+    // is the 2-2.5x CDNA4 loss the v1 epilogue measured (Guideline 13)
 #pragma unroll
     for (int cf = 0; cf < 2; ++cf) {
       const int j = j0 + cf * 16 + frag_col;
 #pragma unroll
-      for (int rf = 0; rf < 2; ++rf)
+      for (int rf = 0; rf < 4; ++rf)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = rf * 16 + (lane >> 4) * 4 + r;
@@ -138,11 +154,12 @@ __global__ void lstm_rec_fwd_kernel(
   }
   __syncthreads();
 
-  // ---- pointwise phase: thread -> (row, 8-col chunk); 2 fully-unrolled
-  // passes so ~30 16-B loads are in flight per lane -----------------------
-#pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const int row = p * 16 + (tid >> 5);  // 0..31
+  // ---- pointwise phase: thread -> (row, 8-col chunk); all global traffic
+  // 16-B and row-contiguous (2 rows x 512 B per wave-instruction).
+  // unroll 1: a 4x unroll keeps ~64 VT8 registers live and spills ---------
+#pragma unroll 1
+  for (int p = 0; p < 4; ++p) {
+    const int row = p * 16 + (tid >> 5);  // 0..63
     const int ch = (tid & 31) * 8;        // hidden-unit chunk base (8 els)
     const long grow = (long)(row0 + row);
     if (grow >= batch) continue;
@@ -246,8 +263,8 @@ __global__ void lstm_rec_bwd_kernel(
   // 8-unit chunks of its j slice (16 rows x 64 B per instruction — the
   // lane->row map of v1 made every load 2048-B-strided / uncoalesced) ----
 #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const int row = p * 16 + (lane >> 2);  // 0..31
+  for (int p = 0; p < 4; ++p) {
+    const int row = p * 16 + (lane >> 2);  // 0..63
     const int cc = lane & 3;
     const int jc = j0 + cc * 8;
     const long grow = (long)(row0 + row);
@@ -314,9 +331,9 @@ __global__ void lstm_rec_bwd_kernel(
   __syncthreads();
 
   // ---- phase 2: grad_h[64,256] = gg[64,1024] @ W_hh[1024,256] ------------
-  rf32x4 acc[2][2];
+  rf32x4 acc[4][2];
 #pragma unroll
-  for (int rf = 0; rf < 2; ++rf)
+  for (int rf = 0; rf < 4; ++rf)
 #pragma unroll
     for (int cf = 0; cf < 2; ++cf) acc[rf][cf] = rf32x4{0.f, 0.f, 0.f, 0.f};
 
@@ -331,16 +348,16 @@ __global__ void lstm_rec_bwd_kernel(
       bfr[cf] = *reinterpret_cast<const rbf16x8*>(
           w_hh_t + (long)orow * REC_G + ks * 32 + kchunk);
     }
-    rbf16x8 afr[2];
+    rbf16x8 afr[4];
 #pragma unroll
-    for (int rf = 0; rf < 2; ++rf) {
+    for (int rf = 0; rf < 4; ++rf) {
       const int arow = rf * 16 + frag_col;
       afr[rf] = *reinterpret_cast<const rbf16x8*>(
           gg_lds + rec_swz(arow * REC_GROW_B, arow,
                            ks * 64 + (lane >> 4) * 16));
     }
 #pragma unroll
-    for (int rf = 0; rf < 2; ++rf)
+    for (int rf = 0; rf < 4; ++rf)
 #pragma unroll
       for (int cf = 0; cf < 2; ++cf)
         acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -352,7 +369,7 @@ __global__ void lstm_rec_bwd_kernel(
   for (int cf = 0; cf < 2; ++cf) {
     const int j = j0 + cf * 16 + frag_col;
 #pragma unroll
-    for (int rf = 0; rf < 2; ++rf) {
+    for (int rf = 0; rf < 4; ++rf) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = rf * 16 + (lane >> 4) * 4 + r;
