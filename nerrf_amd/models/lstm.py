@@ -85,20 +85,20 @@ class BiLSTMDetector(nn.Module):
         ar = torch.arange(t, device=feats.device)
         mask = (ar.unsqueeze(1) < lengths.unsqueeze(0)).to(feats.dtype)  # [T, B]
         h = feats.transpose(0, 1).contiguous()  # time-major [T, B, E]
-        from ..ops.proj import dual_projection
+        from ..ops.lstm_seq import lstm_bilayer2
 
         for layer in self.dirs:
             f, r = layer[0], layer[1]
-            # both directions write one [T, B, 2H] buffer through the
-            # kernels' row-stride args — no torch.cat on the hot path;
-            # the input projections share one A pass (ops/proj.py)
+            # both directions in ONE projection GEMM ([M,K] x [K, 2*4H]:
+            # the A panel is read once instead of twice) and one [T,B,2H]
+            # output buffer; the layer backward then sees a single
+            # contiguous grad tensor, so dgrad and wgrad are single GEMMs
             flat = h.reshape(t * b, -1)
-            xg2_f, xg2_b = dual_projection(flat, f.w_ih, r.w_ih)
-            xg_f = xg2_f.reshape(t, b, 4 * f.hidden)
-            xg_b = xg2_b.reshape(t, b, 4 * r.hidden)
+            w_cat = torch.cat([f.w_ih, r.w_ih], dim=0)  # [2*4H, K]
+            xg2 = torch.matmul(flat, w_cat.t()).reshape(t, b, 8 * f.hidden)
             h0 = h.new_zeros(b, f.hidden)
             c0 = h.new_zeros(b, f.hidden)
-            h = lstm_bilayer(xg_f, xg_b, h0, c0, f.w_hh, f.b, r.w_hh, r.b, mask)
+            h = lstm_bilayer2(xg2, h0, c0, f.w_hh, f.b, r.w_hh, r.b, mask)
         # forward state at t=len-1, backward state at t=0
         idx = (lengths.clamp(min=1) - 1).view(1, b, 1).expand(1, b, self.cfg.hidden)
         h_fwd = h[:, :, : self.cfg.hidden].gather(0, idx).squeeze(0)  # [B, H]

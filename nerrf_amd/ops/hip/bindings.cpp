@@ -26,7 +26,7 @@ void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                bool, hipStream_t);
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
-                               void*, void*, long, int, long, bool,
+                               void*, void*, long, int, long, long, bool,
                                hipStream_t);
 void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
                             const void*, const float*, void*, void*, void*,
@@ -36,7 +36,7 @@ void launch_lstm_rec_fwd(const void*, const void*, const void*, const void*,
                          long, long, long, hipStream_t);
 void launch_lstm_rec_bwd(const void*, const void*, const void*, const void*,
                          const void*, const void*, const float*, void*, void*,
-                         void*, int, long, hipStream_t);
+                         void*, int, long, long, hipStream_t);
 void launch_proj_fwd_dual(const void*, const void*, const void*, void*, void*,
                           long, long, hipStream_t);
 void launch_proj_dgrad_dual(const void*, const void*, const void*, const void*,
@@ -184,7 +184,7 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
   check_gpu_contig(grad_c, "grad_c");
   check_gpu_contig(gates_act, "gates_act");
   check_gpu_contig(c_prev, "c_prev");
-  check_gpu_contig(grad_gates, "grad_gates");
+  const long gg_stride = row_stride_checked(grad_gates, "grad_gates");
   check_gpu_contig(grad_c_prev, "grad_c_prev");
   check_gpu_contig(grad_h_pass, "grad_h_pass");
   const long batch = c_prev.size(0);
@@ -204,7 +204,7 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
       grad_h.data_ptr(), got, grad_c.data_ptr(), gates_act.data_ptr(),
       c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
       grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
-      gout_stride, is_bf16(grad_h), stream.stream());
+      gout_stride, gg_stride, is_bf16(grad_h), stream.stream());
 }
 
 // Fused recurrent step (bf16, H == 256): h_prev @ W_hh^T + LSTM pointwise
@@ -258,7 +258,7 @@ void lstm_rec_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
   check_gpu_contig(gates_act, "gates_act");
   check_gpu_contig(c_prev, "c_prev");
   check_gpu_contig(w_hh_t, "w_hh_t");
-  check_gpu_contig(grad_gates, "grad_gates");
+  const long gg_stride = row_stride_checked(grad_gates, "grad_gates");
   check_gpu_contig(grad_c_prev, "grad_c_prev");
   check_gpu_contig(grad_h_out, "grad_h_out");
   for (auto* t : {&grad_h, &grad_c, &gates_act, &c_prev, &w_hh_t}) {
@@ -284,7 +284,7 @@ void lstm_rec_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
       grad_h.data_ptr(), got, grad_c.data_ptr(), gates_act.data_ptr(),
       c_prev.data_ptr(), w_hh_t.data_ptr(), mask_ptr, grad_gates.data_ptr(),
       grad_c_prev.data_ptr(), grad_h_out.data_ptr(), (int)batch, gout_stride,
-      stream.stream());
+      gg_stride, stream.stream());
 }
 
 // Fully-fused MFMA step (bf16, H == 256). Writes into caller buffers.

@@ -75,16 +75,17 @@ __global__ void lstm_pointwise_bwd_kernel(
     const T* __restrict__ gates_act,  // [B, 4H]
     const T* __restrict__ c_prev,     // [B, H]
     const float* __restrict__ mask,   // [B] or nullptr
-    T* __restrict__ grad_gates,       // [B, 4H]
+    T* __restrict__ grad_gates,       // [B(row-stride gg_stride), 4H]
     T* __restrict__ grad_c_prev,      // [B, H]
     T* __restrict__ grad_h_pass,      // [B, H]
-    long batch, int hdim, long gout_stride) {
+    long batch, int hdim, long gout_stride, long gg_stride) {
   const long total = batch * hdim;
   for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     const long b = t / hdim;
     const int d = (int)(t % hdim);
     const long g0 = b * 4 * hdim + d;
+    const long o0 = b * gg_stride + d;
     const float i = to_f32(gates_act[g0]);
     const float f = to_f32(gates_act[g0 + hdim]);
     const float g = to_f32(gates_act[g0 + 2 * hdim]);
@@ -104,10 +105,10 @@ __global__ void lstm_pointwise_bwd_kernel(
     const float d_g = d_c * i;
     grad_c_prev[t] = from_f32<T>(d_c * f + gc_in * (1.0f - m));
     grad_h_pass[t] = from_f32<T>(gh_in * (1.0f - m));
-    grad_gates[g0] = from_f32<T>(d_i * i * (1.0f - i));
-    grad_gates[g0 + hdim] = from_f32<T>(d_f * f * (1.0f - f));
-    grad_gates[g0 + 2 * hdim] = from_f32<T>(d_g * (1.0f - g * g));
-    grad_gates[g0 + 3 * hdim] = from_f32<T>(d_o * o * (1.0f - o));
+    grad_gates[o0] = from_f32<T>(d_i * i * (1.0f - i));
+    grad_gates[o0 + hdim] = from_f32<T>(d_f * f * (1.0f - f));
+    grad_gates[o0 + 2 * hdim] = from_f32<T>(d_g * (1.0f - g * g));
+    grad_gates[o0 + 3 * hdim] = from_f32<T>(d_o * o * (1.0f - o));
   }
 }
 
@@ -195,7 +196,8 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
     const T* __restrict__ grad_c, const T* __restrict__ gates_act,
     const T* __restrict__ c_prev, const float* __restrict__ mask,
     T* __restrict__ grad_gates, T* __restrict__ grad_c_prev,
-    T* __restrict__ grad_h_pass, long batch, int hdim, long gout_stride) {
+    T* __restrict__ grad_h_pass, long batch, int hdim, long gout_stride,
+    long gg_stride) {
   using VT = VecT<T, V>;
   const int hv = hdim / V;
   const long total = batch * hv;
@@ -204,6 +206,7 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
     const long b = t / hv;
     const int dv = (int)(t % hv) * V;
     const long g0 = b * 4 * hdim + dv;
+    const long o0 = b * gg_stride + dv;
     const long c0 = b * (long)hdim + dv;
     const VT ga_i = *reinterpret_cast<const VT*>(gates_act + g0);
     const VT ga_f = *reinterpret_cast<const VT*>(gates_act + g0 + hdim);
@@ -244,10 +247,10 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
     }
     *reinterpret_cast<VT*>(grad_c_prev + c0) = gcp_v;
     *reinterpret_cast<VT*>(grad_h_pass + c0) = ghp_v;
-    *reinterpret_cast<VT*>(grad_gates + g0) = gg_i;
-    *reinterpret_cast<VT*>(grad_gates + g0 + hdim) = gg_f;
-    *reinterpret_cast<VT*>(grad_gates + g0 + 2 * hdim) = gg_g;
-    *reinterpret_cast<VT*>(grad_gates + g0 + 3 * hdim) = gg_o;
+    *reinterpret_cast<VT*>(grad_gates + o0) = gg_i;
+    *reinterpret_cast<VT*>(grad_gates + o0 + hdim) = gg_f;
+    *reinterpret_cast<VT*>(grad_gates + o0 + 2 * hdim) = gg_g;
+    *reinterpret_cast<VT*>(grad_gates + o0 + 3 * hdim) = gg_o;
   }
 }
 
@@ -262,10 +265,10 @@ template __global__ void lstm_pointwise_fwd_vec_kernel<float, 4>(
 template __global__ void lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
-    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long);
+    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long, long);
 template __global__ void lstm_pointwise_bwd_vec_kernel<float, 4>(
     const float*, const float*, const float*, const float*, const float*,
-    const float*, float*, float*, float*, long, int, long);
+    const float*, float*, float*, float*, long, int, long, long);
 
 template __global__ void lstm_pointwise_fwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
@@ -276,10 +279,10 @@ template __global__ void lstm_pointwise_fwd_kernel<__hip_bfloat16>(
     long, int, long, long, long);
 template __global__ void lstm_pointwise_bwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
-    float*, float*, float*, long, int, long);
+    float*, float*, float*, long, int, long, long);
 template __global__ void lstm_pointwise_bwd_kernel<__hip_bfloat16>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
-    const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long);
+    const __hip_bfloat16*, const float*, __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long, long);
 
 // ---------------------------------------------------------------------------
 // host launchers
@@ -345,10 +348,11 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
                                const void* c_prev, const float* mask,
                                void* grad_gates, void* grad_c_prev,
                                void* grad_h_pass, long batch, int hdim,
-                               long gout_stride, bool bf16, hipStream_t s) {
+                               long gout_stride, long gg_stride, bool bf16,
+                               hipStream_t s) {
   const int block = 256;
   const int v = bf16 ? 8 : 4;
-  const bool vec = hdim % v == 0 && gout_stride % v == 0;
+  const bool vec = hdim % v == 0 && gout_stride % v == 0 && gg_stride % v == 0;
   if (bf16) {
     if (vec) {
       const int grid = grid_elems(batch * (hdim / v), block);
@@ -357,7 +361,7 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
           (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
           (const __hip_bfloat16*)c_prev, mask, (__hip_bfloat16*)grad_gates,
           (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch,
-          hdim, gout_stride);
+          hdim, gout_stride, gg_stride);
       return;
     }
     const int grid = grid_elems(batch * hdim, block);
@@ -366,7 +370,7 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
         (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
         (const __hip_bfloat16*)c_prev, mask, (__hip_bfloat16*)grad_gates,
         (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch, hdim,
-        gout_stride);
+        gout_stride, gg_stride);
   } else {
     if (vec) {
       const int grid = grid_elems(batch * (hdim / v), block);
@@ -374,14 +378,15 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
           (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
           (const float*)gates_act, (const float*)c_prev, mask,
           (float*)grad_gates, (float*)grad_c_prev, (float*)grad_h_pass, batch,
-          hdim, gout_stride);
+          hdim, gout_stride, gg_stride);
       return;
     }
     const int grid = grid_elems(batch * hdim, block);
     lstm_pointwise_bwd_kernel<float><<<grid, block, 0, s>>>(
         (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
         (const float*)gates_act, (const float*)c_prev, mask, (float*)grad_gates,
-        (float*)grad_c_prev, (float*)grad_h_pass, batch, hdim, gout_stride);
+        (float*)grad_c_prev, (float*)grad_h_pass, batch, hdim, gout_stride,
+        gg_stride);
   }
 }
 

@@ -245,10 +245,10 @@ __global__ void lstm_rec_bwd_kernel(
     const __hip_bfloat16* __restrict__ c_prev,     // [B, 256]
     const __hip_bfloat16* __restrict__ w_hh_t,     // [256, 1024] = W_hh^T contig
     const float* __restrict__ mask,                // [B] or nullptr
-    __hip_bfloat16* __restrict__ grad_gates,       // [B, 1024]
+    __hip_bfloat16* __restrict__ grad_gates,  // [B(row-stride gg_stride), 1024]
     __hip_bfloat16* __restrict__ grad_c_prev,      // [B, 256]
     __hip_bfloat16* __restrict__ grad_h_out,       // [B, 256]
-    int batch, long gout_stride) {
+    int batch, long gout_stride, long gg_stride) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* gg_lds = smem;                         // 128 KB, swizzled
   char* ghp_lds = smem + REC_BM * REC_GROW_B;  // 32 KB, linear bf16
@@ -310,7 +310,7 @@ __global__ void lstm_rec_bwd_kernel(
     // grad_gates to HBM (the weight-grad GEMMs read it) AND to LDS (the
     // in-launch grad_h GEMM reads it as swizzled A fragments)
     if (live) {
-      __hip_bfloat16* op = grad_gates + grow * REC_G + jc;
+      __hip_bfloat16* op = grad_gates + grow * gg_stride + jc;
       *reinterpret_cast<rbf16x8*>(op) = gg_i;
       *reinterpret_cast<rbf16x8*>(op + REC_H) = gg_f;
       *reinterpret_cast<rbf16x8*>(op + 2 * REC_H) = gg_g;
@@ -410,7 +410,7 @@ void launch_lstm_rec_bwd(const void* grad_h, const void* grad_out_t,
                          const void* c_prev, const void* w_hh_t,
                          const float* mask, void* grad_gates,
                          void* grad_c_prev, void* grad_h_out, int batch,
-                         long gout_stride, hipStream_t s) {
+                         long gout_stride, long gg_stride, hipStream_t s) {
   const int grid = (batch + REC_BM - 1) / REC_BM;
   const size_t lds = REC_BM * (REC_GROW_B + REC_HROW_B);  // 160 KB
   lstm_rec_bwd_kernel<<<grid, 512, lds, s>>>(
@@ -418,7 +418,7 @@ void launch_lstm_rec_bwd(const void* grad_h, const void* grad_out_t,
       (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
       (const __hip_bfloat16*)c_prev, (const __hip_bfloat16*)w_hh_t, mask,
       (__hip_bfloat16*)grad_gates, (__hip_bfloat16*)grad_c_prev,
-      (__hip_bfloat16*)grad_h_out, batch, gout_stride);
+      (__hip_bfloat16*)grad_h_out, batch, gout_stride, gg_stride);
 }
 
 }  // namespace nerrf

@@ -119,19 +119,26 @@ def _dir_forward(ext, xg, h0, c0, w_hh, bias, mask, reverse, infer, h_all):
 
 
 def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
-                  mask, reverse):
+                  mask, reverse, gg_out=None):
     """One direction's backward.
 
     grad_out: [T, B, H] (rows may be strided — a slab of the dual buffer).
     h_flat:   [T*B, H] flat view of this direction's hidden history with
               uniform row stride (slab of the dual buffer or contiguous).
+    gg_out:   optional [T, B, 4H] target for grad_gates — may be a slab of
+              a wider [T, B, 2*4H] buffer (rows contiguous), so the caller's
+              input-projection backward sees ONE contiguous gradient tensor
+              and runs as a single GEMM.
     Returns (grad_xg, grad_h0, grad_c0, grad_whh, grad_bias).
     """
     t_len, batch, hdim = grad_out.shape
     gdim = 4 * hdim
     dev, dt = c_all.device, c_all.dtype
 
-    grad_gates_all = torch.empty(t_len, batch, gdim, device=dev, dtype=dt)
+    grad_gates_all = (
+        gg_out if gg_out is not None
+        else torch.empty(t_len, batch, gdim, device=dev, dtype=dt)
+    )
     grad_h = torch.zeros(batch, hdim, device=dev, dtype=dt)
     grad_c = torch.zeros(batch, hdim, device=dev, dtype=dt)
     grad_h_pass = torch.empty(batch, hdim, device=dev, dtype=dt)
@@ -300,6 +307,77 @@ class _LSTMBiLayerFn(torch.autograd.Function):
         )
         return (gxf, gxb, gh0f + gh0b, gc0f + gc0b, gwf, gbf, gwb, gbb,
                 None, None)
+
+
+class _LSTMBiLayer2Fn(torch.autograd.Function):
+    """One bidirectional layer whose input projection is a SINGLE
+    [T, B, 2*4H] tensor (forward slab 0:4H, backward slab 4H:8H) — the
+    layout one concatenated-weights GEMM produces.  The backward writes
+    both directions' gate grads into one [T, B, 2*4H] buffer, so the
+    projection's dgrad/wgrad run as single GEMMs too (A panels read once)."""
+
+    @staticmethod
+    def forward(ctx, xg2, h0, c0, w_f, b_f, w_b, b_b, mask, infer: bool = False):
+        t_len, batch, gdim2 = xg2.shape
+        gdim = gdim2 // 2
+        hdim = gdim // 4
+        h2 = torch.empty(t_len, batch, 2 * hdim, device=xg2.device, dtype=xg2.dtype)
+        ext = get_native(xg2)
+        c_f, g_f = _dir_forward(
+            ext, xg2[:, :, :gdim], h0, c0, w_f, b_f, mask, False, infer,
+            h2[:, :, :hdim]
+        )
+        c_b, g_b = _dir_forward(
+            ext, xg2[:, :, gdim:], h0, c0, w_b, b_b, mask, True, infer,
+            h2[:, :, hdim:]
+        )
+        ctx.save_for_backward(
+            g_f, g_b, h2, c_f, c_b, h0, c0, w_f, w_b,
+            mask if mask is not None else torch.empty(0, device=xg2.device),
+        )
+        return h2
+
+    @staticmethod
+    def backward(ctx, grad2):
+        g_f, g_b, h2, c_f, c_b, h0, c0, w_f, w_b, mask_t = ctx.saved_tensors
+        mask = mask_t if mask_t.numel() else None
+        t_len, batch, hdim2 = h2.shape
+        hdim = hdim2 // 2
+        gdim = 4 * hdim
+        ext = get_native(h2)
+        grad2 = grad2.contiguous()
+        h2_flat = h2.reshape(t_len * batch, hdim2)
+        grad_xg2 = torch.empty(t_len, batch, 2 * gdim, device=h2.device,
+                               dtype=h2.dtype)
+        _, gh0f, gc0f, gwf, gbf = _dir_backward(
+            ext, grad2[:, :, :hdim], g_f, h2_flat[:, :hdim], c_f, h0, c0,
+            w_f, mask, False, gg_out=grad_xg2[:, :, :gdim],
+        )
+        _, gh0b, gc0b, gwb, gbb = _dir_backward(
+            ext, grad2[:, :, hdim:], g_b, h2_flat[:, hdim:], c_b, h0, c0,
+            w_b, mask, True, gg_out=grad_xg2[:, :, gdim:],
+        )
+        return (grad_xg2, gh0f + gh0b, gc0f + gc0b, gwf, gbf, gwb, gbb,
+                None, None)
+
+
+def lstm_bilayer2(
+    xg2: torch.Tensor,  # [T, B, 2*4H]: fwd projection 0:4H, bwd 4H:8H
+    h0: torch.Tensor,
+    c0: torch.Tensor,
+    w_f: torch.Tensor,
+    b_f: torch.Tensor,
+    w_b: torch.Tensor,
+    b_b: torch.Tensor,
+    mask: Optional[torch.Tensor] = None,  # [T, B]
+) -> torch.Tensor:
+    """Both directions of one layer from a single concatenated projection
+    -> [T, B, 2H].  Rows of xg2 must be contiguous (the tensor itself may
+    be a reshaped GEMM output)."""
+    m = mask.detach().contiguous().to(torch.float32) if mask is not None else None
+    infer = _infer_mode(xg2, h0, c0, w_f, b_f, w_b, b_b)
+    assert xg2.stride(2) == 1, "xg2 rows must be contiguous"
+    return _LSTMBiLayer2Fn.apply(xg2, h0, c0, w_f, b_f, w_b, b_b, m, infer)
 
 
 def _infer_mode(*tensors) -> bool:

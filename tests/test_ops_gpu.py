@@ -767,3 +767,49 @@ def test_proj_wgrad_matches_matmul(gpu_device):
     # f32 accumulate over bf16 products; tolerance scales with K=m
     assert torch.allclose(dw1, r1, atol=0.5, rtol=2e-2)
     assert torch.allclose(dw2, r2, atol=0.5, rtol=2e-2)
+
+
+def test_lstm_bilayer2_matches_bilayer(gpu_device):
+    """Single-projection-tensor bilayer (cat-GEMM layout, slab grads) vs
+    the two-tensor bilayer, fwd + bwd."""
+    from nerrf_amd.ops.lstm_seq import lstm_bilayer, lstm_bilayer2
+
+    torch.manual_seed(41)
+    t, b, hd = 5, 150, 256
+    gd = 4 * hd
+    xf32 = torch.randn(t, b, gd, device=gpu_device) * 0.4
+    xb32 = torch.randn(t, b, gd, device=gpu_device) * 0.4
+    wf32 = torch.randn(gd, hd, device=gpu_device) * 0.05
+    wb32 = torch.randn(gd, hd, device=gpu_device) * 0.05
+    bf32 = torch.randn(gd, device=gpu_device) * 0.5
+    bb32 = torch.randn(gd, device=gpu_device) * 0.5
+    mask = (torch.rand(t, b, device=gpu_device) > 0.2).float()
+    h0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    c0 = torch.zeros(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    g_out = torch.randn(t, b, 2 * hd, device=gpu_device).to(torch.bfloat16)
+
+    # two-tensor path
+    xf = xf32.to(torch.bfloat16).requires_grad_(True)
+    xb = xb32.to(torch.bfloat16).requires_grad_(True)
+    wf = wf32.to(torch.bfloat16).requires_grad_(True)
+    wb = wb32.to(torch.bfloat16).requires_grad_(True)
+    bf = bf32.to(torch.bfloat16).requires_grad_(True)
+    bb = bb32.to(torch.bfloat16).requires_grad_(True)
+    out_a = lstm_bilayer(xf, xb, h0, c0, wf, bf, wb, bb, mask)
+    out_a.backward(g_out)
+
+    # single-tensor path
+    xg2 = torch.cat([xf32, xb32], dim=-1).to(torch.bfloat16).requires_grad_(True)
+    wf2 = wf32.to(torch.bfloat16).requires_grad_(True)
+    wb2 = wb32.to(torch.bfloat16).requires_grad_(True)
+    bf2 = bf32.to(torch.bfloat16).requires_grad_(True)
+    bb2 = bb32.to(torch.bfloat16).requires_grad_(True)
+    out_b = lstm_bilayer2(xg2, h0, c0, wf2, bf2, wb2, bb2, mask)
+    out_b.backward(g_out)
+
+    assert torch.allclose(out_a.float(), out_b.float(), atol=1e-5)
+    gx2 = torch.cat([xf.grad, xb.grad], dim=-1).float()
+    assert torch.allclose(gx2, xg2.grad.float(), atol=1e-5)
+    for a, c, name in ((wf.grad, wf2.grad, "wf"), (wb.grad, wb2.grad, "wb"),
+                       (bf.grad, bf2.grad, "bf"), (bb.grad, bb2.grad, "bb")):
+        assert torch.allclose(a.float(), c.float(), atol=1e-5), name
