@@ -47,12 +47,15 @@ def _dir_forward(ext, xg, h0, c0, w_hh, bias, mask, reverse, infer, h_all):
     steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
     if ext is not None:
         empty_mask = torch.empty(0, device=dev)
-        # Default bf16/H=256 path: the fused recurrent step
+        # Opt-in bf16/H=256 path: the fused recurrent step
         # (lstm_rec_fused.hip) keeps the [B, 4H] pre-activation slab out of
-        # HBM entirely — one launch per timestep instead of GEMM+pointwise.
+        # HBM entirely (one launch per timestep), but measured at parity
+        # fwd / -13% bwd vs the split path: its LDS transpose stage caps
+        # the kernel at 1 block/CU and the memory phases go latency-bound
+        # (~2.5 TB/s vs the split pointwise's 5.3 — profiles/PROFILES.md).
         rec_fused = (
             hasattr(ext, "lstm_rec_fwd")
-            and os.environ.get("NERRF_REC_FUSED", "1") == "1"
+            and os.environ.get("NERRF_REC_FUSED", "0") == "1"
             and dt == torch.bfloat16 and hdim == 256 and w_hh.is_contiguous()
         )
         if rec_fused:
@@ -150,7 +153,7 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
     rec_fused = (
         ext is not None
         and hasattr(ext, "lstm_rec_bwd")
-        and os.environ.get("NERRF_REC_FUSED", "1") == "1"
+        and os.environ.get("NERRF_REC_FUSED", "0") == "1"
         and dt == torch.bfloat16 and hdim == 256
     )
     if rec_fused:
