@@ -456,7 +456,7 @@ torch::Tensor sage_layer_fwd(torch::Tensor h, torch::Tensor nbr_idx,
 nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
   nerrf::PlannerParamsDev p;
   p.n_groups = d["n_groups"].cast<int>();
-  p.n_actions = p.n_groups + 2;
+  p.n_actions = p.n_groups + 3;  // STOP, KILL, RESTORE, revert g
   p.max_depth = d["max_depth"].cast<int>();
   p.sims_per_tree = d["sims_per_tree"].cast<int>();
   p.downtime_weight = d["downtime_weight"].cast<float>();
@@ -465,6 +465,8 @@ nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
   p.fp_weight = d["fp_weight"].cast<float>();
   p.attack_rate_mbps = d["attack_rate_mbps"].cast<float>();
   p.horizon_s = d["horizon_s"].cast<float>();
+  p.restore_time_s = d["restore_time_s"].cast<float>();
+  p.restore_loss_mb = d["restore_loss_mb"].cast<float>();
   p.ucb_c = d["ucb_c"].cast<float>();
   p.seed = d["seed"].cast<unsigned>();
   TORCH_CHECK(p.n_groups <= 16, "n_groups must be <= 16");

@@ -15,11 +15,12 @@
 namespace nerrf {
 
 #define MAX_GROUPS 16
-#define MAX_ACTIONS (2 + MAX_GROUPS)
+#define MAX_ACTIONS (3 + MAX_GROUPS)
 #define MAX_DEPTH 16
 #define A_STOP 0
 #define A_KILL 1
-#define A_REVERT_BASE 2
+#define A_RESTORE 2
+#define A_REVERT_BASE 3
 
 __device__ __forceinline__ unsigned xorshift32_dev(unsigned s) {
   s ^= s << 13;
@@ -36,8 +37,8 @@ __device__ float eval_plan(const float* gscore, const float* gmb,
                            const PlannerParamsDev& p) {
   float reverted[MAX_GROUPS];
   for (int g = 0; g < p.n_groups; ++g) reverted[g] = 0.0f;
-  float fp_mb = 0.0f, downtime = 0.0f, ongoing = 0.0f;
-  bool alive = true;
+  float fp_mb = 0.0f, downtime = 0.0f, ongoing = 0.0f, staleness = 0.0f;
+  bool alive = true, restored = false;
   for (int i = 0; i < n_act; ++i) {
     const int a = actions[i];
     if (a == A_STOP) break;
@@ -46,6 +47,16 @@ __device__ float eval_plan(const float* gscore, const float* gmb,
         downtime += p.kill_time_s;
         alive = false;
       }
+      continue;
+    }
+    if (a == A_RESTORE) {
+      if (restored) continue;
+      const float dt = p.restore_time_s;
+      if (alive) ongoing += p.attack_rate_mbps * dt * proc_score;
+      downtime += dt;
+      for (int g = 0; g < p.n_groups; ++g) reverted[g] = 1.0f;
+      staleness = p.restore_loss_mb;
+      restored = true;
       continue;
     }
     const int gi = a - A_REVERT_BASE;
@@ -58,12 +69,72 @@ __device__ float eval_plan(const float* gscore, const float* gmb,
   }
   if (alive) {
     ongoing += p.attack_rate_mbps * p.horizon_s * proc_score;
-    downtime += p.horizon_s;
+    // degradation charge scales with belief the process is malicious
+    downtime += p.horizon_s * proc_score;
   }
   float loss = 0.0f;
   for (int g = 0; g < p.n_groups; ++g)
     loss += (1.0f - reverted[g]) * gscore[g] * gmb[g];
   loss += fminf(ongoing, remaining_clean_mb);
+  loss += staleness;
+  return -(loss + p.downtime_weight * downtime + p.fp_weight * fp_mb);
+}
+
+// Wave-vectorized eval_plan: group g's stats live in lane g's registers
+// (loaded once per tree instead of per simulation); per-action group
+// lookups are shuffles; the final loss reduces by a SEQUENTIAL shuffle
+// chain so the fp32 accumulation order matches eval_plan / the CPU
+// reference bit-for-bit.  All 64 lanes execute the (wave-uniform) control
+// flow; only the group-indexed state is lane-local.
+__device__ float eval_plan_wave(float my_score, float my_mb, float my_files,
+                                float proc_score, float remaining_clean_mb,
+                                const int* actions, int n_act,
+                                const PlannerParamsDev& p, int lane) {
+  float my_rev = 0.0f;
+  float fp_mb = 0.0f, downtime = 0.0f, ongoing = 0.0f, staleness = 0.0f;
+  bool alive = true, restored = false;
+  for (int i = 0; i < n_act; ++i) {
+    const int a = actions[i];
+    if (a == A_STOP) break;
+    if (a == A_KILL) {
+      if (alive) {
+        downtime += p.kill_time_s;
+        alive = false;
+      }
+      continue;
+    }
+    if (a == A_RESTORE) {
+      if (restored) continue;
+      const float dt = p.restore_time_s;
+      if (alive) ongoing += p.attack_rate_mbps * dt * proc_score;
+      downtime += dt;
+      my_rev = 1.0f;
+      staleness = p.restore_loss_mb;
+      restored = true;
+      continue;
+    }
+    const int gi = a - A_REVERT_BASE;
+    if (gi < 0 || gi >= p.n_groups) continue;
+    if (__shfl(my_rev, gi, NERRF_WAVE) > 0.0f) continue;
+    const float files_g = __shfl(my_files, gi, NERRF_WAVE);
+    const float score_g = __shfl(my_score, gi, NERRF_WAVE);
+    const float mb_g = __shfl(my_mb, gi, NERRF_WAVE);
+    const float dt = p.revert_time_s * fmaxf(files_g, 1.0f);
+    if (alive) ongoing += p.attack_rate_mbps * dt * proc_score;
+    downtime += dt;
+    if (lane == gi) my_rev = 1.0f;
+    fp_mb += (1.0f - score_g) * mb_g * 0.05f;
+  }
+  if (alive) {
+    ongoing += p.attack_rate_mbps * p.horizon_s * proc_score;
+    downtime += p.horizon_s * proc_score;
+  }
+  const float my_term = (1.0f - my_rev) * my_score * my_mb;
+  float loss = 0.0f;
+  for (int g = 0; g < p.n_groups; ++g)  // sequential order == scalar order
+    loss += __shfl(my_term, g, NERRF_WAVE);
+  loss += fminf(ongoing, remaining_clean_mb);
+  loss += staleness;
   return -(loss + p.downtime_weight * downtime + p.fp_weight * fp_mb);
 }
 
@@ -84,29 +155,36 @@ __global__ void mcts_kernel(
   const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / NERRF_WAVE;
   const int lane = threadIdx.x % NERRF_WAVE;
   if (wave >= n_trees) return;
-  if (lane != 0) return;  // lane 0 drives; tree walk is scalar by nature
 
+  // All 64 lanes run the (wave-uniform) tree walk redundantly; arena WRITES
+  // are lane-0-guarded, and the per-simulation reward evaluates wave-wide
+  // with group state lane-resident (eval_plan_wave) — the v1 kernel parked
+  // 63/64 lanes for the whole search.
   int* parent = arena_i + (long)wave * (3 * cap + cap * p.n_actions);
   int* action = parent + cap;
   int* visits = action + cap;
   int* children = visits + cap;
   float* value = arena_f + (long)wave * cap;
 
-  // init arena
-  for (int i = 0; i < cap; ++i) {
+  const float my_score = (lane < p.n_groups) ? gscore[lane] : 0.0f;
+  const float my_mb = (lane < p.n_groups) ? gmb[lane] : 0.0f;
+  const float my_files = (lane < p.n_groups) ? gfiles[lane] : 0.0f;
+
+  // init arena (lane-parallel)
+  for (int i = lane; i < cap; i += NERRF_WAVE) {
     parent[i] = -1;
     action[i] = -1;
     visits[i] = 0;
     value[i] = 0.0f;
   }
-  for (int i = 0; i < cap * p.n_actions; ++i) children[i] = -1;
+  for (int i = lane; i < cap * p.n_actions; i += NERRF_WAVE) children[i] = -1;
   int n_nodes = 1;
 
   unsigned rng = (p.seed * 2654435761u + (unsigned)wave * 40503u + 1u);
   int acts[MAX_DEPTH + 1];
 
   for (int si = 0; si < p.sims_per_tree; ++si) {
-    // ---- selection + expansion ----
+    // ---- selection + expansion (redundant on every lane) ----
     int node = 0, depth = 0;
     while (depth < p.max_depth) {
       int* kids = children + node * p.n_actions;
@@ -116,11 +194,15 @@ __global__ void mcts_kernel(
       if (untried >= 0) {
         if (n_nodes >= cap) break;  // arena full: treat as leaf
         const int nw = n_nodes++;
-        parent[nw] = node;
-        action[nw] = untried;
-        kids[untried] = nw;
+        if (lane == 0) {
+          parent[nw] = node;
+          action[nw] = untried;
+          kids[untried] = nw;
+        }
+        // lanes track the walk in registers; memory catches up via lane 0
         node = nw;
         ++depth;
+        acts[depth - 1] = untried;  // kept in sync below
         break;
       }
       // UCB1 over fully-expanded children
@@ -141,17 +223,13 @@ __global__ void mcts_kernel(
         if (u > best_u) { best_u = u; best = a; }
       }
       node = kids[best];
+      acts[depth] = best;
       ++depth;
-      if (action[node] == A_STOP) break;
+      if (best == A_STOP) break;
     }
+    const int path_len = depth;  // acts[0:depth] = root->node actions
     // ---- rollout ----
-    int n_act = 0;
-    {  // reconstruct path actions root->node
-      int chain[MAX_DEPTH];
-      int cl = 0;
-      for (int nd = node; nd != 0 && cl < MAX_DEPTH; nd = parent[nd]) chain[cl++] = action[nd];
-      for (int i = cl - 1; i >= 0; --i) acts[n_act++] = chain[i];
-    }
+    int n_act = path_len;
     rng = xorshift32_dev(rng ^ ((unsigned)si * 747796405u + 2891336453u));
     unsigned r = rng;
     while (n_act < p.max_depth && (n_act == 0 || acts[n_act - 1] != A_STOP)) {
@@ -160,21 +238,26 @@ __global__ void mcts_kernel(
       acts[n_act++] = a;
       if (a == A_STOP) break;
     }
-    const float reward = eval_plan(gscore, gmb, gfiles, proc_score,
-                                   remaining_clean_mb, acts, n_act, p);
-    // ---- backup ----
-    for (int nd = node; nd >= 0; nd = parent[nd]) {
-      visits[nd] += 1;
-      value[nd] += reward;
+    const float reward = eval_plan_wave(my_score, my_mb, my_files, proc_score,
+                                        remaining_clean_mb, acts, n_act, p,
+                                        lane);
+    // ---- backup (lane 0 writes; all lanes know the path in registers) ----
+    if (lane == 0) {
+      for (int nd = node; nd >= 0; nd = parent[nd]) {
+        visits[nd] += 1;
+        value[nd] += reward;
+      }
     }
   }
 
   // export root child statistics
-  int* kids = children;  // node 0
-  for (int a = 0; a < p.n_actions; ++a) {
-    const int ch = kids[a];
-    root_visits[(long)wave * p.n_actions + a] = (ch >= 0) ? visits[ch] : 0;
-    root_value[(long)wave * p.n_actions + a] = (ch >= 0) ? value[ch] : 0.0f;
+  if (lane == 0) {
+    int* kids = children;  // node 0
+    for (int a = 0; a < p.n_actions; ++a) {
+      const int ch = kids[a];
+      root_visits[(long)wave * p.n_actions + a] = (ch >= 0) ? visits[ch] : 0;
+      root_value[(long)wave * p.n_actions + a] = (ch >= 0) ? value[ch] : 0.0f;
+    }
   }
 }
 

@@ -14,6 +14,8 @@ struct PlannerParamsDev {
   float fp_weight;
   float attack_rate_mbps;
   float horizon_s;
+  float restore_time_s;
+  float restore_loss_mb;
   float ucb_c;
   unsigned seed;
 };

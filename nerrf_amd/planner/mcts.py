@@ -19,7 +19,8 @@ from typing import List, Optional, Tuple
 
 import numpy as np
 
-from .rewards import A_KILL, A_REVERT_BASE, A_STOP, PlannerParams, PlannerState, simulate_plan
+from .rewards import (A_KILL, A_RESTORE, A_REVERT_BASE, A_STOP, PlannerParams,
+                      PlannerState, simulate_plan)
 
 UCB_C = 1.2
 
@@ -45,6 +46,8 @@ class PlanResult:
         for a in self.plan:
             if a == A_KILL:
                 out.append("kill_process")
+            elif a == A_RESTORE:
+                out.append("restore_from_backup")
             elif a >= A_REVERT_BASE:
                 out.append(f"revert_group_{a - A_REVERT_BASE}")
         return out
@@ -84,7 +87,7 @@ def run_mcts(
 ) -> PlanResult:
     """Root-parallel MCTS: n_sims total simulations over n_sims//sims_per_tree trees."""
     params = params or PlannerParams()
-    n_actions = 2 + state.n_groups  # STOP, KILL, revert g
+    n_actions = 3 + state.n_groups  # STOP, KILL, RESTORE, revert g
     n_trees = max(1, n_sims // sims_per_tree)
     root_visits = np.zeros(n_actions, dtype=np.int64)
     root_value = np.zeros(n_actions, dtype=np.float64)
@@ -164,7 +167,7 @@ def _result_from_root_stats(
     root_value: np.ndarray,
     total_sims: int,
 ) -> PlanResult:
-    n_actions = 2 + state.n_groups
+    n_actions = 3 + state.n_groups
     mean_val = np.where(root_visits > 0, root_value / np.maximum(root_visits, 1), -np.inf)
     ranked = sorted(
         [(int(a), float(mean_val[a]), int(root_visits[a])) for a in range(n_actions)],
@@ -211,6 +214,8 @@ def run_mcts_gpu(
         "fp_weight": params.fp_weight,
         "attack_rate_mbps": params.attack_rate_mbps,
         "horizon_s": params.horizon_s,
+        "restore_time_s": params.restore_time_s,
+        "restore_loss_mb": params.restore_loss_mb,
         "ucb_c": UCB_C,
         "seed": seed & 0xFFFFFFFF,
     }

@@ -25,7 +25,8 @@ import numpy as np
 # action ids (shared with the HIP kernel — keep in sync with mcts.hip)
 A_STOP = 0
 A_KILL = 1
-A_REVERT_BASE = 2  # A_REVERT_BASE + g : revert file group g
+A_RESTORE = 2  # restore everything from backup (reference threat-model.mdx:206-222)
+A_REVERT_BASE = 3  # A_REVERT_BASE + g : revert file group g
 
 
 @dataclass
@@ -38,6 +39,11 @@ class PlannerParams:
     attack_rate_mbps: float = 2.0  # reference simulator rate limit
     horizon_s: float = 60.0
     max_depth: int = 10
+    # restore-from-backup (reference threat-model.mdx:206-222: "cost 100 —
+    # data loss risk, confidence 1.0"): recovers every group at a large
+    # fixed downtime plus the data written since the last backup
+    restore_time_s: float = 30.0
+    restore_loss_mb: float = 16.0
 
 
 @dataclass
@@ -99,6 +105,8 @@ def simulate_plan(
     fp_mb = 0.0
     downtime = 0.0
     alive = True
+    restored = False
+    staleness = 0.0
     ongoing = 0.0  # extra MB encrypted while we act
     for a in actions:
         if a == A_STOP:
@@ -107,6 +115,17 @@ def simulate_plan(
             if alive:
                 downtime += params.kill_time_s
                 alive = False
+            continue
+        if a == A_RESTORE:
+            if restored:
+                continue
+            dt = params.restore_time_s
+            if alive:
+                ongoing += params.attack_rate_mbps * dt * state.proc_score
+            downtime += dt
+            reverted[:] = 1.0
+            staleness = params.restore_loss_mb
+            restored = True
             continue
         gi = a - A_REVERT_BASE
         if gi < 0 or gi >= g or reverted[gi] > 0:
@@ -119,6 +138,12 @@ def simulate_plan(
         fp_mb += (1.0 - state.group_score[gi]) * state.group_mb[gi] * 0.05
     if alive:  # attack keeps running until the horizon
         ongoing += params.attack_rate_mbps * params.horizon_s * state.proc_score
-        downtime += params.horizon_s
-    loss = float(((1.0 - reverted) * unrec).sum()) + min(ongoing, state.remaining_clean_mb)
+        # the service-degradation charge scales with our belief the process
+        # is actually malicious: a clean state pays ~nothing for inaction
+        downtime += params.horizon_s * state.proc_score
+    loss = (
+        float(((1.0 - reverted) * unrec).sum())
+        + min(ongoing, state.remaining_clean_mb)
+        + staleness
+    )
     return -(loss + params.downtime_weight * downtime + params.fp_weight * fp_mb)

@@ -44,6 +44,9 @@ class Detection:
     exfil_destinations: List[str] = field(default_factory=list)
     indicators: Dict[str, float] = field(default_factory=dict)
     window_events: int = 0
+    # counterfactual-refinement context (node features + sampled adjacency +
+    # path->node map) for model-in-the-loop plan scoring; host-side numpy
+    refine_ctx: Optional[dict] = None
 
 
 PRETRAINED_DIR = "checkpoints/pretrained"
@@ -284,6 +287,28 @@ class StreamingEngine:
         seq_max = float(seq_score.max()) if seq_score is not None and len(seq_score) else 0.0
         alarm_score = max(ind_score, min(model_max, seq_max))
         alarm = alarm_score >= self.alarm_threshold
+        # refinement context: lets plan() re-score counterfactual post-plan
+        # graphs through the GNN in one batch (planner/model_eval.py)
+        path_to_node = {
+            strings[k]: i for i, k in enumerate(node_key[:n_files].tolist())
+        }
+        proc_nodes = _np.array(
+            [n_files + i for i, s_ in enumerate(ns[n_files : len(node_kind)])
+             if s_ >= 0.5],
+            dtype=_np.int64,
+        )
+        if len(proc_nodes) == 0 and len(node_kind) > n_files:
+            proc_nodes = _np.array(
+                [n_files + int(_np.argmax(ns[n_files : len(node_kind)]))],
+                dtype=_np.int64,
+            )
+        refine_ctx = {
+            "x": (x.float().cpu().numpy() if hasattr(x, "cpu") else _np.asarray(x)),
+            "nbr_idx": nbr_idx,
+            "nbr_w": nbr_w,
+            "path_to_node": path_to_node,
+            "proc_nodes": proc_nodes,
+        }
         return Detection(
             alarm=alarm,
             t_detect=t_detect,
@@ -294,10 +319,12 @@ class StreamingEngine:
             exfil_destinations=exfil_dests,
             indicators=indicators,
             window_events=len(events),
+            refine_ctx=refine_ctx,
         )
 
     # -------------------------------------------------------------------- plan
-    def plan(self, det: Detection, n_sims: int = 1024, use_gpu: Optional[bool] = None) -> PlanResult:
+    def plan(self, det: Detection, n_sims: int = 1024, use_gpu: Optional[bool] = None,
+             model_refine: bool = True) -> PlanResult:
         paths = list(det.file_scores.keys())
         scores = np.array([det.file_scores[p] for p in paths], dtype=np.float64)
         mb = np.array([max(det.file_mb.get(p, 0.01), 0.01) for p in paths], dtype=np.float64)
@@ -310,8 +337,56 @@ class StreamingEngine:
         if use_gpu is None:
             use_gpu = self.device.type == "cuda"
         if use_gpu:
-            return run_mcts_gpu(state, self.planner_params, n_sims=n_sims, device=str(self.device))
-        return run_mcts(state, self.planner_params, n_sims=n_sims)
+            res = run_mcts_gpu(state, self.planner_params, n_sims=n_sims, device=str(self.device))
+        else:
+            res = run_mcts(state, self.planner_params, n_sims=n_sims)
+        if model_refine and det.refine_ctx is not None and paths:
+            res = self._refine_plan(res, det, state, paths, scores)
+        return res
+
+    def _refine_plan(self, res: PlanResult, det: Detection, state, paths, scores) -> PlanResult:
+        """Batched counterfactual re-scoring of candidate plans through the
+        GNN node head (SURVEY §7 "batched leaf evaluation")."""
+        from ..planner.model_eval import refine_plans_with_model
+        from ..planner.rewards import A_KILL, A_RESTORE, A_STOP
+
+        ctx = det.refine_ctx
+        # planner-group -> node-id map, mirroring build_state's bucketing
+        order = np.argsort(-scores, kind="stable")
+        split = np.array_split(order, self.planner_params.n_groups)
+        p2n = ctx["path_to_node"]
+        group_nodes = [
+            np.array([p2n[paths[i]] for i in ids if paths[i] in p2n], dtype=np.int64)
+            for ids in split
+        ]
+        # candidate set: the MCTS plan plus structured alternatives
+        cands = [list(res.plan)]
+        top_reverts = [a for a, _v, _n in res.ranked_actions if a >= 3][:3]
+        cands.append([A_KILL] + top_reverts)
+        cands.append([A_KILL, A_RESTORE])
+        if A_KILL not in res.plan and res.plan:
+            cands.append([A_KILL] + list(res.plan))
+        uniq, seen = [], set()
+        for c in cands:
+            key = tuple(c)
+            if key not in seen:
+                seen.add(key)
+                uniq.append(c)
+        try:
+            ranked = refine_plans_with_model(
+                self.model, ctx["x"], ctx["nbr_idx"], ctx["nbr_w"], group_nodes,
+                ctx["proc_nodes"], uniq, state, self.planner_params,
+                device=str(self.device), dtype=self.dtype,
+            )
+        except Exception:
+            return res  # refinement is advisory; the MCTS plan stands
+        best_plan, combined, closed, residual = ranked[0]
+        return PlanResult(
+            plan=[a for a in best_plan if a != A_STOP],
+            ranked_actions=res.ranked_actions,
+            root_value=closed,
+            simulations=res.simulations,
+        )
 
     # ----------------------------------------------------------------- respond
     def respond(
