@@ -117,6 +117,7 @@ def build_sequences_torch(
     seq_len: int = SEQ_LEN,
     min_events: int = 2,
     dtype=None,
+    dev_cols=None,
 ):
     """GPU sequence assembly: same semantics as build_sequences, in torch.
 
@@ -139,12 +140,20 @@ def build_sequences_torch(
 
     if not len(events):
         return empty()
-    t_path = torch.from_numpy(events.path_id).to(dev, non_blocking=True)
-    t_newp = torch.from_numpy(events.new_path_id).to(dev, non_blocking=True)
-    t_sc = torch.from_numpy(events.syscall).to(dev, non_blocking=True)
-    t_nb = torch.from_numpy(np.ascontiguousarray(events.nbytes, dtype=np.float32)).to(dev, non_blocking=True)
-    t_ts = torch.from_numpy(events.ts).to(dev, non_blocking=True)  # float64: epoch ts
-    t_pid = torch.from_numpy(events.pid).to(dev, non_blocking=True)
+    if dev_cols is not None:  # HBM delta ring: columns already on-device
+        t_path = dev_cols["path_id"]
+        t_newp = dev_cols["new_path_id"]
+        t_sc = dev_cols["syscall"]
+        t_nb = dev_cols["nbytes_f32"]
+        t_ts = dev_cols["ts"]
+        t_pid = dev_cols["pid"]
+    else:
+        t_path = torch.from_numpy(events.path_id).to(dev, non_blocking=True)
+        t_newp = torch.from_numpy(events.new_path_id).to(dev, non_blocking=True)
+        t_sc = torch.from_numpy(events.syscall).to(dev, non_blocking=True)
+        t_nb = torch.from_numpy(np.ascontiguousarray(events.nbytes, dtype=np.float32)).to(dev, non_blocking=True)
+        t_ts = torch.from_numpy(events.ts).to(dev, non_blocking=True)  # float64: epoch ts
+        t_pid = torch.from_numpy(events.pid).to(dev, non_blocking=True)
     t_sus = torch.from_numpy(sus_np).to(dev, non_blocking=True)
 
     idx = (t_path >= 0).nonzero(as_tuple=True)[0]

@@ -31,6 +31,7 @@ def gpu_window_graph(
     dtype: torch.dtype = torch.float32,
     parts: Optional[dict] = None,
     ed: Optional[dict] = None,
+    dev_cols: Optional[dict] = None,
 ) -> Optional[dict]:
     """Build one window's graph with GPU feature compaction.
 
@@ -72,13 +73,21 @@ def gpu_window_graph(
             t = t.to(dt)
         return t.to(device, non_blocking=True)
 
-    ts_ms = np.round((events.ts - t0) * 1000.0).astype(np.int32)
+    if dev_cols is not None:
+        # HBM delta ring: event columns already resident on-device
+        t_sc = dev_cols["syscall"]
+        t_nb = dev_cols["nbytes_f32"]
+        t_ts_ms = ((dev_cols["ts"] - t0) * 1000.0).round().to(torch.int32)
+    else:
+        t_sc = dev(events.syscall)
+        t_nb = dev(events.nbytes.astype(np.float32))
+        t_ts_ms = dev(np.round((events.ts - t0) * 1000.0).astype(np.int32))
     x = ext.event_features(
         dev(parts["ev_file"]),
         dev(parts["ev_proc"]),
-        dev(events.syscall),
-        dev(events.nbytes.astype(np.float32)),
-        dev(ts_ms),
+        t_sc,
+        t_nb,
+        t_ts_ms,
         n_nodes,
         dev(ed["in_deg"].astype(np.float32)),
         dev(ed["out_deg"].astype(np.float32)),

@@ -135,6 +135,7 @@ def merge_window(
     summaries: List[DeltaSummary],
     causality_tau_s: float = 10.0,
     device=None,
+    dev_cols=None,
 ) -> Tuple[dict, dict]:
     """Merge per-delta summaries into (parts, ed) — drop-in for
     build_graph_parts + build_edges_and_flags over the same events.
@@ -203,7 +204,14 @@ def merge_window(
         # of the derived columns — same bytes over PCIe)
         import torch
 
-        t_pi = torch.from_numpy(pi).to(device, non_blocking=True)
+        # path_id/pid come from the HBM delta ring when available (shipped
+        # once per sealed delta) instead of re-crossing PCIe every tick
+        if dev_cols is not None:
+            t_pi = dev_cols["path_id"]
+            t_pid = dev_cols["pid"]
+        else:
+            t_pi = torch.from_numpy(pi).to(device, non_blocking=True)
+            t_pid = torch.from_numpy(events.pid).to(device, non_blocking=True)
         t_root = torch.from_numpy(path_root).to(device, non_blocking=True)
         t_rtf = torch.from_numpy(root_to_file).to(device, non_blocking=True)
         ev_file = torch.where(
@@ -211,7 +219,6 @@ def merge_window(
             t_rtf[t_root[t_pi.clamp(min=0)]],
             torch.full((), -1, dtype=torch.int64, device=device),
         )
-        t_pid = torch.from_numpy(events.pid).to(device, non_blocking=True)
         if pid_lut is not None:
             t_lut = torch.from_numpy(pid_lut).to(device, non_blocking=True)
             ev_proc = n_files + t_lut[t_pid]

@@ -37,6 +37,10 @@ class DeltaGraphStore:
         self._lock = threading.Lock()
         self.total_events = 0
         self.evicted_events = 0
+        # per-delta DEVICE column cache (HBM-resident delta ring, SURVEY
+        # §2a): sealed deltas ship to the GPU once; the window view is a
+        # device-side cat instead of a host concat + re-upload every tick
+        self._dev_cache: dict = {}
 
     def append(self, ts, pid, syscall, path="", new_path="", nbytes=0, ret_val=0, comm="") -> None:
         """Append one event (shares the store-wide string tables)."""
@@ -182,6 +186,58 @@ class DeltaGraphStore:
                 cols = {k: v[keep] for k, v in cols.items()}
         arr = EventArray(paths=self.paths, comms=self.comms, **cols)
         return arr.sort_by_time(), deltas
+
+    def device_columns(self, deltas: list, device):
+        """Concatenated device columns for `deltas` (the HBM delta ring).
+
+        Each sealed delta's numeric columns are copied to `device` exactly
+        once and cached by delta identity; dead entries evict with the
+        window.  Returns a dict of device tensors in compact_with_deltas'
+        event order, or None when the delta concatenation is not already
+        time-ordered (compact would re-sort and the orders would diverge —
+        out-of-order ingest falls back to the host path).
+        """
+        import torch
+
+        if not deltas:
+            return None
+        prev_end = -np.inf
+        for d in deltas:
+            if len(d) == 0:
+                continue
+            if float(d.ts[0]) < prev_end:
+                return None
+            prev_end = float(d.ts[-1])
+        live = set()
+        cols_list = []
+        for d in deltas:
+            if len(d) == 0:
+                continue
+            k = id(d)
+            live.add(k)
+            ent = self._dev_cache.get(k)
+            if ent is None or ent[0] is not d:
+                ent = (d, {
+                    "ts": torch.from_numpy(d.ts).to(device, non_blocking=True),
+                    "pid": torch.from_numpy(d.pid).to(device, non_blocking=True),
+                    "syscall": torch.from_numpy(d.syscall).to(device, non_blocking=True),
+                    "path_id": torch.from_numpy(d.path_id).to(device, non_blocking=True),
+                    "new_path_id": torch.from_numpy(d.new_path_id).to(device, non_blocking=True),
+                    "nbytes_f32": torch.from_numpy(
+                        np.ascontiguousarray(d.nbytes, dtype=np.float32)
+                    ).to(device, non_blocking=True),
+                })
+                self._dev_cache[k] = ent
+            cols_list.append(ent[1])
+        for k in list(self._dev_cache):
+            if k not in live:
+                del self._dev_cache[k]
+        if not cols_list:
+            return None
+        if len(cols_list) == 1:
+            return dict(cols_list[0])
+        return {key: torch.cat([c[key] for c in cols_list])
+                for key in cols_list[0]}
 
     @property
     def window_event_count(self) -> int:

@@ -135,13 +135,19 @@ class StreamingEngine:
         from ..graph.constructor import build_edges_and_flags, build_graph, build_graph_parts
         from ..graph.sampling import sample_fanout, to_csr
 
+        dev_cols = None
         if use_incremental:
             from ..graph.incremental import merge_window
 
+            if self.device.type == "cuda":
+                # HBM-resident delta ring: sealed deltas' columns ship to
+                # the GPU once; the window is a device-side cat
+                dev_cols = self.store.device_columns(window_deltas, self.device)
             sums = self._inc_state.summaries(window_deltas)
             parts, ed = merge_window(
                 events, sums,
                 device=self.device if self.device.type == "cuda" else None,
+                dev_cols=dev_cols,
             )
         else:
             parts = build_graph_parts(events)
@@ -161,7 +167,8 @@ class StreamingEngine:
         if self.device.type == "cuda":
             from ..graph.gpu_store import gpu_window_graph
 
-            gg = gpu_window_graph(events, self.device, parts=parts, ed=ed, dtype=self.dtype)
+            gg = gpu_window_graph(events, self.device, parts=parts, ed=ed,
+                                  dtype=self.dtype, dev_cols=dev_cols)
             x = gg["x"]
             edge_index = gg["edge_index"]
             edge_weight = gg["edge_weight"]
@@ -169,7 +176,7 @@ class StreamingEngine:
             # sequence assembly on-device too (the numpy build is ~57 ms per
             # 600k-event window; the torch version is a radix sort + scatters)
             seq_feats, seq_lengths_cpu, seq_fids = build_sequences_torch(
-                events, self.device, dtype=self.dtype
+                events, self.device, dtype=self.dtype, dev_cols=dev_cols
             )
             seq_lengths = seq_lengths_cpu.to(self.device)
             seq_fids = seq_fids.numpy()

@@ -813,3 +813,48 @@ def test_lstm_bilayer2_matches_bilayer(gpu_device):
     for a, c, name in ((wf.grad, wf2.grad, "wf"), (wb.grad, wb2.grad, "wb"),
                        (bf.grad, bf2.grad, "bf"), (bb.grad, bb2.grad, "bb")):
         assert torch.allclose(a.float(), c.float(), atol=1e-5), name
+
+
+def test_engine_device_delta_ring_matches_host_path(gpu_device):
+    """Scoring with the HBM-resident delta ring (per-delta device columns,
+    device-side window cat) must match the host-upload path exactly, and
+    the cache must stay bounded across eviction."""
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.models.graphsage import SageConfig
+    from nerrf_amd.models.joint import JointConfig, NerrfJointModel
+    from nerrf_amd.models.lstm import LSTMConfig
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    torch.manual_seed(5)
+    model = NerrfJointModel(JointConfig(sage=SageConfig(layers=3, hidden=64),
+                                        lstm=LSTMConfig(hidden=256)))
+    arr, _ = generate(SynthConfig(seed=4, duration_s=40, benign_rate_hz=200,
+                                  n_victim_files=8))
+
+    def run(use_ring):
+        eng = StreamingEngine(model=model, device=gpu_device,
+                              dtype=torch.bfloat16, window_s=1e9)
+        if not use_ring:
+            eng.store.device_columns = lambda *a, **k: None
+        eng.ingest_events(arr)
+        det = eng.score_window()
+        return det, eng
+
+    det_ring, eng_ring = run(True)
+    det_host, _ = run(False)
+    assert det_ring.alarm == det_host.alarm
+    assert det_ring.window_events == det_host.window_events
+    for p, s in det_host.file_scores.items():
+        assert abs(det_ring.file_scores[p] - s) < 1e-3, p
+    # cache reuse + boundedness: more ticks than deltas
+    eng2 = StreamingEngine(model=model, device=gpu_device,
+                           dtype=torch.bfloat16, window_s=8.0)
+    t0 = float(arr.ts.min())
+    for k in range(6):
+        lo = arr.ts < t0 + 6 * (k + 1)
+        hi = arr.ts >= t0 + 6 * k
+        seg = arr.time_window(t0 + 6 * k, t0 + 6 * (k + 1) - 1e-9)
+        if len(seg):
+            eng2.ingest_events(seg)
+        eng2.score_window()
+    assert len(eng2.store._dev_cache) <= len(eng2.store._deltas) + 1
