@@ -223,9 +223,11 @@ def test_mcts_eval_plans_matches_cpu(gpu_device):
         "downtime_weight": params.downtime_weight, "revert_time_s": params.revert_time_s,
         "kill_time_s": params.kill_time_s, "fp_weight": params.fp_weight,
         "attack_rate_mbps": params.attack_rate_mbps, "horizon_s": params.horizon_s,
+        "restore_time_s": params.restore_time_s,
+        "restore_loss_mb": params.restore_loss_mb,
         "ucb_c": UCB_C, "seed": 0,
     }
-    n_actions = st.n_groups + 2
+    n_actions = st.n_groups + 3  # STOP, KILL, RESTORE, reverts
     plans = rng.integers(0, n_actions, size=(200, 6)).astype(np.int32)
     gs = torch.from_numpy(st.group_score.astype(np.float32)).to(gpu_device)
     gm = torch.from_numpy(st.group_mb.astype(np.float32)).to(gpu_device)
@@ -440,7 +442,7 @@ def test_mcts_search_16_groups(gpu_device):
     st = build_state(rng.random(80), rng.random(80) * 3, 0.9, 40.0, n_groups=16)
     res = run_mcts_gpu(st, PlannerParams(n_groups=16), n_sims=512, device=str(gpu_device))
     assert res.simulations == 512
-    assert len(res.ranked_actions) == 18
+    assert len(res.ranked_actions) == 19  # STOP, KILL, RESTORE + 16 groups
 
 
 def test_lstm_sequence_t1_edge(gpu_device):
