@@ -89,6 +89,32 @@ class RecoveryMetrics:
         }
 
 
+def operating_point(y: np.ndarray, scores: np.ndarray, max_fp_frac: float = 0.05) -> Dict[str, float]:
+    """Fixed-FP-budget operating point (reference README.md:23-27:
+    "FP-undo < 5%"): the lowest threshold whose flagged set stays under
+    `max_fp_frac` false positives, i.e. precision >= 1 - max_fp_frac.
+    Returns threshold, precision, recall (and flagged count) there."""
+    y = np.asarray(y).astype(bool)
+    s = np.asarray(scores, dtype=np.float64)
+    order = np.argsort(-s, kind="stable")
+    ys = y[order]
+    tp = np.cumsum(ys)
+    fp = np.cumsum(~ys)
+    n = len(ys)
+    prec = tp / np.maximum(tp + fp, 1)
+    ok = np.nonzero(prec >= 1.0 - max_fp_frac)[0]
+    if len(ok) == 0 or not y.any():
+        return {"threshold": 1.0, "precision": 1.0, "recall": 0.0, "flagged": 0.0}
+    k = int(ok[-1])  # largest flagged set still within budget
+    thr = float(s[order][k])
+    return {
+        "threshold": thr,
+        "precision": float(prec[k]),
+        "recall": float(tp[k] / max(int(y.sum()), 1)),
+        "flagged": float(k + 1),
+    }
+
+
 def detection_report(
     y_node: np.ndarray,
     node_scores: np.ndarray,

@@ -134,3 +134,23 @@ def test_config_sequence_override_forms():
     ):
         cfg = load_config(None, [ov])
         assert cfg.data.scenario_kinds == ("lockbit", "supply_chain"), ov
+
+
+def test_operating_point_fixed_fp_budget():
+    import numpy as np
+
+    from nerrf_amd.eval import operating_point
+
+    y = np.array([1, 1, 1, 1, 0, 0, 0, 0, 0, 0])
+    s = np.array([0.95, 0.9, 0.85, 0.4, 0.5, 0.3, 0.2, 0.1, 0.05, 0.01])
+    op = operating_point(y, s, max_fp_frac=0.05)
+    # flagging the top 3 keeps precision 1.0; the 4th pick (0.5) is a FP
+    assert op["precision"] >= 0.95
+    assert op["recall"] == 0.75
+    assert op["threshold"] > 0.5
+    # degenerate: no positives
+    op0 = operating_point(np.zeros(5), np.random.default_rng(0).random(5))
+    assert op0["recall"] == 0.0
+    # generous budget flags more
+    op2 = operating_point(y, s, max_fp_frac=0.5)
+    assert op2["recall"] >= op["recall"]
