@@ -53,6 +53,12 @@ void launch_feature_assemble(const float*, const int*, const int*,
 void launch_sage_layer_fwd(const void*, const long*, const float*, const void*,
                            const void*, const void*, const void*, const void*,
                            void*, int, int, hipStream_t);
+void launch_sage_ln_act_fwd(const void*, const void*, const void*, const void*,
+                            const void*, void*, void*, float*, unsigned char*,
+                            long, float, unsigned, hipStream_t);
+void launch_sage_ln_act_bwd(const void*, const void*, const float*,
+                            const void*, void*, float*, float*, long, float,
+                            unsigned, hipStream_t);
 }  // namespace nerrf
 
 namespace {
@@ -453,6 +459,54 @@ torch::Tensor sage_layer_fwd(torch::Tensor h, torch::Tensor nbr_idx,
   return out;
 }
 
+// Fused GraphSAGE training-layer tail: y = h + LN(dropout(GELU(zs+zn)))*g+b
+std::vector<torch::Tensor> sage_ln_act_fwd(torch::Tensor h, torch::Tensor zs,
+                                           torch::Tensor zn, torch::Tensor gamma,
+                                           torch::Tensor beta, double drop_p,
+                                           long seed, bool want_mask) {
+  for (auto* t : {&h, &zs, &zn}) {
+    check_gpu_contig(*t, "sage_ln_act arg");
+    TORCH_CHECK(t->scalar_type() == torch::kBFloat16, "sage_ln_act is bf16-only");
+    TORCH_CHECK(t->size(1) == 128, "sage_ln_act requires D=128");
+  }
+  const long n = h.size(0);
+  auto gc = gamma.contiguous().to(torch::kBFloat16);
+  auto bc = beta.contiguous().to(torch::kBFloat16);
+  auto y = torch::empty_like(h);
+  auto s_save = torch::empty_like(h);
+  auto stats = torch::empty({n, 2}, h.options().dtype(torch::kFloat32));
+  torch::Tensor mask = want_mask
+      ? torch::zeros({n, 128}, h.options().dtype(torch::kUInt8))
+      : torch::empty({0}, h.options().dtype(torch::kUInt8));
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_sage_ln_act_fwd(
+      h.data_ptr(), zs.data_ptr(), zn.data_ptr(), gc.data_ptr(), bc.data_ptr(),
+      y.data_ptr(), s_save.data_ptr(), stats.data_ptr<float>(),
+      want_mask ? mask.data_ptr<unsigned char>() : nullptr, n, (float)drop_p,
+      (unsigned)seed, stream.stream());
+  return {y, s_save, stats, mask};
+}
+
+std::vector<torch::Tensor> sage_ln_act_bwd(torch::Tensor dy, torch::Tensor s_save,
+                                           torch::Tensor stats, torch::Tensor gamma,
+                                           double drop_p, long seed) {
+  check_gpu_contig(dy, "dy");
+  check_gpu_contig(s_save, "s_save");
+  check_gpu_contig(stats, "stats");
+  const long n = dy.size(0);
+  auto gc = gamma.contiguous().to(torch::kBFloat16);
+  auto dz = torch::empty_like(dy);
+  const long grid = (n + 7) / 8;
+  auto dg_part = torch::empty({grid, 128}, dy.options().dtype(torch::kFloat32));
+  auto db_part = torch::empty({grid, 128}, dy.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_sage_ln_act_bwd(
+      dy.data_ptr(), s_save.data_ptr(), stats.data_ptr<float>(), gc.data_ptr(),
+      dz.data_ptr(), dg_part.data_ptr<float>(), db_part.data_ptr<float>(), n,
+      (float)drop_p, (unsigned)seed, stream.stream());
+  return {dz, dg_part.sum(0), db_part.sum(0)};
+}
+
 nerrf::PlannerParamsDev params_from_dict(const pybind11::dict& d) {
   nerrf::PlannerParamsDev p;
   p.n_groups = d["n_groups"].cast<int>();
@@ -521,6 +575,9 @@ torch::Tensor mcts_eval_plans(torch::Tensor gscore, torch::Tensor gmb,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sage_ln_act_fwd", &sage_ln_act_fwd,
+        "fused GNN layer tail fwd (add+GELU+dropout+LN+residual)");
+  m.def("sage_ln_act_bwd", &sage_ln_act_bwd, "fused GNN layer tail bwd");
   m.def("sage_layer_fwd", &sage_layer_fwd,
         "fused GraphSAGE-T layer forward (MFMA, inference)");
   m.def("event_features", &event_features,

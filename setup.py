@@ -24,6 +24,7 @@ sources = [
     os.path.join(HIP_DIR, "mcts.hip"),
     os.path.join(HIP_DIR, "event_scatter.hip"),
     os.path.join(HIP_DIR, "sage_fused.hip"),
+    os.path.join(HIP_DIR, "sage_ln_act.hip"),
 ]
 sources = [s for s in sources if os.path.exists(s)]
 

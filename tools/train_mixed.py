@@ -29,11 +29,12 @@ for kind in ("lockbit", "supply_chain", "supply_chain_net"):
     print(f"family={kind}: " + json.dumps({k: round(float(v), 4) for k, v in rep.items() if "auc" in k or k.endswith("f1")}))
     # fixed-FP-undo operating point (reference README.md:23-27: FP < 5%)
     ys, ss = [], []
-    for b in hb:
-        tb = b.to_torch("cuda", torch.bfloat16)
-        nl, _, sl = model(tb)
-        ys.append(tb["y_node"].cpu().numpy())
-        ss.append(torch.sigmoid(nl.float()).cpu().numpy())
+    with torch.no_grad():
+        for b in hb:
+            tb = b.to_torch("cuda", torch.bfloat16)
+            nl, _, sl = model(tb)
+            ys.append(tb["y_node"].cpu().numpy())
+            ss.append(torch.sigmoid(nl.float()).cpu().numpy())
     import numpy as _op_np
     op = operating_point(_op_np.concatenate(ys), _op_np.concatenate(ss), 0.05)
     print(f"  operating_point(FP<5%)={json.dumps({k: round(v, 4) for k, v in op.items()})}")
@@ -43,10 +44,11 @@ for kind in ("benign_rotate", "benign_backup", "benign_build"):
     hb = synth_window_batches(n_scenarios=2, attack_fraction=0.0, base_seed=777000,
                               benign_kinds=(kind,))
     mx = []
-    for b in hb:
-        tb = b.to_torch("cuda", torch.bfloat16)
-        nl, _, sl = model(tb)
-        mx.append(float(torch.sigmoid(nl.float()).max()))
-        if sl is not None and sl.numel():
-            mx.append(float(torch.sigmoid(sl.float()).max()))
+    with torch.no_grad():
+        for b in hb:
+            tb = b.to_torch("cuda", torch.bfloat16)
+            nl, _, sl = model(tb)
+            mx.append(float(torch.sigmoid(nl.float()).max()))
+            if sl is not None and sl.numel():
+                mx.append(float(torch.sigmoid(sl.float()).max()))
     print(f"negative={kind}: max_score={max(mx):.4f}")
