@@ -47,6 +47,10 @@ class Detection:
     # counterfactual-refinement context (node features + sampled adjacency +
     # path->node map) for model-in-the-loop plan scoring; host-side numpy
     refine_ctx: Optional[dict] = None
+    # raw per-head window maxima (before score folding) — the alarm gate's
+    # inputs, exposed for calibration and observability
+    node_max: float = 0.0
+    seq_max: float = 0.0
 
 
 PRETRAINED_DIR = "checkpoints/pretrained"
@@ -86,6 +90,7 @@ class StreamingEngine:
         planner_params: Optional[PlannerParams] = None,
         shard_id: int = 0,
         world: int = 1,
+        calibration: Optional[str] = "auto",
     ) -> None:
         self.device = torch.device(device)
         self.dtype = dtype
@@ -98,6 +103,23 @@ class StreamingEngine:
 
         self._inc_state = IncrementalWindowState()
         self.alarm_threshold = alarm_threshold
+        # calibrated per-channel thresholds (tools/calibrate_alarm.py sweep
+        # over off-distribution window configs; benign FP = 0 across the
+        # sweep — docs/threat-model.md has the trade table).  "auto" loads
+        # the vendored table; None keeps the single-threshold rule.
+        self.calibrated = None
+        if calibration == "auto":
+            import json as _json
+            from pathlib import Path as _Path
+
+            cal_p = _Path(__file__).parent / "alarm_calibration.json"
+            if cal_p.exists():
+                try:
+                    c = _json.loads(cal_p.read_text())
+                    self.calibrated = {"ind_thr": float(c["ind_thr"]),
+                                       "model_thr": float(c["model_thr"])}
+                except (ValueError, KeyError):
+                    self.calibrated = None
         self.planner_params = planner_params or PlannerParams()
         self.shard_id = shard_id
         self.world = world
@@ -293,7 +315,11 @@ class StreamingEngine:
         model_max = float(node_score.max()) if len(node_score) else 0.0
         seq_max = float(seq_score.max()) if seq_score is not None and len(seq_score) else 0.0
         alarm_score = max(ind_score, min(model_max, seq_max))
-        alarm = alarm_score >= self.alarm_threshold
+        if self.calibrated is not None:
+            alarm = (ind_score >= self.calibrated["ind_thr"]
+                     or min(model_max, seq_max) >= self.calibrated["model_thr"])
+        else:
+            alarm = alarm_score >= self.alarm_threshold
         # refinement context: lets plan() re-score counterfactual post-plan
         # graphs through the GNN in one batch (planner/model_eval.py)
         path_to_node = {
@@ -327,6 +353,8 @@ class StreamingEngine:
             indicators=indicators,
             window_events=len(events),
             refine_ctx=refine_ctx,
+            node_max=model_max,
+            seq_max=seq_max,
         )
 
     # -------------------------------------------------------------------- plan

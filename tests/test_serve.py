@@ -451,3 +451,27 @@ def test_indicators_do_not_latch_after_window_eviction():
     assert det_clean.indicators["exfil_dest_count"] == 0
     assert det_clean.encrypted_paths == []
     assert not det_clean.alarm
+
+
+def test_calibrated_alarm_thresholds_loaded_and_applied():
+    """The vendored calibration drives the default alarm rule; disabling it
+    restores the single-threshold rule."""
+    import json
+    from pathlib import Path
+
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    cal = json.loads(
+        (Path("nerrf_amd/serve/alarm_calibration.json")).read_text()
+    )
+    assert 0.0 < cal["ind_thr"] < cal["model_thr"] < 1.0
+    assert cal["sweep"]["benign_alarmed_at_thr"] == 0
+    eng = _small_engine(device="cpu")
+    assert eng.calibrated == {"ind_thr": cal["ind_thr"], "model_thr": cal["model_thr"]}
+    eng2 = _small_engine(device="cpu", calibration=None)
+    assert eng2.calibrated is None
+    # attack window alarms under the calibrated rule
+    arr, _ = generate(SynthConfig(seed=2, duration_s=40, benign_rate_hz=40, n_victim_files=8))
+    eng.store.window_s = 1e9
+    eng.ingest_events(arr)
+    assert eng.score_window().alarm
