@@ -209,3 +209,53 @@ def test_native_decoder_rejects_crafted_length_overflow():
     frame2 = bytes([0x0A, len(inner2)]) + inner2
     with pytest.raises(Exception):
         dec.decode([frame2])
+
+
+def test_fd_resolver_exact_and_heuristic():
+    """Userspace fd->path fallback (reference M2 spec): exact (pid, fd)
+    binding when openat ret_val carries the fd; last-open heuristic when
+    fds are absent (the upstream artifact schema)."""
+    import numpy as np
+
+    from nerrf_amd.data.fdresolve import resolve_fd_paths
+    from nerrf_amd.data.trace import EventArrayBuilder, StringTable
+
+    b = EventArrayBuilder(StringTable(), StringTable())
+    # pid 1: open a.txt (fd 3), open b.txt (fd 4), write fd 3, write fd 4,
+    #        close fd 3, write fd 4
+    b.add(ts=1.0, pid=1, syscall="openat", path="/a.txt", ret_val=3)
+    b.add(ts=2.0, pid=1, syscall="openat", path="/b.txt", ret_val=4)
+    b.add(ts=3.0, pid=1, syscall="write", nbytes=10, ret_val=3)
+    b.add(ts=4.0, pid=1, syscall="write", nbytes=20, ret_val=4)
+    b.add(ts=5.0, pid=1, syscall="close", ret_val=3)
+    b.add(ts=6.0, pid=1, syscall="write", nbytes=30, ret_val=4)
+    # pid 2: interleaved, must not leak pid 1's table
+    b.add(ts=3.5, pid=2, syscall="openat", path="/c.txt", ret_val=3)
+    b.add(ts=4.5, pid=2, syscall="write", nbytes=5, ret_val=3)
+    arr = b.build()
+    res = resolve_fd_paths(arr)
+    names = [res.paths.lookup(int(i)) if i >= 0 else None for i in res.path_id]
+    by = {(float(t), int(p), int(nb)): nm
+          for t, p, nb, nm in zip(res.ts, res.pid, res.nbytes, names)}
+    assert by[(3.0, 1, 10)] == "/a.txt"   # exact fd 3
+    assert by[(4.0, 1, 20)] == "/b.txt"   # exact fd 4
+    assert by[(6.0, 1, 30)] == "/b.txt"   # fd 3 closed; fd 4 still bound
+    assert by[(4.5, 2, 5)] == "/c.txt"    # per-pid isolation
+
+    # heuristic mode: no fds recorded (ret_val = 0 means fd 0 is unknown
+    # here — disable exact matching)
+    b2 = EventArrayBuilder(StringTable(), StringTable())
+    b2.add(ts=1.0, pid=7, syscall="openat", path="/x.dat")
+    b2.add(ts=2.0, pid=7, syscall="write", nbytes=64)
+    b2.add(ts=3.0, pid=7, syscall="openat", path="/y.dat")
+    b2.add(ts=4.0, pid=7, syscall="write", nbytes=65)
+    r2 = resolve_fd_paths(b2.build(), use_ret_fd=False)
+    n2 = [r2.paths.lookup(int(i)) if i >= 0 else None for i in r2.path_id]
+    writes = [nm for sc, nm in zip(r2.syscall, n2) if sc == 2]
+    assert writes == ["/x.dat", "/y.dat"]  # last-open-wins
+
+    # already-pathed writes are untouched
+    b3 = EventArrayBuilder(StringTable(), StringTable())
+    b3.add(ts=1.0, pid=9, syscall="write", path="/known.log", nbytes=1)
+    r3 = resolve_fd_paths(b3.build())
+    assert r3.paths.lookup(int(r3.path_id[0])) == "/known.log"

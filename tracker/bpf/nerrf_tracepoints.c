@@ -26,15 +26,54 @@ static __u64 (*bpf_get_current_pid_tgid)(void) = (void *)14;
 static long (*bpf_get_current_comm)(void *buf, __u32 size) = (void *)16;
 static long (*bpf_probe_read_user_str)(void *dst, __u32 size,
                                        const void *unsafe_ptr) = (void *)114;
+static void *(*bpf_map_lookup_elem)(void *map, const void *key) = (void *)1;
+static long (*bpf_map_update_elem)(void *map, const void *key,
+                                   const void *value, __u64 flags) = (void *)2;
+static long (*bpf_map_delete_elem)(void *map, const void *key) = (void *)3;
 
 char LICENSE[] SEC("license") = "GPL";
 
 /* BTF-style ring buffer map (loader: libbpf >= 0.x with BTF support) */
 #define __uint(name, val) int(*name)[val]
+#define __type(name, val) typeof(val) *name
 struct {
   __uint(type, BPF_MAP_TYPE_RINGBUF);
   __uint(max_entries, NERRF_RINGBUF_BYTES);
 } events SEC(".maps");
+
+/* fd -> path resolution for write/read (reference M2 spec,
+ * implementation.mdx:520-563: "per-thread fd->path map, registered at
+ * openat time").  Two hash maps:
+ *   pending_open : tid -> path, written at sys_enter_openat, bound to the
+ *                  returned fd at sys_exit_openat;
+ *   fd_paths     : (tgid << 32 | fd) -> path, consumed by write/read and
+ *                  dropped on close.
+ * Bounded (LRU semantics via plain hash + delete-on-close; stale entries
+ * for long-lived fds are the accepted trade — same as the upstream plan).
+ */
+struct fd_path_val {
+  char path[NERRF_PATH_MAX];
+};
+
+struct {
+  __uint(type, BPF_MAP_TYPE_HASH);
+  __uint(max_entries, 8192);
+  __type(key, __u32);
+  __type(value, struct fd_path_val);
+} pending_open SEC(".maps");
+
+struct {
+  __uint(type, BPF_MAP_TYPE_HASH);
+  __uint(max_entries, 65536);
+  __type(key, __u64);
+  __type(value, struct fd_path_val);
+} fd_paths SEC(".maps");
+
+struct sys_exit_ctx {
+  __u64 _pad;
+  long id;
+  long ret;
+};
 
 /* raw tracepoint context for syscalls:sys_enter_* */
 struct sys_enter_ctx {
@@ -67,17 +106,52 @@ int nerrf_trace_openat(struct sys_enter_ctx *ctx) {
   if (!ev) return 0;
   bpf_probe_read_user_str(ev->path, sizeof(ev->path), (void *)ctx->args[1]);
   ev->flags = (int)(ctx->args[2] & 3); /* O_ACCMODE */
+  /* stage the path for fd binding at sys_exit_openat */
+  {
+    __u32 tid = (__u32)bpf_get_current_pid_tgid();
+    struct fd_path_val v = {};
+    bpf_probe_read_user_str(v.path, sizeof(v.path), (void *)ctx->args[1]);
+    bpf_map_update_elem(&pending_open, &tid, &v, 0 /* BPF_ANY */);
+  }
   bpf_ringbuf_submit(ev, 0);
   return 0;
+}
+
+SEC("tracepoint/syscalls/sys_exit_openat")
+int nerrf_trace_openat_exit(struct sys_exit_ctx *ctx) {
+  __u64 id = bpf_get_current_pid_tgid();
+  __u32 tid = (__u32)id;
+  struct fd_path_val *v = bpf_map_lookup_elem(&pending_open, &tid);
+  if (!v) return 0;
+  if (ctx->ret >= 0) {
+    __u64 key = ((id >> 32) << 32) | (__u64)(__u32)ctx->ret;
+    bpf_map_update_elem(&fd_paths, &key, v, 0);
+  }
+  bpf_map_delete_elem(&pending_open, &tid);
+  return 0;
+}
+
+SEC("tracepoint/syscalls/sys_enter_close")
+int nerrf_trace_close(struct sys_enter_ctx *ctx) {
+  __u64 id = bpf_get_current_pid_tgid();
+  __u64 key = ((id >> 32) << 32) | (__u64)(__u32)ctx->args[0];
+  bpf_map_delete_elem(&fd_paths, &key);
+  return 0;
+}
+
+static __always_inline void fill_fd_path(struct nerrf_event *ev, __u64 fd) {
+  __u64 id = bpf_get_current_pid_tgid();
+  __u64 key = ((id >> 32) << 32) | (fd & 0xFFFFFFFF);
+  struct fd_path_val *v = bpf_map_lookup_elem(&fd_paths, &key);
+  if (v) __builtin_memcpy(ev->path, v->path, NERRF_PATH_MAX);
 }
 
 SEC("tracepoint/syscalls/sys_enter_write")
 int nerrf_trace_write(struct sys_enter_ctx *ctx) {
   struct nerrf_event *ev = reserve_event(NERRF_SYS_WRITE);
   if (!ev) return 0;
-  /* fd->path resolution needs a kprobe-side fd table map (future work,
-   * mirrors the upstream limitation); record the byte count */
   ev->bytes = (__u64)ctx->args[2];
+  fill_fd_path(ev, ctx->args[0]); /* resolved via the openat-time fd map */
   bpf_ringbuf_submit(ev, 0);
   return 0;
 }
@@ -87,6 +161,7 @@ int nerrf_trace_read(struct sys_enter_ctx *ctx) {
   struct nerrf_event *ev = reserve_event(NERRF_SYS_READ);
   if (!ev) return 0;
   ev->bytes = (__u64)ctx->args[2];
+  fill_fd_path(ev, ctx->args[0]);
   bpf_ringbuf_submit(ev, 0);
   return 0;
 }
