@@ -59,3 +59,27 @@ def test_bench_torchrun_world2_gloo():
     assert res["n_gpus"] == 2
     assert res["config"]["parallelism"] == "dp2"
     assert res["value"] > 0
+
+
+def test_bench_torchrun_world4_gloo():
+    """World-4 launch contract: the 8-GPU driver run
+    differs from world-2 only in degree; exercise a deeper rank fan-out on
+    gloo so bucket-order determinism holds beyond the pairwise case."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    proc = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "4",
+            "--master-addr", "127.0.0.1", "--master-port", "29519",
+            "bench.py", "--gpus", "4", "--steps", "2", "--warmup", "1",
+            "--scale", "tiny", "--windows", "1", "--batch-windows", "1",
+        ],
+        cwd=ROOT, capture_output=True, text=True, timeout=1200, env=env,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    res = _last_json_line(proc.stdout)
+    assert res["n_gpus"] == 4
+    assert res["config"]["parallelism"] == "dp4"
+    assert res["value"] > 0
