@@ -131,6 +131,7 @@ class _LSTMCellFn(torch.autograd.Function):
                 grad_h, torch.empty(0, device=grad_h.device), grad_c, gates_act, c,
                 mask if mask is not None else torch.empty(0, device=grad_h.device),
                 grad_gates, grad_c_prev, grad_h_pass,
+                torch.empty(0, device=grad_h.device),
             )
         else:
             grad_gates, grad_c_prev, grad_h_pass = _ref.lstm_pointwise_bwd_ref(

@@ -26,8 +26,8 @@ void launch_lstm_pointwise_fwd(const void*, const void*, const void*,
                                bool, hipStream_t);
 void launch_lstm_pointwise_bwd(const void*, const void*, const void*,
                                const void*, const void*, const float*, void*,
-                               void*, void*, long, int, long, long, bool,
-                               hipStream_t);
+                               void*, void*, float*, long, int, long, long,
+                               bool, hipStream_t);
 void launch_lstm_step_fused(const void*, const void*, const void*, const void*,
                             const void*, const float*, void*, void*, void*,
                             int, bool, hipStream_t);
@@ -181,11 +181,15 @@ void lstm_pointwise_fwd(torch::Tensor hg, torch::Tensor xg, torch::Tensor bias,
 
 // grad_out_t (may be empty): this timestep's dL/dh, folded in-kernel so the
 // sequence backward needs no separate elementwise add per step.
-void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
+// bias_accum (optional, may be empty): f32 [4H] accumulator the kernel
+// folds sum_b(grad_gates) into (vectorised path only) — callers then skip
+// the separate whole-tensor gg.sum(0) reduction.  Returns true when the
+// fused accumulation ran.
+bool lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
                         torch::Tensor grad_c, torch::Tensor gates_act,
                         torch::Tensor c_prev, torch::Tensor mask,
                         torch::Tensor grad_gates, torch::Tensor grad_c_prev,
-                        torch::Tensor grad_h_pass) {
+                        torch::Tensor grad_h_pass, torch::Tensor bias_accum) {
   check_gpu_contig(grad_h, "grad_h");
   check_gpu_contig(grad_c, "grad_c");
   check_gpu_contig(gates_act, "gates_act");
@@ -206,11 +210,22 @@ void lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
   const void* got = grad_out_t.numel() ? grad_out_t.data_ptr() : nullptr;
   long gout_stride = hdim;
   if (grad_out_t.numel()) gout_stride = row_stride_checked(grad_out_t, "grad_out_t");
+  float* bias_ptr = nullptr;
+  const int v = is_bf16(grad_h) ? 8 : 4;
+  const bool vec = hdim % v == 0 && gout_stride % v == 0 && gg_stride % v == 0;
+  if (bias_accum.numel() > 0 && vec) {
+    check_gpu_contig(bias_accum, "bias_accum");
+    TORCH_CHECK(bias_accum.scalar_type() == torch::kFloat32 &&
+                    bias_accum.numel() == 4 * hdim,
+                "bias_accum must be f32 [4H]");
+    bias_ptr = bias_accum.data_ptr<float>();
+  }
   nerrf::launch_lstm_pointwise_bwd(
       grad_h.data_ptr(), got, grad_c.data_ptr(), gates_act.data_ptr(),
       c_prev.data_ptr(), mask_ptr, grad_gates.data_ptr(),
-      grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), batch, hdim,
+      grad_c_prev.data_ptr(), grad_h_pass.data_ptr(), bias_ptr, batch, hdim,
       gout_stride, gg_stride, is_bf16(grad_h), stream.stream());
+  return bias_ptr != nullptr;
 }
 
 // Fused recurrent step (bf16, H == 256): h_prev @ W_hh^T + LSTM pointwise

@@ -196,9 +196,16 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
     const T* __restrict__ grad_c, const T* __restrict__ gates_act,
     const T* __restrict__ c_prev, const float* __restrict__ mask,
     T* __restrict__ grad_gates, T* __restrict__ grad_c_prev,
-    T* __restrict__ grad_h_pass, long batch, int hdim, long gout_stride,
-    long gg_stride) {
+    T* __restrict__ grad_h_pass, float* __restrict__ bias_accum,
+    long batch, int hdim, long gout_stride, long gg_stride) {
   using VT = VecT<T, V>;
+  // optional fused bias-grad: per-block LDS partial of sum_b(grad_gates),
+  // atomically folded into bias_accum[4H] at block end — removes the
+  // separate 13-GB-per-direction gg re-read (gg2.sum(0)) from the step
+  extern __shared__ float bias_lds[];
+  if (bias_accum != nullptr)
+    for (int i = threadIdx.x; i < 4 * hdim; i += blockDim.x) bias_lds[i] = 0.0f;
+  if (bias_accum != nullptr) __syncthreads();
   const int hv = hdim / V;
   const long total = batch * hv;
   for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
@@ -251,6 +258,20 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
     *reinterpret_cast<VT*>(grad_gates + o0 + hdim) = gg_f;
     *reinterpret_cast<VT*>(grad_gates + o0 + 2 * hdim) = gg_g;
     *reinterpret_cast<VT*>(grad_gates + o0 + 3 * hdim) = gg_o;
+    if (bias_accum != nullptr) {
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        atomicAdd(bias_lds + dv + j, to_f32(gg_i.v[j]));
+        atomicAdd(bias_lds + hdim + dv + j, to_f32(gg_f.v[j]));
+        atomicAdd(bias_lds + 2 * hdim + dv + j, to_f32(gg_g.v[j]));
+        atomicAdd(bias_lds + 3 * hdim + dv + j, to_f32(gg_o.v[j]));
+      }
+    }
+  }
+  if (bias_accum != nullptr) {
+    __syncthreads();
+    for (int i = threadIdx.x; i < 4 * hdim; i += blockDim.x)
+      if (bias_lds[i] != 0.0f) atomicAdd(bias_accum + i, bias_lds[i]);
   }
 }
 
@@ -265,10 +286,11 @@ template __global__ void lstm_pointwise_fwd_vec_kernel<float, 4>(
 template __global__ void lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8>(
     const __hip_bfloat16*, const __hip_bfloat16*, const __hip_bfloat16*,
     const __hip_bfloat16*, const __hip_bfloat16*, const float*,
-    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, long, int, long, long);
+    __hip_bfloat16*, __hip_bfloat16*, __hip_bfloat16*, float*, long, int,
+    long, long);
 template __global__ void lstm_pointwise_bwd_vec_kernel<float, 4>(
     const float*, const float*, const float*, const float*, const float*,
-    const float*, float*, float*, float*, long, int, long, long);
+    const float*, float*, float*, float*, float*, long, int, long, long);
 
 template __global__ void lstm_pointwise_fwd_kernel<float>(
     const float*, const float*, const float*, const float*, const float*, const float*,
@@ -347,21 +369,22 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
                                const void* grad_c, const void* gates_act,
                                const void* c_prev, const float* mask,
                                void* grad_gates, void* grad_c_prev,
-                               void* grad_h_pass, long batch, int hdim,
-                               long gout_stride, long gg_stride, bool bf16,
-                               hipStream_t s) {
+                               void* grad_h_pass, float* bias_accum,
+                               long batch, int hdim, long gout_stride,
+                               long gg_stride, bool bf16, hipStream_t s) {
   const int block = 256;
   const int v = bf16 ? 8 : 4;
   const bool vec = hdim % v == 0 && gout_stride % v == 0 && gg_stride % v == 0;
   if (bf16) {
     if (vec) {
       const int grid = grid_elems(batch * (hdim / v), block);
-      lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8><<<grid, block, 0, s>>>(
+      lstm_pointwise_bwd_vec_kernel<__hip_bfloat16, 8>
+          <<<grid, block, bias_accum ? 4 * hdim * sizeof(float) : 0, s>>>(
           (const __hip_bfloat16*)grad_h, (const __hip_bfloat16*)grad_out_t,
           (const __hip_bfloat16*)grad_c, (const __hip_bfloat16*)gates_act,
           (const __hip_bfloat16*)c_prev, mask, (__hip_bfloat16*)grad_gates,
-          (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass, batch,
-          hdim, gout_stride, gg_stride);
+          (__hip_bfloat16*)grad_c_prev, (__hip_bfloat16*)grad_h_pass,
+          bias_accum, batch, hdim, gout_stride, gg_stride);
       return;
     }
     const int grid = grid_elems(batch * hdim, block);
@@ -374,11 +397,12 @@ void launch_lstm_pointwise_bwd(const void* grad_h, const void* grad_out_t,
   } else {
     if (vec) {
       const int grid = grid_elems(batch * (hdim / v), block);
-      lstm_pointwise_bwd_vec_kernel<float, 4><<<grid, block, 0, s>>>(
+      lstm_pointwise_bwd_vec_kernel<float, 4>
+          <<<grid, block, bias_accum ? 4 * hdim * sizeof(float) : 0, s>>>(
           (const float*)grad_h, (const float*)grad_out_t, (const float*)grad_c,
           (const float*)gates_act, (const float*)c_prev, mask,
-          (float*)grad_gates, (float*)grad_c_prev, (float*)grad_h_pass, batch,
-          hdim, gout_stride, gg_stride);
+          (float*)grad_gates, (float*)grad_c_prev, (float*)grad_h_pass,
+          bias_accum, batch, hdim, gout_stride, gg_stride);
       return;
     }
     const int grid = grid_elems(batch * hdim, block);

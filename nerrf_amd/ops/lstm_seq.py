@@ -188,6 +188,11 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
             grad_whh = torch.mm(gg2.t(), h0.to(dt))
         grad_bias = gg2.sum(dim=0)
         return grad_gates_all, grad_h, grad_c, grad_whh, grad_bias
+    bias_accum = (
+        torch.zeros(gdim, device=dev, dtype=torch.float32)
+        if ext is not None else None
+    )
+    bias_fused = ext is not None
     for ti in steps:
         # c/h input of step ti = previous step's output (or h0/c0 at start)
         first = (ti == t_len - 1) if reverse else (ti == 0)
@@ -196,13 +201,16 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
         else:
             c_in = c_all[ti + 1] if reverse else c_all[ti - 1]
         if ext is not None:
-            # grad_out[ti] is folded inside the kernel (no separate add)
-            ext.lstm_pointwise_bwd(
+            # grad_out[ti] is folded inside the kernel (no separate add),
+            # and so is this step's bias-grad partial (sum over batch) —
+            # the big gg2.sum(0) re-read disappears when every step fused
+            used = ext.lstm_pointwise_bwd(
                 grad_h.contiguous(), grad_out[ti], grad_c.contiguous(),
                 gates_all[ti], c_in.contiguous(),
                 mask[ti] if mask is not None else empty_mask,
-                grad_gates_all[ti], grad_c_prev, grad_h_pass,
+                grad_gates_all[ti], grad_c_prev, grad_h_pass, bias_accum,
             )
+            bias_fused = bias_fused and bool(used)
             grad_h = torch.addmm(grad_h_pass, grad_gates_all[ti], w_hh)
         else:
             gg, gcp, ghp = _ref.lstm_pointwise_bwd_ref(
@@ -233,7 +241,11 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
         grad_whh = torch.addmm(grad_whh, edge_gg.t(), h0.to(dt))
     else:
         grad_whh = torch.mm(gg2.t(), h0.to(dt))
-    grad_bias = gg2.sum(dim=0)
+    grad_bias = (
+        bias_accum.to(dt)
+        if (bias_accum is not None and bias_fused)
+        else gg2.sum(dim=0)
+    )
     return grad_gates_all, grad_h, grad_c, grad_whh, grad_bias
 
 

@@ -955,3 +955,30 @@ def test_sage_layer_training_fused_matches_eager_path(gpu_device):
     assert torch.allclose(out.float(), out2.float(), atol=3e-2, rtol=3e-2)
     scale = h2.grad.float().abs().max().clamp(min=1.0)
     assert torch.allclose(gf / scale, h2.grad.float() / scale, atol=4e-2)
+
+
+def test_lstm_bwd_fused_bias_accum_matches_sum(gpu_device):
+    """The in-kernel bias-grad accumulation equals gg.sum(0)."""
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(61)
+    b, hd = 3000, 256
+    g = 4 * hd
+    gh = (torch.randn(b, hd, device=gpu_device) * 0.2).to(torch.bfloat16)
+    gout = (torch.randn(b, hd, device=gpu_device) * 0.2).to(torch.bfloat16)
+    gc = (torch.randn(b, hd, device=gpu_device) * 0.2).to(torch.bfloat16)
+    gacts = torch.sigmoid(torch.randn(b, g, device=gpu_device)).to(torch.bfloat16)
+    c = (torch.randn(b, hd, device=gpu_device) * 0.2).to(torch.bfloat16)
+    mask = (torch.rand(b, device=gpu_device) > 0.2).float()
+    gg = torch.empty(b, g, device=gpu_device, dtype=torch.bfloat16)
+    gcp = torch.empty(b, hd, device=gpu_device, dtype=torch.bfloat16)
+    ghp = torch.empty_like(gcp)
+    accum = torch.zeros(g, device=gpu_device, dtype=torch.float32)
+    used = ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp, accum)
+    assert used
+    ref = gg.float().sum(0)
+    assert torch.allclose(accum, ref, atol=0.5, rtol=1e-2)
+    # accumulates ACROSS calls (the sequence loop reuses one accumulator)
+    ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp, accum)
+    assert torch.allclose(accum, 2 * ref, atol=1.0, rtol=1e-2)

@@ -43,7 +43,7 @@ def fwd_fused():
 
 
 def bwd_split():
-    ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp)
+    ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp, torch.empty(0, device=dev))
     torch.addmm(ghp, gg, w, out=gh_out)
 
 

@@ -45,7 +45,7 @@ def run():
         torch.mm(h, w_t, out=hg)
         ext.lstm_pointwise_fwd(hg, xgt, bias, c, h, mask, h_out, c_out, gates)
     elif which == "bwd_split":
-        ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp)
+        ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp, torch.empty(0, device=dev))
         torch.addmm(ghp, gg, w, out=gh_out)
     elif which == "proj":
         ext.proj_fwd_dual(x, w1, w2, c1, c2)
