@@ -216,8 +216,8 @@ bool lstm_pointwise_bwd(torch::Tensor grad_h, torch::Tensor grad_out_t,
   if (bias_accum.numel() > 0 && vec) {
     check_gpu_contig(bias_accum, "bias_accum");
     TORCH_CHECK(bias_accum.scalar_type() == torch::kFloat32 &&
-                    bias_accum.numel() == 4 * hdim,
-                "bias_accum must be f32 [4H]");
+                    bias_accum.numel() == 64 * 4 * hdim,
+                "bias_accum must be f32 [64, 4H] (striped replicas)");
     bias_ptr = bias_accum.data_ptr<float>();
   }
   nerrf::launch_lstm_pointwise_bwd(

@@ -270,8 +270,13 @@ __global__ void lstm_pointwise_bwd_vec_kernel(
   }
   if (bias_accum != nullptr) {
     __syncthreads();
+    // 64 striped replicas: ~8k blocks all adding to ONE [4H] vector
+    // serialize on the per-address atomic queues (~100 us/launch measured);
+    // striping by block id cuts each address's chain 64x.  The caller sums
+    // the [64, 4H] buffer once per direction.
+    float* rep = bias_accum + (long)(blockIdx.x & 63) * (4 * hdim);
     for (int i = threadIdx.x; i < 4 * hdim; i += blockDim.x)
-      if (bias_lds[i] != 0.0f) atomicAdd(bias_accum + i, bias_lds[i]);
+      if (bias_lds[i] != 0.0f) atomicAdd(rep + i, bias_lds[i]);
   }
 }
 

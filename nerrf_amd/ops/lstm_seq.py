@@ -189,7 +189,7 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
         grad_bias = gg2.sum(dim=0)
         return grad_gates_all, grad_h, grad_c, grad_whh, grad_bias
     bias_accum = (
-        torch.zeros(gdim, device=dev, dtype=torch.float32)
+        torch.zeros(64, gdim, device=dev, dtype=torch.float32)
         if ext is not None else None
     )
     bias_fused = ext is not None
@@ -242,7 +242,7 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
     else:
         grad_whh = torch.mm(gg2.t(), h0.to(dt))
     grad_bias = (
-        bias_accum.to(dt)
+        bias_accum.sum(dim=0).to(dt)
         if (bias_accum is not None and bias_fused)
         else gg2.sum(dim=0)
     )
