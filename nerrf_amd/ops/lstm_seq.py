@@ -190,9 +190,10 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
         return grad_gates_all, grad_h, grad_c, grad_whh, grad_bias
     bias_accum = (
         torch.zeros(64, gdim, device=dev, dtype=torch.float32)
-        if ext is not None else None
+        if ext is not None and os.environ.get("NERRF_FUSED_BIAS", "1") == "1"
+        else None
     )
-    bias_fused = ext is not None
+    bias_fused = bias_accum is not None
     for ti in steps:
         # c/h input of step ti = previous step's output (or h0/c0 at start)
         first = (ti == t_len - 1) if reverse else (ti == 0)
@@ -208,7 +209,8 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
                 grad_h.contiguous(), grad_out[ti], grad_c.contiguous(),
                 gates_all[ti], c_in.contiguous(),
                 mask[ti] if mask is not None else empty_mask,
-                grad_gates_all[ti], grad_c_prev, grad_h_pass, bias_accum,
+                grad_gates_all[ti], grad_c_prev, grad_h_pass,
+                bias_accum if bias_accum is not None else empty_mask,
             )
             bias_fused = bias_fused and bool(used)
             grad_h = torch.addmm(grad_h_pass, grad_gates_all[ti], w_hh)
