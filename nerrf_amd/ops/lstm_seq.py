@@ -188,9 +188,13 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
             grad_whh = torch.mm(gg2.t(), h0.to(dt))
         grad_bias = gg2.sum(dim=0)
         return grad_gates_all, grad_h, grad_c, grad_whh, grad_bias
+    # opt-in: in-kernel bias-grad accumulation (NERRF_FUSED_BIAS=1) removes
+    # the gg.sum(0) re-reads but the LDS atomicAdds cost the bwd pointwise
+    # kernel ~+105 ms/step at production shapes (A/B: 244.0 vs 349.3 ms) —
+    # a measured negative on CDNA4; the separate reduction stays default
     bias_accum = (
         torch.zeros(64, gdim, device=dev, dtype=torch.float32)
-        if ext is not None and os.environ.get("NERRF_FUSED_BIAS", "1") == "1"
+        if ext is not None and os.environ.get("NERRF_FUSED_BIAS", "0") == "1"
         else None
     )
     bias_fused = bias_accum is not None
