@@ -14,6 +14,7 @@ it.
 """
 from __future__ import annotations
 
+import os
 import re
 import time
 from dataclasses import dataclass, field
@@ -144,9 +145,13 @@ class StreamingEngine:
         # delta set) merges cached per-delta summaries instead of re-scanning
         # every event; an explicit `now` trims at event granularity, which
         # invalidates the summaries, so that path takes the full rebuild
+        timing = {} if os.environ.get("NERRF_SERVE_TIMING") else None
+        _t0 = time.perf_counter() if timing is not None else 0.0
         events, window_deltas = self.store.compact_with_deltas(now)
         use_incremental = now is None and len(window_deltas) > 0
         t_detect = time.time()
+        if timing is not None:
+            timing["compact"] = time.perf_counter() - _t0
         if len(events) == 0:
             return Detection(False, t_detect, {}, {}, {}, window_events=0)
         # ---- graph build: host identity/edges once; GPU feature compaction
@@ -183,8 +188,12 @@ class StreamingEngine:
         node_key = _np.concatenate(
             [parts["touched_roots"], parts["upids"].astype(_np.int64)]
         )
+        if timing is not None:
+            timing["graph_merge"] = time.perf_counter() - _t0 - sum(timing.values())
         csr = to_csr(ed["edge_index"], parts["n_nodes"], ed["edge_weight"])
         nbr_idx, nbr_w = sample_fanout(csr, 16, seed=self.scored_windows)
+        if timing is not None:
+            timing["csr_fanout"] = time.perf_counter() - _t0 - sum(timing.values())
 
         if self.device.type == "cuda":
             from ..graph.gpu_store import gpu_window_graph
@@ -224,6 +233,8 @@ class StreamingEngine:
         }
         node_logit, _, seq_logit = self.model(batch)
         node_score = torch.sigmoid(node_logit.float()).cpu().numpy()
+        if timing is not None:
+            timing["model_fwd_sync"] = time.perf_counter() - _t0 - sum(timing.values())
         seq_score = (
             torch.sigmoid(seq_logit.float()).cpu().numpy() if seq_logit is not None else None
         )
@@ -263,6 +274,8 @@ class StreamingEngine:
         for i in np.nonzero(mb_by_id)[0]:
             file_mb[events.paths.lookup(int(i))] = float(mb_by_id[i])
 
+        if timing is not None:
+            timing["score_maps"] = time.perf_counter() - _t0 - sum(timing.values())
         # ---- rule indicators --------------------------------------------
         sc = events.syscall
         from ..data.trace import SYSCALL_IDS
@@ -320,6 +333,8 @@ class StreamingEngine:
                      or min(model_max, seq_max) >= self.calibrated["model_thr"])
         else:
             alarm = alarm_score >= self.alarm_threshold
+        if timing is not None:
+            timing["indicators"] = time.perf_counter() - _t0 - sum(timing.values())
         # refinement context: lets plan() re-score counterfactual post-plan
         # graphs through the GNN in one batch (planner/model_eval.py)
         path_to_node = {
@@ -342,6 +357,10 @@ class StreamingEngine:
             "path_to_node": path_to_node,
             "proc_nodes": proc_nodes,
         }
+        if timing is not None:
+            timing["refine_ctx_rest"] = time.perf_counter() - _t0 - sum(timing.values())
+            timing["total"] = time.perf_counter() - _t0
+            self.last_timing = dict(timing)
         return Detection(
             alarm=alarm,
             t_detect=t_detect,
@@ -356,6 +375,7 @@ class StreamingEngine:
             node_max=model_max,
             seq_max=seq_max,
         )
+
 
     # -------------------------------------------------------------------- plan
     def plan(self, det: Detection, n_sims: int = 1024, use_gpu: Optional[bool] = None,

@@ -37,7 +37,7 @@ t0 = time.perf_counter()
 plan = engine.plan(det, n_sims=1024, use_gpu=(dev != "cpu"))
 t_plan = time.perf_counter() - t0
 
-print(json.dumps({
+out = {
     "window_events": det.window_events,
     "ingest_s": t_ingest,
     "score_s_per_window": t_score,
@@ -46,4 +46,7 @@ print(json.dumps({
     "plan_s": t_plan,
     "plan_sims": plan.simulations,
     "plan": plan.describe(engine.planner_params.n_groups)[:4],
-}))
+}
+if getattr(engine, "last_timing", None):
+    out["stage_ms"] = {k: round(v * 1e3, 2) for k, v in engine.last_timing.items()}
+print(json.dumps(out))
