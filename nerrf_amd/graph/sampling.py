@@ -113,3 +113,50 @@ def edge_reverse_index(endpoints: np.ndarray) -> Tuple[np.ndarray, np.ndarray, n
         order.astype(np.int64),
         np.ones(len(order), dtype=np.float32),
     )
+
+
+def sample_fanout_torch(edge_index, edge_weight, num_nodes: int, fanout: int,
+                        seed: int = 0, symmetric: bool = True):
+    """Device-side CSR build + fixed-fanout sampling (torch tensors in/out).
+
+    The serving tick was spending ~13.6 ms per 600k-event window on the
+    HOST argsort inside to_csr/sample_fanout while every consumer of the
+    [N, K] neighbor matrices lives on the GPU; this computes them where
+    they are used.  Same algorithm as the numpy pair (dst-indexed CSR,
+    random offsets modulo degree, self-fill for isolated nodes); the RNG
+    stream differs (torch generator), which only changes WHICH valid
+    neighbors are drawn.
+    """
+    import torch
+
+    dev = edge_index.device
+    src, dst = edge_index[0], edge_index[1]
+    w = edge_weight.float()
+    if symmetric:
+        src = torch.cat([src, edge_index[1]])
+        dst = torch.cat([dst, edge_index[0]])
+        w = torch.cat([w, w])
+    order = torch.argsort(dst, stable=True)
+    src_s, w_s = src[order], w[order]
+    counts = torch.bincount(dst[order], minlength=num_nodes)
+    indptr = torch.zeros(num_nodes + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    deg = counts
+    k = fanout
+    if src_s.numel() == 0:
+        idx = torch.arange(num_nodes, device=dev).unsqueeze(1).expand(num_nodes, k).contiguous()
+        return idx, torch.ones(num_nodes, k, device=dev)
+    g = torch.Generator(device=dev)
+    g.manual_seed(int(seed) & 0x7FFFFFFF)
+    rand = torch.randint(0, 1 << 62, (num_nodes, k), generator=g, device=dev)
+    safe_deg = deg.clamp(min=1)
+    offs = rand % safe_deg.unsqueeze(1)
+    flat = (indptr[:-1].unsqueeze(1) + offs).clamp(0, src_s.numel() - 1)
+    idx = src_s[flat]
+    wk = w_s[flat]
+    isolated = deg == 0
+    if bool(isolated.any()):
+        own = torch.arange(num_nodes, device=dev)
+        idx[isolated] = own[isolated].unsqueeze(1)
+        wk[isolated] = 1.0
+    return idx, wk

@@ -982,3 +982,35 @@ def test_lstm_bwd_fused_bias_accum_matches_sum(gpu_device):
     # accumulates ACROSS calls (the sequence loop reuses one accumulator)
     ext.lstm_pointwise_bwd(gh, gout, gc, gacts, c, mask, gg, gcp, ghp, accum)
     assert torch.allclose(accum.sum(0), 2 * ref, atol=1.0, rtol=1e-2)
+
+
+def test_sample_fanout_torch_structural(gpu_device):
+    """Device-side fanout sampler: every drawn neighbor is a valid CSR
+    in-neighbor with its edge weight; isolated nodes self-fill."""
+    import numpy as np
+
+    from nerrf_amd.graph.sampling import sample_fanout_torch, to_csr
+
+    rng = np.random.default_rng(9)
+    n, e = 500, 3000
+    ei = np.stack([rng.integers(0, n - 5, e), rng.integers(0, n - 5, e)]).astype(np.int64)
+    ew = rng.random(e).astype(np.float32) + 0.01
+    t_ei = torch.from_numpy(ei).to(gpu_device)
+    t_ew = torch.from_numpy(ew).to(gpu_device)
+    idx, w = sample_fanout_torch(t_ei, t_ew, n, 16, seed=3)
+    assert idx.shape == (n, 16) and w.shape == (n, 16)
+    csr = to_csr(ei, n, ew)
+    idx_np = idx.cpu().numpy()
+    w_np = w.float().cpu().numpy()
+    deg = np.diff(csr.indptr)
+    for node in list(range(0, n, 37)) + [n - 1]:
+        nbrs = set(csr.indices[csr.indptr[node]:csr.indptr[node + 1]].tolist())
+        wmap = {}
+        for j, nb in enumerate(csr.indices[csr.indptr[node]:csr.indptr[node + 1]]):
+            wmap.setdefault(int(nb), set()).add(round(float(csr.weights[csr.indptr[node] + j]), 5))
+        if deg[node] == 0:
+            assert (idx_np[node] == node).all() and (w_np[node] == 1.0).all()
+        else:
+            for nb, wv in zip(idx_np[node], w_np[node]):
+                assert int(nb) in nbrs
+                assert round(float(wv), 5) in wmap[int(nb)]
