@@ -44,6 +44,29 @@ _SOCKET_PREFIXES = ("tcp://", "udp://")
 DEST_ALLOWLIST = ("tcp://10.", "tcp://192.168.", "tcp://backup.", "udp://10.")
 
 
+def rename_path_roots(ra: np.ndarray, rb: np.ndarray, n_paths: int) -> np.ndarray:
+    """path_root from rename pairs: connected components restricted to the
+    paths that actually appear in a rename (the full-path-space CC walked
+    every interned string — ~12 ms/tick at 600k paths for a few hundred
+    rename endpoints).  Root = minimum original path id of the component."""
+    if ra.size == 0:
+        return np.arange(n_paths, dtype=np.int64)
+    from scipy.sparse import coo_matrix
+    from scipy.sparse.csgraph import connected_components
+
+    nodes = np.unique(np.concatenate([ra, rb]))
+    m = len(nodes)
+    la = np.searchsorted(nodes, ra)
+    lb = np.searchsorted(nodes, rb)
+    g = coo_matrix((np.ones(ra.size, dtype=np.int8), (la, lb)), shape=(m, m))
+    n_comp, label = connected_components(g, directed=False)
+    min_id = np.full(n_comp, np.iinfo(np.int64).max, dtype=np.int64)
+    np.minimum.at(min_id, label, nodes)
+    path_root = np.arange(n_paths, dtype=np.int64)
+    path_root[nodes] = min_id[label]
+    return path_root
+
+
 def aggregate_sparse_keys(
     key: np.ndarray, wsum: np.ndarray, ts_v: np.ndarray, key_space: int
 ) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
@@ -243,19 +266,7 @@ def build_graph_parts(
     rb = events.new_path_id[ren_mask]
     m = (ra >= 0) & (rb >= 0)
     ra, rb = ra[m], rb[m]
-    if ra.size:
-        from scipy.sparse import coo_matrix
-        from scipy.sparse.csgraph import connected_components
-
-        g = coo_matrix(
-            (np.ones(ra.size, dtype=np.int8), (ra, rb)), shape=(n_paths, n_paths)
-        )
-        n_comp, label = connected_components(g, directed=False)
-        min_id = np.full(n_comp, np.iinfo(np.int64).max, dtype=np.int64)
-        np.minimum.at(min_id, label, np.arange(n_paths, dtype=np.int64))
-        path_root = min_id[label]
-    else:
-        path_root = np.arange(n_paths, dtype=np.int64)
+    path_root = rename_path_roots(ra, rb, n_paths)
 
     # file nodes = distinct roots actually touched (presence mask instead of
     # a concatenate + sort-based unique over ~2x the event count)

@@ -30,7 +30,8 @@ from typing import Dict, List, Tuple
 import numpy as np
 
 from ..data.trace import SYSCALL_IDS, EventArray
-from .constructor import _READ_LIKE, _path_flags, aggregate_sparse_keys
+from .constructor import (_READ_LIKE, _path_flags, aggregate_sparse_keys,
+                          rename_path_roots)
 
 _SCALE_GUARD = 60.0  # max (t1 - t_d0)/tau before a partial sum underflows
 
@@ -157,17 +158,7 @@ def merge_window(
         rb = np.concatenate([s.ren_b for s in summaries])
     else:
         ra = rb = np.empty(0, np.int64)
-    if ra.size:
-        from scipy.sparse import coo_matrix
-        from scipy.sparse.csgraph import connected_components
-
-        g = coo_matrix((np.ones(ra.size, dtype=np.int8), (ra, rb)), shape=(n_paths, n_paths))
-        n_comp, label = connected_components(g, directed=False)
-        min_id = np.full(n_comp, np.iinfo(np.int64).max, dtype=np.int64)
-        np.minimum.at(min_id, label, np.arange(n_paths, dtype=np.int64))
-        path_root = min_id[label]
-    else:
-        path_root = np.arange(n_paths, dtype=np.int64)
+    path_root = rename_path_roots(ra, rb, n_paths)
 
     # ---- node tables from merged unique sets ------------------------------
     seen = np.zeros(n_paths, dtype=bool)
