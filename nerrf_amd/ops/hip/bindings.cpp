@@ -42,7 +42,7 @@ void launch_proj_fwd_dual(const void*, const void*, const void*, void*, void*,
 void launch_proj_dgrad_dual(const void*, const void*, const void*, const void*,
                             void*, long, hipStream_t);
 void launch_proj_wgrad(const void*, const void*, const void*, float*, float*,
-                       long, int, hipStream_t);
+                       long, long, int, hipStream_t);
 void launch_event_scatter(const long*, const long*, const signed char*,
                           const float*, const int*, float*, int*, int*, long,
                           hipStream_t);
@@ -392,13 +392,17 @@ void proj_dgrad_dual(torch::Tensor a1, torch::Tensor a2, torch::Tensor w1t,
 // directions in one launch (transposed LDS tiles, f32 atomic partials).
 std::vector<torch::Tensor> proj_wgrad(torch::Tensor g1, torch::Tensor g2,
                                       torch::Tensor x, long n_mchunks) {
-  check_gpu_contig(g1, "g1");
+  // g1/g2 may be column slabs of one [M, 2048] cat-layout grad (rows
+  // contiguous, shared row stride)
+  const long g_stride = row_stride_checked(g1, "g1");
   check_gpu_contig(x, "x");
   TORCH_CHECK(g1.scalar_type() == torch::kBFloat16, "proj_wgrad is bf16-only");
   TORCH_CHECK(g1.size(1) == 1024 && x.size(1) == 512,
               "proj_wgrad requires g [M,1024], x [M,512]");
   const bool dual = g2.numel() > 0;
-  if (dual) check_gpu_contig(g2, "g2");
+  if (dual)
+    TORCH_CHECK(row_stride_checked(g2, "g2") == g_stride,
+                "g1/g2 must share a row stride");
   auto dw1 = torch::zeros({1024, 512}, x.options().dtype(torch::kFloat32));
   auto dw2 = dual ? torch::zeros({1024, 512}, x.options().dtype(torch::kFloat32))
                   : torch::empty({0}, x.options().dtype(torch::kFloat32));
@@ -406,7 +410,7 @@ std::vector<torch::Tensor> proj_wgrad(torch::Tensor g1, torch::Tensor g2,
   nerrf::launch_proj_wgrad(g1.data_ptr(), dual ? g2.data_ptr() : nullptr,
                            x.data_ptr(), dw1.data_ptr<float>(),
                            dual ? dw2.data_ptr<float>() : nullptr, g1.size(0),
-                           (int)n_mchunks, stream.stream());
+                           g_stride, (int)n_mchunks, stream.stream());
   return {dw1, dw2};
 }
 

@@ -321,14 +321,13 @@ void launch_proj_dgrad_dual(const void* a1, const void* a2, const void* w1t,
 // ---------------------------------------------------------------------------
 
 // stage a [32 m][256 col] global panel as a transposed [256 col][32 m]
-// bf16 LDS tile (row stride 80 B).  256 threads; returns nothing staged for
-// out-of-range rows (zeros).
-__device__ __forceinline__ void wg_stage_t(
+// bf16 LDS tile (row stride 80 B), split T14-style: issue the loads BEFORE
+// the compute phase they overlap, transpose+write after the barrier.
+__device__ __forceinline__ void wg_stage_load(
     const __hip_bfloat16* __restrict__ src, long src_stride, long m0,
-    long m_rows, int col0, char* dst, int task) {
+    long m_rows, int col0, int task, uint4 (&in)[4]) {
   const int mg = task >> 5;       // 0..7: group of 4 m rows
   const int cg = task & 31;       // 0..31: group of 8 columns
-  uint4 in[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const long m = m0 + mg * 4 + i;
@@ -337,6 +336,12 @@ __device__ __forceinline__ void wg_stage_t(
       in[i] = *reinterpret_cast<const uint4*>(src + m * src_stride + col0 +
                                               cg * 8);
   }
+}
+
+__device__ __forceinline__ void wg_stage_write(char* dst, int task,
+                                               const uint4 (&in)[4]) {
+  const int mg = task >> 5;
+  const int cg = task & 31;
   // 4x8 micro-transpose: out[j] = {in[0].h[j], in[1].h[j], in[2].h[j], in[3].h[j]}
   const ushort* h0 = reinterpret_cast<const ushort*>(&in[0]);
   const ushort* h1 = reinterpret_cast<const ushort*>(&in[1]);
@@ -353,12 +358,12 @@ __device__ __forceinline__ void wg_stage_t(
 
 __launch_bounds__(512)
 __global__ void proj_wgrad_kernel(
-    const __hip_bfloat16* __restrict__ g1,  // [M, 1024] grad_xg dir 0
-    const __hip_bfloat16* __restrict__ g2,  // [M, 1024] dir 1 or nullptr
+    const __hip_bfloat16* __restrict__ g1,  // [M(g_stride), 1024] grad_xg dir 0
+    const __hip_bfloat16* __restrict__ g2,  // [M(g_stride), 1024] dir 1 or null
     const __hip_bfloat16* __restrict__ x,   // [M, 512]
     float* __restrict__ dw1,                // [1024, 512] f32 (zeroed)
     float* __restrict__ dw2,                // [1024, 512] f32
-    long m_rows, int n_mchunks) {
+    long m_rows, long g_stride, int n_mchunks) {
   constexpr int TILE_BYTES = 256 * 80;  // 20 KB per transposed tile
   extern __shared__ __attribute__((aligned(16))) char smem[];
   auto ta = [&](int i) -> char* { return smem + i * TILE_BYTES; };          // A^T dbuf
@@ -390,10 +395,16 @@ __global__ void proj_wgrad_kernel(
   const int nrow0 = wave * 32;  // wave's 32 output rows (n of dW)
 
   // prologue: stage first 32-m panel of both operands
-  if (tid < 256)
-    wg_stage_t(g, 1024, m_lo, m_hi, n0, ta(0), tid);
-  else
-    wg_stage_t(x, 512, m_lo, m_hi, k0, tb(0), tid - 256);
+  {
+    uint4 pre[4];
+    if (tid < 256) {
+      wg_stage_load(g, g_stride, m_lo, m_hi, n0, tid, pre);
+      wg_stage_write(ta(0), tid, pre);
+    } else {
+      wg_stage_load(x, 512, m_lo, m_hi, k0, tid - 256, pre);
+      wg_stage_write(tb(0), tid - 256, pre);
+    }
+  }
   __syncthreads();
 
   sf32x4 acc[2][16];
@@ -406,10 +417,19 @@ __global__ void proj_wgrad_kernel(
   const int mchunk_b = (lane >> 4) * 16;
   const long n_steps = (m_hi - m_lo + 31) / 32;
 
+  uint4 pre[4];
   for (long st = 0; st < n_steps; ++st) {
     const int cur = (int)(st & 1);
-    // compute on cur while the other half of the block... (single phase:
-    // stage next AFTER compute, standard 2-buffer with two barriers)
+    // T14 split: issue the NEXT panel's global loads now — their ~900-cycle
+    // HBM latency hides under this step's 32 MFMAs (the fused load+write
+    // stage between two barriers serialized the whole kernel: 51 ms)
+    const bool has_next = st + 1 < n_steps;
+    if (has_next) {
+      if (tid < 256)
+        wg_stage_load(g, g_stride, m_lo + (st + 1) * 32, m_hi, n0, tid, pre);
+      else
+        wg_stage_load(x, 512, m_lo + (st + 1) * 32, m_hi, k0, tid - 256, pre);
+    }
     const char* a_t = ta(cur);
     const char* b_t = tb(cur);
 #pragma unroll
@@ -431,13 +451,13 @@ __global__ void proj_wgrad_kernel(
               afr, bfr[c8], acc[rf][half * 8 + c8], 0, 0, 0);
       }
     }
-    if (st + 1 < n_steps) {
+    if (has_next) {
       __syncthreads();  // everyone done with the buffer we are overwriting
       const int nxt = (int)((st + 1) & 1);
       if (tid < 256)
-        wg_stage_t(g, 1024, m_lo + (st + 1) * 32, m_hi, n0, ta(nxt), tid);
+        wg_stage_write(ta(nxt), tid, pre);
       else
-        wg_stage_t(x, 512, m_lo + (st + 1) * 32, m_hi, k0, tb(nxt), tid - 256);
+        wg_stage_write(tb(nxt), tid - 256, pre);
       __syncthreads();
     }
   }
@@ -457,14 +477,14 @@ __global__ void proj_wgrad_kernel(
 }
 
 void launch_proj_wgrad(const void* g1, const void* g2, const void* x,
-                       float* dw1, float* dw2, long m_rows, int n_mchunks,
-                       hipStream_t s) {
+                       float* dw1, float* dw2, long m_rows, long g_stride,
+                       int n_mchunks, hipStream_t s) {
   const int n_tiles = (g2 != nullptr) ? 16 : 8;
   const int grid = n_mchunks * n_tiles;
   const size_t lds = 4 * 256 * 80;  // 80 KB
   proj_wgrad_kernel<<<grid, 512, lds, s>>>(
       (const __hip_bfloat16*)g1, (const __hip_bfloat16*)g2,
-      (const __hip_bfloat16*)x, dw1, dw2, m_rows, n_mchunks);
+      (const __hip_bfloat16*)x, dw1, dw2, m_rows, g_stride, n_mchunks);
 }
 
 }  // namespace nerrf
