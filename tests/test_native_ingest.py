@@ -259,3 +259,24 @@ def test_fd_resolver_exact_and_heuristic():
     b3.add(ts=1.0, pid=9, syscall="write", path="/known.log", nbytes=1)
     r3 = resolve_fd_paths(b3.build())
     assert r3.paths.lookup(int(r3.path_id[0])) == "/known.log"
+
+
+def test_codec_negative_nanos_cross_language():
+    """negative Timestamp nanos must round-trip
+    identically through BOTH codecs (C++ sign-extends on encode; both
+    decoders truncate to int32 like the protobuf runtime)."""
+    ev = codec.Event(ts_sec=5, ts_nsec=-7, pid=1, tid=1, comm="x",
+                     syscall="write", path="/f", ret_val=-2, bytes=3)
+    # python encode -> C++ decode
+    d = _ingest.decode_batch(codec.encode_event_batch([ev]))[0]
+    assert (d["ts_sec"], d["ts_nsec"]) == (5, -7)
+    # C++ encode -> python decode
+    frame = _ingest.encode_batch([dict(
+        ts_sec=5, ts_nsec=-7, pid=1, tid=1, comm="x", syscall="write",
+        path="/f", new_path="", ret_val=-2, bytes=3)])
+    back = codec.decode_event_batch(frame)[0]
+    assert (back.ts_sec, back.ts_nsec) == (5, -7)
+    # and byte-identical frames from the two encoders
+    assert frame == codec.encode_event_batch([codec.Event(
+        ts_sec=5, ts_nsec=-7, pid=1, tid=1, comm="x", syscall="write",
+        path="/f", ret_val=-2, bytes=3)])
