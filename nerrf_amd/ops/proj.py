@@ -10,7 +10,11 @@ backward's input gradient sums both directions in one pass
 (`proj_dgrad_dual`).  Weight grads stay on hipBLASLt for now (huge-K
 reduction — own kernel is round-2 follow-up work).
 
-Falls back to torch.matmul off-GPU, off-shape, or when NERRF_STREAM_PROJ=0.
+Measured round 2: the custom kernels run at 0.88x (fwd) / 0.77x (dgrad)
+of tuned hipBLASLt at production shapes (profiles/PROFILES.md ladder), so
+they are OPT-IN (NERRF_STREAM_PROJ=1) and the model uses the concatenated
+single-GEMM layout instead (models/lstm.py).  Kept as the validated
+baseline for the round-3 pipelined rewrite.
 """
 from __future__ import annotations
 
@@ -25,7 +29,7 @@ def _use_stream(x: torch.Tensor, w_f: torch.Tensor, w_b: torch.Tensor) -> bool:
     return (
         x.is_cuda
         and x.dtype == torch.bfloat16
-        and os.environ.get("NERRF_STREAM_PROJ", "1") == "1"
+        and os.environ.get("NERRF_STREAM_PROJ", "0") == "1"
         and x.dim() == 2
         and x.shape[1] == 512
         and tuple(w_f.shape) == (1024, 512)
