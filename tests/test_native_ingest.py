@@ -280,3 +280,45 @@ def test_codec_negative_nanos_cross_language():
     assert frame == codec.encode_event_batch([codec.Event(
         ts_sec=5, ts_nsec=-7, pid=1, tid=1, comm="x", syscall="write",
         path="/f", ret_val=-2, bytes=3)])
+
+
+@pytest.mark.skipif(not DAEMON.exists(), reason="nerrfd not built")
+def test_nerrfd_filter_prefix(tmp_path):
+    """Capture-side path filtering (upstream M2 plan): only events under the
+    configured prefixes (plus pathless ones) leave the daemon."""
+    from nerrf_amd.graph.store import DeltaGraphStore
+    from nerrf_amd.serve.daemon_bridge import pump_daemon_into_store
+
+    trace = tmp_path / "t.jsonl"
+    with open(trace, "w") as fh:
+        for i in range(90):
+            path = ["/app/uploads/a.dat", "/var/log/syslog", "/tmp/x"][i % 3]
+            fh.write(json.dumps({
+                "timestamp": 10.0 + i * 0.01, "event": "openat",
+                "path": path, "size": 1, "pid": 7,
+            }) + "\n")
+        fh.write(json.dumps({  # pathless event must pass the filter
+            "timestamp": 11.0, "event": "write", "path": "", "size": 64,
+            "pid": 7,
+        }) + "\n")
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    proc = subprocess.Popen(
+        [str(DAEMON), "--replay", str(trace), "--port", str(port), "--once",
+         "--batch", "16", "--filter-prefix", "/app/uploads/",
+         "--filter-prefix", "/tmp/"],
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+    )
+    try:
+        time.sleep(0.3)
+        store = DeltaGraphStore(window_s=1e9)
+        n = pump_daemon_into_store("127.0.0.1", port, store, timeout_s=10.0)
+        assert n == 61  # 30 uploads + 30 tmp + 1 pathless
+        arr = store.compact()
+        names = {arr.paths.lookup(int(i)) for i in arr.path_id if i >= 0}
+        assert "/var/log/syslog" not in names
+        assert {"/app/uploads/a.dat", "/tmp/x"} <= names
+    finally:
+        proc.terminate()
+        proc.wait(timeout=5)

@@ -21,6 +21,7 @@
 static void *(*bpf_ringbuf_reserve)(void *ringbuf, __u64 size, __u64 flags) =
     (void *)131;
 static void (*bpf_ringbuf_submit)(void *data, __u64 flags) = (void *)132;
+static void (*bpf_ringbuf_discard)(void *data, __u64 flags) = (void *)133;
 static __u64 (*bpf_ktime_get_ns)(void) = (void *)5;
 static __u64 (*bpf_get_current_pid_tgid)(void) = (void *)14;
 static long (*bpf_get_current_comm)(void *buf, __u32 size) = (void *)16;
@@ -75,6 +76,54 @@ struct sys_exit_ctx {
   long ret;
 };
 
+/* BPF-side path filtering (upstream M2 plan, overview.mdx:262-269): when
+ * the daemon programs prefixes into this map, events whose paths match
+ * none of them are DISCARDED in-kernel — the ring buffer and userspace
+ * never see off-scope traffic.  Pathless events (plain write/read before
+ * fd resolution) always pass. */
+#define NERRF_FILTER_MAX 8
+#define NERRF_FILTER_LEN 64
+struct nerrf_filter_cfg {
+  __u32 n;
+  char prefix[NERRF_FILTER_MAX][NERRF_FILTER_LEN];
+};
+
+struct {
+  __uint(type, BPF_MAP_TYPE_ARRAY);
+  __uint(max_entries, 1);
+  __type(key, __u32);
+  __type(value, struct nerrf_filter_cfg);
+} path_filter SEC(".maps");
+
+static __always_inline int nerrf_path_allowed(const char *path) {
+  __u32 zero = 0;
+  struct nerrf_filter_cfg *cfg = bpf_map_lookup_elem(&path_filter, &zero);
+  if (!cfg || cfg->n == 0) return 1;
+  if (path[0] == 0) return 1;
+#pragma unroll
+  for (int i = 0; i < NERRF_FILTER_MAX; i++) {
+    if (i >= (int)cfg->n) break;
+    int match = 1;
+#pragma unroll
+    for (int j = 0; j < NERRF_FILTER_LEN; j++) {
+      char c = cfg->prefix[i][j];
+      if (c == 0) break;
+      if (path[j] != c) { match = 0; break; }
+    }
+    if (match) return 1;
+  }
+  return 0;
+}
+
+/* submit unless the filter rejects the event's path(s) */
+static __always_inline void submit_filtered(struct nerrf_event *ev) {
+  if (nerrf_path_allowed(ev->path) ||
+      (ev->new_path[0] && nerrf_path_allowed(ev->new_path)))
+    bpf_ringbuf_submit(ev, 0);
+  else
+    bpf_ringbuf_discard(ev, 0);
+}
+
 /* raw tracepoint context for syscalls:sys_enter_* */
 struct sys_enter_ctx {
   __u64 _pad;
@@ -113,7 +162,7 @@ int nerrf_trace_openat(struct sys_enter_ctx *ctx) {
     bpf_probe_read_user_str(v.path, sizeof(v.path), (void *)ctx->args[1]);
     bpf_map_update_elem(&pending_open, &tid, &v, 0 /* BPF_ANY */);
   }
-  bpf_ringbuf_submit(ev, 0);
+  submit_filtered(ev);
   return 0;
 }
 
@@ -173,7 +222,7 @@ int nerrf_trace_rename(struct sys_enter_ctx *ctx) {
   bpf_probe_read_user_str(ev->path, sizeof(ev->path), (void *)ctx->args[0]);
   bpf_probe_read_user_str(ev->new_path, sizeof(ev->new_path),
                           (void *)ctx->args[1]);
-  bpf_ringbuf_submit(ev, 0);
+  submit_filtered(ev);
   return 0;
 }
 
@@ -184,7 +233,7 @@ int nerrf_trace_renameat2(struct sys_enter_ctx *ctx) {
   bpf_probe_read_user_str(ev->path, sizeof(ev->path), (void *)ctx->args[1]);
   bpf_probe_read_user_str(ev->new_path, sizeof(ev->new_path),
                           (void *)ctx->args[3]);
-  bpf_ringbuf_submit(ev, 0);
+  submit_filtered(ev);
   return 0;
 }
 
@@ -193,7 +242,7 @@ int nerrf_trace_unlinkat(struct sys_enter_ctx *ctx) {
   struct nerrf_event *ev = reserve_event(NERRF_SYS_UNLINK);
   if (!ev) return 0;
   bpf_probe_read_user_str(ev->path, sizeof(ev->path), (void *)ctx->args[1]);
-  bpf_ringbuf_submit(ev, 0);
+  submit_filtered(ev);
   return 0;
 }
 
@@ -202,6 +251,6 @@ int nerrf_trace_fchmodat(struct sys_enter_ctx *ctx) {
   struct nerrf_event *ev = reserve_event(NERRF_SYS_CHMOD);
   if (!ev) return 0;
   bpf_probe_read_user_str(ev->path, sizeof(ev->path), (void *)ctx->args[1]);
-  bpf_ringbuf_submit(ev, 0);
+  submit_filtered(ev);
   return 0;
 }

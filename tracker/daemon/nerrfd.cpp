@@ -209,20 +209,40 @@ int main(int argc, char** argv) {
   double rate = 0.0;  // 0 = as fast as possible
   bool wait_client = true;
   bool once = false;
+  // capture-side path-prefix filtering (upstream M2 plan,
+  // overview.mdx:262-269: cut event volume before it leaves the node) —
+  // the same prefix set the eBPF program's path_filter map enforces
+  // in-kernel on a live host applies here to replay/live streams
+  std::vector<std::string> filter_prefixes;
   for (int i = 1; i < argc; ++i) {
     std::string a = argv[i];
     if (a == "--replay" && i + 1 < argc) replay_path = argv[++i];
     else if (a == "--port" && i + 1 < argc) port = atoi(argv[++i]);
     else if (a == "--batch" && i + 1 < argc) batch_size = atoi(argv[++i]);
     else if (a == "--rate" && i + 1 < argc) rate = atof(argv[++i]);
+    else if (a == "--filter-prefix" && i + 1 < argc)
+      filter_prefixes.push_back(argv[++i]);
     else if (a == "--no-wait") wait_client = false;
     else if (a == "--once") once = true;  // exit after one replay pass
     else {
       std::cerr << "usage: nerrfd [--replay trace.jsonl] [--port P] "
-                   "[--batch N] [--rate X] [--no-wait] [--once]\n";
+                   "[--batch N] [--rate X] [--filter-prefix P]... "
+                   "[--no-wait] [--once]\n";
       return 2;
     }
   }
+  auto passes_filter = [&](const Event& ev) {
+    if (filter_prefixes.empty()) return true;
+    // pathless events (e.g. unresolved writes) always pass: dropping them
+    // would hide byte-volume signal the detector consumes
+    if (ev.path.empty() && ev.new_path.empty()) return true;
+    for (const auto& p : filter_prefixes) {
+      if (ev.path.compare(0, p.size(), p) == 0) return true;
+      if (!ev.new_path.empty() && ev.new_path.compare(0, p.size(), p) == 0)
+        return true;
+    }
+    return false;
+  };
 
   int srv = socket(AF_INET, SOCK_STREAM, 0);
   int one = 1;
@@ -270,6 +290,7 @@ int main(int argc, char** argv) {
               std::min(dt, 0.5)));
       }
       t_prev = ev.timestamp();
+      if (!passes_filter(ev)) continue;
       batch.push_back(std::move(ev));
       if (int(batch.size()) >= batch_size) {
         bc.broadcast(nerrf::wire::encode_event_batch(batch));
