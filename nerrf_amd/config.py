@@ -25,6 +25,7 @@ class DataConfig:
     stride_s: float = 15.0
     attack_fraction: float = 0.6
     scenario_kinds: tuple = ("lockbit", "supply_chain")
+    config_jitter: bool = False  # vary duration/rate per scenario (see dataset)
     fanout: int = 16
     seq_len: int = 100
     seed: int = 0

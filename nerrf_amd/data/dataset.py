@@ -124,22 +124,33 @@ def synth_window_batches(
     base_seed: int = 0,
     kinds: tuple = ("lockbit", "supply_chain"),
     benign_kinds: tuple = ("lockbit", "benign_rotate", "benign_backup", "benign_build"),
+    config_jitter: bool = False,
 ) -> List[WindowBatch]:
     """Prebuild window batches from synthetic scenarios.
 
     Attack scenarios alternate over `kinds`; clean scenarios alternate over
     `benign_kinds` (plain background plus the hard negatives — log rotation
     and backup daemons — so the model learns to NOT fire on attack
-    lookalikes: the FP-undo < 5% target)."""
+    lookalikes: the FP-undo < 5% target).
+
+    `config_jitter` varies duration and event rate per scenario (x0.5-x2
+    around the given values) so the heads stay calibrated on
+    off-distribution window configurations — the round-2 calibration
+    sweep showed head maxima drifting at unseen duration/rate combos."""
     batches: List[WindowBatch] = []
     for i in range(n_scenarios):
         # interleave attack/benign at the requested ratio over ANY
         # scenario count, spread so small n still gets both classes
         _rank = [0, 2, 4, 6, 8, 1, 3, 5, 7, 9]  # rank of i%10 in a strided order
         is_attack = _rank[i % 10] < round(attack_fraction * 10)
+        dur_i, rate_i = duration_s, benign_rate_hz
+        if config_jitter:
+            # deterministic per-scenario jitter over a 4x span
+            dur_i = duration_s * (0.5 + 1.5 * ((i * 29) % 10) / 9.0)
+            rate_i = benign_rate_hz * (0.5 + 1.5 * ((i * 37) % 10) / 9.0)
         cfg = SynthConfig(
-            duration_s=duration_s,
-            benign_rate_hz=benign_rate_hz,
+            duration_s=dur_i,
+            benign_rate_hz=rate_i,
             attack=is_attack,
             seed=base_seed + 7919 * i,
             attack_start_frac=0.2 + 0.5 * ((i * 13) % 10) / 10.0,

@@ -107,6 +107,7 @@ def run_training(cfg: TrainConfig, resume: str | None = None) -> Dict[str, float
         seq_len=d.seq_len,
         base_seed=d.seed + 100000 * rank,
         kinds=tuple(d.scenario_kinds),
+        config_jitter=bool(getattr(d, "config_jitter", False)),
     )
     holdout = synth_window_batches(
         n_scenarios=cfg.run.eval_holdout,

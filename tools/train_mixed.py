@@ -13,6 +13,7 @@ cfg = load_config(None, [
     # (rotate/backup/build) come from dataset defaults and now include
     # benign_build (round-1 NEXT gap 7)
     "data.scenario_kinds=(lockbit,supply_chain,supply_chain_net)",
+    "data.config_jitter=true",
     "run.eval_holdout=6", "run.checkpoint_dir=gpurun_out/ckpt_mixed",
     "run.log_every=100",
 ])
