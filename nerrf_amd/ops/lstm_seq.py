@@ -47,15 +47,17 @@ def _dir_forward(ext, xg, h0, c0, w_hh, bias, mask, reverse, infer, h_all):
     steps = range(t_len - 1, -1, -1) if reverse else range(t_len)
     if ext is not None:
         empty_mask = torch.empty(0, device=dev)
-        # Opt-in bf16/H=256 path: the fused recurrent step
-        # (lstm_rec_fused.hip) keeps the [B, 4H] pre-activation slab out of
-        # HBM entirely (one launch per timestep), but measured at parity
-        # fwd / -13% bwd vs the split path: its LDS transpose stage caps
-        # the kernel at 1 block/CU and the memory phases go latency-bound
-        # (~2.5 TB/s vs the split pointwise's 5.3 — profiles/PROFILES.md).
+        # Fused recurrent step (lstm_rec_fused.hip): keeps the [B, 4H]
+        # pre-activation slab out of HBM (one launch per timestep).
+        # Measured: TRAINING at parity fwd / -13% bwd vs split (LDS
+        # transpose stage caps occupancy — profiles/PROFILES.md ladder),
+        # but INFERENCE 16% faster per serving window (no gate store:
+        # 394 -> ~100 MB/timestep).  Default: fused on the infer path,
+        # split for training; NERRF_REC_FUSED=1/0 forces either.
+        mode = os.environ.get("NERRF_REC_FUSED", "infer")
         rec_fused = (
             hasattr(ext, "lstm_rec_fwd")
-            and os.environ.get("NERRF_REC_FUSED", "0") == "1"
+            and (mode == "1" or (mode == "infer" and infer))
             and dt == torch.bfloat16 and hdim == 256 and w_hh.is_contiguous()
         )
         if rec_fused:
@@ -153,7 +155,7 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
     rec_fused = (
         ext is not None
         and hasattr(ext, "lstm_rec_bwd")
-        and os.environ.get("NERRF_REC_FUSED", "0") == "1"
+        and os.environ.get("NERRF_REC_FUSED", "infer") == "1"
         and dt == torch.bfloat16 and hdim == 256
     )
     if rec_fused:
