@@ -420,3 +420,30 @@ def test_incremental_merge_radix_guard_wide_window():
     assert np.array_equal(ed["edge_index"], ed_ref["edge_index"])
     assert np.allclose(ed["edge_weight"], ed_ref["edge_weight"], atol=1e-5)
     assert np.allclose(ed["edge_ts"], ed_ref["edge_ts"], atol=1e-6)
+
+
+def test_bulk_append_small_batches_coalesce_deltas():
+    """Many small append_array calls must not fragment the delta ring: the
+    trailing delta keeps absorbing events until its delta_s span closes
+    (same boundary rule as scalar append)."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    st = DeltaGraphStore(window_s=1000.0, delta_s=5.0)
+    t = 0.0
+    for batch in range(60):  # 60 batches of 10 events over 30 s
+        b = EventArrayBuilder(st.paths, st.comms)
+        for _ in range(10):
+            b.add(ts=t, pid=1, syscall="write", path=f"/f{int(t) % 5}", nbytes=1)
+            t += 0.05
+        st.append_array(b.build())
+    # 30 s of events at delta_s=5 -> ~6 deltas, not 60
+    assert len(st._deltas) <= 8
+    arr = st.compact()
+    assert len(arr) == 600
+    assert bool(np.all(np.diff(arr.ts) >= 0))
+    # boundary rule: every delta spans < delta_s from its first event
+    for d in st._deltas:
+        assert float(d.ts[-1]) - float(d.ts[0]) < 5.0
