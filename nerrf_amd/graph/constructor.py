@@ -43,6 +43,69 @@ _DOUBLE_EXT = re.compile(r"\.\w{1,5}\.\w{1,9}$")
 _SOCKET_PREFIXES = ("tcp://", "udp://")
 DEST_ALLOWLIST = ("tcp://10.", "tcp://192.168.", "tcp://backup.", "udp://10.")
 
+# process-identity channel (x[:, 27], reserved in round 1): comms a deployment
+# considers "known-good daemons".  A process node is trusted only when EVERY
+# event it emitted in the window carries an allowlisted comm — a single
+# unknown or off-list comm (execve into a payload, masquerading child)
+# clears the flag for the whole window.  Deployment overrides this module
+# constant from config, like DEST_ALLOWLIST above.
+TRUSTED_COMMS = frozenset({
+    # system daemons
+    "systemd", "systemd-journal", "journald", "dbus-daemon", "sshd",
+    "cron", "crond", "rsyslogd", "agetty", "chronyd",
+    # container / orchestration
+    "containerd", "dockerd", "kubelet", "containerd-shim",
+    # common benign workload comms (matches data/synth.py's benign mix)
+    "nginx", "postgres", "node", "python3", "redis-server",
+})
+
+
+def proc_identity_enabled() -> bool:
+    """Gate for the x[:, 27] trusted-process channel.
+
+    Default OFF: the vendored checkpoint and the calibrated alarm were
+    trained/swept with the channel all-zero, and its input weight column is
+    untrained (zero input -> zero gradient), so enabling it without a
+    retrain injects random-init noise into benign process nodes.  Flip
+    NERRF_PROC_IDENTITY=1 together with a retrain (tools/train_mixed.py)
+    and a re-calibration (tools/calibrate_alarm.py).
+    """
+    import os
+
+    return os.environ.get("NERRF_PROC_IDENTITY", "0") == "1"
+
+
+def trusted_proc_flags(
+    events: EventArray,
+    ev_proc: np.ndarray,
+    n_files: int,
+    n_procs: int,
+    n_nodes: int,
+) -> np.ndarray:
+    """[n_nodes] float32: 1.0 on process nodes whose every window event has
+    an allowlisted comm (TRUSTED_COMMS); 0.0 on file nodes, on processes
+    with any off-list/unknown comm, and on processes with no comm'd events.
+    Host string-domain work, like the path flags."""
+    out = np.zeros(n_nodes, dtype=np.float32)
+    if n_procs == 0:
+        return out
+    n_comms = len(events.comms)
+    allowed = np.zeros(n_comms + 1, dtype=bool)
+    for c in TRUSTED_COMMS:
+        i = events.comms.get(c)
+        if i is not None:
+            allowed[i] = True
+    sel = ev_proc >= 0
+    if not sel.any():
+        return out
+    pidx = (ev_proc[sel] - n_files).astype(np.int64)
+    cid = events.comm_id[sel]
+    ok = (cid >= 0) & allowed[np.clip(cid, 0, n_comms)]
+    n_ok = np.bincount(pidx, weights=ok.astype(np.float64), minlength=n_procs)
+    n_all = np.bincount(pidx, minlength=n_procs)
+    out[n_files:] = ((n_ok == n_all) & (n_all > 0)).astype(np.float32)
+    return out
+
 
 def rename_path_roots(ra: np.ndarray, rb: np.ndarray, n_paths: int) -> np.ndarray:
     """path_root from rename pairs: connected components restricted to the
@@ -392,6 +455,7 @@ def build_edges_and_flags(parts: dict, causality_tau_s: float = 10.0) -> dict:
         "note": np.concatenate([note, pad]),
         "recon": np.concatenate([recon, pad]),
         "double_ext": np.concatenate([double_ext, pad]),
+        "trusted_proc": trusted_proc_flags(events, ev_proc, n_files, n_procs, n_nodes),
     }
 
 
@@ -496,7 +560,9 @@ def build_graph(
     x[:, 24] = np.log1p(cnt_total)
     x[:, 25] = np.log1p(bytes_write / np.maximum(cnt_write, 1.0)) / 16.0
     x[:, 26] = np.log1p(cnt_exec)
-    # 27..31 reserved
+    if proc_identity_enabled():
+        x[:, 27] = ed["trusted_proc"]
+    # 28..31 reserved
 
     # ---- labels -----------------------------------------------------------
     y_node = None

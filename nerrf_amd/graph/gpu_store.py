@@ -20,7 +20,8 @@ from typing import Optional
 import numpy as np
 import torch
 
-from .constructor import build_edges_and_flags, build_graph_parts, file_node_kinds
+from .constructor import (build_edges_and_flags, build_graph_parts,
+                          file_node_kinds, proc_identity_enabled)
 from ..data.trace import EventArray
 
 
@@ -63,6 +64,10 @@ def gpu_window_graph(
         | (ed["recon"].astype(np.uint8) << 2)
         | (ed["double_ext"].astype(np.uint8) << 3)
     )
+    if proc_identity_enabled() and "trusted_proc" in ed:
+        # bit 4 -> x[:, 27] (feature_assemble_kernel); gate is host-side so
+        # the kernel mapping stays unconditional
+        flags = flags | (ed["trusted_proc"].astype(np.uint8) << 4)
 
     def dev(a, dt=None):
         if isinstance(a, torch.Tensor):  # already staged on-device (merge_window)

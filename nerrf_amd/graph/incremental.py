@@ -31,7 +31,8 @@ import numpy as np
 
 from ..data.trace import SYSCALL_IDS, EventArray
 from .constructor import (_READ_LIKE, _path_flags, aggregate_sparse_keys,
-                          rename_path_roots)
+                          proc_identity_enabled, rename_path_roots,
+                          trusted_proc_flags)
 
 _SCALE_GUARD = 60.0  # max (t1 - t_d0)/tau before a partial sum underflows
 
@@ -296,6 +297,13 @@ def merge_window(
     suspicious, note, recon, double_ext = _path_flags(
         events.paths, path_root, root_to_file, n_files
     )
+    if proc_identity_enabled() and n_ev:
+        # host per-event proc map (the device branch keeps ev_proc on-GPU,
+        # but the comm allowlist is string-domain host work either way)
+        ev_proc_host = n_files + pid_to_local(events.pid)
+        trusted = trusted_proc_flags(events, ev_proc_host, n_files, n_procs, n_nodes)
+    else:
+        trusted = np.zeros(n_nodes, dtype=np.float32)
     pad = np.zeros(n_procs, dtype=np.float32)
     ed = {
         "edge_index": edge_index,
@@ -308,5 +316,6 @@ def merge_window(
         "note": np.concatenate([note, pad]),
         "recon": np.concatenate([recon, pad]),
         "double_ext": np.concatenate([double_ext, pad]),
+        "trusted_proc": trusted,
     }
     return parts, ed

@@ -55,7 +55,7 @@ __global__ void feature_assemble_kernel(
     const float* __restrict__ in_deg,    // [M] (edge-domain, from host)
     const float* __restrict__ out_deg,   // [M]
     const float* __restrict__ peer,      // [M]
-    const unsigned char* __restrict__ flags,  // [M] bit0 susp, 1 note, 2 recon, 3 dbl_ext
+    const unsigned char* __restrict__ flags,  // [M] bit0 susp, 1 note, 2 recon, 3 dbl_ext, 4 trusted_proc
     const signed char* __restrict__ node_kind,  // [M] 0 proc, 1 file
     float* __restrict__ x,               // [M, 32]
     float span_s, int m_nodes) {
@@ -97,7 +97,11 @@ __global__ void feature_assemble_kernel(
   o[24] = log1pf(cnt_total);
   o[25] = log1pf(bytes_write / fmaxf(cnt_write, 1.0f)) / 16.0f;
   o[26] = log1pf(cnt_exec);
-  o[27] = 0.0f; o[28] = 0.0f; o[29] = 0.0f; o[30] = 0.0f; o[31] = 0.0f;
+  // process-identity channel (constructor.trusted_proc_flags); the host
+  // leaves bit 4 unset unless NERRF_PROC_IDENTITY=1, so the default x
+  // matches the vendored checkpoint's training distribution
+  o[27] = (fl & 16) ? 1.0f : 0.0f;
+  o[28] = 0.0f; o[29] = 0.0f; o[30] = 0.0f; o[31] = 0.0f;
 }
 
 void launch_event_scatter(const long* ev_file, const long* ev_proc,

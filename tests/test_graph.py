@@ -447,3 +447,74 @@ def test_bulk_append_small_batches_coalesce_deltas():
     # boundary rule: every delta spans < delta_s from its first event
     for d in st._deltas:
         assert float(d.ts[-1]) - float(d.ts[0]) < 5.0
+
+
+# ---------------------------------------------------------------------------
+# process-identity channel (x[:, 27], NERRF_PROC_IDENTITY gate)
+# ---------------------------------------------------------------------------
+
+def _comm_events():
+    """pid 10: all-allowlisted comm; pid 11: execs into an off-list payload;
+    pid 12: no comm recorded."""
+    b = EventArrayBuilder()
+    b.add(ts=1.0, pid=10, syscall="write", path="/d/a.dat", nbytes=10, comm="nginx")
+    b.add(ts=2.0, pid=10, syscall="read", path="/d/a.dat", nbytes=5, comm="nginx")
+    b.add(ts=3.0, pid=11, syscall="write", path="/d/b.dat", nbytes=10, comm="python3")
+    b.add(ts=4.0, pid=11, syscall="write", path="/d/b.dat", nbytes=10, comm="evilbin")
+    b.add(ts=5.0, pid=12, syscall="read", path="/d/c.dat", nbytes=1)
+    return b.build()
+
+
+def test_trusted_proc_flags_semantics():
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+
+    arr = _comm_events()
+    parts = build_graph_parts(arr)
+    ed = build_edges_and_flags(parts)
+    n_files = parts["n_files"]
+    trusted = ed["trusted_proc"]
+    assert trusted.shape == (parts["n_nodes"],)
+    assert trusted[:n_files].sum() == 0  # file nodes never trusted
+    # upids sorted -> proc nodes are pids 10, 11, 12 in order
+    assert trusted[n_files + 0] == 1.0  # every event allowlisted
+    assert trusted[n_files + 1] == 0.0  # one off-list comm clears the flag
+    assert trusted[n_files + 2] == 0.0  # unknown comm is not trusted
+
+
+def test_proc_identity_channel_default_off():
+    from nerrf_amd.graph.constructor import build_graph
+
+    g = build_graph(_comm_events())
+    assert np.all(g.x[:, 27] == 0.0)
+
+
+def test_proc_identity_channel_enabled(monkeypatch):
+    from nerrf_amd.graph.constructor import build_graph, build_graph_parts
+
+    monkeypatch.setenv("NERRF_PROC_IDENTITY", "1")
+    arr = _comm_events()
+    g = build_graph(arr)
+    n_files = build_graph_parts(arr)["n_files"]
+    assert g.x[:, 27].sum() == 1.0
+    assert g.x[n_files + 0, 27] == 1.0
+
+
+def test_proc_identity_incremental_matches_full(monkeypatch):
+    """merge_window's trusted_proc == build_edges_and_flags' on the same
+    window, including comms arriving across delta boundaries."""
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    monkeypatch.setenv("NERRF_PROC_IDENTITY", "1")
+    st = DeltaGraphStore(window_s=100.0, delta_s=2.0)
+    st.append(ts=0.1, pid=7, syscall="write", path="/d/a", nbytes=8, comm="postgres")
+    st.append(ts=3.0, pid=7, syscall="write", path="/d/a", nbytes=8, comm="lockbit")
+    st.append(ts=3.1, pid=8, syscall="read", path="/d/a", nbytes=4, comm="sshd")
+    st.append(ts=6.0, pid=9, syscall="read", path="/d/a", nbytes=4)
+    ev, deltas = st.compact_with_deltas(None)
+    ref_ed = build_edges_and_flags(build_graph_parts(ev))
+    _, ed = merge_window(ev, IncrementalWindowState().summaries(deltas))
+    assert np.array_equal(ed["trusted_proc"], ref_ed["trusted_proc"])
+    # pid 7 crossed deltas with one off-list comm -> 0; pid 8 trusted
+    assert ed["trusted_proc"].sum() == 1.0

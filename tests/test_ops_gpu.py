@@ -391,6 +391,27 @@ def test_gpu_window_graph_matches_cpu(gpu_device):
     assert np.array_equal(g_gpu["node_kind"].cpu().numpy(), g_cpu.node_kind)
 
 
+def test_gpu_proc_identity_channel(gpu_device, monkeypatch):
+    """x[:, 27] (trusted-process channel, flags bit 4) CPU/GPU parity with
+    the gate enabled, and all-zero with it off (default)."""
+    import numpy as np
+
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.graph.constructor import build_graph
+    from nerrf_amd.graph.gpu_store import gpu_window_graph
+
+    arr, _ = generate(SynthConfig(seed=22, duration_s=30, benign_rate_hz=300, n_victim_files=8))
+    g_off = gpu_window_graph(arr, gpu_device)
+    assert float(g_off["x"][:, 27].abs().sum()) == 0.0
+
+    monkeypatch.setenv("NERRF_PROC_IDENTITY", "1")
+    g_cpu = build_graph(arr)
+    g_gpu = gpu_window_graph(arr, gpu_device)
+    x27 = g_gpu["x"][:, 27].cpu().numpy()
+    assert np.array_equal(x27, g_cpu.x[:, 27])
+    assert x27.sum() > 0  # benign daemon comms present in the synth mix
+
+
 def test_streaming_engine_gpu_scores(gpu_device):
     """Engine scoring on GPU (delta compaction + model + planner kernels)."""
     from nerrf_amd.data.synth import SynthConfig, generate
