@@ -175,6 +175,7 @@ class StreamingEngine:
                 events, sums,
                 device=self.device if self.device.type == "cuda" else None,
                 dev_cols=dev_cols,
+                state=self._inc_state,
             )
         else:
             parts = build_graph_parts(events)

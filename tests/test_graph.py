@@ -518,3 +518,48 @@ def test_proc_identity_incremental_matches_full(monkeypatch):
     assert np.array_equal(ed["trusted_proc"], ref_ed["trusted_proc"])
     # pid 7 crossed deltas with one off-list comm -> 0; pid 8 trusted
     assert ed["trusted_proc"].sum() == 1.0
+
+
+def test_edge_cache_parity_across_ticks():
+    """Stable-prefix edge cache: repeated merges with a growing open delta
+    (cache hits), new renames (sig invalidation), and a seal (token
+    invalidation) all reproduce the full rebuild exactly."""
+    from nerrf_amd.data.synth import SynthConfig, generate
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    arr, _ = generate(SynthConfig(duration_s=12.0, benign_rate_hz=800.0,
+                                  n_benign_files=300, seed=31))
+    st = DeltaGraphStore(window_s=1e9, delta_s=2.0)
+    state = IncrementalWindowState()
+    # replay in 9 tick-sized chunks; each tick merges and checks parity
+    bounds = np.linspace(0, len(arr), 10).astype(int)
+    hits = 0
+    for i in range(9):
+        st.append_array(arr.slice(bounds[i], bounds[i + 1]))
+        if i == 4:  # new rename mid-stream -> sig invalidation path
+            t = float(arr.ts[bounds[i + 1] - 1]) + 0.001
+            st.append(ts=t, pid=999, syscall="rename", path="/x/q.dat",
+                      new_path="/x/q.dat.enc")
+        ev, deltas = st.compact_with_deltas(None)
+        sums = state.summaries(deltas)
+        before = state._edge_cache
+        parts, ed = merge_window(ev, sums, state=state)
+        if before is not None and state._edge_cache is before:
+            hits += 1
+        ref_ed = build_edges_and_flags(build_graph_parts(ev))
+        assert np.array_equal(ed["edge_index"], ref_ed["edge_index"]), f"tick {i}"
+        assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-5), f"tick {i}"
+        assert np.allclose(ed["edge_ts"], ref_ed["edge_ts"], atol=1e-6), f"tick {i}"
+        for k in ("in_deg", "out_deg", "peer"):
+            assert np.array_equal(ed[k], ref_ed[k]), f"tick {i} {k}"
+    # same-delta-set re-merge must be a pure cache hit with identical output
+    ev, deltas = st.compact_with_deltas(None)
+    sums = state.summaries(deltas)
+    a = merge_window(ev, sums, state=state)[1]
+    cache_obj = state._edge_cache
+    b = merge_window(ev, sums, state=state)[1]
+    assert state._edge_cache is cache_obj  # hit, not rebuilt
+    assert np.array_equal(a["edge_index"], b["edge_index"])
+    assert np.array_equal(a["edge_weight"], b["edge_weight"])
