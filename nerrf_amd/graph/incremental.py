@@ -126,6 +126,9 @@ class IncrementalWindowState:
         self.tau = causality_tau_s
         self._cache: Dict[int, Tuple[EventArray, DeltaSummary]] = {}
         self._edge_cache: Optional[dict] = None
+        # (ren_a, ren_b, n_paths, path_root) — rename CC memo; path_root is
+        # shared read-only with callers via parts["path_root"]
+        self._root_cache: Optional[tuple] = None
 
     @staticmethod
     def _token(prefix: list) -> tuple:
@@ -195,7 +198,16 @@ def merge_window(
         rb = np.concatenate([s.ren_b for s in summaries])
     else:
         ra = rb = np.empty(0, np.int64)
-    path_root = rename_path_roots(ra, rb, n_paths)
+    if state is not None:
+        rc = state._root_cache
+        if (rc is not None and rc[2] == n_paths
+                and np.array_equal(rc[0], ra) and np.array_equal(rc[1], rb)):
+            path_root = rc[3]
+        else:
+            path_root = rename_path_roots(ra, rb, n_paths)
+            state._root_cache = (ra.copy(), rb.copy(), n_paths, path_root)
+    else:
+        path_root = rename_path_roots(ra, rb, n_paths)
 
     # ---- node tables from merged unique sets ------------------------------
     seen = np.zeros(n_paths, dtype=bool)
