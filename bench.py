@@ -20,6 +20,7 @@ from __future__ import annotations
 
 import argparse
 import json
+import os
 import sys
 import time
 
@@ -124,8 +125,12 @@ def main() -> None:
     model = NerrfJointModel(JointConfig(sage=SageConfig(), lstm=LSTMConfig())).to(
         device=device, dtype=dtype
     )
+    # fused AdamW (one multi-tensor HIP kernel) vs foreach (one launch per
+    # op across tensor groups); NERRF_FUSED_ADAM=0 falls back to foreach
+    fused_ok = has_gpu and os.environ.get("NERRF_FUSED_ADAM", "1") == "1"
     opt = torch.optim.AdamW(
-        model.parameters(), lr=1e-3, weight_decay=1e-4, foreach=True,
+        model.parameters(), lr=1e-3, weight_decay=1e-4,
+        fused=fused_ok, foreach=not fused_ok,
         capturable=use_graphs,
     )
     reducer = GradAllReducer(model)
