@@ -125,9 +125,10 @@ def main() -> None:
     model = NerrfJointModel(JointConfig(sage=SageConfig(), lstm=LSTMConfig())).to(
         device=device, dtype=dtype
     )
-    # fused AdamW (one multi-tensor HIP kernel) vs foreach (one launch per
-    # op across tensor groups); NERRF_FUSED_ADAM=0 falls back to foreach
-    fused_ok = has_gpu and os.environ.get("NERRF_FUSED_ADAM", "1") == "1"
+    # fused AdamW measured neutral-to-slightly-slower than foreach on
+    # MI355X at this step (243.7 vs 242.5 ms, gpurun_out/bench_fused.log) —
+    # foreach stays the default; NERRF_FUSED_ADAM=1 to re-test
+    fused_ok = has_gpu and os.environ.get("NERRF_FUSED_ADAM", "0") == "1"
     opt = torch.optim.AdamW(
         model.parameters(), lr=1e-3, weight_decay=1e-4,
         fused=fused_ok, foreach=not fused_ok,

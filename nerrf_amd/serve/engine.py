@@ -278,15 +278,25 @@ class StreamingEngine:
                         file_scores[path] = sc
         # bytes per file (window-local) — vectorised: a per-event Python loop
         # here costs ~0.4 s per 600k-event window, which dominates the host
-        # path once the model forward is on the GPU
-        valid = events.path_id >= 0
-        mb_by_id = np.bincount(
-            events.path_id[valid],
-            weights=events.nbytes[valid].astype(np.float64),
-            minlength=len(events.paths.strings),
-        ) / 1e6
-        for i in np.nonzero(mb_by_id)[0]:
-            file_mb[events.paths.lookup(int(i))] = float(mb_by_id[i])
+        # path once the model forward is on the GPU.  When the incremental
+        # merge ran, the per-delta summaries already hold sparse per-path
+        # byte sums (summarize_delta), so the O(window events) bincount
+        # collapses to O(touched paths).
+        if use_incremental and sums:
+            ids = _np.concatenate([s.mb_ids for s in sums])
+            vals = _np.concatenate([s.mb_vals for s in sums])
+            mb_by_id = _np.bincount(
+                ids, weights=vals, minlength=len(strings)) / 1e6
+        else:
+            valid = events.path_id >= 0
+            mb_by_id = np.bincount(
+                events.path_id[valid],
+                weights=events.nbytes[valid].astype(np.float64),
+                minlength=len(strings),
+            ) / 1e6
+        nz = _np.nonzero(mb_by_id)[0]
+        file_mb = dict(zip((strings[i] for i in nz.tolist()),
+                           mb_by_id[nz].tolist()))
 
         if timing is not None:
             timing["score_maps"] = time.perf_counter() - _t0 - sum(timing.values())

@@ -15,10 +15,12 @@ import torch
 _ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, _ROOT)
 
-from nerrf_amd.checkpoint import load_model_from_checkpoint  # noqa: E402
 from nerrf_amd.data.synth import SynthConfig, generate  # noqa: E402
-from nerrf_amd.serve.engine import StreamingEngine  # noqa: E402
+from nerrf_amd.perf import enable_tuned_gemms  # noqa: E402
+from nerrf_amd.serve.engine import (StreamingEngine,  # noqa: E402
+                                    load_model_from_checkpoint)
 
+enable_tuned_gemms()
 dev = "cuda" if torch.cuda.is_available() else "cpu"
 model = load_model_from_checkpoint(os.path.join(_ROOT, "checkpoints", "pretrained"))
 engine = StreamingEngine(model=model, device=dev,
