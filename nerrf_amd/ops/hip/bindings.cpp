@@ -37,6 +37,8 @@ void launch_lstm_rec_fwd(const void*, const void*, const void*, const void*,
 void launch_lstm_rec_bwd(const void*, const void*, const void*, const void*,
                          const void*, const void*, const float*, void*, void*,
                          void*, int, long, long, hipStream_t);
+void launch_rec_gemm_fwd(const void*, const void*, void*, long, long, long,
+                         hipStream_t);
 void launch_proj_fwd_dual(const void*, const void*, const void*, void*, void*,
                           long, long, hipStream_t);
 void launch_proj_dgrad_dual(const void*, const void*, const void*, const void*,
@@ -340,6 +342,23 @@ void lstm_step_fused(torch::Tensor h_prev, torch::Tensor w_hh, torch::Tensor xg,
       gates_act.data_ptr(), batch, raw, stream.stream());
 }
 
+// Recurrent-step GEMM: c = a @ w^T with [M,256] x [1024,256] (bf16; A and C
+// may be row-strided column slabs of wider buffers).
+void rec_gemm_fwd(torch::Tensor a, torch::Tensor w, torch::Tensor c) {
+  const long a_stride = row_stride_checked(a, "a");
+  const long c_stride = row_stride_checked(c, "c");
+  check_gpu_contig(w, "w");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16, "rec_gemm_fwd is bf16-only");
+  TORCH_CHECK(a.size(1) == 256 && w.size(0) == 1024 && w.size(1) == 256 &&
+                  c.size(1) == 1024 && c.size(0) == a.size(0),
+              "rec_gemm_fwd requires [M,256] x [1024,256] -> [M,1024]");
+  TORCH_CHECK(a_stride % 8 == 0 && c_stride % 8 == 0,
+              "rec_gemm_fwd needs 16-B aligned rows");
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_rec_gemm_fwd(a.data_ptr(), w.data_ptr(), c.data_ptr(),
+                             a.size(0), a_stride, c_stride, stream.stream());
+}
+
 // Dual-direction LSTM input projection: c1 = a @ w1^T, c2 = a @ w2^T in one
 // launch (A staged in LDS once; K=512, N=1024 per direction, bf16).
 void proj_fwd_dual(torch::Tensor a, torch::Tensor w1, torch::Tensor w2,
@@ -612,6 +631,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused recurrent GEMM + LSTM pointwise fwd (bf16, H=256)");
   m.def("lstm_rec_bwd", &lstm_rec_bwd,
         "fused LSTM gate grads + grad_h GEMM bwd (bf16, H=256)");
+  m.def("rec_gemm_fwd", &rec_gemm_fwd,
+        "recurrent-step GEMM [M,256]x[1024,256]^T (bf16, strided rows)");
   m.def("proj_fwd_dual", &proj_fwd_dual,
         "dual-direction LSTM input projection (A read once)");
   m.def("proj_dgrad_dual", &proj_dgrad_dual,

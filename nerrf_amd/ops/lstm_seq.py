@@ -100,8 +100,22 @@ def _dir_forward(ext, xg, h0, c0, w_hh, bias, mask, reverse, infer, h_all):
         else:
             w_hh_t = w_hh.t().contiguous()
             hg = torch.empty(batch, gdim, device=dev, dtype=dt)
+            # dedicated NT MFMA kernel for the [B,256]x[1024,256]^T step
+            # (rec_gemm.hip): hipBLASLt's tuned tiles run this shape at
+            # ~2.2 TB/s effective (profiles/train_kstats_r02.txt) —
+            # K=256 is too short to hide latency and the C epilogue
+            # dominates.  NERRF_REC_GEMM=0 falls back to hipBLASLt.
+            use_rg = (
+                hasattr(ext, "rec_gemm_fwd")
+                and os.environ.get("NERRF_REC_GEMM", "1") == "1"
+                and dt == torch.bfloat16 and hdim == 256 and gdim == 1024
+                and w_hh.is_contiguous()
+            )
             for ti in steps:
-                torch.mm(h, w_hh_t, out=hg)
+                if use_rg and h.stride(1) == 1 and h.stride(0) % 8 == 0:
+                    ext.rec_gemm_fwd(h, w_hh, hg)
+                else:
+                    torch.mm(h, w_hh_t, out=hg)
                 ext.lstm_pointwise_fwd(
                     hg, xg[ti], bias, c, h,
                     mask[ti] if mask is not None else empty_mask,

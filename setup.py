@@ -21,6 +21,7 @@ sources = [
     os.path.join(HIP_DIR, "lstm_step_fused.hip"),
     os.path.join(HIP_DIR, "lstm_rec_fused.hip"),
     os.path.join(HIP_DIR, "stream_gemm.hip"),
+    os.path.join(HIP_DIR, "rec_gemm.hip"),
     os.path.join(HIP_DIR, "mcts.hip"),
     os.path.join(HIP_DIR, "event_scatter.hip"),
     os.path.join(HIP_DIR, "sage_fused.hip"),

@@ -713,6 +713,56 @@ def test_lstm_bilayer_rec_fused_strided_slabs(gpu_device, monkeypatch):
         )
 
 
+def test_rec_gemm_fwd_matches_matmul(gpu_device):
+    """rec_gemm.hip NT step GEMM vs fp32 matmul, contiguous + strided rows
+    + tail guard (M not a multiple of 128)."""
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(41)
+    for m in (1000, 4096):
+        a = (torch.randn(m, 256, device=gpu_device) * 0.3).to(torch.bfloat16)
+        w = (torch.randn(1024, 256, device=gpu_device) * 0.05).to(torch.bfloat16)
+        c = torch.empty(m, 1024, device=gpu_device, dtype=torch.bfloat16)
+        ext.rec_gemm_fwd(a, w, c)
+        ref = a.float() @ w.float().t()
+        assert torch.allclose(c.float(), ref, atol=3e-2, rtol=3e-2), f"M={m}"
+        # column-slab A (the recurrence reads h out of the [T,B,2H] buffer)
+        slab = (torch.randn(m, 512, device=gpu_device) * 0.3).to(torch.bfloat16)
+        a_s = slab[:, 256:]
+        ext.rec_gemm_fwd(a_s, w, c)
+        refs = a_s.float() @ w.float().t()
+        assert torch.allclose(c.float(), refs, atol=3e-2, rtol=3e-2), f"M={m} strided"
+
+
+def test_rec_gemm_in_recurrence_matches_reference(gpu_device):
+    """lstm_bilayer2 with NERRF_REC_GEMM=1 == reference recurrence (the
+    step GEMM swap must not move training numerics)."""
+    import os
+
+    from nerrf_amd.models.lstm import BiLSTMDetector, LSTMConfig
+
+    torch.manual_seed(42)
+    t_len, batch = 12, 96
+    model = BiLSTMDetector(LSTMConfig(in_dim=32, hidden=256, layers=2)).to(
+        gpu_device, torch.bfloat16
+    )
+    x = (torch.randn(batch, t_len, 32, device=gpu_device) * 0.3).to(torch.bfloat16)
+    lengths = torch.full((batch,), t_len, device=gpu_device, dtype=torch.int64)
+    old = os.environ.get("NERRF_REC_GEMM")
+    try:
+        os.environ["NERRF_REC_GEMM"] = "1"
+        y1 = model(x, lengths)
+        os.environ["NERRF_REC_GEMM"] = "0"
+        y0 = model(x, lengths)
+    finally:
+        if old is None:
+            os.environ.pop("NERRF_REC_GEMM", None)
+        else:
+            os.environ["NERRF_REC_GEMM"] = old
+    assert torch.allclose(y1.float(), y0.float(), atol=3e-2, rtol=3e-2)
+
+
 def test_proj_fwd_dual_matches_matmul(gpu_device):
     from nerrf_amd.ops.native import load_extension
 
