@@ -47,15 +47,7 @@ __device__ __forceinline__ unsigned rg_swz(unsigned row, unsigned byte_col,
 // ---------------------------------------------------------------------------
 // rec_gemm_fwd: C[M, 1024] = A[M, 256] @ W[1024, 256]^T
 // 512 threads = 8 waves; wave w owns a 32-col slice of each 256-col chunk.
-//
-// v2: A fragments load DIRECTLY from global — the NT layout means every
-// fragment is one k-contiguous 16-B load, and the 64-KB A strip stays
-// L2-resident across the four n-chunks, so LDS staging (v1) only added a
-// block barrier and ds-traffic.  The C tile is the only LDS user: each
-// wave dumps its whole [128, 32] bf16 slice (8 KB), one barrier, then
-// re-reads row-major for 16-B global stores (v1 staged per 16-row tile =
-// 16 barriers per chunk; measured 1.17x blas at M=64k, this removes its
-// dominant stall).
+// LDS: 64 KB A strip + 8 KB/wave C staging (reused per rf tile).
 // ---------------------------------------------------------------------------
 __launch_bounds__(512)
 __global__ void rec_gemm_fwd_kernel(
@@ -64,29 +56,36 @@ __global__ void rec_gemm_fwd_kernel(
     __hip_bfloat16* __restrict__ c,        // [M(row-stride c_stride), 1024]
     long m_rows, long a_stride, long c_stride) {
   constexpr int K = 256;
-  constexpr int ROW_B = K * 2;  // 512 B per A/W row
+  constexpr int ROW_B = K * 2;  // 512 B per A row
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* a_lds = smem;                       // 64 KB swizzled A strip
+  char* st_lds = smem + RG_BM * ROW_B;      // 8 waves x 1 KB C staging
 
   const long row0 = (long)blockIdx.x * RG_BM;
   const int tid = threadIdx.x;
   const int lane = tid % NERRF_WAVE;
   const int wave = tid / NERRF_WAVE;
 
+  // ---- stage the A strip once: 128 rows x 512 B --------------------------
+  {
+    const int r = tid >> 2;      // 0..127
+    const int c0 = tid & 3;      // interleaved 16-B chunks within the row
+    const long grow = row0 + r;
+#pragma unroll
+    for (int cc = 0; cc < 8; ++cc) {
+      const int chunk = c0 + cc * 4;
+      uint4 v = make_uint4(0, 0, 0, 0);
+      if (grow < m_rows)
+        v = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const char*>(a + grow * a_stride) + chunk * 16);
+      *reinterpret_cast<uint4*>(a_lds + rg_swz(r, chunk * 16, ROW_B)) = v;
+    }
+  }
+  __syncthreads();
+
   const int frag_col = lane & 15;
   const int kchunk_b = (lane >> 4) * 16;  // byte offset within 64-B k-step
-  char* my_st = smem + wave * 8192;       // [128 rows][32 cols] bf16
-
-  // A addressing: one vector base + uniform per-fragment offsets (the
-  // fragment rows differ by rf*16 rows, a scalar quantity — keeping the
-  // offsets uniform saves a VGPR-resident pointer array; tail blocks
-  // clamp each row individually, garbage rows are never stored)
-  const char* arow[8];
-#pragma unroll
-  for (int rf = 0; rf < 8; ++rf) {
-    long r = row0 + rf * 16 + frag_col;
-    if (r >= m_rows) r = m_rows - 1;  // tail: garbage rows never stored
-    arow[rf] = reinterpret_cast<const char*>(a + r * a_stride) + kchunk_b;
-  }
+  char* my_st = st_lds + wave * 1024;     // [16 rows][32 cols] bf16
 
 #pragma unroll 1
   for (int nc = 0; nc < 4; ++nc) {  // 256-col chunks of N = 1024
@@ -99,58 +98,65 @@ __global__ void rec_gemm_fwd_kernel(
     const char* wrow0 = reinterpret_cast<const char*>(w) +
                         (long)(col0 + frag_col) * ROW_B + kchunk_b;
     const char* wrow1 = wrow0 + 16 * ROW_B;
+    rbf16x8 b_cur[2], b_nxt[2];
+    b_cur[0] = *reinterpret_cast<const rbf16x8*>(wrow0);
+    b_cur[1] = *reinterpret_cast<const rbf16x8*>(wrow1);
 #pragma unroll 2
     for (int ks = 0; ks < K / 32; ++ks) {
-      rbf16x8 bfr[2];
-      bfr[0] = *reinterpret_cast<const rbf16x8*>(wrow0 + ks * 64);
-      bfr[1] = *reinterpret_cast<const rbf16x8*>(wrow1 + ks * 64);
+      if (ks + 1 < K / 32) {
+        b_nxt[0] = *reinterpret_cast<const rbf16x8*>(wrow0 + (ks + 1) * 64);
+        b_nxt[1] = *reinterpret_cast<const rbf16x8*>(wrow1 + (ks + 1) * 64);
+      }
       rbf16x8 afr[8];
 #pragma unroll
       for (int rf = 0; rf < 8; ++rf)
-        afr[rf] = *reinterpret_cast<const rbf16x8*>(arow[rf] + ks * 64);
-      // let all 10 VMEM issue, then the MFMA block (partial vmcnt waits)
-      __builtin_amdgcn_sched_group_barrier(0x20, 10, 0);  // VMEM read x10
+        afr[rf] = *reinterpret_cast<const rbf16x8*>(
+            a_lds + rg_swz(rf * 16 + frag_col, ks * 64 + kchunk_b, ROW_B));
+      __builtin_amdgcn_sched_group_barrier(0x100, 8, 0);  // DS_READ x8
       __builtin_amdgcn_sched_group_barrier(0x8, 16, 0);   // MFMA x16
 #pragma unroll
       for (int rf = 0; rf < 8; ++rf)
 #pragma unroll
         for (int cf = 0; cf < 2; ++cf)
           acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[rf], bfr[cf], acc[rf][cf], 0, 0, 0);
+              afr[rf], b_cur[cf], acc[rf][cf], 0, 0, 0);
+      b_cur[0] = b_nxt[0];
+      b_cur[1] = b_nxt[1];
     }
-    // ---- store: acc layout -> per-wave LDS slice -> 16-B row stores ------
+    // ---- store: acc layout -> per-wave LDS -> 16-B row-vector stores -----
     // acc element r of (rf, cf) is C row rf*16 + (lane>>4)*4 + r, col
-    // col0 + cf*16 + (lane&15); the full [128, 32] bf16 slice stages in
-    // 8 KB of this wave's LDS, then lane l stores rows l>>2 (+16 per rr)
-    // as 16-B chunks — 64-B segments per 4 lanes vs 2-B scalar stores.
-    if (nc) __syncthreads();  // previous chunk's reads done before overwrite
+    // col0 + cf*16 + (lane&15); staging one rf tile ([16, 32] bf16 = 1 KB)
+    // at a time lets lane l re-read row l>>2, 16-B chunk l&3 and write a
+    // full 64-B segment per 4 lanes (vs 2-B scalar stores straight from
+    // the acc layout).  The buffers are per-wave but the hazard fences are
+    // __syncthreads() (uniform control flow), the proven idiom from
+    // lstm_rec_fused.hip's gpre staging.
 #pragma unroll
-    for (int rf = 0; rf < 8; ++rf)
+    for (int rf = 0; rf < 8; ++rf) {
 #pragma unroll
-      for (int cf = 0; cf < 2; ++cf)
+      for (int cf = 0; cf < 2; ++cf) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int srow = rf * 16 + (lane >> 4) * 4 + r;
-          const int cb = (cf * 16 + frag_col) * 2;
-          // 16-B-granular XOR swizzle: the four srows alive in one write
-          // instruction differ by 4, so (srow>>2)&3 spreads their bank
-          // groups (without it they all land on banks 0-7)
-          const int cbs = cb ^ (((srow >> 2) & 3) << 4);
-          *reinterpret_cast<__hip_bfloat16*>(my_st + srow * 64 + cbs) =
+          const int srow = (lane >> 4) * 4 + r;
+          *reinterpret_cast<__hip_bfloat16*>(
+              my_st + srow * 64 + (cf * 16 + frag_col) * 2) =
               __float2bfloat16(acc[rf][cf][r]);
         }
-    __syncthreads();
-#pragma unroll
-    for (int rr = 0; rr < 8; ++rr) {
-      const int srow = rr * 16 + (lane >> 2);
-      const long grow = row0 + srow;
+      }
+      // per-wave private buffer: the write->read hazard is intra-wave, so
+      // a scheduling fence suffices (hipcc inserts the lgkmcnt wait from
+      // the may-alias dependence); no block barrier needed
+      __builtin_amdgcn_sched_barrier(0);
+      const int srow = lane >> 2;
+      const long grow = row0 + rf * 16 + srow;
       if (grow < m_rows) {
-        const int cbs = ((lane & 3) * 16) ^ (((srow >> 2) & 3) << 4);
-        uint4 v = *reinterpret_cast<const uint4*>(my_st + srow * 64 + cbs);
+        uint4 v = *reinterpret_cast<const uint4*>(
+            my_st + srow * 64 + (lane & 3) * 16);
         *reinterpret_cast<uint4*>(
             reinterpret_cast<char*>(c + grow * c_stride) +
             (col0 + (lane & 3) * 8) * 2) = v;
       }
+      __builtin_amdgcn_sched_barrier(0);  // staging reused by next rf tile
     }
   }
 }
@@ -158,7 +164,7 @@ __global__ void rec_gemm_fwd_kernel(
 void launch_rec_gemm_fwd(const void* a, const void* w, void* c, long m_rows,
                          long a_stride, long c_stride, hipStream_t s) {
   const int grid = (int)((m_rows + RG_BM - 1) / RG_BM);
-  const size_t lds = 8 * 8192;  // 8 waves x 8 KB C staging
+  const size_t lds = RG_BM * 512 + 8 * 1024;  // 64 KB A + 8 KB staging
   rec_gemm_fwd_kernel<<<grid, 512, lds, s>>>(
       (const __hip_bfloat16*)a, (const __hip_bfloat16*)w, (__hip_bfloat16*)c,
       m_rows, a_stride, c_stride);
