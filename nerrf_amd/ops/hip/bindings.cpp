@@ -39,6 +39,8 @@ void launch_lstm_rec_bwd(const void*, const void*, const void*, const void*,
                          void*, int, long, long, hipStream_t);
 void launch_rec_gemm_fwd(const void*, const void*, void*, long, long, long,
                          hipStream_t);
+void launch_rec_gemm_dgrad(const void*, const void*, const void*, void*,
+                           long, long, hipStream_t);
 void launch_proj_fwd_dual(const void*, const void*, const void*, void*, void*,
                           long, long, hipStream_t);
 void launch_proj_dgrad_dual(const void*, const void*, const void*, const void*,
@@ -359,6 +361,29 @@ void rec_gemm_fwd(torch::Tensor a, torch::Tensor w, torch::Tensor c) {
                              a.size(0), a_stride, c_stride, stream.stream());
 }
 
+// Recurrence backward dgrad: c = a @ wt^T + d with [M,1024] x [256,1024]
+// (wt = W_hh^T contiguous; d optional addend, pass an empty tensor to skip).
+void rec_gemm_dgrad(torch::Tensor a, torch::Tensor wt, torch::Tensor d,
+                    torch::Tensor c) {
+  const long a_stride = row_stride_checked(a, "a");
+  check_gpu_contig(wt, "wt");
+  check_gpu_contig(c, "c");
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16, "rec_gemm_dgrad is bf16-only");
+  TORCH_CHECK(a.size(1) == 1024 && wt.size(0) == 256 && wt.size(1) == 1024 &&
+                  c.size(1) == 256 && c.size(0) == a.size(0),
+              "rec_gemm_dgrad requires [M,1024] x [256,1024] -> [M,256]");
+  TORCH_CHECK(a_stride % 8 == 0, "rec_gemm_dgrad needs 16-B aligned rows");
+  const bool has_d = d.numel() > 0;
+  if (has_d) {
+    check_gpu_contig(d, "d");
+    TORCH_CHECK(d.sizes() == c.sizes(), "addend shape must match output");
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  nerrf::launch_rec_gemm_dgrad(a.data_ptr(), wt.data_ptr(),
+                               has_d ? d.data_ptr() : nullptr, c.data_ptr(),
+                               a.size(0), a_stride, stream.stream());
+}
+
 // Dual-direction LSTM input projection: c1 = a @ w1^T, c2 = a @ w2^T in one
 // launch (A staged in LDS once; K=512, N=1024 per direction, bf16).
 void proj_fwd_dual(torch::Tensor a, torch::Tensor w1, torch::Tensor w2,
@@ -633,6 +658,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused LSTM gate grads + grad_h GEMM bwd (bf16, H=256)");
   m.def("rec_gemm_fwd", &rec_gemm_fwd,
         "recurrent-step GEMM [M,256]x[1024,256]^T (bf16, strided rows)");
+  m.def("rec_gemm_dgrad", &rec_gemm_dgrad,
+        "recurrence backward dgrad [M,1024]x[256,1024]^T + addend (bf16)");
   m.def("proj_fwd_dual", &proj_fwd_dual,
         "dual-direction LSTM input projection (A read once)");
   m.def("proj_dgrad_dual", &proj_dgrad_dual,

@@ -170,4 +170,137 @@ void launch_rec_gemm_fwd(const void* a, const void* w, void* c, long m_rows,
       m_rows, a_stride, c_stride);
 }
 
+
+// ---------------------------------------------------------------------------
+// rec_gemm_dgrad: C[M, 256] = A[M, 1024] @ Wt[256, 1024]^T + D[M, 256]
+// (the recurrence backward's grad_h = grad_h_pass + gg @ W_hh; Wt = W_hh^T
+// contiguous, so this is the same NT form as the forward with K/N swapped.)
+//
+// K=1024 means the A strip cannot sit in LDS whole (256 KB); it streams
+// through a double-buffered [128, 128] chunk (2 x 32 KB).  N=256 is covered
+// by the 8 waves in ONE pass (wave w owns cols w*32..w*32+31), so each A
+// element is read from HBM exactly once and every fragment is k-contiguous.
+// The D addend rides the vectorized epilogue: the C tile round-trips LDS
+// as bf16 rows and D is read with the same 16-B pattern.
+// ---------------------------------------------------------------------------
+__launch_bounds__(512)
+__global__ void rec_gemm_dgrad_kernel(
+    const __hip_bfloat16* __restrict__ a,   // [M(row-stride a_stride), 1024]
+    const __hip_bfloat16* __restrict__ wt,  // [256, 1024] (= W_hh^T)
+    const __hip_bfloat16* __restrict__ d,   // [M, 256] addend or nullptr
+    __hip_bfloat16* __restrict__ c,         // [M, 256]
+    long m_rows, long a_stride) {
+  constexpr int K = 1024;
+  constexpr int WROW_B = K * 2;   // 2048 B per Wt row
+  constexpr int CK = 128;         // k-chunk staged per round
+  constexpr int CH_B = CK * 2;    // 256 B per LDS row
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto ch_lds = [&](int buf) -> char* {            // 2 x 32 KB chunk bufs
+    return smem + buf * (RG_BM * CH_B);
+  };
+  char* st_lds = smem + 2 * RG_BM * CH_B;          // 8 waves x 1 KB
+
+  const long row0 = (long)blockIdx.x * RG_BM;
+  const int tid = threadIdx.x;
+  const int lane = tid % NERRF_WAVE;
+  const int wave = tid / NERRF_WAVE;
+  const int frag_col = lane & 15;
+  const int kchunk_b = (lane >> 4) * 16;
+  char* my_st = st_lds + wave * 1024;
+
+  // chunk-staging map: 4 threads per row, each covering 4 interleaved
+  // 16-B chunks of the row's 256-B chunk slice
+  const int s_r = tid >> 2;
+  const int s_c0 = tid & 3;
+  const long s_row = (row0 + s_r < m_rows) ? row0 + s_r : m_rows - 1;
+  const char* s_base = reinterpret_cast<const char*>(a + s_row * a_stride);
+
+  auto stage = [&](int buf, int kc) {
+#pragma unroll
+    for (int cc = 0; cc < 4; ++cc) {
+      const int chunk = s_c0 + cc * 4;
+      uint4 v = *reinterpret_cast<const uint4*>(
+          s_base + kc * CH_B + chunk * 16);
+      *reinterpret_cast<uint4*>(ch_lds(buf) + rg_swz(s_r, chunk * 16, CH_B)) = v;
+    }
+  };
+
+  stage(0, 0);
+
+  rf32x4 acc[8][2];
+#pragma unroll
+  for (int rf = 0; rf < 8; ++rf)
+#pragma unroll
+    for (int cf = 0; cf < 2; ++cf) acc[rf][cf] = rf32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int col0 = wave * 32;
+  const char* wrow0 = reinterpret_cast<const char*>(wt) +
+                      (long)(col0 + frag_col) * WROW_B + kchunk_b;
+  const char* wrow1 = wrow0 + 16 * WROW_B;
+
+  for (int kc = 0; kc < K / CK; ++kc) {
+    __syncthreads();  // chunk kc staged (and prior reads of this buf done)
+    if (kc + 1 < K / CK) stage((kc + 1) & 1, kc + 1);
+    const char* cur = ch_lds(kc & 1);
+#pragma unroll
+    for (int ks = 0; ks < CK / 32; ++ks) {
+      rbf16x8 bfr[2];
+      bfr[0] = *reinterpret_cast<const rbf16x8*>(wrow0 + kc * CH_B + ks * 64);
+      bfr[1] = *reinterpret_cast<const rbf16x8*>(wrow1 + kc * CH_B + ks * 64);
+      rbf16x8 afr[8];
+#pragma unroll
+      for (int rf = 0; rf < 8; ++rf)
+        afr[rf] = *reinterpret_cast<const rbf16x8*>(
+            cur + rg_swz(rf * 16 + frag_col, ks * 64 + kchunk_b, CH_B));
+#pragma unroll
+      for (int rf = 0; rf < 8; ++rf)
+#pragma unroll
+        for (int cf = 0; cf < 2; ++cf)
+          acc[rf][cf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[rf], bfr[cf], acc[rf][cf], 0, 0, 0);
+    }
+    __syncthreads();  // this buf's reads done before it restages at kc+2
+  }
+
+  // ---- epilogue: stage per-16-row tiles, add D, 16-B stores --------------
+#pragma unroll 1
+  for (int rf = 0; rf < 8; ++rf) {
+#pragma unroll
+    for (int cf = 0; cf < 2; ++cf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int srow = (lane >> 4) * 4 + r;
+        *reinterpret_cast<__hip_bfloat16*>(
+            my_st + srow * 64 + (cf * 16 + frag_col) * 2) =
+            __float2bfloat16(acc[rf][cf][r]);
+      }
+    __builtin_amdgcn_sched_barrier(0);
+    const int srow = lane >> 2;
+    const long grow = row0 + rf * 16 + srow;
+    if (grow < m_rows) {
+      rbf16x8 v = *reinterpret_cast<const rbf16x8*>(
+          my_st + srow * 64 + (lane & 3) * 16);
+      const long cb = grow * 256 + col0 + (lane & 3) * 8;
+      if (d != nullptr) {
+        const rbf16x8 dv = *reinterpret_cast<const rbf16x8*>(d + cb);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          v[i] = (__bf16)((float)v[i] + (float)dv[i]);
+      }
+      *reinterpret_cast<rbf16x8*>(c + cb) = v;
+    }
+    __builtin_amdgcn_sched_barrier(0);
+  }
+}
+
+void launch_rec_gemm_dgrad(const void* a, const void* wt, const void* d,
+                           void* c, long m_rows, long a_stride,
+                           hipStream_t s) {
+  const int grid = (int)((m_rows + RG_BM - 1) / RG_BM);
+  const size_t lds = 2 * RG_BM * 256 + 8 * 1024;  // 64 KB chunks + 8 KB st
+  rec_gemm_dgrad_kernel<<<grid, 512, lds, s>>>(
+      (const __hip_bfloat16*)a, (const __hip_bfloat16*)wt,
+      (const __hip_bfloat16*)d, (__hip_bfloat16*)c, m_rows, a_stride);
+}
+
 }  // namespace nerrf

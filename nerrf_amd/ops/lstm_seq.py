@@ -214,6 +214,18 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
         else None
     )
     bias_fused = bias_accum is not None
+    # dedicated dgrad kernel for grad_h = grad_h_pass + gg @ W_hh
+    # (rec_gemm.hip, same NT form as the forward with K/N swapped); W^T
+    # materialises once per direction vs 100 per-step addmm allocations
+    use_rg_d = (
+        ext is not None
+        and hasattr(ext, "rec_gemm_dgrad")
+        and os.environ.get("NERRF_REC_GEMM", "1") == "1"
+        and dt == torch.bfloat16 and hdim == 256 and gdim == 1024
+    )
+    if use_rg_d:
+        w_hh_t_d = w_hh.t().contiguous()
+        gh_buf = torch.empty(batch, hdim, device=dev, dtype=dt)
     for ti in steps:
         # c/h input of step ti = previous step's output (or h0/c0 at start)
         first = (ti == t_len - 1) if reverse else (ti == 0)
@@ -233,7 +245,14 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
                 bias_accum if bias_accum is not None else empty_mask,
             )
             bias_fused = bias_fused and bool(used)
-            grad_h = torch.addmm(grad_h_pass, grad_gates_all[ti], w_hh)
+            gg_t = grad_gates_all[ti]
+            if use_rg_d and gg_t.stride(1) == 1 and gg_t.stride(0) % 8 == 0:
+                # stream order makes one buffer safe: this iteration's
+                # pointwise (the only reader of grad_h) completed first
+                ext.rec_gemm_dgrad(gg_t, w_hh_t_d, grad_h_pass, gh_buf)
+                grad_h = gh_buf
+            else:
+                grad_h = torch.addmm(grad_h_pass, gg_t, w_hh)
         else:
             gg, gcp, ghp = _ref.lstm_pointwise_bwd_ref(
                 grad_h + grad_out[ti], grad_c, gates_all[ti], c_in,

@@ -60,3 +60,44 @@ for m in (16384, 65536):
     print(f"M={m}: blas {t_blas:.1f} us ({gb / (t_blas * 1e-6):.2f} TB/s eff)"
           f" | rec_gemm {t_k:.1f} us ({gb / (t_k * 1e-6):.2f} TB/s eff)"
           f" | speedup {t_blas / t_k:.2f}x")
+
+# ---- dgrad: grad_h = gg @ W_hh (+ grad_h_pass) -----------------------------
+for m in (16384, 65536):
+    gg = (torch.randn(m, 1024, device=dev) * 0.1).to(torch.bfloat16)
+    w = (torch.randn(1024, 256, device=dev) * 0.1).to(torch.bfloat16)
+    wt = w.t().contiguous()
+    dpass = (torch.randn(m, 256, device=dev) * 0.1).to(torch.bfloat16)
+    out = torch.empty(m, 256, device=dev, dtype=torch.bfloat16)
+    empty = torch.empty(0, device=dev, dtype=torch.bfloat16)
+
+    ext.rec_gemm_dgrad(gg, wt, dpass, out)
+    ref32 = torch.addmm(dpass.float(), gg.float(), w.float())
+    err = (out.float() - ref32).abs().max().item()
+    rel = err / ref32.abs().max().item()
+    print(f"dgrad M={m}: max abs err {err:.4e} (rel {rel:.2e})")
+    assert rel < 2e-2, "dgrad numerics FAIL"
+    # no-addend + strided-A (gg slab, row stride 2048)
+    slab = (torch.randn(m, 2048, device=dev) * 0.1).to(torch.bfloat16)
+    gg_s = slab[:, :1024]
+    ext.rec_gemm_dgrad(gg_s, wt, empty, out)
+    refs = torch.matmul(gg_s.float(), w.float())
+    errs = (out.float() - refs).abs().max().item()
+    print(f"dgrad M={m} strided/no-d: max abs err {errs:.4e}")
+    assert errs / refs.abs().max().item() < 2e-2, "dgrad strided FAIL"
+
+    def bench2(fn, n=200):
+        for _ in range(20):
+            fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(n):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / n * 1e6
+
+    t_blas = bench2(lambda: torch.addmm(dpass, gg, w, out=out))
+    t_k = bench2(lambda: ext.rec_gemm_dgrad(gg, wt, dpass, out))
+    gb = (m * 1024 * 2 + 2 * m * 256 * 2 + 1024 * 256 * 2) / 1e9
+    print(f"dgrad M={m}: blas(addmm) {t_blas:.1f} us ({gb / (t_blas * 1e-6) / 1e3:.2f} TB/s)"
+          f" | rec_dgrad {t_k:.1f} us ({gb / (t_k * 1e-6) / 1e3:.2f} TB/s)"
+          f" | speedup {t_blas / t_k:.2f}x")
