@@ -215,12 +215,15 @@ def _dir_backward(ext, grad_out, gates_all, h_flat, c_all, h0, c0, w_hh,
     )
     bias_fused = bias_accum is not None
     # dedicated dgrad kernel for grad_h = grad_h_pass + gg @ W_hh
-    # (rec_gemm.hip, same NT form as the forward with K/N swapped); W^T
-    # materialises once per direction vs 100 per-step addmm allocations
+    # (rec_gemm.hip, NT form with K/N swapped).  Measured 0.31x of
+    # hipBLASLt addmm at M=64k (tools/rec_gemm_ab.py: 172.9 vs 54.1 us —
+    # the K=1024 chunk-staging barriers serialise what blas pipelines, and
+    # addmm already runs at 3.7 TB/s on this shape), so OPT-IN
+    # (NERRF_REC_DGRAD=1) as the in-tree baseline for a pipelined rewrite.
     use_rg_d = (
         ext is not None
         and hasattr(ext, "rec_gemm_dgrad")
-        and os.environ.get("NERRF_REC_GEMM", "1") == "1"
+        and os.environ.get("NERRF_REC_DGRAD", "0") == "1"
         and dt == torch.bfloat16 and hdim == 256 and gdim == 1024
     )
     if use_rg_d:
