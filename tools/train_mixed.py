@@ -1,5 +1,12 @@
 import sys, os, json
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+# --proc-identity: train WITH the x[:,27] trusted-comm channel active
+# (constructor.proc_identity_enabled) — the round-3 flip is then this
+# retrain + tools/calibrate_alarm.py with the same env + updating the
+# default in constructor.py.  The env must be set before graph building.
+if "--proc-identity" in sys.argv:
+    os.environ["NERRF_PROC_IDENTITY"] = "1"
+    print("# training with NERRF_PROC_IDENTITY=1 (x[:,27] active)")
 import torch
 from nerrf_amd.config import load_config
 from nerrf_amd.train import run_training, evaluate
