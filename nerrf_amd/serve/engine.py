@@ -92,6 +92,7 @@ class StreamingEngine:
         shard_id: int = 0,
         world: int = 1,
         calibration: Optional[str] = "auto",
+        content_probe_root: Optional[str] = None,
     ) -> None:
         self.device = torch.device(device)
         self.dtype = dtype
@@ -103,6 +104,10 @@ class StreamingEngine:
         from ..graph.incremental import IncrementalWindowState
 
         self._inc_state = IncrementalWindowState()
+        # first-N-KB partial-encrypt byte analysis (reference
+        # threat-model.mdx:246-262): only meaningful when the engine shares
+        # a filesystem with the workload — None keeps the probe off
+        self.content_probe_root = content_probe_root
         self.alarm_threshold = alarm_threshold
         # calibrated per-channel thresholds (tools/calibrate_alarm.py sweep
         # over off-distribution window configs; benign FP = 0 across the
@@ -329,17 +334,31 @@ class StreamingEngine:
         # socket egress to destinations outside the allowlist (the policy
         # channel syscall data alone cannot provide — threat-model.md)
         exfil_dests = [strings[i] for i in _np.nonzero(bits & 32)[0].tolist()]
+        recon_paths = int((bits & 4).sum())
         indicators = {
             "write_to_rename": w2r,
             "suspicious_ext_count": float(len(encrypted_paths)),
             "ransom_note": float(note),
             "exfil_dest_count": float(len(exfil_dests)),
+            "recon_burst": float(recon_paths),
         }
         ind_score = min(
             1.0,
             0.6 * float(len(encrypted_paths) > 0) + 0.3 * float(note)
             + 0.4 * float(w2r > 0.1) + 0.7 * float(len(exfil_dests) > 0),
         )
+        if self.content_probe_root is not None:
+            # partial-encrypt byte analysis on the flagged (or, absent
+            # flags, the highest-write) files that exist on this filesystem
+            from .content_probe import probe_encrypted_fraction
+
+            cand = encrypted_paths or [
+                p for p, _ in sorted(file_mb.items(), key=lambda kv: -kv[1])[:16]
+            ]
+            probe = probe_encrypted_fraction(cand, root=self.content_probe_root)
+            indicators.update(probe)
+            if probe["encrypted_content_frac"] >= 0.5 and probe["probed"] >= 2:
+                ind_score = min(1.0, ind_score + 0.6)
         # boost file scores for files with suspicious aliases
         for p in encrypted_paths:
             base = p

@@ -292,6 +292,43 @@ def test_incremental_random_streams_match_full():
         assert np.array_equal(ed["suspicious"], ref_ed["suspicious"]), trial
 
 
+def test_edge_cache_random_streams_match_full():
+    """Property check: random streams merged TICK BY TICK with a persistent
+    IncrementalWindowState (edge cache live, renames invalidating mid-stream)
+    == full rebuild at every tick."""
+    import numpy as np
+
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    rng = np.random.default_rng(123)
+    names = [f"/x/f{i}" for i in range(40)]
+    calls = ["write", "read", "openat", "unlink", "chmod", "exec"]
+    for trial in range(2):
+        st = DeltaGraphStore(window_s=1000.0, delta_s=1.0)
+        state = IncrementalWindowState()
+        t = 0.0
+        for tick in range(8):
+            for _ in range(120):
+                t += float(rng.random() * 0.02)
+                if rng.random() < 0.08:
+                    i, j = rng.integers(0, len(names), 2)
+                    st.append(ts=t, pid=int(10 + rng.integers(0, 5)),
+                              syscall="rename", path=names[i], new_path=names[j])
+                else:
+                    st.append(ts=t, pid=int(10 + rng.integers(0, 5)),
+                              syscall=str(rng.choice(calls)),
+                              path=str(rng.choice(names)),
+                              nbytes=int(rng.integers(0, 4096)))
+            ev, deltas = st.compact_with_deltas(None)
+            ref_ed = build_edges_and_flags(build_graph_parts(ev))
+            _, ed = merge_window(ev, state.summaries(deltas), state=state)
+            assert np.array_equal(ed["edge_index"], ref_ed["edge_index"]), (trial, tick)
+            assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-6), (trial, tick)
+            assert np.allclose(ed["edge_ts"], ref_ed["edge_ts"], atol=1e-6), (trial, tick)
+
+
 def test_bulk_append_array_matches_scalar_append():
     """store.append_array == per-event append: same delta boundaries, same
     compacted window, same string interning."""

@@ -475,3 +475,75 @@ def test_calibrated_alarm_thresholds_loaded_and_applied():
     eng.store.window_s = 1e9
     eng.ingest_events(arr)
     assert eng.score_window().alarm
+
+
+# ---------------------------------------------------------------------------
+# first-N-KB partial-encrypt content probe (reference threat-model indicator)
+# ---------------------------------------------------------------------------
+
+def test_content_probe_entropy(tmp_path):
+    import os
+
+    from nerrf_amd.serve.content_probe import first_kb_entropy, probe_encrypted_fraction
+
+    text = tmp_path / "plain.dat"
+    text.write_bytes((b"The quick brown fox jumps over the lazy dog.\n" * 200))
+    rnd = tmp_path / "enc.dat"
+    rnd.write_bytes(os.urandom(8192))
+    e_text = first_kb_entropy(str(text))
+    e_rnd = first_kb_entropy(str(rnd))
+    assert e_text is not None and e_text < 5.5
+    assert e_rnd is not None and e_rnd > 7.5
+    assert first_kb_entropy(str(tmp_path / "missing.dat")) is None
+    probe = probe_encrypted_fraction([str(text), str(rnd), str(tmp_path / "nope")])
+    assert probe["probed"] == 2.0
+    assert probe["high_entropy"] == 1.0
+    assert probe["encrypted_content_frac"] == 0.5
+
+
+def test_content_probe_flags_harness_encryption(tmp_path):
+    """run_attack's XOR-keystream output reads as encrypted; the plaintext
+    originals do not."""
+    from nerrf_amd.harness.attack_sim import run_attack
+    from nerrf_amd.serve.content_probe import probe_encrypted_fraction
+
+    for i in range(4):
+        (tmp_path / f"doc_{i:04d}.dat").write_bytes(
+            f"report {i}: quarterly numbers and notes\n".encode() * 300
+        )
+    plain = [str(p) for p in sorted(tmp_path.glob("doc_*.dat"))]
+    assert probe_encrypted_fraction(plain)["encrypted_content_frac"] == 0.0
+    run_attack(tmp_path)
+    enc = [str(p) for p in sorted(tmp_path.glob("*.lockbit3"))]
+    assert len(enc) == 4
+    probe = probe_encrypted_fraction(enc)
+    assert probe["probed"] == 4.0
+    assert probe["encrypted_content_frac"] == 1.0
+
+
+def test_engine_content_probe_indicator(tmp_path):
+    """Engine with content_probe_root set probes the flagged paths of the
+    window and raises the indicator score on real encrypted content."""
+    import numpy as np
+
+    from nerrf_amd.data.trace import EventArrayBuilder
+    from nerrf_amd.harness.attack_sim import run_attack
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    for i in range(4):
+        (tmp_path / f"doc_{i:04d}.dat").write_bytes(b"plain text business data\n" * 400)
+    run_attack(tmp_path)
+    b = EventArrayBuilder()
+    t = 1.0
+    for i, p in enumerate(sorted(tmp_path.glob("*.lockbit3"))):
+        orig = str(p)[: -len(".lockbit3")]
+        b.add(ts=t, pid=77, syscall="write", path=orig, nbytes=4096)
+        b.add(ts=t + 0.1, pid=77, syscall="rename", path=orig, new_path=str(p))
+        t += 0.3
+    eng = StreamingEngine(device="cpu", content_probe_root="")
+    eng.ingest_events(b.build())
+    det = eng.score_window()
+    assert det.indicators["probed"] >= 2.0
+    assert det.indicators["encrypted_content_frac"] == 1.0
+    assert det.indicators["suspicious_ext_count"] >= 4.0
+    assert det.alarm
