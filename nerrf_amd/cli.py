@@ -256,11 +256,16 @@ def main(argv=None) -> int:
     p.add_argument("--dtype", default="bf16")
 
     sub.add_parser("train", help="training entrypoint (args passed through)")
+    sub.add_parser("tune", help="hparam search (args passed through to nerrf_amd.tune)")
 
     if argv is None:
         argv = sys.argv[1:]
     if argv and argv[0] == "train":
         return cmd_train(None, argv[1:])
+    if argv and argv[0] == "tune":
+        from .tune import main as tune_main
+
+        return tune_main(argv[1:])
     args = ap.parse_args(argv)
     return {
         "status": cmd_status,

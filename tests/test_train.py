@@ -109,3 +109,41 @@ def test_pretrained_checkpoint_loads_and_detects():
     # the trained model itself scores victim files high (not just indicators)
     hot = [p for p, s in det.file_scores.items() if s > 0.5 and "/app/uploads/" in p]
     assert len(hot) >= 10
+
+
+def test_hparam_random_search(tmp_path):
+    """tune.random_search: deterministic sampling, per-trial checkpoints,
+    best-by-composite selection, resumable search.json artifact."""
+    import json
+
+    from nerrf_amd.tune import DEFAULT_SPACE, random_search, sample_trial
+
+    # deterministic sampling: same (seed, trial) -> same overrides
+    a = sample_trial(DEFAULT_SPACE, seed=7, trial=3)
+    b = sample_trial(DEFAULT_SPACE, seed=7, trial=3)
+    assert a == b
+    assert set(a) == set(DEFAULT_SPACE)
+    assert a != sample_trial(DEFAULT_SPACE, seed=7, trial=4)
+    # ranged keys sample inside their bounds
+    assert 2e-4 <= float(a["optim.lr"]) <= 5e-3
+
+    res = random_search(
+        n_trials=2,
+        space={"optim.lr": (1e-3, 3e-3), "model.pos_weight": [2.0, 4.0]},
+        base_overrides=[
+            "optim.epochs=1", "data.n_scenarios=2", "data.duration_s=40",
+            "data.benign_rate_hz=120", "model.sage.layers=2",
+            "model.sage.hidden=32", "model.lstm.hidden=32",
+            "run.eval_holdout=1", "run.log_every=1000",
+        ],
+        out_dir=str(tmp_path / "hs"),
+        seed=1,
+    )
+    assert len(res.trials) == 2
+    best = res.best
+    assert best.score == max(t.score for t in res.trials)
+    assert (tmp_path / "hs" / "trial_00").is_dir()
+    summary = json.loads((tmp_path / "hs" / "search.json").read_text())
+    assert summary["best"] == best.trial
+    assert len(summary["trials"]) == 2
+    assert "seq_f1" in res.trials[0].metrics or "node_auc" in res.trials[0].metrics
