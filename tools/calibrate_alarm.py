@@ -59,7 +59,11 @@ def window_maxima(model, kind, duration, rate, seed, attack=True):
 
 
 def main():
-    model = load_model_from_checkpoint(os.path.join(ROOT, "checkpoints", "pretrained"))
+    # NERRF_CAL_CKPT / NERRF_CAL_OUT: calibrate an alternate checkpoint
+    # (e.g. the proc-identity retrain) without touching the vendored rule
+    ckpt = os.environ.get(
+        "NERRF_CAL_CKPT", os.path.join(ROOT, "checkpoints", "pretrained"))
+    model = load_model_from_checkpoint(ckpt)
     rows = []
     for kind, dur, rate, seed in itertools.product(FAMILIES, DURATIONS, RATES, SEEDS[:2]):
         rows.append(window_maxima(model, kind, dur, rate, seed, attack=True))
@@ -100,7 +104,9 @@ def main():
         },
         "rows": rows,
     }
-    out = os.path.join(ROOT, "nerrf_amd", "serve", "alarm_calibration.json")
+    out = os.environ.get(
+        "NERRF_CAL_OUT",
+        os.path.join(ROOT, "nerrf_amd", "serve", "alarm_calibration.json"))
     with open(out, "w") as fh:
         json.dump(cal, fh, indent=1)
     print(json.dumps({k: v for k, v in cal.items() if k != "rows"}, indent=2))
