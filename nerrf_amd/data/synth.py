@@ -51,6 +51,9 @@ class SynthConfig:
     encrypted_ext: str = ".lockbit3"
     seed: int = 0
     kind: str = "lockbit"  # lockbit | supply_chain | supply_chain_net | benign_rotate | benign_backup | benign_build
+    # adversarial: attack process reports an allowlisted comm (defeats naive
+    # trust in the process-identity channel; behavior channels unaffected)
+    comm_masquerade: bool = False
 
 
 _BENIGN_DIRS = ["/var/www/html", "/home/svc/data", "/app/cache", "/var/log/app"]
@@ -126,7 +129,10 @@ def generate(cfg: SynthConfig) -> Tuple[EventArray, Optional[AttackWindow]]:
     if cfg.attack:
         t0 = cfg.attack_start_frac * cfg.duration_s
         atk_pid = np.int64(6666)
-        atk_comm = comms.intern("lockbit")
+        # comm-masquerade variant: the payload reports an allowlisted daemon
+        # comm (process-identity adversarial case — the trusted-comm channel
+        # must not become a bypass; behavioral channels still fire)
+        atk_comm = comms.intern("nginx" if cfg.comm_masquerade else "lockbit")
 
         # phase 1 — recon burst (exec of enumeration binaries)
         n_recon = cfg.recon_burst

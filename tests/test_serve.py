@@ -547,3 +547,27 @@ def test_engine_content_probe_indicator(tmp_path):
     assert det.indicators["encrypted_content_frac"] == 1.0
     assert det.indicators["suspicious_ext_count"] >= 4.0
     assert det.alarm
+
+
+def test_comm_masquerade_attack_still_detected(monkeypatch):
+    """Adversarial process-identity case: the attacker reports an
+    allowlisted comm, so the trusted-comm channel marks it trusted — and
+    the behavioral channels must still alarm (the channel is evidence,
+    never a bypass)."""
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    arr, win = generate(SynthConfig(seed=4, duration_s=45.0, benign_rate_hz=400.0,
+                                    n_victim_files=12, comm_masquerade=True))
+    # the masqueraded payload IS trusted by the comm channel...
+    monkeypatch.setenv("NERRF_PROC_IDENTITY", "1")
+    parts = build_graph_parts(arr)
+    ed = build_edges_and_flags(parts)
+    atk_local = int(np.searchsorted(parts["upids"], 6666))
+    assert ed["trusted_proc"][parts["n_files"] + atk_local] == 1.0
+    # ...and detection still fires on behavior (indicators + rename storm)
+    eng = StreamingEngine(device="cpu")
+    eng.ingest_events(arr)
+    det = eng.score_window()
+    assert det.alarm
+    assert det.indicators["suspicious_ext_count"] > 0
