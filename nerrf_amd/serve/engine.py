@@ -114,11 +114,18 @@ class StreamingEngine:
         # sweep — docs/threat-model.md has the trade table).  "auto" loads
         # the vendored table; None keeps the single-threshold rule.
         self.calibrated = None
-        if calibration == "auto":
+        if calibration is not None and calibration != "off":
+            # "auto" -> the vendored default; anything else is a path to a
+            # calibration json (e.g. the proc-identity checkpoint ships its
+            # own next to its weights: checkpoints/pretrained_procid/)
             import json as _json
             from pathlib import Path as _Path
 
-            cal_p = _Path(__file__).parent / "alarm_calibration.json"
+            cal_p = (
+                _Path(__file__).parent / "alarm_calibration.json"
+                if calibration == "auto"
+                else _Path(calibration)
+            )
             if cal_p.exists():
                 try:
                     c = _json.loads(cal_p.read_text())

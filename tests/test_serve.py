@@ -571,3 +571,36 @@ def test_comm_masquerade_attack_still_detected(monkeypatch):
     det = eng.score_window()
     assert det.alarm
     assert det.indicators["suspicious_ext_count"] > 0
+
+
+def test_proc_identity_stack_end_to_end(monkeypatch):
+    """The vendored proc-identity stack (checkpoints/pretrained_procid +
+    its own calibration, NERRF_PROC_IDENTITY=1): detects attacks including
+    the comm-masquerade variant, stays quiet on benign hard negatives."""
+    import os as _os
+
+    from nerrf_amd.serve.engine import StreamingEngine, load_model_from_checkpoint
+
+    root = _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__)))
+    ckpt = _os.path.join(root, "checkpoints", "pretrained_procid")
+    if not _os.path.isdir(ckpt):
+        pytest.skip("proc-identity checkpoint not vendored")
+    monkeypatch.setenv("NERRF_PROC_IDENTITY", "1")
+    model = load_model_from_checkpoint(ckpt)
+    cal = _os.path.join(ckpt, "alarm_calibration.json")
+
+    def score(arr):
+        eng = StreamingEngine(model=model, device="cpu", window_s=1e9,
+                              calibration=cal)
+        eng.ingest_events(arr)
+        return eng.score_window()
+
+    atk, _ = generate(SynthConfig(seed=11, duration_s=45.0, benign_rate_hz=300.0,
+                                  n_victim_files=10))
+    assert score(atk).alarm
+    masq, _ = generate(SynthConfig(seed=12, duration_s=45.0, benign_rate_hz=300.0,
+                                   n_victim_files=10, comm_masquerade=True))
+    assert score(masq).alarm
+    ben, _ = generate(SynthConfig(seed=13, duration_s=45.0, benign_rate_hz=300.0,
+                                  kind="benign_backup"))
+    assert not score(ben).alarm
