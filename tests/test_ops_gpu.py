@@ -735,6 +735,31 @@ def test_rec_gemm_fwd_matches_matmul(gpu_device):
         assert torch.allclose(c.float(), refs, atol=3e-2, rtol=3e-2), f"M={m} strided"
 
 
+def test_rec_gemm_dgrad_matches_addmm(gpu_device):
+    """Opt-in dgrad kernel (rec_gemm.hip): c = a @ wt^T + d vs fp32 addmm,
+    with and without the addend, strided gate-slab rows, tail guard."""
+    from nerrf_amd.ops.native import load_extension
+
+    ext = load_extension(required=True)
+    torch.manual_seed(43)
+    for m in (1000, 4096):
+        gg = (torch.randn(m, 1024, device=gpu_device) * 0.1).to(torch.bfloat16)
+        w = (torch.randn(1024, 256, device=gpu_device) * 0.1).to(torch.bfloat16)
+        wt = w.t().contiguous()
+        d = (torch.randn(m, 256, device=gpu_device) * 0.1).to(torch.bfloat16)
+        out = torch.empty(m, 256, device=gpu_device, dtype=torch.bfloat16)
+        ext.rec_gemm_dgrad(gg, wt, d, out)
+        ref = torch.addmm(d.float(), gg.float(), w.float())
+        assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), f"M={m}"
+        # no addend + strided A (gate slab of the bilayer [T,B,2*4H] buffer)
+        slab = (torch.randn(m, 2048, device=gpu_device) * 0.1).to(torch.bfloat16)
+        gg_s = slab[:, 1024:]
+        empty = torch.empty(0, device=gpu_device, dtype=torch.bfloat16)
+        ext.rec_gemm_dgrad(gg_s, wt, empty, out)
+        refs = gg_s.float() @ w.float()
+        assert torch.allclose(out.float(), refs, atol=3e-2, rtol=3e-2), f"M={m} strided"
+
+
 def test_rec_gemm_in_recurrence_matches_reference(gpu_device):
     """lstm_bilayer2 with NERRF_REC_GEMM=1 == reference recurrence (the
     step GEMM swap must not move training numerics)."""
