@@ -600,3 +600,36 @@ def test_edge_cache_parity_across_ticks():
     assert state._edge_cache is cache_obj  # hit, not rebuilt
     assert np.array_equal(a["edge_index"], b["edge_index"])
     assert np.array_equal(a["edge_weight"], b["edge_weight"])
+
+
+def test_edge_cache_survives_window_expiry():
+    """Delta expiry (window trim drops sealed deltas from the front)
+    invalidates the stable-prefix cache token; every post-expiry merge
+    still equals the full rebuild."""
+    from nerrf_amd.graph.constructor import build_edges_and_flags, build_graph_parts
+    from nerrf_amd.graph.incremental import IncrementalWindowState, merge_window
+    from nerrf_amd.graph.store import DeltaGraphStore
+
+    st = DeltaGraphStore(window_s=6.0, delta_s=1.0)
+    state = IncrementalWindowState()
+    rng = np.random.default_rng(17)
+    t = 0.0
+    n_checked = 0
+    for tick in range(12):
+        for _ in range(60):
+            t += float(rng.random() * 0.05)
+            st.append(ts=t, pid=int(20 + rng.integers(0, 3)),
+                      syscall=["write", "read", "rename"][int(rng.integers(0, 3))],
+                      path=f"/w/f{int(rng.integers(0, 25))}",
+                      new_path=f"/w/g{int(rng.integers(0, 25))}"
+                      if rng.random() < 0.15 else "",
+                      nbytes=int(rng.integers(0, 2048)))
+        ev, deltas = st.compact_with_deltas(None)
+        if not len(ev):
+            continue
+        ref_ed = build_edges_and_flags(build_graph_parts(ev))
+        _, ed = merge_window(ev, state.summaries(deltas), state=state)
+        assert np.array_equal(ed["edge_index"], ref_ed["edge_index"]), tick
+        assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-6), tick
+        n_checked += 1
+    assert n_checked >= 10
