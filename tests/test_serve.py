@@ -604,3 +604,26 @@ def test_proc_identity_stack_end_to_end(monkeypatch):
     ben, _ = generate(SynthConfig(seed=13, duration_s=45.0, benign_rate_hz=300.0,
                                   kind="benign_backup"))
     assert not score(ben).alarm
+
+
+def test_metrics_instrumentation():
+    """serve.metrics: instrument_engine wraps score/plan without changing
+    behavior (works with or without prometheus_client installed)."""
+    from nerrf_amd.serve import metrics
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    arr, _ = generate(SynthConfig(seed=21, duration_s=40.0, benign_rate_hz=200.0,
+                                  n_victim_files=8))
+    eng = StreamingEngine(device="cpu")
+    eng.ingest_events(arr)
+    base = eng.score_window()
+    metrics.instrument_engine(eng)
+    det = eng.score_window()
+    assert det.alarm == base.alarm
+    assert det.window_events == base.window_events
+    plan = eng.plan(det, n_sims=64, use_gpu=False)
+    assert plan.simulations == 64
+    # endpoint helper returns the port (prom installed) or None (absent) —
+    # both are valid contracts; calling it must not raise
+    port = metrics.serve_metrics(port=0) if metrics._HAVE_PROM else metrics.serve_metrics()
+    assert port is None or isinstance(port, int)
