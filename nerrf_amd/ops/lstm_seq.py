@@ -382,14 +382,44 @@ class _LSTMBiLayer2Fn(torch.autograd.Function):
         hdim = gdim // 4
         h2 = torch.empty(t_len, batch, 2 * hdim, device=xg2.device, dtype=xg2.dtype)
         ext = get_native(xg2)
-        c_f, g_f = _dir_forward(
-            ext, xg2[:, :, :gdim], h0, c0, w_f, b_f, mask, False, infer,
-            h2[:, :, :hdim]
+        # the two directions are independent recurrence chains; at inference
+        # (no autograd stream bookkeeping) they can overlap on a side HIP
+        # stream — each chain is 100 serial small-grid kernels, so the
+        # other direction's work fills the idle CUs.  Opt-in
+        # (NERRF_STREAM_DIRS=1) pending an in-context A/B.
+        overlap = (
+            infer
+            and xg2.is_cuda
+            and os.environ.get("NERRF_STREAM_DIRS", "0") == "1"
         )
-        c_b, g_b = _dir_forward(
-            ext, xg2[:, :, gdim:], h0, c0, w_b, b_b, mask, True, infer,
-            h2[:, :, hdim:]
-        )
+        if overlap:
+            main = torch.cuda.current_stream()
+            side = torch.cuda.Stream()
+            side.wait_stream(main)  # xg2/h0/c0 ready
+            c_f, g_f = _dir_forward(
+                ext, xg2[:, :, :gdim], h0, c0, w_f, b_f, mask, False, infer,
+                h2[:, :, :hdim]
+            )
+            with torch.cuda.stream(side):
+                c_b, g_b = _dir_forward(
+                    ext, xg2[:, :, gdim:], h0, c0, w_b, b_b, mask, True, infer,
+                    h2[:, :, hdim:]
+                )
+            main.wait_stream(side)
+            # cross-stream allocator safety: h2 (allocated on main) was
+            # written on the side stream; c_b (allocated on side) is
+            # consumed/freed on main — record each tensor's FOREIGN stream
+            h2.record_stream(side)
+            c_b.record_stream(main)
+        else:
+            c_f, g_f = _dir_forward(
+                ext, xg2[:, :, :gdim], h0, c0, w_f, b_f, mask, False, infer,
+                h2[:, :, :hdim]
+            )
+            c_b, g_b = _dir_forward(
+                ext, xg2[:, :, gdim:], h0, c0, w_b, b_b, mask, True, infer,
+                h2[:, :, hdim:]
+            )
         ctx.save_for_backward(
             g_f, g_b, h2, c_f, c_b, h0, c0, w_f, w_b,
             mask if mask is not None else torch.empty(0, device=xg2.device),

@@ -735,6 +735,26 @@ def test_rec_gemm_fwd_matches_matmul(gpu_device):
         assert torch.allclose(c.float(), refs, atol=3e-2, rtol=3e-2), f"M={m} strided"
 
 
+def test_stream_dirs_inference_matches_sequential(gpu_device, monkeypatch):
+    """Opt-in dual-stream inference (NERRF_STREAM_DIRS=1): both directions
+    overlapped on a side HIP stream must produce bit-identical output to
+    the sequential path."""
+    from nerrf_amd.models.lstm import BiLSTMDetector, LSTMConfig
+
+    torch.manual_seed(44)
+    model = BiLSTMDetector(LSTMConfig(in_dim=32, hidden=256, layers=2)).to(
+        gpu_device, torch.bfloat16
+    ).eval()
+    x = (torch.randn(128, 20, 32, device=gpu_device) * 0.3).to(torch.bfloat16)
+    lengths = torch.full((128,), 20, device=gpu_device, dtype=torch.int64)
+    with torch.no_grad():
+        y_seq = model(x, lengths)
+        monkeypatch.setenv("NERRF_STREAM_DIRS", "1")
+        y_ovl = model(x, lengths)
+        torch.cuda.synchronize()
+    assert torch.equal(y_seq, y_ovl)
+
+
 def test_rec_gemm_dgrad_matches_addmm(gpu_device):
     """Opt-in dgrad kernel (rec_gemm.hip): c = a @ wt^T + d vs fp32 addmm,
     with and without the addend, strided gate-slab rows, tail guard."""
