@@ -35,6 +35,10 @@ def cmd_status(args) -> int:
     }
     if args.checkpoint and Path(args.checkpoint, "checkpoint.json").exists():
         info["checkpoint"] = json.loads(Path(args.checkpoint, "checkpoint.json").read_text())
+    from .serve.registry import DEFAULT_STATE_DIR, DetectionRegistry
+
+    info["detections"] = DetectionRegistry(
+        getattr(args, "state_dir", None) or DEFAULT_STATE_DIR).list()[:20]
     print(json.dumps(info, indent=2))
     return 0
 
@@ -65,6 +69,41 @@ def cmd_undo(args) -> int:
     from .serve.engine import StreamingEngine
 
     t0 = time.time()
+    if args.id:
+        # replay a RECORDED response (reference contract: nerrf undo --id):
+        # the monitor already scored, planned and persisted the attack —
+        # execute its rollback without re-scoring
+        from .serve.registry import DEFAULT_STATE_DIR, DetectionRegistry
+        from .serve.rollback import execute_rollback
+
+        reg = DetectionRegistry(args.state_dir or DEFAULT_STATE_DIR)
+        rec = reg.load(args.id)
+        if rec is None:
+            print(json.dumps({"error": f"unknown attack id {args.id}",
+                              "known": [r["attack_id"] for r in reg.list()]}))
+            return 1
+        target = args.dir or rec.get("target_dir") or ""
+        if not target:
+            print(json.dumps({"error": "record has no target_dir; pass --dir"}))
+            return 1
+        manifest = None
+        mpath = Path(target, ".nerrf_manifest.json")
+        if mpath.exists():
+            manifest = json.loads(mpath.read_text())
+        result = execute_rollback(target, manifest=manifest, decrypt=True,
+                                  validate_in_sandbox=True)
+        out = {
+            "attack_id": rec["attack_id"],
+            "recorded_plan": rec.get("plan", {}).get("actions"),
+            "indicators": rec.get("indicators"),
+            "mttr_s": time.time() - t0,
+            **result.as_dict(),
+        }
+        print(json.dumps(out, indent=2))
+        return 0 if result.files_failed == 0 else 2
+    if not args.dir:
+        print(json.dumps({"error": "--dir is required without --id"}))
+        return 1
     model = None
     if args.checkpoint:
         from .serve.engine import load_model_from_checkpoint
@@ -125,10 +164,14 @@ def cmd_serve(args) -> int:
         # bf16 on GPU enables the fused-MFMA scoring path
         dtype = torch.bfloat16 if args.device != "cpu" else torch.float32
         engine = StreamingEngine(model=model, device=args.device, dtype=dtype)
+        from .serve.registry import DEFAULT_STATE_DIR, DetectionRegistry
+
         for status in engine.run_monitor(
             interval_s=args.interval,
             max_iterations=args.iterations,
             tracker_address=args.tracker,
+            registry=DetectionRegistry(args.state_dir or DEFAULT_STATE_DIR),
+            target_dir=args.target_dir,
         ):
             print(json.dumps(status), flush=True)
         return 0
@@ -221,8 +264,9 @@ def main(argv=None) -> int:
     p.add_argument("--trace-out", default=None)
 
     p = sub.add_parser("undo", help="detect + plan + rollback a directory")
-    p.add_argument("--dir", required=True)
-    p.add_argument("--id", default=None, help="attack id (informational)")
+    p.add_argument("--dir", default=None, help="target directory (defaults to the recorded one with --id)")
+    p.add_argument("--id", default=None, help="recorded attack id (see `nerrf status`)")
+    p.add_argument("--state-dir", default=None, help="detection registry dir")
     p.add_argument("--trace", default=None, help="trace file to ingest first")
     p.add_argument("--device", default="cpu")
     p.add_argument("--sims", type=int, default=1024)
@@ -246,6 +290,9 @@ def main(argv=None) -> int:
     p.add_argument("--max-events", type=int, default=None)
     p.add_argument("--timeout", type=float, default=15.0)
     p.add_argument("--device", default="cpu")
+    p.add_argument("--state-dir", default=None,
+                   help="persist alarms for `nerrf undo --id` (default .nerrf/detections)")
+    p.add_argument("--target-dir", default="", help="recorded rollback target")
 
     p = sub.add_parser("eval", help="detection-quality report on synthetic scenarios")
     p.add_argument("--checkpoint", default="checkpoints/pretrained")

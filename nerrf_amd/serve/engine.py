@@ -574,6 +574,8 @@ class StreamingEngine:
         tracker_address: Optional[str] = None,
         stop_event=None,
         sims: int = 1024,
+        registry=None,  # serve.registry.DetectionRegistry: persist alarms
+        target_dir: str = "",
     ):
         """Continuous detection loop: ingest -> score -> (on alarm) plan.
 
@@ -601,14 +603,21 @@ class StreamingEngine:
             t0 = time.perf_counter()
             det = self.score_window()
             plan = None
+            attack_id = None
             if det.alarm:
                 plan = self.plan(det, n_sims=sims)
+                if registry is not None:
+                    attack_id = registry.record(
+                        det, plan, target_dir=target_dir,
+                        n_groups=self.planner_params.n_groups,
+                    )
                 if on_alarm is not None:
                     on_alarm(det, plan)
             status = {
                 "iteration": it,
                 "window_events": det.window_events,
                 "alarm": det.alarm,
+                "attack_id": attack_id,
                 "score_s": time.perf_counter() - t0,
                 "plan": None if plan is None else plan.describe(self.planner_params.n_groups),
             }

@@ -627,3 +627,85 @@ def test_metrics_instrumentation():
     # both are valid contracts; calling it must not raise
     port = metrics.serve_metrics(port=0) if metrics._HAVE_PROM else metrics.serve_metrics()
     assert port is None or isinstance(port, int)
+
+
+# ---------------------------------------------------------------------------
+# detection registry + `nerrf undo --id` (reference CLI contract)
+# ---------------------------------------------------------------------------
+
+def test_detection_registry_roundtrip(tmp_path):
+    from nerrf_amd.serve.engine import StreamingEngine
+    from nerrf_amd.serve.registry import DetectionRegistry
+
+    arr, _ = generate(SynthConfig(seed=31, duration_s=40.0, benign_rate_hz=200.0,
+                                  n_victim_files=8))
+    eng = StreamingEngine(device="cpu")
+    eng.ingest_events(arr)
+    det = eng.score_window()
+    assert det.alarm
+    plan = eng.plan(det, n_sims=64, use_gpu=False)
+    reg = DetectionRegistry(str(tmp_path / "state"))
+    aid = reg.record(det, plan, target_dir="/app/uploads",
+                     n_groups=eng.planner_params.n_groups)
+    assert aid.startswith("atk-")
+    rec = reg.load(aid)
+    assert rec["alarm"] is True
+    assert rec["target_dir"] == "/app/uploads"
+    assert rec["plan"]["simulations"] == 64
+    assert len(rec["file_scores"]) <= 200
+    # listing: newest first, summary fields
+    lst = reg.list()
+    assert lst[0]["attack_id"] == aid
+    assert reg.load("atk-nope") is None
+
+
+def test_monitor_records_alarms(tmp_path):
+    from nerrf_amd.serve.engine import StreamingEngine
+    from nerrf_amd.serve.registry import DetectionRegistry
+
+    arr, _ = generate(SynthConfig(seed=32, duration_s=40.0, benign_rate_hz=200.0,
+                                  n_victim_files=8))
+    eng = StreamingEngine(device="cpu")
+    eng.ingest_events(arr)
+    reg = DetectionRegistry(str(tmp_path / "state"))
+    statuses = list(eng.run_monitor(interval_s=0.0, max_iterations=2, sims=64,
+                                    registry=reg, target_dir="/t"))
+    alarmed = [s for s in statuses if s["alarm"]]
+    assert alarmed and all(s["attack_id"] for s in alarmed)
+    assert len(reg.list()) == len(alarmed)
+
+
+def test_cli_undo_by_id(tmp_path, capsys, monkeypatch):
+    """nerrf undo --id executes the recorded rollback (sha256-verified)."""
+    from nerrf_amd.cli import main
+    from nerrf_amd.harness.attack_sim import run_attack, seed_files, verify_manifest
+    from nerrf_amd.serve.engine import StreamingEngine
+    from nerrf_amd.serve.registry import DetectionRegistry
+
+    victim = tmp_path / "v"
+    manifest = seed_files(victim, n_files=5, file_kb=4, seed=9)
+    (victim / ".nerrf_manifest.json").write_text(json.dumps(manifest))
+    report = run_attack(victim, trace_path=victim / "t.jsonl")
+    assert len(report.files_attacked) == 5
+
+    from nerrf_amd.data.trace import load_trace
+
+    eng = StreamingEngine(device="cpu")
+    eng.ingest_events(load_trace(victim / "t.jsonl"))
+    det = eng.score_window()
+    assert det.alarm
+    plan = eng.plan(det, n_sims=64, use_gpu=False)
+    reg = DetectionRegistry(str(tmp_path / "state"))
+    aid = reg.record(det, plan, target_dir=str(victim),
+                     n_groups=eng.planner_params.n_groups)
+
+    rc = main(["undo", "--id", aid, "--state-dir", str(tmp_path / "state")])
+    out = json.loads(capsys.readouterr().out)
+    assert rc == 0
+    assert out["attack_id"] == aid
+    assert out["files_restored"] == 5
+    assert all(verify_manifest(manifest).values())
+    # unknown id fails gracefully with the known list
+    rc = main(["undo", "--id", "atk-bogus", "--state-dir", str(tmp_path / "state")])
+    err = json.loads(capsys.readouterr().out)
+    assert rc == 1 and aid in err["known"]
