@@ -297,7 +297,9 @@ def generate_supply_chain(cfg: SynthConfig) -> Tuple[EventArray, Optional[Attack
         return base, None
     paths, comms = base.paths, base.comms
     atk_pid = np.int64(7777)
-    atk_comm = comms.intern("postinstall")
+    # comm_masquerade: the trojaned postinstall reports an allowlisted
+    # daemon comm (see the lockbit variant above)
+    atk_comm = comms.intern("node" if cfg.comm_masquerade else "postinstall")
     t0 = cfg.attack_start_frac * cfg.duration_s
     n_deps = max(cfg.n_victim_files, 4)
     dep_ids = np.array(
