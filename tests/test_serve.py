@@ -709,3 +709,25 @@ def test_cli_undo_by_id(tmp_path, capsys, monkeypatch):
     rc = main(["undo", "--id", "atk-bogus", "--state-dir", str(tmp_path / "state")])
     err = json.loads(capsys.readouterr().out)
     assert rc == 1 and aid in err["known"]
+
+
+def test_registry_same_second_ids_unique(tmp_path):
+    from nerrf_amd.serve.engine import Detection
+    from nerrf_amd.serve.registry import DetectionRegistry
+
+    det = Detection(True, 1.0, {"/a": 0.9}, {}, {})
+    reg = DetectionRegistry(str(tmp_path))
+    ids = {reg.record(det) for _ in range(4)}
+    assert len(ids) == 4  # seq suffix disambiguates same-second alarms
+    assert len(reg.list()) == 4
+
+
+def test_engine_missing_calibration_falls_back(tmp_path):
+    from nerrf_amd.serve.engine import StreamingEngine
+
+    eng = StreamingEngine(device="cpu", calibration=str(tmp_path / "nope.json"))
+    assert eng.calibrated is None  # single-threshold rule still works
+    arr, _ = generate(SynthConfig(seed=33, duration_s=40.0, benign_rate_hz=150.0,
+                                  n_victim_files=6))
+    eng.ingest_events(arr)
+    assert eng.score_window().alarm
