@@ -190,11 +190,22 @@ def cmd_serve(args) -> int:
     n = engine.ingest_from_tracker(server.address, max_events=args.max_events, timeout_s=args.timeout)
     det = engine.score_window()
     server.stop()
+    attack_id = None
+    if det.alarm and args.state_dir:
+        # persist for `nerrf undo --id` (same record the live monitor makes)
+        from .serve.registry import DetectionRegistry
+
+        plan = engine.plan(det, n_sims=256, use_gpu=(args.device != "cpu"))
+        attack_id = DetectionRegistry(args.state_dir).record(
+            det, plan, target_dir=args.target_dir,
+            n_groups=engine.planner_params.n_groups,
+        )
     print(
         json.dumps(
             {
                 "events_ingested": n,
                 "alarm": det.alarm,
+                "attack_id": attack_id,
                 "indicators": det.indicators,
                 "suspicious_files": det.encrypted_paths[:10],
             },

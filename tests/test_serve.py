@@ -731,3 +731,29 @@ def test_engine_missing_calibration_falls_back(tmp_path):
                                   n_victim_files=6))
     eng.ingest_events(arr)
     assert eng.score_window().alarm
+
+
+def test_cli_simulate_serve_undo_workflow(tmp_path, capsys):
+    """The full CLI story: simulate an attack on disk, serve the trace
+    (recording the alarm), then undo by the recorded id — byte-exact."""
+    from nerrf_amd.cli import main
+    from nerrf_amd.harness.attack_sim import verify_manifest
+
+    victim = str(tmp_path / "v")
+    state = str(tmp_path / "state")
+    trace = str(tmp_path / "t.jsonl")
+    rc = main(["simulate", "--dir", victim, "--seed-files", "--n-files", "5",
+               "--file-kb", "4", "--trace-out", trace])
+    capsys.readouterr()
+    assert rc == 0
+    rc = main(["serve", "--trace", trace, "--state-dir", state,
+               "--target-dir", victim, "--timeout", "10"])
+    out = capsys.readouterr().out
+    payload = json.loads(out[out.index("{"):])
+    assert rc == 0 and payload["alarm"] and payload["attack_id"]
+    rc = main(["undo", "--id", payload["attack_id"], "--state-dir", state])
+    undo = json.loads(capsys.readouterr().out)
+    assert rc == 0
+    assert undo["files_restored"] == 5
+    manifest = json.loads((Path(victim) / ".nerrf_manifest.json").read_text())
+    assert all(verify_manifest(manifest).values())
