@@ -3,8 +3,9 @@
 Commands (spec: reference ROADMAP.md:86, README.md:82 — `nerrf undo`,
 `nerrf status` — plus the operational verbs this engine adds):
 
-  nerrf status                       engine/store/checkpoint status
-  nerrf undo --dir D [--id ID]       detect + plan + sandbox-validate + restore
+  nerrf status                       engine/store/checkpoint status + recorded detections
+  nerrf undo --dir D                 detect + plan + sandbox-validate + restore
+  nerrf undo --id ATK                replay a recorded response (serve/registry.py)
   nerrf simulate --dir D             run the reversible LockBit attack sim
   nerrf scenario --dir D             full e2e: seed -> attack -> detect ->
                                      plan -> rollback -> sha256 verify
