@@ -6,4 +6,4 @@ nerrf.trace gRPC contract, 30-60 s temporal dependency graphs in HBM,
 GraphSAGE-T + BiLSTM anomaly detection with hand-written HIP kernels,
 batched MCTS rollback planning, and RCCL data-parallel training over xGMI.
 """
-__version__ = "0.1.0"
+__version__ = "0.2.0"
