@@ -126,7 +126,7 @@ def main() -> None:
         device=device, dtype=dtype
     )
     # fused AdamW measured neutral-to-slightly-slower than foreach on
-    # MI355X at this step (243.7 vs 242.5 ms, gpurun_out/bench_fused.log) —
+    # MI355X at this step (243.7 vs 242.5 ms, profiles/PROFILES.md) —
     # foreach stays the default; NERRF_FUSED_ADAM=1 to re-test
     fused_ok = has_gpu and os.environ.get("NERRF_FUSED_ADAM", "0") == "1"
     opt = torch.optim.AdamW(
