@@ -154,3 +154,27 @@ def test_operating_point_fixed_fp_budget():
     # generous budget flags more
     op2 = operating_point(y, s, max_fp_frac=0.5)
     assert op2["recall"] >= op["recall"]
+
+
+def test_flags_doc_covers_all_env_flags():
+    """docs/flags.md must document every NERRF_* runtime flag in the code
+    (guards the catalogue against rot as flags are added)."""
+    import os
+    import re
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    flags = set()
+    scan_dirs = ["nerrf_amd", "tools"]
+    for d in scan_dirs:
+        for base, _dirs, files in os.walk(os.path.join(root, d)):
+            if "__pycache__" in base:
+                continue
+            for fn in files:
+                if fn.endswith(".py"):
+                    src = open(os.path.join(base, fn)).read()
+                    flags.update(re.findall(r"NERRF_[A-Z_]+", src))
+    flags.update(re.findall(r"NERRF_[A-Z_]+", open(os.path.join(root, "bench.py")).read()))
+    doc = open(os.path.join(root, "docs", "flags.md")).read()
+    documented = set(re.findall(r"NERRF_[A-Z_]+", doc))
+    missing = flags - documented
+    assert not missing, f"undocumented runtime flags: {sorted(missing)}"
