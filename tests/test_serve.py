@@ -262,7 +262,7 @@ def test_monitor_loop_detects_and_plans(tmp_path):
     assert alarms and alarms[0][1].plan  # a non-empty undo plan was produced
 
 
-def test_cli_serve_live_monitor_mode(capsys):
+def test_cli_serve_live_monitor_mode(capsys, tmp_path):
     """`nerrf serve --tracker` consumes a live stream and monitors."""
     from nerrf_amd.cli import main
     from nerrf_amd.serve.tracker_sim import TrackerSimServer
@@ -271,11 +271,14 @@ def test_cli_serve_live_monitor_mode(capsys):
     server = TrackerSimServer(arr, batch_size=64)
     server.start()
     try:
-        rc = main(["serve", "--tracker", server.address, "--interval", "1.5", "--iterations", "2"])
+        rc = main(["serve", "--tracker", server.address, "--interval", "1.5",
+                   "--iterations", "2", "--state-dir", str(tmp_path / "st")])
         assert rc == 0
         lines = [json.loads(l) for l in capsys.readouterr().out.strip().splitlines() if l.startswith("{")]
         assert len(lines) == 2
         assert lines[-1]["window_events"] > 0
+        if any(l["alarm"] for l in lines):  # alarms persisted hermetically
+            assert list((tmp_path / "st").glob("atk-*.json"))
         assert lines[-1]["alarm"]  # .lockbit3 traffic in the stream
     finally:
         server.stop()
