@@ -84,3 +84,22 @@ def test_detection_with_vendored_checkpoint(tmp_path):
     rep = detect_on_recorded_run(REF / "m1" / "results", model=model)
     assert rep["alarm_within_window"]
     assert rep["encrypted_file_recall"] == 1.0
+
+
+@needs_reference
+def test_detection_with_proc_identity_stack(monkeypatch):
+    """The proc-identity stack (channel active + its checkpoint) also
+    detects the recorded m0/m1 attacks inside their ground-truth windows —
+    the future default-swap is CPU-validated against the only real data."""
+    ckpt = Path("checkpoints/pretrained_procid")
+    if not (ckpt / "checkpoint.json").exists():
+        pytest.skip("no proc-identity checkpoint")
+    from nerrf_amd.harness.reference_parity import detect_on_recorded_run
+    from nerrf_amd.serve.engine import load_model_from_checkpoint
+
+    monkeypatch.setenv("NERRF_PROC_IDENTITY", "1")
+    model = load_model_from_checkpoint(str(ckpt))
+    for stem in ("m0", "m1"):
+        rep = detect_on_recorded_run(REF / stem / "results", model=model)
+        assert rep["alarm_within_window"], stem
+        assert rep["encrypted_file_recall"] == 1.0, stem
