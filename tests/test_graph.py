@@ -633,3 +633,29 @@ def test_edge_cache_survives_window_expiry():
         assert np.allclose(ed["edge_weight"], ref_ed["edge_weight"], atol=1e-6), tick
         n_checked += 1
     assert n_checked >= 10
+
+
+def test_aggregate_sparse_keys_branches_agree():
+    """Property: the bincount branch and the radix branch of
+    aggregate_sparse_keys produce identical (keys, sums, maxes) — checked
+    by forcing both on the same data via the key_space guard."""
+    from nerrf_amd.graph.constructor import aggregate_sparse_keys
+
+    rng = np.random.default_rng(77)
+    for trial in range(5):
+        n = int(rng.integers(1, 5000))
+        space = int(rng.integers(10, 1 << 20))
+        keys = rng.integers(0, space, size=n).astype(np.int64)
+        w = rng.random(n)
+        ts = rng.random(n) * 100
+        # small key_space -> bincount branch
+        uk_a, s_a, m_a = aggregate_sparse_keys(keys, w, ts, space)
+        # huge declared key_space -> radix branch (same keys)
+        uk_b, s_b, m_b = aggregate_sparse_keys(keys, w, ts, 1 << 40)
+        assert np.array_equal(uk_a, uk_b), trial
+        assert np.allclose(s_a, s_b, atol=1e-9), trial
+        assert np.array_equal(m_a, m_b), trial
+        # against np.unique reference
+        uk_r, inv = np.unique(keys, return_inverse=True)
+        assert np.array_equal(uk_a, uk_r)
+        assert np.allclose(s_a, np.bincount(inv, weights=w), atol=1e-9)
