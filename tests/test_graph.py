@@ -659,3 +659,30 @@ def test_aggregate_sparse_keys_branches_agree():
         uk_r, inv = np.unique(keys, return_inverse=True)
         assert np.array_equal(uk_a, uk_r)
         assert np.allclose(s_a, np.bincount(inv, weights=w), atol=1e-9)
+
+
+def test_sample_fanout_torch_validity_cpu():
+    """Property: the device-side sampler (runs on CPU tensors too) draws
+    only true in-neighbors, self-fills isolated nodes, and is
+    deterministic per seed — same contract as the numpy sampler."""
+    import torch
+
+    from nerrf_amd.graph.sampling import sample_fanout_torch, to_csr
+
+    arr, _ = generate(SynthConfig(seed=8, duration_s=40, benign_rate_hz=150))
+    g = build_graph(arr)
+    ei = torch.from_numpy(g.edge_index)
+    ew = torch.from_numpy(g.edge_weight)
+    idx, w = sample_fanout_torch(ei, ew, g.num_nodes, 16, seed=3)
+    idx2, _ = sample_fanout_torch(ei, ew, g.num_nodes, 16, seed=3)
+    assert torch.equal(idx, idx2)  # deterministic per seed
+    csr = to_csr(g.edge_index, g.num_nodes, g.edge_weight)
+    deg = np.diff(csr.indptr)
+    idx_np = idx.numpy()
+    for n in range(0, g.num_nodes, max(1, g.num_nodes // 23)):
+        if deg[n] == 0:
+            assert (idx_np[n] == n).all()
+        else:
+            nbrs = set(csr.indices[csr.indptr[n]:csr.indptr[n + 1]].tolist())
+            assert set(idx_np[n].tolist()) <= nbrs
+    assert (w >= 0).all()
